@@ -1,0 +1,55 @@
+// pybind bindings for colossalai_amd._C (gfx950 HIP kernels).
+#include <torch/extension.h>
+#include <vector>
+
+namespace cai {
+
+// multi_tensor_adam.hip
+void multi_tensor_adam(std::vector<at::Tensor> grads, std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+                       std::vector<at::Tensor> param_outs, double lr, double beta1, double beta2,
+                       double eps, long step, bool adamw_mode, bool bias_correction, double weight_decay,
+                       double div_scale, long chunk_size);
+void multi_tensor_scale(std::vector<at::Tensor> inputs, std::vector<at::Tensor> outputs, double scale,
+                        long chunk_size);
+at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> inputs, long chunk_size);
+
+// rmsnorm.hip
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor input, at::Tensor weight, double eps, bool save_inv_rms);
+std::vector<at::Tensor> rmsnorm_fused_add_fwd(at::Tensor input, at::Tensor residual, at::Tensor weight,
+                                              double eps, bool save_inv_rms);
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor inv_rms);
+
+// rope.hip
+void rope_inplace(at::Tensor q, c10::optional<at::Tensor> k, at::Tensor table,
+                  c10::optional<at::Tensor> positions, bool backward);
+
+// swiglu.hip
+at::Tensor swiglu_fwd(at::Tensor gate_up);
+at::Tensor swiglu_bwd(at::Tensor dout, at::Tensor gate_up);
+
+// mfma_selftest.hip
+std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor A32, at::Tensor B32);
+
+// flash_attn.hip
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale);
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
+                                       at::Tensor out, at::Tensor lse, bool causal, double scale,
+                                       at::Tensor dq, at::Tensor dk, at::Tensor dv);
+
+}  // namespace cai
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("multi_tensor_adam", &cai::multi_tensor_adam, "fused multi-tensor Adam/AdamW (gfx950)");
+  m.def("multi_tensor_scale", &cai::multi_tensor_scale, "fused multi-tensor scale");
+  m.def("multi_tensor_l2norm", &cai::multi_tensor_l2norm, "fused multi-tensor L2 norm");
+  m.def("rmsnorm_fwd", &cai::rmsnorm_fwd, "RMSNorm forward");
+  m.def("rmsnorm_fused_add_fwd", &cai::rmsnorm_fused_add_fwd, "fused residual-add RMSNorm forward");
+  m.def("rmsnorm_bwd", &cai::rmsnorm_bwd, "RMSNorm backward");
+  m.def("rope_inplace", &cai::rope_inplace, "in-place rotary embedding (fwd/bwd)");
+  m.def("swiglu_fwd", &cai::swiglu_fwd, "fused SwiGLU forward");
+  m.def("swiglu_bwd", &cai::swiglu_bwd, "fused SwiGLU backward");
+  m.def("mfma_selftest", &cai::mfma_selftest, "MFMA layout self-test probes");
+  m.def("flash_attn_fwd", &cai::flash_attn_fwd, "flash attention forward (bf16, causal, GQA)");
+  m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward");
+}

@@ -1,0 +1,825 @@
+// Flash attention forward + backward for gfx950 (CDNA4), bf16, causal, GQA.
+//
+// MI355X-native design (not a CUDA port):
+//  - mfma_f32_32x32x16_bf16 tiles; 64-wide waves; 8 waves per workgroup.
+//  - forward: each wave owns 32 q-rows (block tile = 256 q-rows); K/V tiles
+//    of 64 rows are staged in LDS (K row-major XOR-swizzled, V transposed)
+//    and shared by all waves. QK^T is computed OPERAND-SWAPPED
+//    (S^T = K·Q^T) so the online-softmax row reduction is lane-local:
+//    the MFMA C-layout puts one q-row on lanes (r, r+32) — row max/sum is
+//    an in-register reduce + one shfl_xor(32) (guide §B "fused attention").
+//  - P is restaged per-wave through a 4 KB swizzled LDS tile to convert the
+//    C-layout into the A-fragment layout for the P·V MFMAs.
+//  - all LDS tiles carry the ((row&7)<<4) byte-XOR swizzle: a row-major
+//    [*][128] bf16 tile read column-wise is otherwise a 32-way bank
+//    conflict, measured as ~50% of attention kernel time (guide §6 G4).
+//  - backward is FA2-style: delta = rowsum(dO⊙O) preprocess; one kernel
+//    recomputes P^T per kv-tile and accumulates dK/dV (GQA group summed
+//    in-register); a second kernel recomputes P per q-tile for dQ.
+//
+// Shapes: q [B,S,Hq,D], k/v [B,S,Hkv,D] ("bshd" — no transposes in the
+// model), D in {64,128}. lse/delta are [B,Hq,S] fp32.
+// Reference behavioral equivalent: flash-attn 2 package as consumed by
+// colossalai/shardformer/layer/attn.py (ColoAttention).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace cai {
+
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+
+namespace fa {
+
+constexpr int KVB = 64;     // kv rows per tile
+constexpr int QW = 32;      // q rows per wave (fwd / dq); kv rows per wave (dkdv)
+constexpr int NW = 8;       // waves per block (forward)
+constexpr int NT = 512;     // threads per block (forward)
+// Backward kernels hold dK+dV (or dQ) accumulators plus operand fragments in
+// registers (~250 VGPR): 8-wave blocks would exceed the 2048-VGPR/CU pool,
+// so backward runs 4-wave blocks (128 kv/q rows per block).
+constexpr int NWB = 4;
+constexpr int NTB = 256;
+
+DEV_INLINE bf16x8_t ld_g16(const unsigned short* p) {
+  return *reinterpret_cast<const bf16x8_t*>(p);
+}
+
+DEV_INLINE void st_lds16(char* lds, int byte, bf16x8_t v) {
+  *reinterpret_cast<bf16x8_t*>(lds + byte) = v;
+}
+
+DEV_INLINE bf16x8_t ld_lds16(const char* lds, int byte) {
+  return *reinterpret_cast<const bf16x8_t*>(lds + byte);
+}
+
+DEV_INLINE unsigned short bf_raw(__bf16 b) {
+  union { __bf16 b; unsigned short u; } c;
+  c.b = b;
+  return c.u;
+}
+
+DEV_INLINE __bf16 f2b(float f) {
+  union { unsigned short u; __bf16 b; } c;
+  c.u = f2bf(f);
+  return c.b;
+}
+
+// C-layout row for mfma_f32_32x32x16 register j (guide §3, HW-verified).
+DEV_INLINE constexpr int crow(int j, int half) { return (j & 3) + 8 * (j >> 2) + 4 * half; }
+
+// Swizzled byte offset inside a row-major LDS tile with ROWB bytes per row.
+// XOR spreads a column access over 8 distinct 16-byte slots (guide §6 G4).
+template <int ROWB>
+DEV_INLINE int swz(int row, int byte_in_row) {
+  return row * ROWB + (byte_in_row ^ ((row & 7) << 4));
+}
+
+// ---- cooperative staging helpers (512 threads) ----------------------------
+
+// Row-major [ROWS][D] bf16 tile, swizzled. Source rows clamped to < S.
+template <int D, int ROWS>
+DEV_INLINE void stage_rowmajor(char* lds, const unsigned short* base, long tok_stride, int row0, int S) {
+  constexpr int CPR = D / 8;  // 16B chunks per row
+  constexpr int TOTAL = ROWS * CPR;
+  for (int c = threadIdx.x; c < TOTAL; c += blockDim.x) {
+    const int row = c / CPR, ch = c % CPR;
+    const int grow = min(row0 + row, S - 1);
+    bf16x8_t v = ld_g16(base + (long)grow * tok_stride + ch * 8);
+    st_lds16(lds, swz<D * 2>(row, ch * 16), v);
+  }
+}
+
+// Transposed [D][ROWS] bf16 tile, swizzled: ldsT[d][row] = src[row][d].
+template <int D, int ROWS>
+DEV_INLINE void stage_transposed(char* ldsT, const unsigned short* base, long tok_stride, int row0, int S) {
+  constexpr int CPR = D / 8;
+  constexpr int TOTAL = ROWS * CPR;
+  for (int c = threadIdx.x; c < TOTAL; c += blockDim.x) {
+    const int row = c / CPR, ch = c % CPR;
+    const int grow = min(row0 + row, S - 1);
+    bf16x8_t v = ld_g16(base + (long)grow * tok_stride + ch * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int d = ch * 8 + e;
+      *reinterpret_cast<unsigned short*>(ldsT + swz<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
+    }
+  }
+}
+
+// Both layouts from one global read (used for Q/dO in the dkdv kernel).
+template <int D, int ROWS>
+DEV_INLINE void stage_both(char* lds, char* ldsT, const unsigned short* base, long tok_stride, int row0, int S) {
+  constexpr int CPR = D / 8;
+  constexpr int TOTAL = ROWS * CPR;
+  for (int c = threadIdx.x; c < TOTAL; c += blockDim.x) {
+    const int row = c / CPR, ch = c % CPR;
+    const int grow = min(row0 + row, S - 1);
+    bf16x8_t v = ld_g16(base + (long)grow * tok_stride + ch * 8);
+    st_lds16(lds, swz<D * 2>(row, ch * 16), v);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int d = ch * 8 + e;
+      *reinterpret_cast<unsigned short*>(ldsT + swz<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
+    }
+  }
+}
+
+}  // namespace fa
+
+// ============================================================ forward kernel
+//
+// grid: (ceil(S/256), B*Hq). LDS: K[KVB][D] + V_T[D][KVB] + P[NW][32][KVB].
+template <int D>
+__global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
+    const unsigned short* __restrict__ Q,
+    const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V,
+    unsigned short* __restrict__ O,
+    float* __restrict__ LSE,
+    long qbs, long qts, long kbs, long kts,
+    int B, int S, int Hq, int Hkv, float scale, int causal) {
+  using namespace fa;
+  extern __shared__ char smem[];
+  constexpr int KB_BYTES = KVB * D * 2;
+  constexpr int VT_BYTES = D * KVB * 2;
+  char* Klds = smem;
+  char* VT = smem + KB_BYTES;
+  char* Pw = smem + KB_BYTES + VT_BYTES;  // + wave*QW*KVB*2
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int half = lane >> 5;
+  const int ln = lane & 31;
+
+  const int bh = blockIdx.y;
+  const int b = bh / Hq, h = bh % Hq;
+  const int hk = h / (Hq / Hkv);
+  const int q0 = blockIdx.x * (NW * QW);
+  const int qw = q0 + w * QW;  // this wave's first q row
+
+  const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
+  const unsigned short* k_base = K + (long)b * kbs + (long)hk * D;
+  const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
+  const long o_stride = (long)Hq * D;  // O/LSE are always packed
+
+  constexpr int DSL = D / 16;  // MFMA K-slices over the head dim
+  char* P = Pw + w * (QW * KVB * 2);
+
+  // ---- Q fragments (B-operand of S^T = K·Q^T): lane holds q-col (ln),
+  //      d-rows half*8+[0..7] per 16-d slice — one 16 B load per slice.
+  const int q_my = min(qw + ln, S - 1);
+  bf16x8_t qf[DSL];
+  {
+    const unsigned short* qp = q_base + (long)q_my * qts + half * 8;
+#pragma unroll
+    for (int sl = 0; sl < DSL; ++sl) qf[sl] = ld_g16(qp + sl * 16);
+  }
+
+  f32x16 oacc[D / 32];
+#pragma unroll
+  for (int nb = 0; nb < D / 32; ++nb)
+#pragma unroll
+    for (int j = 0; j < 16; ++j) oacc[nb][j] = 0.0f;
+  float m_run = -INFINITY, l_run = 0.0f;
+
+  const int kv_tiles_all = (S + KVB - 1) / KVB;
+  const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NW * QW + KVB - 1) / KVB) : kv_tiles_all;
+
+  for (int t = 0; t < kv_tiles; ++t) {
+    const int k0 = t * KVB;
+    __syncthreads();
+    stage_rowmajor<D, KVB>(Klds, k_base, kts, k0, S);
+    stage_transposed<D, KVB>(VT, v_base, kts, k0, S);
+    __syncthreads();
+
+    // waves entirely above the diagonal produce nothing
+    if (causal && k0 > qw + QW - 1) continue;
+    if (qw >= S) continue;
+
+    // ---- S^T = K · Q^T : two 32-k blocks. C: col = q (ln), row = k (crow).
+    float p[2][16];
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb) {
+      f32x16 acc;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+#pragma unroll
+      for (int sl = 0; sl < DSL; ++sl) {
+        // A-operand: K rows. lane: row = kb*32+ln, cols d = sl*16 + half*8 + [0..7]
+        bf16x8_t kf = ld_lds16(Klds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[sl], acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int j = 0; j < 16; ++j) p[kb][j] = acc[j];
+    }
+
+    // ---- online softmax over this tile's 64 scores of q-row (qw+ln)
+    const int q_abs = qw + ln;
+    float rowmax = -INFINITY;
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int k_abs = k0 + kb * 32 + crow(j, half);
+        const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs);
+        p[kb][j] = valid ? p[kb][j] * scale : -INFINITY;
+        rowmax = fmaxf(rowmax, p[kb][j]);
+      }
+    rowmax = fmaxf(rowmax, __shfl_xor(rowmax, 32));
+    const float m_new = fmaxf(m_run, rowmax);
+    const float msafe = (m_new == -INFINITY) ? 0.0f : m_new;
+    const float corr = (m_run == -INFINITY) ? ((m_new == -INFINITY) ? 1.0f : 0.0f) : __expf(m_run - m_new);
+    float rowsum = 0.0f;
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        p[kb][j] = __expf(p[kb][j] - msafe);
+        rowsum += p[kb][j];
+      }
+    rowsum += __shfl_xor(rowsum, 32);
+    m_run = m_new;
+    l_run = l_run * corr + rowsum;
+
+    // ---- stage P (bf16) into swizzled per-wave LDS: P[q=ln][k]
+    // C-regs j=0..3 within a group are 4 consecutive k values → 8 B packed.
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        const int kcol = kb * 32 + 8 * jj + 4 * half;
+        short4v pk;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) pk[e] = (short)f2bf(p[kb][jj * 4 + e]);
+        *reinterpret_cast<short4v*>(P + swz<KVB * 2>(ln, kcol * 2)) = pk;
+      }
+
+    // ---- rescale O accumulator by corr (per q-row, via shfl broadcast)
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int r = crow(j, half);
+      const float c = __shfl(corr, r);  // corr for q-row r lives on lanes r, r+32
+#pragma unroll
+      for (int nb = 0; nb < D / 32; ++nb) oacc[nb][j] *= c;
+    }
+
+    // ---- O += P · V  (A = P from LDS, B = V_T reads)
+#pragma unroll
+    for (int ks = 0; ks < KVB / 16; ++ks) {
+      // A: lane row = q (ln), k-cols = ks*16 + half*8 + [0..7]
+      bf16x8_t pa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
+#pragma unroll
+      for (int nb = 0; nb < D / 32; ++nb) {
+        // B: lane col = d (nb*32+ln), k-rows = ks*16 + half*8 + [0..7] → V_T[d][k]
+        bf16x8_t vb = ld_lds16(VT, swz<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        oacc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, oacc[nb], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: O /= l, store O and LSE
+  if (qw >= S) return;
+  unsigned short* o_base = O + ((long)b * S * Hq + h) * D;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const int r = crow(j, half);
+    const float l_r = __shfl(l_run, r);
+    const float inv_l = (l_r > 0.0f) ? 1.0f / l_r : 0.0f;
+    const int q_abs = qw + r;
+    if (q_abs < S) {
+#pragma unroll
+      for (int nb = 0; nb < D / 32; ++nb) {
+        o_base[(long)q_abs * o_stride + nb * 32 + ln] = f2bf(oacc[nb][j] * inv_l);
+      }
+    }
+  }
+  if (lane < QW && qw + lane < S) {
+    const float lse = (l_run > 0.0f) ? m_run + __logf(l_run) : -INFINITY;
+    LSE[((long)b * Hq + h) * S + qw + lane] = lse;
+  }
+}
+
+// ==================================================== delta = rowsum(dO ⊙ O)
+__global__ __launch_bounds__(256) void fa_delta_kernel(
+    const unsigned short* __restrict__ dO,
+    const unsigned short* __restrict__ Oin,
+    float* __restrict__ delta,  // [B, H, S]
+    long rows,  // B*S*H
+    int S, int H, int D) {
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int sub = lane >> 4;       // 4 rows per wave
+  const int e16 = lane & 15;       // 16 lanes per row
+  const int epl = D / 16;          // elems per lane (8 for D=128)
+  const long waves_total = (long)gridDim.x * 4;  // 4 waves per block
+  for (long r4 = blockIdx.x * 4L + w; r4 * 4 < rows; r4 += waves_total) {
+    const long row = r4 * 4 + sub;
+    if (row >= rows) continue;
+    float acc = 0.0f;
+    const unsigned short* dop = dO + row * D + e16 * epl;
+    const unsigned short* op = Oin + row * D + e16 * epl;
+    for (int k = 0; k < epl; k += 8) {
+      short8 a = *reinterpret_cast<const short8*>(dop + k);
+      short8 b = *reinterpret_cast<const short8*>(op + k);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += bf2f((unsigned short)a[j]) * bf2f((unsigned short)b[j]);
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
+    if (e16 == 0) {
+      const long bidx = row / ((long)S * H);
+      const long rem = row % ((long)S * H);
+      const long s = rem / H;
+      const long hh = rem % H;
+      delta[(bidx * H + hh) * S + s] = acc;
+    }
+  }
+}
+
+// ================================================= backward dK/dV kernel
+//
+// grid: (ceil(S/256) kv-tiles, B*Hkv). Each wave owns 32 kv rows; loops over
+// the GQA group's q-heads and that head's q-tiles (64 q rows staged in LDS
+// in both layouts). dK/dV accumulate in registers across the whole loop.
+template <int D>
+__global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
+    const unsigned short* __restrict__ Q,
+    const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V,
+    const unsigned short* __restrict__ dO,
+    const float* __restrict__ LSE,
+    const float* __restrict__ DELTA,
+    unsigned short* __restrict__ dK,
+    unsigned short* __restrict__ dV,
+    long qbs, long qts, long kbs, long kts, long dkbs, long dkts,
+    int B, int S, int Hq, int Hkv, float scale, int causal) {
+  using namespace fa;
+  constexpr int QB = 64;  // q rows per staged tile
+  extern __shared__ char smem[];
+  constexpr int TB = QB * D * 2;  // 16 KB (D=128)
+  char* Qlds = smem;
+  char* QT = smem + TB;
+  char* dOlds = smem + 2 * TB;
+  char* dOT = smem + 3 * TB;
+  char* Pw = smem + 4 * TB;                      // NWB * 32*QB*2
+  float* lse_lds = reinterpret_cast<float*>(smem + 4 * TB + NWB * (QW * QB * 2));
+  float* dta_lds = lse_lds + QB;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int half = lane >> 5;
+  const int ln = lane & 31;
+
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv, hk = bh % Hkv;
+  const int G = Hq / Hkv;
+  const int kv0 = blockIdx.x * (NWB * QW);
+  const int kvw = kv0 + w * QW;  // this wave's first kv row
+
+  const unsigned short* k_base = K + (long)b * kbs + (long)hk * D;
+  const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
+
+  constexpr int DSL = D / 16;
+  char* P = Pw + w * (QW * QB * 2);
+
+  // ---- K, V fragments (A-operands): lane row = kv (ln), d = sl*16+half*8+[0..7]
+  const int kv_my = min(kvw + ln, S - 1);
+  bf16x8_t kf[DSL], vf[DSL];
+  {
+    const unsigned short* kp = k_base + (long)kv_my * kts + half * 8;
+    const unsigned short* vp = v_base + (long)kv_my * kts + half * 8;
+#pragma unroll
+    for (int sl = 0; sl < DSL; ++sl) {
+      kf[sl] = ld_g16(kp + sl * 16);
+      vf[sl] = ld_g16(vp + sl * 16);
+    }
+  }
+
+  f32x16 dk_acc[D / 32], dv_acc[D / 32];
+#pragma unroll
+  for (int nb = 0; nb < D / 32; ++nb)
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      dk_acc[nb][j] = 0.0f;
+      dv_acc[nb][j] = 0.0f;
+    }
+
+  const int qt_first = causal ? (kv0 / QB) : 0;
+  const int qt_last = (S + QB - 1) / QB;
+
+  for (int g = 0; g < G; ++g) {
+    const int h = hk * G + g;
+    const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
+    const unsigned short* do_base = dO + ((long)b * S * Hq + h) * D;  // dO packed
+    const long do_stride = (long)Hq * D;
+    const float* lse_base = LSE + ((long)b * Hq + h) * S;
+    const float* dta_base = DELTA + ((long)b * Hq + h) * S;
+
+    for (int qt = qt_first; qt < qt_last; ++qt) {
+      const int qt0 = qt * QB;
+      __syncthreads();
+      stage_both<D, QB>(Qlds, QT, q_base, qts, qt0, S);
+      stage_both<D, QB>(dOlds, dOT, do_base, do_stride, qt0, S);
+      if (threadIdx.x < QB) {
+        const int qr = min(qt0 + (int)threadIdx.x, S - 1);
+        lse_lds[threadIdx.x] = lse_base[qr];
+        dta_lds[threadIdx.x] = dta_base[qr];
+      }
+      __syncthreads();
+
+      if (causal && qt0 + QB - 1 < kvw) continue;  // wave fully above diagonal
+      if (kvw >= S) continue;
+
+      // ---- S'^T[kv][q] = K · Q^T : C col = q (ln within qb), row = kv.
+      float p[2][16], dp[2][16];
+#pragma unroll
+      for (int qb = 0; qb < 2; ++qb) {
+        f32x16 acc;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+#pragma unroll
+        for (int sl = 0; sl < DSL; ++sl) {
+          // B: lane col = q (qb*32+ln), rows d → Qlds[q][d]
+          bf16x8_t qb_frag = ld_lds16(Qlds, swz<D * 2>(qb * 32 + ln, (sl * 16 + half * 8) * 2));
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[sl], qb_frag, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int j = 0; j < 16; ++j) p[qb][j] = acc[j];
+      }
+
+      // ---- P' = exp(scale*s - lse[q]), masked
+#pragma unroll
+      for (int qb = 0; qb < 2; ++qb) {
+        const int q_abs = qt0 + qb * 32 + ln;
+        const float lse = lse_lds[qb * 32 + ln];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int k_abs = kvw + crow(j, half);
+          const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+          p[qb][j] = valid ? __expf(p[qb][j] * scale - lse) : 0.0f;
+        }
+      }
+
+      // ---- stage P' to LDS: P[kv local][q] bf16 swizzled (scalar stores)
+#pragma unroll
+      for (int qb = 0; qb < 2; ++qb)
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int r = crow(j, half);
+          *reinterpret_cast<unsigned short*>(P + swz<QB * 2>(r, (qb * 32 + ln) * 2)) = f2bf(p[qb][j]);
+        }
+
+      // ---- dV += P' · dO   (A = P' from LDS, B = dO_T reads)
+#pragma unroll
+      for (int ks = 0; ks < QB / 16; ++ks) {
+        bf16x8_t pa = ld_lds16(P, swz<QB * 2>(ln, (ks * 16 + half * 8) * 2));
+#pragma unroll
+        for (int nb = 0; nb < D / 32; ++nb) {
+          bf16x8_t dob = ld_lds16(dOT, swz<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+          dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[nb], 0, 0, 0);
+        }
+      }
+
+      // ---- dP'[kv][q] = V · dO^T  (A = V regs, B = dOlds[q][d] reads)
+#pragma unroll
+      for (int qb = 0; qb < 2; ++qb) {
+        f32x16 acc;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+#pragma unroll
+        for (int sl = 0; sl < DSL; ++sl) {
+          bf16x8_t dob = ld_lds16(dOlds, swz<D * 2>(qb * 32 + ln, (sl * 16 + half * 8) * 2));
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[sl], dob, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int j = 0; j < 16; ++j) dp[qb][j] = acc[j];
+      }
+
+      // ---- dS' = scale * P' ⊙ (dP' - delta[q]) → overwrite P LDS
+#pragma unroll
+      for (int qb = 0; qb < 2; ++qb) {
+        const float dta = dta_lds[qb * 32 + ln];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int r = crow(j, half);
+          const float ds = scale * p[qb][j] * (dp[qb][j] - dta);
+          *reinterpret_cast<unsigned short*>(P + swz<QB * 2>(r, (qb * 32 + ln) * 2)) = f2bf(ds);
+        }
+      }
+
+      // ---- dK += dS' · Q   (A = dS' from LDS, B = Q_T reads)
+#pragma unroll
+      for (int ks = 0; ks < QB / 16; ++ks) {
+        bf16x8_t dsa = ld_lds16(P, swz<QB * 2>(ln, (ks * 16 + half * 8) * 2));
+#pragma unroll
+        for (int nb = 0; nb < D / 32; ++nb) {
+          bf16x8_t qtb = ld_lds16(QT, swz<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+          dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[nb], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // ---- epilogue: write dK, dV (each kv row owned by exactly one block)
+  if (kvw >= S) return;
+  unsigned short* dk_base = dK + (long)b * dkbs + (long)hk * D;
+  unsigned short* dv_base = dV + (long)b * dkbs + (long)hk * D;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const int r = crow(j, half);
+    const int kv_abs = kvw + r;
+    if (kv_abs < S) {
+#pragma unroll
+      for (int nb = 0; nb < D / 32; ++nb) {
+        dk_base[(long)kv_abs * dkts + nb * 32 + ln] = f2bf(dk_acc[nb][j]);
+        dv_base[(long)kv_abs * dkts + nb * 32 + ln] = f2bf(dv_acc[nb][j]);
+      }
+    }
+  }
+}
+
+// ===================================================== backward dQ kernel
+//
+// grid: (ceil(S/256) q-tiles, B*Hq). Each wave owns 32 q rows; kv tiles of
+// 64 staged in LDS (K both layouts, V row-major).
+template <int D>
+__global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
+    const unsigned short* __restrict__ Q,
+    const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V,
+    const unsigned short* __restrict__ dO,
+    const float* __restrict__ LSE,
+    const float* __restrict__ DELTA,
+    unsigned short* __restrict__ dQ,
+    long qbs, long qts, long kbs, long kts, long dqbs, long dqts,
+    int B, int S, int Hq, int Hkv, float scale, int causal) {
+  using namespace fa;
+  extern __shared__ char smem[];
+  constexpr int KB_BYTES = KVB * D * 2;
+  constexpr int KT_BYTES = D * KVB * 2;
+  char* Klds = smem;
+  char* KT = smem + KB_BYTES;
+  char* Vlds = smem + KB_BYTES + KT_BYTES;
+  char* Pw = smem + 2 * KB_BYTES + KT_BYTES;  // NWB * 32*KVB*2
+  float* lse_lds = reinterpret_cast<float*>(Pw + NWB * (QW * KVB * 2));
+  float* dta_lds = lse_lds + NW * QW;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int half = lane >> 5;
+  const int ln = lane & 31;
+
+  const int bh = blockIdx.y;
+  const int b = bh / Hq, h = bh % Hq;
+  const int hk = h / (Hq / Hkv);
+  const int q0 = blockIdx.x * (NWB * QW);
+  const int qw = q0 + w * QW;
+
+  const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
+  const unsigned short* do_base = dO + ((long)b * S * Hq + h) * D;  // dO packed
+  const long do_stride = (long)Hq * D;
+  const unsigned short* k_base = K + (long)b * kbs + (long)hk * D;
+  const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
+
+  constexpr int DSL = D / 16;
+  char* P = Pw + w * (QW * KVB * 2);
+
+  // lse/delta for the block's q rows
+  {
+    const float* lse_base = LSE + ((long)b * Hq + h) * S;
+    const float* dta_base = DELTA + ((long)b * Hq + h) * S;
+    for (int i = threadIdx.x; i < NWB * QW; i += NTB) {
+      const int qr = min(q0 + i, S - 1);
+      lse_lds[i] = lse_base[qr];
+      dta_lds[i] = dta_base[qr];
+    }
+  }
+
+  // ---- Q, dO fragments (A-operands): lane row = q (ln), d cols
+  const int q_my = min(qw + ln, S - 1);
+  bf16x8_t qa[DSL], doa[DSL];
+  {
+    const unsigned short* qp = q_base + (long)q_my * qts + half * 8;
+    const unsigned short* dop = do_base + (long)q_my * do_stride + half * 8;
+#pragma unroll
+    for (int sl = 0; sl < DSL; ++sl) {
+      qa[sl] = ld_g16(qp + sl * 16);
+      doa[sl] = ld_g16(dop + sl * 16);
+    }
+  }
+
+  f32x16 dq_acc[D / 32];
+#pragma unroll
+  for (int nb = 0; nb < D / 32; ++nb)
+#pragma unroll
+    for (int j = 0; j < 16; ++j) dq_acc[nb][j] = 0.0f;
+
+  const int kv_tiles_all = (S + KVB - 1) / KVB;
+  const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NWB * QW + KVB - 1) / KVB) : kv_tiles_all;
+
+  for (int t = 0; t < kv_tiles; ++t) {
+    const int k0 = t * KVB;
+    __syncthreads();
+    stage_both<D, KVB>(Klds, KT, k_base, kts, k0, S);
+    stage_rowmajor<D, KVB>(Vlds, v_base, kts, k0, S);
+    __syncthreads();
+
+    if (causal && k0 > qw + QW - 1) continue;
+    if (qw >= S) continue;
+
+    // ---- S[q][kv] = Q · K^T : C col = kv (ln within kb), row = q (crow).
+    float p[2][16], dp[2][16];
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb) {
+      f32x16 acc, acc2;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) { acc[j] = 0.0f; acc2[j] = 0.0f; }
+#pragma unroll
+      for (int sl = 0; sl < DSL; ++sl) {
+        // B = K^T: lane col = kv (kb*32+ln), rows d → Klds[kv][d]
+        bf16x8_t kb_frag = ld_lds16(Klds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa[sl], kb_frag, acc, 0, 0, 0);
+        // B = V^T: same pattern from Vlds → dP = dO · V^T
+        bf16x8_t vb_frag = ld_lds16(Vlds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa[sl], vb_frag, acc2, 0, 0, 0);
+      }
+#pragma unroll
+      for (int j = 0; j < 16; ++j) { p[kb][j] = acc[j]; dp[kb][j] = acc2[j]; }
+    }
+
+    // ---- dS = scale * P ⊙ (dP - delta[q]) with P = exp(scale*s - lse[q])
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int r = crow(j, half);
+        const int q_abs = qw + r;
+        const int k_abs = k0 + kb * 32 + ln;
+        const float lse = lse_lds[w * QW + r];
+        const float dta = dta_lds[w * QW + r];
+        const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+        const float pv = valid ? __expf(p[kb][j] * scale - lse) : 0.0f;
+        const float ds = scale * pv * (dp[kb][j] - dta);
+        // stage dS to P LDS: [q local][kv] bf16 swizzled
+        *reinterpret_cast<unsigned short*>(P + swz<KVB * 2>(r, (kb * 32 + ln) * 2)) = f2bf(ds);
+      }
+
+    // ---- dQ += dS · K   (A = dS from LDS, B = K_T reads)
+#pragma unroll
+    for (int ks = 0; ks < KVB / 16; ++ks) {
+      bf16x8_t dsa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
+#pragma unroll
+      for (int nb = 0; nb < D / 32; ++nb) {
+        bf16x8_t ktb = ld_lds16(KT, swz<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, ktb, dq_acc[nb], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: write dQ
+  if (qw >= S) return;
+  unsigned short* dq_base = dQ + (long)b * dqbs + (long)h * D;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const int r = crow(j, half);
+    const int q_abs = qw + r;
+    if (q_abs < S) {
+#pragma unroll
+      for (int nb = 0; nb < D / 32; ++nb) {
+        dq_base[(long)q_abs * dqts + nb * 32 + ln] = f2bf(dq_acc[nb][j]);
+      }
+    }
+  }
+}
+
+// =============================================================== host side
+
+namespace {
+
+void set_lds_limit(const void* kernel, size_t bytes) {
+  if (bytes > 65536) {
+    (void)hipFuncSetAttribute(kernel, hipFuncAttributeMaxDynamicSharedMemorySize, (int)bytes);
+  }
+}
+
+// q/k/v are [B,S,H,D] views that may be strided in dims 0/1 (e.g. views into
+// a packed QKV GEMM output); dims 2/3 must be dense.
+void check_fa_view(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.dim() == 4, "flash_attn: ", name, " must be [B,S,H,D]");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, "flash_attn: bf16 only");
+  TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == t.size(3), "flash_attn: ", name,
+              " must be dense in [H,D] (strided B/S allowed)");
+}
+
+void check_fa_inputs(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v) {
+  check_fa_view(q, "q");
+  check_fa_view(k, "k");
+  check_fa_view(v, "v");
+  const int D = (int)q.size(3);
+  TORCH_CHECK(D == 64 || D == 128, "flash_attn: head dim must be 64 or 128");
+  TORCH_CHECK(k.size(3) == D && v.size(3) == D);
+  TORCH_CHECK(k.sizes() == v.sizes() && k.strides() == v.strides(), "flash_attn: k/v must match");
+  TORCH_CHECK((int)q.size(2) % (int)k.size(2) == 0, "flash_attn: Hq must be a multiple of Hkv (GQA)");
+}
+
+}  // namespace
+
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale) {
+  using namespace fa;
+  check_fa_inputs(q, k, v);
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2), D = (int)q.size(3);
+  const int Hkv = (int)k.size(2);
+  auto o = at::empty({B, S, Hq, D}, q.options());
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  const dim3 grid((S + NW * QW - 1) / (NW * QW), B * Hq);
+  const long qbs = q.stride(0), qts = q.stride(1), kbs = k.stride(0), kts = k.stride(1);
+
+#define LAUNCH_FWD(DD)                                                                            \
+  do {                                                                                            \
+    const size_t lds = KVB * DD * 2 + DD * KVB * 2 + NW * (QW * KVB * 2);                         \
+    set_lds_limit((const void*)fa_fwd_kernel<DD>, lds);                                           \
+    hipLaunchKernelGGL((fa_fwd_kernel<DD>), grid, dim3(NT), lds, stream.stream(),                 \
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
+                       (const unsigned short*)v.data_ptr(), (unsigned short*)o.data_ptr(),        \
+                       lse.data_ptr<float>(), qbs, qts, kbs, kts,                                 \
+                       B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                              \
+  } while (0)
+
+  if (D == 128) LAUNCH_FWD(128);
+  else LAUNCH_FWD(64);
+#undef LAUNCH_FWD
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
+
+// dq/dk/dv may be caller-provided strided views (e.g. into a packed dQKV
+// buffer) to avoid post-hoc gathers; pass empty tensors to allocate.
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
+                                       at::Tensor out, at::Tensor lse, bool causal, double scale,
+                                       at::Tensor dq, at::Tensor dk, at::Tensor dv) {
+  using namespace fa;
+  check_fa_inputs(q, k, v);
+  TORCH_CHECK(dout.is_contiguous() && out.is_contiguous() && lse.is_contiguous(),
+              "flash_attn_bwd: dout/out/lse must be contiguous");
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2), D = (int)q.size(3);
+  const int Hkv = (int)k.size(2);
+  if (dq.numel() == 0) dq = at::empty({B, S, Hq, D}, q.options());
+  if (dk.numel() == 0) dk = at::empty({B, S, Hkv, D}, k.options());
+  if (dv.numel() == 0) dv = at::empty({B, S, Hkv, D}, v.options());
+  check_fa_view(dq, "dq");
+  check_fa_view(dk, "dk");
+  TORCH_CHECK(dk.strides() == dv.strides(), "flash_attn_bwd: dk/dv strides must match");
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+
+  {  // delta preprocess
+    const long rows = (long)B * S * Hq;
+    const int grid = capped_grid((rows + 15) / 16, 4096);
+    hipLaunchKernelGGL(fa_delta_kernel, dim3(grid), dim3(256), 0, stream.stream(),
+                       (const unsigned short*)dout.data_ptr(), (const unsigned short*)out.data_ptr(),
+                       delta.data_ptr<float>(), rows, S, Hq, D);
+    HIP_CHECK_LAST();
+  }
+
+  const dim3 grid_kv((S + NWB * QW - 1) / (NWB * QW), B * Hkv);
+  const dim3 grid_q((S + NWB * QW - 1) / (NWB * QW), B * Hq);
+  const long qbs = q.stride(0), qts = q.stride(1), kbs = k.stride(0), kts = k.stride(1);
+  const long dqbs = dq.stride(0), dqts = dq.stride(1), dkbs = dk.stride(0), dkts = dk.stride(1);
+
+#define LAUNCH_BWD(DD)                                                                              \
+  do {                                                                                              \
+    constexpr int QB = 64;                                                                          \
+    const size_t lds_kv = 4 * (QB * DD * 2) + NWB * (QW * QB * 2) + 2 * QB * sizeof(float);         \
+    set_lds_limit((const void*)fa_bwd_dkdv_kernel<DD>, lds_kv);                                     \
+    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD>), grid_kv, dim3(NTB), lds_kv, stream.stream(),       \
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
+                       (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
+                       (unsigned short*)dk.data_ptr(), (unsigned short*)dv.data_ptr(),              \
+                       qbs, qts, kbs, kts, dkbs, dkts,                                              \
+                       B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
+    HIP_CHECK_LAST();                                                                               \
+    const size_t lds_q = 2 * (KVB * DD * 2) + DD * KVB * 2 + NWB * (QW * KVB * 2)                   \
+                         + 2 * NWB * QW * sizeof(float);                                            \
+    set_lds_limit((const void*)fa_bwd_dq_kernel<DD>, lds_q);                                        \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), grid_q, dim3(NTB), lds_q, stream.stream(),           \
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
+                       (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
+                       (unsigned short*)dq.data_ptr(),                                              \
+                       qbs, qts, kbs, kts, dqbs, dqts,                                              \
+                       B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
+    HIP_CHECK_LAST();                                                                               \
+  } while (0)
+
+  if (D == 128) LAUNCH_BWD(128);
+  else LAUNCH_BWD(64);
+#undef LAUNCH_BWD
+  return {dq, dk, dv};
+}
+
+}  // namespace cai

@@ -1,0 +1,217 @@
+"""MI355X-native Llama implementation.
+
+Design differences vs the HF module graph (all MI355X-motivated):
+- packed QKV GEMM ([B,S,(Hq+2Hkv)·D] in one hipBLASLt call) feeding
+  ``fused_rope_attention`` (RoPE in-place + flash attention on strided
+  views — zero layout copies; attention stays bshd end-to-end).
+- packed gate+up GEMM feeding the fused SwiGLU kernel.
+- every residual add is fused into the next RMSNorm
+  (``fused_add_rms_norm``) — one HBM pass instead of two per junction.
+- norms/softmax accumulate fp32 inside HIP kernels; the model itself runs
+  in bf16 without autocast.
+
+State-dict keys match HF (`model.layers.N.self_attn.q_proj.weight`, ...)
+via parameter aliasing of the packed weights, so HF checkpoints load/save
+through checkpoint_io. Reference model parity target:
+transformers LlamaForCausalLM as sharded by
+colossalai/shardformer/policies/llama.py.
+"""
+
+import math
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops import build_rope_table, fused_add_rms_norm, fused_rope_attention, rms_norm, swiglu
+
+__all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_token"]
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 32000
+    hidden_size: int = 4096
+    intermediate_size: int = 11008
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: Optional[int] = None
+    max_position_embeddings: int = 4096
+    rms_norm_eps: float = 1e-5
+    rope_theta: float = 10000.0
+    tie_word_embeddings: bool = False
+    initializer_range: float = 0.02
+    gradient_checkpointing: bool = False
+
+    def __post_init__(self):
+        if self.num_key_value_heads is None:
+            self.num_key_value_heads = self.num_attention_heads
+
+    @property
+    def head_dim(self) -> int:
+        return self.hidden_size // self.num_attention_heads
+
+
+# model table mirrors the reference benchmark configs
+# (/root/reference/examples/language/llama/benchmark.py:33-59)
+LLAMA_CONFIGS = {
+    "llama-100m": LlamaConfig(hidden_size=768, intermediate_size=2048, num_hidden_layers=12,
+                              num_attention_heads=12, num_key_value_heads=12, max_position_embeddings=4096),
+    "llama-7b": LlamaConfig(hidden_size=4096, intermediate_size=11008, num_hidden_layers=32,
+                            num_attention_heads=32, num_key_value_heads=32, max_position_embeddings=4096),
+    "llama-13b": LlamaConfig(hidden_size=5120, intermediate_size=13824, num_hidden_layers=40,
+                             num_attention_heads=40, num_key_value_heads=40, max_position_embeddings=4096),
+    "llama3-8b": LlamaConfig(vocab_size=128256, hidden_size=4096, intermediate_size=14336,
+                             num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
+                             max_position_embeddings=8192, rope_theta=500000.0),
+    "llama-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
+                             num_attention_heads=64, num_key_value_heads=8, max_position_embeddings=4096),
+}
+
+
+def llama_flops_per_token(cfg: LlamaConfig, seq_len: int, grad_ckpt: bool = False) -> float:
+    """Model FLOPs per token for one fwd+bwd step — the reference's formula
+    (examples/language/performance_evaluator.py:164-165):
+    flop = numel * 2 * (3 + grad_ckpt) + attention term."""
+    numel = sum(
+        [
+            cfg.vocab_size * cfg.hidden_size * (1 if cfg.tie_word_embeddings else 2),
+            cfg.num_hidden_layers
+            * (
+                cfg.hidden_size * (cfg.num_attention_heads + 2 * cfg.num_key_value_heads) * cfg.head_dim
+                + cfg.hidden_size * cfg.hidden_size
+                + 3 * cfg.hidden_size * cfg.intermediate_size
+            ),
+        ]
+    )
+    dense = 2 * numel * (4 if grad_ckpt else 3)
+    # causal attention: 2 matmuls * 2 flop * S/2 (causal) per layer, fwd+bwd(2x [2.5 in practice])
+    attn = 2 * 2 * (seq_len / 2) * cfg.hidden_size * cfg.num_hidden_layers * (4 if grad_ckpt else 3)
+    return dense + attn
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        D, Hq, Hkv = cfg.head_dim, cfg.num_attention_heads, cfg.num_key_value_heads
+        self.qkv_proj = nn.Linear(cfg.hidden_size, (Hq + 2 * Hkv) * D, bias=False)
+        self.o_proj = nn.Linear(Hq * D, cfg.hidden_size, bias=False)
+        self.scale = 1.0 / math.sqrt(D)
+
+    def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        B, S, _ = hidden.shape
+        qkv = self.qkv_proj(hidden)
+        attn = fused_rope_attention(
+            qkv, rope_table, cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim,
+            causal=True, scale=self.scale,
+        )
+        return self.o_proj(attn.reshape(B, S, -1))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * cfg.intermediate_size, bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+
+    def forward(self, hidden: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(swiglu(self.gate_up_proj(hidden)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.self_attn = LlamaAttention(cfg)
+        self.mlp = LlamaMLP(cfg)
+        self.input_layernorm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.post_attention_layernorm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.eps = cfg.rms_norm_eps
+
+    def forward(self, hidden: torch.Tensor, residual: Optional[torch.Tensor], rope_table: torch.Tensor):
+        """hidden = normed input to attention; residual = running stream.
+
+        Returns (mlp_out, residual') where residual' = residual + attn_out:
+        the NEXT junction (residual'' = residual' + mlp_out, then norm) is
+        fused by the caller so norm weights stay owned by their layer.
+        """
+        attn_out = self.self_attn(hidden, rope_table)
+        hidden, residual = fused_add_rms_norm(attn_out, residual, self.post_attention_layernorm_weight, self.eps)
+        mlp_out = self.mlp(hidden)
+        return mlp_out, residual
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(LlamaDecoderLayer(cfg) for _ in range(cfg.num_hidden_layers))
+        self.norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.eps = cfg.rms_norm_eps
+        self.gradient_checkpointing = cfg.gradient_checkpointing
+        self._rope_table = None
+
+    def rope_table(self, device) -> torch.Tensor:
+        if self._rope_table is None or self._rope_table.device != device:
+            self._rope_table = build_rope_table(
+                self.cfg.max_position_embeddings, self.cfg.head_dim, self.cfg.rope_theta, device
+            )
+        return self._rope_table
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        table = self.rope_table(input_ids.device)
+        residual = self.embed_tokens(input_ids)
+        hidden = rms_norm(residual, self.layers[0].input_layernorm_weight, self.eps)
+        n = len(self.layers)
+        for i, layer in enumerate(self.layers):
+            if self.gradient_checkpointing and self.training:
+                out, residual = torch.utils.checkpoint.checkpoint(
+                    layer, hidden, residual, table, use_reentrant=False
+                )
+            else:
+                out, residual = layer(hidden, residual, table)
+            next_w = self.layers[i + 1].input_layernorm_weight if i + 1 < n else self.norm_weight
+            hidden, residual = fused_add_rms_norm(out, residual, next_w, self.eps)
+        return hidden
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.config = cfg
+        self.model = LlamaModel(cfg)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.config.initializer_range
+        if isinstance(module, nn.Linear):
+            module.weight.data.normal_(0.0, std)
+        elif isinstance(module, nn.Embedding):
+            module.weight.data.normal_(0.0, std)
+
+    def gradient_checkpointing_enable(self):
+        self.model.gradient_checkpointing = True
+
+    def gradient_checkpointing_disable(self):
+        self.model.gradient_checkpointing = False
+
+    def forward(self, input_ids: torch.Tensor, labels: Optional[torch.Tensor] = None):
+        hidden = self.model(input_ids)
+        logits = self.lm_head(hidden)
+        loss = None
+        if labels is not None:
+            shift_logits = logits[:, :-1, :].contiguous().float()
+            shift_labels = labels[:, 1:].contiguous()
+            loss = F.cross_entropy(shift_logits.view(-1, shift_logits.size(-1)), shift_labels.view(-1))
+        return {"logits": logits, "loss": loss}
+
+    @property
+    def num_parameters(self) -> int:
+        return sum(p.numel() for p in self.parameters())

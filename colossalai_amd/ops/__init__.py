@@ -1,0 +1,21 @@
+from ._kernels import has_kernels, kernels, use_hip
+from .attention import attention_ref, flash_attention, fused_rope_attention
+from .norm import fused_add_rms_norm, rms_norm, rms_norm_ref
+from .rope import apply_rope, apply_rope_ref, build_rope_table
+from .swiglu import swiglu, swiglu_ref
+
+__all__ = [
+    "has_kernels",
+    "kernels",
+    "use_hip",
+    "flash_attention",
+    "attention_ref",
+    "rms_norm",
+    "fused_add_rms_norm",
+    "rms_norm_ref",
+    "apply_rope",
+    "apply_rope_ref",
+    "build_rope_table",
+    "swiglu",
+    "swiglu_ref",
+]

@@ -1,0 +1,52 @@
+import torch
+
+from colossalai_amd.models import LLAMA_CONFIGS, LlamaConfig, LlamaForCausalLM
+
+
+def tiny_cfg(**kw):
+    base = dict(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    base.update(kw)
+    return LlamaConfig(**base)
+
+
+def test_forward_backward_cpu():
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(tiny_cfg())
+    x = torch.randint(0, 128, (2, 32))
+    out = m(x, labels=x)
+    assert out["logits"].shape == (2, 32, 128)
+    out["loss"].backward()
+    for n, p in m.named_parameters():
+        assert p.grad is not None, n
+        assert torch.isfinite(p.grad).all(), n
+
+
+def test_grad_checkpoint_equivalence():
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(tiny_cfg())
+    x = torch.randint(0, 128, (2, 32))
+    loss1 = m(x, labels=x)["loss"]
+    loss1.backward()
+    g1 = {n: p.grad.clone() for n, p in m.named_parameters()}
+    for p in m.parameters():
+        p.grad = None
+    m.gradient_checkpointing_enable()
+    loss2 = m(x, labels=x)["loss"]
+    loss2.backward()
+    torch.testing.assert_close(loss1, loss2)
+    for n, p in m.named_parameters():
+        torch.testing.assert_close(p.grad, g1[n], rtol=1e-5, atol=1e-6, msg=lambda s: f"{n}: {s}")
+
+
+def test_configs_table():
+    c = LLAMA_CONFIGS["llama-7b"]
+    assert c.hidden_size == 4096 and c.num_hidden_layers == 32
+    # 7B param count sanity (~6.7e9)
+    n = (
+        c.vocab_size * c.hidden_size * 2
+        + c.num_hidden_layers
+        * (4 * c.hidden_size * c.hidden_size + 3 * c.hidden_size * c.intermediate_size + 2 * c.hidden_size)
+        + c.hidden_size
+    )
+    assert 6.5e9 < n < 7.1e9

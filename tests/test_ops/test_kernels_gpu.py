@@ -1,0 +1,249 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from colossalai_amd.ops import kernels
+
+    _C = kernels()
+
+
+def _bf16_close(a, b, rtol=2e-2, atol=2e-2, frac=0.0):
+    """bf16 comparison: elementwise tolerance with optional allowed outlier fraction."""
+    a = a.float().cpu()
+    b = b.float().cpu()
+    err = (a - b).abs()
+    tol = atol + rtol * b.abs()
+    bad = (err > tol).float().mean().item()
+    assert bad <= frac, f"bf16 mismatch: {bad * 100:.3f}% elements out of tol (max err {err.max():.4f})"
+
+
+# ---------------------------------------------------------------- MFMA layout
+def test_mfma_layouts():
+    torch.manual_seed(0)
+    A16 = torch.randn(16, 32, device="cuda")
+    B16 = torch.randn(32, 16, device="cuda")
+    A32 = torch.randn(32, 16, device="cuda")
+    B32 = torch.randn(16, 32, device="cuda")
+    C16, C32, p0, p1 = _C.mfma_selftest(A16, B16, A32, B32)
+    # bf16 inputs -> compare against bf16-rounded matmul
+    ref16 = (A16.bfloat16().float() @ B16.bfloat16().float())
+    ref32 = (A32.bfloat16().float() @ B32.bfloat16().float())
+    torch.testing.assert_close(C16.cpu(), ref16.cpu(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(C32.cpu(), ref32.cpu(), rtol=3e-2, atol=3e-2)
+    # record permlane semantics (assert it moves data between lane halves)
+    p0, p1 = p0.cpu(), p1.cpu()
+    print("permlane32_swap out0:", p0.tolist())
+    print("permlane32_swap out1:", p1.tolist())
+
+
+# ------------------------------------------------------------------- rmsnorm
+@pytest.mark.parametrize("rows,H", [(256, 4096), (1000, 2048), (64, 8192)])
+def test_rmsnorm_fwd_bwd(rows, H):
+    torch.manual_seed(1)
+    x = torch.randn(rows, H, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    eps = 1e-5
+    out, inv_rms = _C.rmsnorm_fwd(x, w, eps, True)
+    xf = x.float()
+    inv_ref = torch.rsqrt(xf.pow(2).mean(-1) + eps)
+    ref = (xf * inv_ref.unsqueeze(-1) * w.float()).bfloat16()
+    _bf16_close(out, ref)
+    torch.testing.assert_close(inv_rms.cpu(), inv_ref.cpu(), rtol=1e-4, atol=1e-5)
+
+    dy = torch.randn_like(x)
+    dx, dw = _C.rmsnorm_bwd(dy, x, w, inv_rms)
+    dyf, wf = dy.float(), w.float()
+    inv = inv_ref.unsqueeze(-1)
+    dyw = dyf * wf
+    dot = (dyw * xf).sum(-1, keepdim=True)
+    dx_ref = inv * (dyw - xf * dot * inv * inv / H)
+    dw_ref = (dyf * xf * inv).sum(0)
+    _bf16_close(dx, dx_ref, frac=1e-4)
+    torch.testing.assert_close(dw.cpu(), dw_ref.cpu(), rtol=2e-2, atol=2e-1)
+
+
+def test_rmsnorm_fused_add():
+    torch.manual_seed(2)
+    rows, H = 512, 4096
+    x = torch.randn(rows, H, device="cuda", dtype=torch.bfloat16)
+    res = torch.randn(rows, H, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    res_work = res.clone()
+    out, inv_rms = _C.rmsnorm_fused_add_fwd(x, res_work, w, 1e-5, True)
+    hf = x.float() + res.float()
+    h_bf = hf.bfloat16()  # kernel stores h in bf16
+    inv_ref = torch.rsqrt(h_bf.float().pow(2).mean(-1, keepdim=True) + 1e-5)
+    ref = (h_bf.float() * inv_ref * w.float()).bfloat16()
+    _bf16_close(res_work, h_bf)
+    _bf16_close(out, ref)
+
+
+# ---------------------------------------------------------------------- rope
+def test_rope():
+    from colossalai_amd.ops import build_rope_table
+    from colossalai_amd.ops.rope import apply_rope_ref
+
+    torch.manual_seed(3)
+    B, S, Hq, Hkv, D = 2, 128, 4, 2, 128
+    table = build_rope_table(S, D, 10000.0, "cuda")
+    q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    q2, k2 = q.clone(), k.clone()
+    _C.rope_inplace(q2, k2, table, None, False)
+    qr, kr = apply_rope_ref(q.float(), k.float(), table, None, S, False)
+    _bf16_close(q2, qr)
+    _bf16_close(k2, kr)
+    # fwd then bwd rotation = identity
+    _C.rope_inplace(q2, k2, table, None, True)
+    _bf16_close(q2, q.float(), rtol=3e-2, atol=3e-2)
+
+    # strided views into a packed tensor
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, device="cuda", dtype=torch.bfloat16)
+    qv = qkv[:, :, : Hq * D].view(B, S, Hq, D)
+    kv = qkv[:, :, Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+    q_ref = qv.clone()
+    k_ref = kv.clone()
+    _C.rope_inplace(qv, kv, table, None, False)
+    qr2, kr2 = apply_rope_ref(q_ref.float(), k_ref.float(), table, None, S, False)
+    _bf16_close(qv, qr2)
+    _bf16_close(kv, kr2)
+
+
+# -------------------------------------------------------------------- swiglu
+def test_swiglu():
+    torch.manual_seed(4)
+    t, I = 1024, 2048
+    gu = torch.randn(t, 2 * I, device="cuda", dtype=torch.bfloat16)
+    out = _C.swiglu_fwd(gu)
+    g, u = gu[:, :I].float(), gu[:, I:].float()
+    ref = torch.nn.functional.silu(g) * u
+    _bf16_close(out, ref)
+    dout = torch.randn(t, I, device="cuda", dtype=torch.bfloat16)
+    dgu = _C.swiglu_bwd(dout, gu)
+    sig = torch.sigmoid(g)
+    d = dout.float()
+    dg_ref = d * u * sig * (1 + g * (1 - sig))
+    du_ref = d * g * sig
+    _bf16_close(dgu[:, :I], dg_ref, frac=1e-5)
+    _bf16_close(dgu[:, I:], du_ref, frac=1e-5)
+
+
+# ---------------------------------------------------------------------- adam
+def test_fused_adam():
+    from colossalai_amd.nn.optimizer.fused_adam import fused_adam_step_cpu
+
+    torch.manual_seed(5)
+    shapes = [(1000,), (333,), (4096, 128)]
+    params, grads, ms, vs, outs = [], [], [], [], []
+    refs = []
+    for sh in shapes:
+        p = torch.randn(sh, device="cuda", dtype=torch.float32)
+        g = torch.randn(sh, device="cuda", dtype=torch.bfloat16)
+        m = torch.randn(sh, device="cuda").abs() * 0.1
+        v = torch.randn(sh, device="cuda").abs() * 0.01
+        o = torch.zeros(sh, device="cuda", dtype=torch.bfloat16)
+        refs.append((p.clone(), g.clone(), m.clone(), v.clone()))
+        params.append(p); grads.append(g); ms.append(m); vs.append(v); outs.append(o)
+    lr, b1, b2, eps, wd, step = 1e-2, 0.9, 0.95, 1e-8, 0.1, 7
+    _C.multi_tensor_adam(grads, params, ms, vs, outs, lr, b1, b2, eps, step, True, True, wd, 1.0, 2048)
+    for (p0, g0, m0, v0), p, m, v, o in zip(refs, params, ms, vs, outs):
+        fused_adam_step_cpu(p0, g0, m0, v0, lr, b1, b2, eps, wd, step, True, True)
+        torch.testing.assert_close(p.cpu(), p0.cpu(), rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(m.cpu(), m0.cpu(), rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(v.cpu(), v0.cpu(), rtol=1e-5, atol=1e-6)
+        _bf16_close(o, p0)  # fused bf16 write-back
+
+
+def test_multi_tensor_l2norm_scale():
+    torch.manual_seed(6)
+    xs = [torch.randn(10007, device="cuda"), torch.randn(64, 64, device="cuda")]
+    n = _C.multi_tensor_l2norm(xs, 4096)
+    ref = torch.sqrt(sum(x.float().pow(2).sum() for x in xs))
+    torch.testing.assert_close(n.cpu(), ref.cpu(), rtol=1e-5, atol=1e-6)
+    ys = [torch.empty_like(x) for x in xs]
+    _C.multi_tensor_scale(xs, ys, 0.5, 4096)
+    for x, y in zip(xs, ys):
+        torch.testing.assert_close(y.cpu(), (x * 0.5).cpu())
+
+
+# ------------------------------------------------------------ flash attention
+@pytest.mark.parametrize("B,S,Hq,Hkv,D", [
+    (2, 256, 4, 4, 128),
+    (1, 512, 8, 2, 128),   # GQA
+    (2, 333, 4, 4, 128),   # ragged S
+    (1, 1024, 4, 4, 64),   # D=64
+])
+@pytest.mark.parametrize("causal", [True, False])
+def test_flash_attention_fwd(B, S, Hq, Hkv, D, causal):
+    from colossalai_amd.ops.attention import attention_ref
+
+    torch.manual_seed(7)
+    q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    out, lse = _C.flash_attn_fwd(q, k, v, causal, 1.0 / math.sqrt(D))
+    ref = attention_ref(q, k, v, causal)
+    _bf16_close(out, ref.float(), rtol=3e-2, atol=3e-2, frac=1e-5)
+
+
+@pytest.mark.parametrize("B,S,Hq,Hkv,D", [
+    (2, 256, 4, 4, 128),
+    (1, 512, 8, 2, 128),
+    (2, 320, 4, 4, 128),
+])
+def test_flash_attention_bwd(B, S, Hq, Hkv, D):
+    from colossalai_amd.ops.attention import attention_ref
+
+    torch.manual_seed(8)
+    causal = True
+    scale = 1.0 / math.sqrt(D)
+    q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16, requires_grad=False)
+    k = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    dout = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+
+    out, lse = _C.flash_attn_fwd(q, k, v, causal, scale)
+    e = torch.empty(0, device="cuda", dtype=torch.bfloat16)
+    dq, dk, dv = _C.flash_attn_bwd(dout, q, k, v, out, lse, causal, scale, e.clone(), e.clone(), e.clone())
+
+    qf = q.float().requires_grad_(True)
+    kf = k.float().requires_grad_(True)
+    vf = v.float().requires_grad_(True)
+    ref = attention_ref(qf, kf, vf, causal, scale)
+    ref.backward(dout.float())
+    _bf16_close(dq, qf.grad, rtol=4e-2, atol=4e-2, frac=1e-4)
+    _bf16_close(dk, kf.grad, rtol=4e-2, atol=4e-2, frac=1e-4)
+    _bf16_close(dv, vf.grad, rtol=4e-2, atol=4e-2, frac=1e-4)
+
+
+def test_fused_rope_attention_autograd():
+    """End-to-end packed-QKV fused op vs the CPU/fp32 composition."""
+    from colossalai_amd.ops import build_rope_table, fused_rope_attention
+    from colossalai_amd.ops.attention import attention_ref
+    from colossalai_amd.ops.rope import apply_rope_ref
+
+    torch.manual_seed(9)
+    B, S, Hq, Hkv, D = 2, 256, 4, 2, 128
+    table = build_rope_table(S, D, 10000.0, "cuda")
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, device="cuda", dtype=torch.bfloat16)
+    qkv_hip = qkv.clone().requires_grad_(True)
+    out = fused_rope_attention(qkv_hip.clone() if False else qkv_hip * 1.0, table, Hq, Hkv, D)
+    loss = out.float().square().mean()
+    loss.backward()
+
+    qkv_ref = qkv.float().requires_grad_(True)
+    q = qkv_ref[:, :, : Hq * D].view(B, S, Hq, D)
+    k = qkv_ref[:, :, Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+    v = qkv_ref[:, :, (Hq + Hkv) * D :].view(B, S, Hkv, D)
+    qr, kr = apply_rope_ref(q, k, table.cpu().cuda(), None, S, False)
+    ref = attention_ref(qr, kr, v, True)
+    loss_ref = ref.float().square().mean()
+    loss_ref.backward()
+    _bf16_close(out, ref, rtol=3e-2, atol=3e-2, frac=1e-5)
+    _bf16_close(qkv_hip.grad, qkv_ref.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
