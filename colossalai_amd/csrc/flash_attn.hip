@@ -565,7 +565,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
   char* Vlds = smem + KB_BYTES + KT_BYTES;
   char* Pw = smem + 2 * KB_BYTES + KT_BYTES;  // NWB * 32*KVB*2
   float* lse_lds = reinterpret_cast<float*>(Pw + NWB * (QW * KVB * 2));
-  float* dta_lds = lse_lds + NW * QW;
+  float* dta_lds = lse_lds + NWB * QW;
 
   const int lane = threadIdx.x & 63;
   const int w = threadIdx.x >> 6;
