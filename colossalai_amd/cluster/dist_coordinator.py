@@ -13,9 +13,14 @@ class DistCoordinator:
     """Convenience wrapper around torch.distributed rank/master logic."""
 
     def __init__(self):
-        assert dist.is_initialized(), "DistCoordinator requires torch.distributed to be initialized."
-        self._rank = dist.get_rank()
-        self._world_size = dist.get_world_size()
+        # Degrades to a single-process coordinator when torch.distributed is
+        # not initialized (single-GPU scripts without a launcher).
+        if dist.is_available() and dist.is_initialized():
+            self._rank = dist.get_rank()
+            self._world_size = dist.get_world_size()
+        else:
+            self._rank = 0
+            self._world_size = 1
         self._local_rank = int(os.environ.get("LOCAL_RANK", 0))
 
     @property
@@ -35,6 +40,8 @@ class DistCoordinator:
         return self._world_size > 1
 
     def is_master(self, process_group=None) -> bool:
+        if not (dist.is_available() and dist.is_initialized()):
+            return True
         return dist.get_rank(group=process_group) == 0
 
     def is_node_master(self) -> bool:
@@ -52,7 +59,8 @@ class DistCoordinator:
             print(msg, flush=True)
 
     def block_all(self, process_group=None) -> None:
-        dist.barrier(group=process_group)
+        if dist.is_available() and dist.is_initialized():
+            dist.barrier(group=process_group)
 
     @contextmanager
     def priority_execution(self, executor_rank: int = 0, process_group=None):
