@@ -38,8 +38,9 @@ constexpr int QW = 32;      // q rows per wave (fwd / dq); kv rows per wave (dkd
 constexpr int NW = 8;       // waves per block (forward)
 constexpr int NT = 512;     // threads per block (forward)
 // Backward kernels hold dK+dV (or dQ) accumulators plus operand fragments in
-// registers (~250 VGPR): 8-wave blocks would exceed the 2048-VGPR/CU pool,
-// so backward runs 4-wave blocks (128 kv/q rows per block).
+// registers (~210 VGPR measured): 4-wave blocks keep the allocator
+// unconstrained (8-wave blocks force a 256-VGPR cap -> spills, measured 21%
+// slower end-to-end).
 constexpr int NWB = 4;
 constexpr int NTB = 256;
 
@@ -127,6 +128,46 @@ DEV_INLINE void stage_both(char* lds, char* ldsT, const unsigned short* base, lo
   }
 }
 
+// Register-staged tile for async-split staging (guide T14 / G15): global
+// loads for tile t+1 are ISSUED before tile t's compute (results land in
+// registers under the MFMAs), LDS writes happen after the barrier. NT_ must
+// divide ROWS*CPR.
+template <int D, int ROWS, int NT_>
+struct RegStage {
+  static constexpr int CPR = D / 8;
+  static constexpr int NCH = ROWS * CPR / NT_;
+  static_assert(ROWS * CPR % NT_ == 0, "RegStage: uneven chunking");
+  bf16x8_t r[NCH];
+
+  DEV_INLINE void load(const unsigned short* base, long tok_stride, int row0, int S) {
+#pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = (int)threadIdx.x + i * NT_;
+      const int row = c / CPR, ch = c % CPR;
+      const int grow = min(row0 + row, S - 1);
+      r[i] = ld_g16(base + (long)grow * tok_stride + ch * 8);
+    }
+  }
+  DEV_INLINE void store_rowmajor(char* lds) const {
+#pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = (int)threadIdx.x + i * NT_;
+      st_lds16(lds, swz<D * 2>(c / CPR, (c % CPR) * 16), r[i]);
+    }
+  }
+  DEV_INLINE void store_transposed(char* ldsT) const {
+#pragma unroll
+    for (int i = 0; i < NCH; ++i) {
+      const int c = (int)threadIdx.x + i * NT_;
+      const int row = c / CPR, ch = c % CPR;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        *reinterpret_cast<unsigned short*>(ldsT + swz<ROWS * 2>(ch * 8 + e, row * 2)) = bf_raw(r[i][e]);
+      }
+    }
+  }
+};
+
 }  // namespace fa
 
 // ============================================================ forward kernel
@@ -188,17 +229,25 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   const int kv_tiles_all = (S + KVB - 1) / KVB;
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NW * QW + KVB - 1) / KVB) : kv_tiles_all;
 
+  // async-split staging (T14): tile 0 staged synchronously, tile t+1's global
+  // loads issue before tile t's compute and land in LDS after the barrier.
+  RegStage<D, KVB, NT> kstage, vstage;
+  kstage.load(k_base, kts, 0, S);
+  vstage.load(v_base, kts, 0, S);
+  kstage.store_rowmajor(Klds);
+  vstage.store_transposed(VT);
+  __syncthreads();
+
   for (int t = 0; t < kv_tiles; ++t) {
     const int k0 = t * KVB;
-    __syncthreads();
-    stage_rowmajor<D, KVB>(Klds, k_base, kts, k0, S);
-    stage_transposed<D, KVB>(VT, v_base, kts, k0, S);
-    __syncthreads();
-
-    // waves entirely above the diagonal produce nothing
-    if (causal && k0 > qw + QW - 1) continue;
-    if (qw >= S) continue;
-
+    const bool has_next = (t + 1 < kv_tiles);
+    if (has_next) {
+      kstage.load(k_base, kts, k0 + KVB, S);
+      vstage.load(v_base, kts, k0 + KVB, S);
+    }
+    // waves entirely above the diagonal produce nothing (barriers stay uniform)
+    const bool active = !(causal && k0 > qw + QW - 1) && (qw < S);
+    if (active) {
     // ---- S^T = K · Q^T : two 32-k blocks. C: col = q (ln), row = k (crow).
     float p[2][16];
 #pragma unroll
@@ -210,7 +259,9 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
       for (int sl = 0; sl < DSL; ++sl) {
         // A-operand: K rows. lane: row = kb*32+ln, cols d = sl*16 + half*8 + [0..7]
         bf16x8_t kf = ld_lds16(Klds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        __builtin_amdgcn_s_setprio(1);
         acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[sl], acc, 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
       }
 #pragma unroll
       for (int j = 0; j < 16; ++j) p[kb][j] = acc[j];
@@ -275,9 +326,19 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
       for (int nb = 0; nb < D / 32; ++nb) {
         // B: lane col = d (nb*32+ln), k-rows = ks*16 + half*8 + [0..7] → V_T[d][k]
         bf16x8_t vb = ld_lds16(VT, swz<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        __builtin_amdgcn_s_setprio(1);
         oacc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, oacc[nb], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
       }
     }
+    }  // active
+
+    __syncthreads();  // all waves done reading Klds/VT for tile t
+    if (has_next) {
+      kstage.store_rowmajor(Klds);
+      vstage.store_transposed(VT);
+    }
+    __syncthreads();
   }
 
   // ---- epilogue: O /= l, store O and LSE
@@ -410,6 +471,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
   const int qt_first = causal ? (kv0 / QB) : 0;
   const int qt_last = (S + QB - 1) / QB;
 
+  RegStage<D, QB, NTB> qstage, dostage;
   for (int g = 0; g < G; ++g) {
     const int h = hk * G + g;
     const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
@@ -418,20 +480,30 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     const float* lse_base = LSE + ((long)b * Hq + h) * S;
     const float* dta_base = DELTA + ((long)b * Hq + h) * S;
 
+    // prologue: stage the first q-tile of this head synchronously
+    qstage.load(q_base, qts, qt_first * QB, S);
+    dostage.load(do_base, do_stride, qt_first * QB, S);
+    __syncthreads();  // previous head's compute done before overwriting LDS
+    qstage.store_rowmajor(Qlds);
+    qstage.store_transposed(QT);
+    dostage.store_rowmajor(dOlds);
+    dostage.store_transposed(dOT);
+    if (threadIdx.x < QB) {
+      const int qr = min(qt_first * QB + (int)threadIdx.x, S - 1);
+      lse_lds[threadIdx.x] = lse_base[qr];
+      dta_lds[threadIdx.x] = dta_base[qr];
+    }
+    __syncthreads();
+
     for (int qt = qt_first; qt < qt_last; ++qt) {
       const int qt0 = qt * QB;
-      __syncthreads();
-      stage_both<D, QB>(Qlds, QT, q_base, qts, qt0, S);
-      stage_both<D, QB>(dOlds, dOT, do_base, do_stride, qt0, S);
-      if (threadIdx.x < QB) {
-        const int qr = min(qt0 + (int)threadIdx.x, S - 1);
-        lse_lds[threadIdx.x] = lse_base[qr];
-        dta_lds[threadIdx.x] = dta_base[qr];
+      const bool has_next = (qt + 1 < qt_last);
+      if (has_next) {
+        qstage.load(q_base, qts, qt0 + QB, S);
+        dostage.load(do_base, do_stride, qt0 + QB, S);
       }
-      __syncthreads();
-
-      if (causal && qt0 + QB - 1 < kvw) continue;  // wave fully above diagonal
-      if (kvw >= S) continue;
+      const bool active = !(causal && qt0 + QB - 1 < kvw) && (kvw < S);
+      if (active) {
 
       // ---- S'^T[kv][q] = K · Q^T : C col = q (ln within qb), row = kv.
       float p[2][16], dp[2][16];
@@ -444,7 +516,9 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
         for (int sl = 0; sl < DSL; ++sl) {
           // B: lane col = q (qb*32+ln), rows d → Qlds[q][d]
           bf16x8_t qb_frag = ld_lds16(Qlds, swz<D * 2>(qb * 32 + ln, (sl * 16 + half * 8) * 2));
+          __builtin_amdgcn_s_setprio(1);
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[sl], qb_frag, acc, 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
         }
 #pragma unroll
         for (int j = 0; j < 16; ++j) p[qb][j] = acc[j];
@@ -479,7 +553,9 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
 #pragma unroll
         for (int nb = 0; nb < D / 32; ++nb) {
           bf16x8_t dob = ld_lds16(dOT, swz<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+          __builtin_amdgcn_s_setprio(1);
           dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[nb], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
         }
       }
 
@@ -517,9 +593,26 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
 #pragma unroll
         for (int nb = 0; nb < D / 32; ++nb) {
           bf16x8_t qtb = ld_lds16(QT, swz<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+          __builtin_amdgcn_s_setprio(1);
           dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[nb], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
         }
       }
+      }  // active
+
+      __syncthreads();  // all waves done with this q-tile's LDS
+      if (has_next) {
+        qstage.store_rowmajor(Qlds);
+        qstage.store_transposed(QT);
+        dostage.store_rowmajor(dOlds);
+        dostage.store_transposed(dOT);
+        if (threadIdx.x < QB) {
+          const int qr = min(qt0 + QB + (int)threadIdx.x, S - 1);
+          lse_lds[threadIdx.x] = lse_base[qr];
+          dta_lds[threadIdx.x] = dta_base[qr];
+        }
+      }
+      __syncthreads();
     }
   }
 
@@ -620,15 +713,23 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
   const int kv_tiles_all = (S + KVB - 1) / KVB;
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NWB * QW + KVB - 1) / KVB) : kv_tiles_all;
 
+  RegStage<D, KVB, NTB> kstage, vstage;
+  kstage.load(k_base, kts, 0, S);
+  vstage.load(v_base, kts, 0, S);
+  kstage.store_rowmajor(Klds);
+  kstage.store_transposed(KT);
+  vstage.store_rowmajor(Vlds);
+  __syncthreads();
+
   for (int t = 0; t < kv_tiles; ++t) {
     const int k0 = t * KVB;
-    __syncthreads();
-    stage_both<D, KVB>(Klds, KT, k_base, kts, k0, S);
-    stage_rowmajor<D, KVB>(Vlds, v_base, kts, k0, S);
-    __syncthreads();
-
-    if (causal && k0 > qw + QW - 1) continue;
-    if (qw >= S) continue;
+    const bool has_next = (t + 1 < kv_tiles);
+    if (has_next) {
+      kstage.load(k_base, kts, k0 + KVB, S);
+      vstage.load(v_base, kts, k0 + KVB, S);
+    }
+    const bool active = !(causal && k0 > qw + QW - 1) && (qw < S);
+    if (active) {
 
     // ---- S[q][kv] = Q · K^T : C col = kv (ln within kb), row = q (crow).
     float p[2][16], dp[2][16];
@@ -674,9 +775,20 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
         bf16x8_t ktb = ld_lds16(KT, swz<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        __builtin_amdgcn_s_setprio(1);
         dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, ktb, dq_acc[nb], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
       }
     }
+    }  // active
+
+    __syncthreads();
+    if (has_next) {
+      kstage.store_rowmajor(Klds);
+      kstage.store_transposed(KT);
+      vstage.store_rowmajor(Vlds);
+    }
+    __syncthreads();
   }
 
   // ---- epilogue: write dQ
