@@ -93,20 +93,23 @@ def llama_flops_per_token(cfg: LlamaConfig, seq_len: int, grad_ckpt: bool = Fals
 
 
 class LlamaAttention(nn.Module):
+    # head counts are instance attributes (not read from the shared config) so
+    # the Shardformer policy can rewrite them per-module under TP/SP.
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
-        self.cfg = cfg
-        D, Hq, Hkv = cfg.head_dim, cfg.num_attention_heads, cfg.num_key_value_heads
+        self.num_heads = cfg.num_attention_heads
+        self.num_kv_heads = cfg.num_key_value_heads
+        self.head_dim = cfg.head_dim
+        D, Hq, Hkv = self.head_dim, self.num_heads, self.num_kv_heads
         self.qkv_proj = nn.Linear(cfg.hidden_size, (Hq + 2 * Hkv) * D, bias=False)
         self.o_proj = nn.Linear(Hq * D, cfg.hidden_size, bias=False)
         self.scale = 1.0 / math.sqrt(D)
 
     def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
-        cfg = self.cfg
         B, S, _ = hidden.shape
         qkv = self.qkv_proj(hidden)
         attn = fused_rope_attention(
-            qkv, rope_table, cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim,
+            qkv, rope_table, self.num_heads, self.num_kv_heads, self.head_dim,
             causal=True, scale=self.scale,
         )
         return self.o_proj(attn.reshape(B, S, -1))
@@ -207,9 +210,15 @@ class LlamaForCausalLM(nn.Module):
         logits = self.lm_head(hidden)
         loss = None
         if labels is not None:
-            shift_logits = logits[:, :-1, :].contiguous().float()
             shift_labels = labels[:, 1:].contiguous()
-            loss = F.cross_entropy(shift_logits.view(-1, shift_logits.size(-1)), shift_labels.view(-1))
+            if getattr(self, "tp_group", None) is not None and getattr(self, "parallel_logits", False):
+                # vocab-parallel CE over the tp group (logits stay sharded)
+                from ..shardformer.layer.loss import dist_cross_entropy
+
+                loss = dist_cross_entropy(logits[:, :-1, :], shift_labels, group=self.tp_group)
+            else:
+                shift_logits = logits[:, :-1, :].contiguous().float()
+                loss = F.cross_entropy(shift_logits.view(-1, shift_logits.size(-1)), shift_labels.view(-1))
         return {"logits": logits, "loss": loss}
 
     @property

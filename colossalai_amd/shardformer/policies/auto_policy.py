@@ -1,0 +1,40 @@
+"""Model-class → policy registry (reference: colossalai/shardformer/policies/auto_policy.py)."""
+
+import importlib
+from typing import Dict
+
+import torch.nn as nn
+
+__all__ = ["get_autopolicy", "register_policy", "POLICY_REGISTRY"]
+
+# class name -> (module path, policy class name)
+POLICY_REGISTRY: Dict[str, tuple] = {
+    # native models
+    "LlamaForCausalLM": ("colossalai_amd.shardformer.policies.llama", "LlamaForCausalLMPolicy"),
+    "LlamaModel": ("colossalai_amd.shardformer.policies.llama", "LlamaPolicy"),
+    # HF transformers models
+    "GPT2LMHeadModel": ("colossalai_amd.shardformer.policies.gpt2", "GPT2LMHeadModelPolicy"),
+    "MixtralForCausalLM": ("colossalai_amd.shardformer.policies.mixtral", "MixtralForCausalLMPolicy"),
+}
+
+
+def register_policy(model_cls_name: str, module_path: str, policy_name: str) -> None:
+    POLICY_REGISTRY[model_cls_name] = (module_path, policy_name)
+
+
+def get_autopolicy(model: nn.Module):
+    # HF models can share class names with native ones: try
+    # "<root_package>.<ClassName>" first, then the bare class name.
+    name = type(model).__name__
+    root = type(model).__module__.split(".")[0]
+    qualified = f"{root}.{name}"
+    if qualified in POLICY_REGISTRY:
+        name = qualified
+    if name not in POLICY_REGISTRY:
+        raise NotImplementedError(
+            f"No shard policy registered for {name}. Register one with "
+            "colossalai_amd.shardformer.policies.auto_policy.register_policy."
+        )
+    module_path, policy_name = POLICY_REGISTRY[name]
+    mod = importlib.import_module(module_path)
+    return getattr(mod, policy_name)()

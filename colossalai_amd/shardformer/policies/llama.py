@@ -1,0 +1,113 @@
+"""TP/SP/PP policy for the native Llama model
+(reference policy shape: colossalai/shardformer/policies/llama.py:26).
+
+TP sharding map (tp = tensor-parallel degree):
+- ``self_attn.qkv_proj``   → Linear1D_Col with split_sizes [Hq·D, Hkv·D, Hkv·D]
+  (each packed segment sharded separately so q/k/v head slices stay aligned)
+- ``self_attn.o_proj``     → Linear1D_Row
+- ``mlp.gate_up_proj``     → Linear1D_Col with split_sizes [I, I]
+- ``mlp.down_proj``        → Linear1D_Row
+- ``model.embed_tokens``   → VocabParallelEmbedding1D
+- ``lm_head``              → Linear1D_Col (parallel logits + DistCrossEntropy)
+- head counts on each attention rewritten to per-rank values
+- RMSNorm weights stay replicated (grads identical across tp ranks because
+  every norm input is replicated post-all-reduce).
+"""
+
+from typing import Dict
+
+from ...models.llama import LlamaAttention, LlamaDecoderLayer, LlamaForCausalLM, LlamaMLP, LlamaModel
+from ..layer.embedding import VocabParallelEmbedding1D
+from ..layer.linear import Linear1D_Col, Linear1D_Row
+from .base_policy import ModulePolicyDescription, Policy, SubModuleReplacementDescription
+
+__all__ = ["LlamaPolicy", "LlamaForCausalLMPolicy"]
+
+
+class LlamaPolicy(Policy):
+    def preprocess(self):
+        return self.model
+
+    def config_sanity_check(self):
+        pass
+
+    def module_policy(self) -> Dict:
+        policy = {}
+        tp = self.shard_config.tensor_parallel_size
+        if self.shard_config.enable_tensor_parallelism and tp > 1:
+            model = self.model.model if hasattr(self.model, "model") else self.model
+            attn0 = model.layers[0].self_attn
+            Hq, Hkv, D = attn0.num_heads, attn0.num_kv_heads, attn0.head_dim
+            assert Hq % tp == 0 and Hkv % tp == 0, f"heads ({Hq},{Hkv}) must divide tp={tp}"
+            qkv_splits = [Hq * D, Hkv * D, Hkv * D]
+            inter = model.layers[0].mlp.gate_up_proj.out_features // 2
+            sp_mode = self.shard_config.sequence_parallelism_mode if self.shard_config.enable_sequence_parallelism else None
+
+            policy[LlamaAttention] = ModulePolicyDescription(
+                attribute_replacement={
+                    "num_heads": Hq // tp,
+                    "num_kv_heads": Hkv // tp,
+                },
+                sub_module_replacement=[
+                    SubModuleReplacementDescription(
+                        suffix="qkv_proj",
+                        target_module=Linear1D_Col,
+                        kwargs={"split_sizes": qkv_splits, "seq_parallel_mode": sp_mode},
+                    ),
+                    SubModuleReplacementDescription(
+                        suffix="o_proj",
+                        target_module=Linear1D_Row,
+                        kwargs={"seq_parallel_mode": sp_mode},
+                    ),
+                ],
+            )
+            policy[LlamaMLP] = ModulePolicyDescription(
+                sub_module_replacement=[
+                    SubModuleReplacementDescription(
+                        suffix="gate_up_proj",
+                        target_module=Linear1D_Col,
+                        kwargs={"split_sizes": [inter, inter], "seq_parallel_mode": sp_mode},
+                    ),
+                    SubModuleReplacementDescription(
+                        suffix="down_proj",
+                        target_module=Linear1D_Row,
+                        kwargs={"seq_parallel_mode": sp_mode},
+                    ),
+                ],
+            )
+            policy[LlamaModel] = ModulePolicyDescription(
+                sub_module_replacement=[
+                    SubModuleReplacementDescription(
+                        suffix="embed_tokens",
+                        target_module=VocabParallelEmbedding1D,
+                    ),
+                ],
+            )
+        return policy
+
+    def postprocess(self):
+        return self.model
+
+
+class LlamaForCausalLMPolicy(LlamaPolicy):
+    def module_policy(self) -> Dict:
+        policy = super().module_policy()
+        tp = self.shard_config.tensor_parallel_size
+        if self.shard_config.enable_tensor_parallelism and tp > 1:
+            self.append_or_create_submodule_replacement(
+                SubModuleReplacementDescription(
+                    suffix="lm_head",
+                    target_module=Linear1D_Col,
+                    kwargs={"gather_output": not self.shard_config.parallel_output},
+                ),
+                policy,
+                LlamaForCausalLM,
+            )
+        return policy
+
+    def postprocess(self):
+        tp = self.shard_config.tensor_parallel_size
+        if self.shard_config.enable_tensor_parallelism and tp > 1 and self.shard_config.parallel_output:
+            self.model.tp_group = self.shard_config.tensor_parallel_process_group
+            self.model.parallel_logits = True
+        return self.model
