@@ -1,0 +1,355 @@
+"""HybridParallelPlugin — dp × pp × tp × sp on one mesh
+(reference: colossalai/booster/plugin/hybrid_parallel_plugin.py:928).
+
+Mesh axes (dp, pp, tp, sp). Shardformer applies TP/SP; pipeline stages own a
+contiguous layer range and exchange the residual stream via RCCL P2P; the dp
+dimension runs ZeRO-0/1/2 (ZeRO via LowLevelZeroOptimizer with the dp group).
+"""
+
+from contextlib import contextmanager, nullcontext
+from typing import Any, Callable, Iterator, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+from torch.optim import Optimizer
+from torch.optim.lr_scheduler import _LRScheduler as LRScheduler
+from torch.utils.data import DataLoader
+
+from ...checkpoint_io import CheckpointIO
+from ...cluster import ProcessGroupMesh
+from ...interface import ModelWrapper, OptimizerWrapper
+from ...amp import MixedPrecisionOptimizer
+from ...pipeline import PipelineStageManager
+from ...pipeline.schedule import OneForwardOneBackwardSchedule
+from ...shardformer import ShardConfig, ShardFormer
+from ...zero import LowLevelZeroOptimizer
+from .plugin_base import Plugin
+from .torch_ddp_plugin import TorchDDPCheckpointIO
+
+__all__ = ["HybridParallelPlugin", "HybridParallelModule"]
+
+DP_AXIS, PP_AXIS, TP_AXIS, SP_AXIS = 0, 1, 2, 3
+
+_PRECISION_DTYPE = {"fp16": torch.float16, "bf16": torch.bfloat16, "fp32": torch.float32}
+
+
+class HybridParallelModule(ModelWrapper):
+    def __init__(self, module: nn.Module, dtype: torch.dtype, dp_group, tp_group, sp_group):
+        super().__init__(module)
+        self.dtype = dtype
+        self.dp_group = dp_group
+        self.tp_group = tp_group
+        self.sp_group = sp_group
+        self.require_grad_sync = True
+
+    def forward(self, *args, **kwargs):
+        args = [a.to(self.dtype) if isinstance(a, torch.Tensor) and a.is_floating_point() else a for a in args]
+        kwargs = {
+            k: (v.to(self.dtype) if isinstance(v, torch.Tensor) and v.is_floating_point() else v)
+            for k, v in kwargs.items()
+        }
+        return self.module(*args, **kwargs)
+
+    @contextmanager
+    def no_sync(self):
+        old = self.require_grad_sync
+        self.require_grad_sync = False
+        try:
+            yield
+        finally:
+            self.require_grad_sync = old
+
+    def sync_dp_grads(self):
+        """ZeRO-0 path: average grads over the dp group (bucket-coalesced)."""
+        if self.dp_group is None or dist.get_world_size(self.dp_group) == 1:
+            return
+        world = dist.get_world_size(self.dp_group)
+        bucket: List[torch.Tensor] = []
+        size = 0
+        BUCKET_BYTES = 128 * 1024 * 1024  # xGMI-sized
+        for p in self.module.parameters():
+            if p.grad is None:
+                continue
+            bucket.append(p.grad)
+            size += p.grad.numel() * p.grad.element_size()
+            if size >= BUCKET_BYTES:
+                self._flush(bucket, world)
+                bucket, size = [], 0
+        if bucket:
+            self._flush(bucket, world)
+
+    def _flush(self, grads: List[torch.Tensor], world: int):
+        flat = torch.cat([g.reshape(-1) for g in grads])
+        dist.all_reduce(flat, group=self.dp_group)
+        flat /= world
+        off = 0
+        for g in grads:
+            g.copy_(flat[off : off + g.numel()].view_as(g))
+            off += g.numel()
+
+
+class HybridParallelNaiveOptimizer(MixedPrecisionOptimizer):
+    """ZeRO-0 optimizer: bf16/fp16 masters + manual dp grad sync in backward."""
+
+    def __init__(self, optim: Optimizer, model: HybridParallelModule, precision: str, max_norm: float = 0.0, **kw):
+        self.model_wrapper = model
+        super().__init__(optim, precision=precision, max_norm=max_norm, **kw)
+
+    def backward(self, loss, inputs=None, retain_graph=False, **kwargs):
+        super().backward(loss, inputs=inputs, retain_graph=retain_graph, **kwargs)
+        if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_dp_grads()
+
+    def backward_by_grad(self, tensor, grad, inputs=None, retain_graph=False):
+        super().backward_by_grad(tensor, grad, inputs=inputs, retain_graph=retain_graph)
+        if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_dp_grads()
+
+
+class HybridParallelPlugin(Plugin):
+    def __init__(
+        self,
+        tp_size: int = 1,
+        pp_size: int = 1,
+        sp_size: Optional[int] = None,
+        precision: str = "bf16",
+        zero_stage: int = 0,
+        enable_flash_attention: bool = True,
+        enable_fused_normalization: bool = True,
+        enable_sequence_parallelism: bool = False,
+        sequence_parallelism_mode: Optional[str] = None,
+        num_microbatches: Optional[int] = None,
+        microbatch_size: Optional[int] = None,
+        initial_scale: float = 2**16,
+        min_scale: float = 1,
+        growth_factor: float = 2,
+        backoff_factor: float = 0.5,
+        growth_interval: int = 1000,
+        hysteresis: int = 2,
+        max_scale: float = 2**32,
+        max_norm: float = 0.0,
+        zero_bucket_size_in_m: int = 32,
+        overlap_communication: bool = True,
+        parallel_output: bool = True,
+        pp_style: str = "1f1b",
+        **kwargs,
+    ):
+        assert dist.is_initialized(), "launch colossalai_amd before creating HybridParallelPlugin"
+        assert zero_stage in (0, 1, 2)
+        assert pp_style == "1f1b", "only 1f1b pipeline style is implemented so far"
+        world = dist.get_world_size()
+        if sp_size is None:
+            sp_size = 1
+        assert world % (tp_size * pp_size * sp_size) == 0, (
+            f"world {world} not divisible by tp{tp_size}*pp{pp_size}*sp{sp_size}"
+        )
+        dp_size = world // (tp_size * pp_size * sp_size)
+        self.tp_size, self.pp_size, self.dp_size, self.sp_size = tp_size, pp_size, dp_size, sp_size
+        self.precision = precision
+        self.zero_stage = zero_stage
+        self.max_norm = max_norm
+        self.amp_kwargs = dict(
+            initial_scale=initial_scale, min_scale=min_scale, growth_factor=growth_factor,
+            backoff_factor=backoff_factor, growth_interval=growth_interval, hysteresis=hysteresis,
+            max_scale=max_scale,
+        )
+        self.zero_kwargs = dict(
+            reduce_bucket_size=zero_bucket_size_in_m * 1024 * 1024,
+            overlap_communication=overlap_communication,
+            partition_grad=(zero_stage == 2),
+            clip_grad_norm=max_norm,
+        )
+
+        self.pg_mesh = ProcessGroupMesh(dp_size, pp_size, tp_size, sp_size)
+        self.dp_group = self.pg_mesh.get_group_along_axis(DP_AXIS)
+        self.pp_group = self.pg_mesh.get_group_along_axis(PP_AXIS)
+        self.tp_group = self.pg_mesh.get_group_along_axis(TP_AXIS)
+        self.sp_group = self.pg_mesh.get_group_along_axis(SP_AXIS)
+        # grads of dp×sp-replicated params sync over the flattened group
+        self.dp_sp_group = self.pg_mesh.get_group_along_axis([DP_AXIS, SP_AXIS])
+
+        self.stage_manager = None
+        self.scheduler = None
+        if pp_size > 1:
+            assert num_microbatches is not None or microbatch_size is not None, (
+                "pipeline parallelism requires num_microbatches or microbatch_size"
+            )
+            self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
+            self.scheduler = OneForwardOneBackwardSchedule(
+                self.stage_manager, num_microbatches=num_microbatches, microbatch_size=microbatch_size
+            )
+
+        self.shard_config = ShardConfig(
+            tensor_parallel_process_group=self.tp_group if tp_size > 1 else None,
+            sequence_parallel_process_group=self.sp_group if sp_size > 1 else None,
+            pipeline_stage_manager=self.stage_manager,
+            enable_tensor_parallelism=tp_size > 1,
+            enable_sequence_parallelism=enable_sequence_parallelism,
+            sequence_parallelism_mode=sequence_parallelism_mode,
+            enable_flash_attention=enable_flash_attention,
+            enable_fused_normalization=enable_fused_normalization,
+            parallel_output=parallel_output,
+        )
+
+    # -------------------------------------------------------------- metadata
+    def supported_devices(self) -> List[str]:
+        return ["cuda", "cpu"]
+
+    def supported_precisions(self) -> List[str]:
+        return ["fp16", "bf16", "fp32"]
+
+    def control_device(self) -> bool:
+        return True
+
+    def control_precision(self) -> bool:
+        return True
+
+    def support_no_sync(self) -> bool:
+        return True
+
+    def support_lora(self) -> bool:
+        return False
+
+    def control_checkpoint_io(self) -> bool:
+        return True
+
+    def get_checkpoint_io(self) -> CheckpointIO:
+        from ...checkpoint_io.hybrid_parallel_checkpoint_io import HybridParallelCheckpointIO
+
+        return HybridParallelCheckpointIO(self.dp_group, self.pp_group, self.tp_group, self.sp_size)
+
+    # -------------------------------------------------------------- configure
+    def configure(
+        self,
+        model: nn.Module,
+        optimizer: Optional[Optimizer] = None,
+        criterion: Optional[Callable] = None,
+        dataloader: Optional[DataLoader] = None,
+        lr_scheduler: Optional[LRScheduler] = None,
+    ) -> Tuple[nn.Module, OptimizerWrapper, Callable, DataLoader, LRScheduler]:
+        dtype = _PRECISION_DTYPE[self.precision]
+
+        if not isinstance(model, ModelWrapper):
+            # pipeline layer assignment BEFORE sharding (policies may use it)
+            if self.stage_manager is not None:
+                self._assign_pipeline_stage(model)
+            if self.tp_size > 1 or self.shard_config.enable_sequence_parallelism:
+                shardformer = ShardFormer(self.shard_config)
+                model, _ = shardformer.optimize(model)
+            model = model.to(dtype)
+            if torch.cuda.is_available():
+                model = model.to("cuda")
+            model = HybridParallelModule(model, dtype, self.dp_sp_group, self.tp_group, self.sp_group)
+
+        if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
+            self._rebuild_param_groups(optimizer, model.module)
+            if self.zero_stage == 0:
+                if self.precision == "fp32":
+                    optimizer = OptimizerWrapper(optimizer)
+                else:
+                    optimizer = HybridParallelNaiveOptimizer(
+                        optimizer, model, precision=self.precision, max_norm=self.max_norm, **self.amp_kwargs
+                    )
+            else:
+                optimizer = LowLevelZeroOptimizer(
+                    optimizer,
+                    dp_process_group=self.dp_sp_group,
+                    forced_dtype=dtype if self.precision != "fp32" else None,
+                    **self.zero_kwargs,
+                    **({} if self.precision != "fp16" else self.amp_kwargs),
+                )
+        return model, optimizer, criterion, dataloader, lr_scheduler
+
+    def _assign_pipeline_stage(self, model: nn.Module) -> None:
+        """Set model.stage_range and free layers outside this stage.
+
+        Convention: the model exposes ``.model.layers`` (decoder stack) — the
+        native model family contract; HF models get stage-aware policies.
+        """
+        inner = model.model if hasattr(model, "model") else model
+        assert hasattr(inner, "layers"), "pipeline parallelism needs a .layers decoder stack"
+        n_layers = len(inner.layers)
+        start, end = self.stage_manager.stage_index(n_layers)
+        model.stage_range = (start, end)
+        for i in range(n_layers):
+            if not (start <= i < end):
+                inner.layers[i] = _StageStub()
+        if not self.stage_manager.is_first_stage() and hasattr(inner, "embed_tokens"):
+            inner.embed_tokens = _StageStub()
+        if not self.stage_manager.is_last_stage() and hasattr(model, "lm_head"):
+            model.lm_head = _StageStub()
+
+    @staticmethod
+    def _rebuild_param_groups(optimizer: Optimizer, model: nn.Module) -> None:
+        """Sharding/stage-release replaced parameter objects; re-point the
+        optimizer's param groups at the live model parameters."""
+        live = list(model.parameters())
+        current = [p for g in optimizer.param_groups for p in g["params"]]
+        if len(current) == len(live) and all(a is b for a, b in zip(current, live)):
+            return
+        assert len(optimizer.param_groups) == 1, (
+            "after sharding, only single-param-group optimizers can be re-pointed automatically; "
+            "construct the optimizer after booster.boost for multiple groups"
+        )
+        optimizer.param_groups[0]["params"] = live
+        optimizer.state.clear()
+
+    # ---------------------------------------------------------------- runtime
+    def execute_pipeline(
+        self,
+        data_iter: Iterator,
+        model: HybridParallelModule,
+        criterion: Callable[[Any, Any], torch.Tensor],
+        optimizer: Optional[OptimizerWrapper] = None,
+        return_loss: bool = True,
+        return_outputs: bool = False,
+    ) -> dict:
+        assert self.scheduler is not None, "execute_pipeline requires pp_size > 1"
+        if isinstance(optimizer, LowLevelZeroOptimizer):
+            ctx = optimizer.no_sync()
+        else:
+            ctx = model.no_sync()
+        with ctx:
+            result = self.scheduler.forward_backward_step(
+                model, data_iter, criterion, optimizer, return_loss, return_outputs
+            )
+        # grad sync after all microbatches
+        if isinstance(optimizer, LowLevelZeroOptimizer):
+            optimizer.sync_dp_grads()
+        else:
+            model.sync_dp_grads()
+        return result
+
+    def no_sync(self, model: nn.Module, optimizer: OptimizerWrapper = None) -> Iterator[None]:
+        if isinstance(optimizer, LowLevelZeroOptimizer):
+            return optimizer.no_sync()
+        return model.no_sync()
+
+    def prepare_dataloader(self, dataset, batch_size, shuffle=False, seed=1024, drop_last=False,
+                           pin_memory=False, num_workers=0, **kwargs):
+        import numpy as np
+        import random
+        from torch.utils.data import DataLoader
+        from torch.utils.data.distributed import DistributedSampler
+
+        sampler = DistributedSampler(
+            dataset,
+            num_replicas=self.dp_size,
+            rank=self.pg_mesh.coordinate(DP_AXIS),
+            shuffle=shuffle,
+        )
+
+        def seed_worker(worker_id):
+            np.random.seed(seed)
+            random.seed(seed)
+
+        return DataLoader(dataset, batch_size=batch_size, sampler=sampler, worker_init_fn=seed_worker,
+                          drop_last=drop_last, pin_memory=pin_memory, num_workers=num_workers, **kwargs)
+
+
+class _StageStub(nn.Module):
+    """Placeholder for modules owned by other pipeline stages."""
+
+    def forward(self, *args, **kwargs):
+        raise RuntimeError("This module belongs to another pipeline stage")

@@ -165,21 +165,47 @@ class LlamaModel(nn.Module):
             )
         return self._rope_table
 
-    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
-        table = self.rope_table(input_ids.device)
-        residual = self.embed_tokens(input_ids)
-        hidden = rms_norm(residual, self.layers[0].input_layernorm_weight, self.eps)
-        n = len(self.layers)
-        for i, layer in enumerate(self.layers):
+    def forward(
+        self,
+        input_ids: Optional[torch.Tensor] = None,
+        hidden_states: Optional[torch.Tensor] = None,
+        stage_range: Optional[tuple] = None,
+    ) -> torch.Tensor:
+        """Full forward, or a pipeline-stage slice when ``stage_range`` is set.
+
+        Pipeline stage IO is the raw residual stream (one tensor); each stage
+        keeps the fused residual-add+norm chain internally and hands the
+        un-normed residual to the next stage, which applies its first layer's
+        input norm (fusion lost only at the stage boundary).
+        """
+        start, end = stage_range if stage_range is not None else (0, len(self.layers))
+        is_first = start == 0
+        is_last = end == len(self.layers)
+        if is_first:
+            assert input_ids is not None
+            residual = self.embed_tokens(input_ids)
+            device = input_ids.device
+        else:
+            assert hidden_states is not None
+            residual = hidden_states
+            device = hidden_states.device
+        table = self.rope_table(device)
+        hidden = rms_norm(residual, self.layers[start].input_layernorm_weight, self.eps)
+        for i in range(start, end):
+            layer = self.layers[i]
             if self.gradient_checkpointing and self.training:
                 out, residual = torch.utils.checkpoint.checkpoint(
                     layer, hidden, residual, table, use_reentrant=False
                 )
             else:
                 out, residual = layer(hidden, residual, table)
-            next_w = self.layers[i + 1].input_layernorm_weight if i + 1 < n else self.norm_weight
-            hidden, residual = fused_add_rms_norm(out, residual, next_w, self.eps)
-        return hidden
+            if i + 1 < end:
+                hidden, residual = fused_add_rms_norm(out, residual, self.layers[i + 1].input_layernorm_weight, self.eps)
+            elif is_last:
+                hidden, residual = fused_add_rms_norm(out, residual, self.norm_weight, self.eps)
+            else:
+                residual = residual + out  # stage boundary: next stage norms
+        return hidden if is_last else residual
 
 
 class LlamaForCausalLM(nn.Module):
@@ -205,8 +231,17 @@ class LlamaForCausalLM(nn.Module):
     def gradient_checkpointing_disable(self):
         self.model.gradient_checkpointing = False
 
-    def forward(self, input_ids: torch.Tensor, labels: Optional[torch.Tensor] = None):
-        hidden = self.model(input_ids)
+    def forward(
+        self,
+        input_ids: Optional[torch.Tensor] = None,
+        labels: Optional[torch.Tensor] = None,
+        hidden_states: Optional[torch.Tensor] = None,
+    ):
+        stage_range = getattr(self, "stage_range", None)
+        out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range)
+        if stage_range is not None and stage_range[1] < len(self.model.layers):
+            return {"hidden_states": out}
+        hidden = out
         logits = self.lm_head(hidden)
         loss = None
         if labels is not None:

@@ -36,11 +36,14 @@ class LlamaPolicy(Policy):
         tp = self.shard_config.tensor_parallel_size
         if self.shard_config.enable_tensor_parallelism and tp > 1:
             model = self.model.model if hasattr(self.model, "model") else self.model
-            attn0 = model.layers[0].self_attn
+            # under pipeline parallelism some layers are stage stubs — use the
+            # first real decoder layer on this stage
+            first = next(l for l in model.layers if isinstance(l, LlamaDecoderLayer))
+            attn0 = first.self_attn
             Hq, Hkv, D = attn0.num_heads, attn0.num_kv_heads, attn0.head_dim
             assert Hq % tp == 0 and Hkv % tp == 0, f"heads ({Hq},{Hkv}) must divide tp={tp}"
             qkv_splits = [Hq * D, Hkv * D, Hkv * D]
-            inter = model.layers[0].mlp.gate_up_proj.out_features // 2
+            inter = first.mlp.gate_up_proj.out_features // 2
             sp_mode = self.shard_config.sequence_parallelism_mode if self.shard_config.enable_sequence_parallelism else None
 
             policy[LlamaAttention] = ModulePolicyDescription(

@@ -76,6 +76,8 @@ class ModelSharder:
                     if rep.ignore_if_not_exist:
                         continue
                     raise AttributeError(f"{type(module).__name__} has no submodule {rep.suffix}")
+                if type(child).__name__ == "_StageStub":
+                    continue  # owned by another pipeline stage
                 new_child = rep.target_module.from_native_module(
                     child, process_group=self.shard_config.tensor_parallel_process_group, **rep.kwargs
                 )

@@ -256,6 +256,10 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         if self.require_grad_sync:
             self._sync_unreduced()
 
+    def sync_dp_grads(self):
+        """Public: reduce all pending buckets (pipeline / grad-accumulation)."""
+        self._sync_unreduced()
+
     def _sync_unreduced(self):
         """After a no_sync accumulation phase, the final (sync) backward's
         hooks reduce normally; but grads accumulated during no_sync for
