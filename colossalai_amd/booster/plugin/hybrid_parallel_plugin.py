@@ -107,6 +107,24 @@ class HybridParallelNaiveOptimizer(MixedPrecisionOptimizer):
             self.model_wrapper.sync_dp_grads()
 
 
+class HybridParallelFP32Optimizer(OptimizerWrapper):
+    """ZeRO-0 fp32: plain optimizer + dp×sp grad averaging after backward."""
+
+    def __init__(self, optim: Optimizer, model: HybridParallelModule):
+        super().__init__(optim)
+        self.model_wrapper = model
+
+    def backward(self, loss, inputs=None, retain_graph=False, **kwargs):
+        loss.backward(inputs=inputs, retain_graph=retain_graph, **kwargs)
+        if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_dp_grads()
+
+    def backward_by_grad(self, tensor, grad, inputs=None, retain_graph=False):
+        torch.autograd.backward(tensor, grad, inputs=inputs, retain_graph=retain_graph)
+        if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_dp_grads()
+
+
 class HybridParallelPlugin(Plugin):
     def __init__(
         self,
@@ -246,7 +264,7 @@ class HybridParallelPlugin(Plugin):
             self._rebuild_param_groups(optimizer, model.module)
             if self.zero_stage == 0:
                 if self.precision == "fp32":
-                    optimizer = OptimizerWrapper(optimizer)
+                    optimizer = HybridParallelFP32Optimizer(optimizer, model)
                 else:
                     optimizer = HybridParallelNaiveOptimizer(
                         optimizer, model, precision=self.precision, max_norm=self.max_norm, **self.amp_kwargs

@@ -108,6 +108,27 @@ class LlamaAttention(nn.Module):
     def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
         B, S, _ = hidden.shape
         qkv = self.qkv_proj(hidden)
+        sp_mode = getattr(self, "sp_mode", None)
+        if sp_mode == "all_to_all":
+            # Ulysses: scatter heads / gather sequence around attention
+            import torch.distributed as dist
+
+            from ..shardformer.layer import all_to_all_comm
+            from ..ops import apply_rope, flash_attention
+
+            sp_group = self.sp_group
+            sp = dist.get_world_size(sp_group)
+            Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+            q = qkv[:, :, : Hq * D].reshape(B, S, Hq, D)
+            k = qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D)
+            v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
+            q = all_to_all_comm(q, sp_group, scatter_dim=2, gather_dim=1)  # [B, S*sp, Hq/sp, D]
+            k = all_to_all_comm(k, sp_group, scatter_dim=2, gather_dim=1)
+            v = all_to_all_comm(v, sp_group, scatter_dim=2, gather_dim=1)
+            q, k = apply_rope(q.contiguous(), k.contiguous(), rope_table)
+            attn = flash_attention(q, k, v.contiguous(), causal=True, scale=self.scale)
+            attn = all_to_all_comm(attn, sp_group, scatter_dim=1, gather_dim=2)  # back to [B, S, Hq, D]
+            return self.o_proj(attn.reshape(B, S, -1))
         attn = fused_rope_attention(
             qkv, rope_table, self.num_heads, self.num_kv_heads, self.head_dim,
             causal=True, scale=self.scale,
@@ -238,6 +259,23 @@ class LlamaForCausalLM(nn.Module):
         hidden_states: Optional[torch.Tensor] = None,
     ):
         stage_range = getattr(self, "stage_range", None)
+        sp_group = getattr(self, "sp_group", None)
+        if sp_group is not None and input_ids is not None:
+            # Ulysses SP: each rank runs its sequence shard; labels are shifted
+            # globally first so the boundary token is not lost.
+            import torch.distributed as dist
+
+            sp = dist.get_world_size(sp_group)
+            rank = dist.get_rank(sp_group)
+            S = input_ids.shape[1]
+            assert S % sp == 0, f"seq len {S} must divide sp size {sp}"
+            shard = S // sp
+            if labels is not None:
+                shifted = torch.full_like(labels, -100)
+                shifted[:, :-1] = labels[:, 1:]
+                labels = shifted[:, rank * shard : (rank + 1) * shard]
+                self._sp_labels_shifted = True
+            input_ids = input_ids[:, rank * shard : (rank + 1) * shard]
         out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range)
         if stage_range is not None and stage_range[1] < len(self.model.layers):
             return {"hidden_states": out}
@@ -245,15 +283,21 @@ class LlamaForCausalLM(nn.Module):
         logits = self.lm_head(hidden)
         loss = None
         if labels is not None:
-            shift_labels = labels[:, 1:].contiguous()
+            if getattr(self, "_sp_labels_shifted", False):
+                shift_logits, shift_labels = logits, labels.contiguous()
+            else:
+                shift_logits, shift_labels = logits[:, :-1, :], labels[:, 1:].contiguous()
             if getattr(self, "tp_group", None) is not None and getattr(self, "parallel_logits", False):
                 # vocab-parallel CE over the tp group (logits stay sharded)
                 from ..shardformer.layer.loss import dist_cross_entropy
 
-                loss = dist_cross_entropy(logits[:, :-1, :], shift_labels, group=self.tp_group)
+                loss = dist_cross_entropy(shift_logits, shift_labels, group=self.tp_group)
             else:
-                shift_logits = logits[:, :-1, :].contiguous().float()
-                loss = F.cross_entropy(shift_logits.view(-1, shift_logits.size(-1)), shift_labels.view(-1))
+                loss = F.cross_entropy(
+                    shift_logits.contiguous().float().view(-1, shift_logits.size(-1)),
+                    shift_labels.view(-1),
+                    ignore_index=-100,
+                )
         return {"logits": logits, "loss": loss}
 
     @property

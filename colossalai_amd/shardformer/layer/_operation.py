@@ -76,6 +76,13 @@ def _all_to_all(x: torch.Tensor, scatter_dim: int, gather_dim: int, group) -> to
     world = dist.get_world_size(group)
     if world == 1:
         return x
+    if dist.get_backend(group) == "gloo":
+        # gloo has no alltoall: emulate with all_gather (CPU tests only)
+        rank = dist.get_rank(group)
+        gathered = [torch.empty_like(x) for _ in range(world)]
+        dist.all_gather(gathered, x.contiguous(), group=group)
+        parts = [g.chunk(world, dim=scatter_dim)[rank] for g in gathered]
+        return torch.cat(parts, dim=gather_dim)
     inputs = [t.contiguous() for t in x.chunk(world, dim=scatter_dim)]
     outputs = [torch.empty_like(inputs[0]) for _ in range(world)]
     dist.all_to_all(outputs, inputs, group=group)

@@ -33,6 +33,17 @@ class LlamaPolicy(Policy):
 
     def module_policy(self) -> Dict:
         policy = {}
+        sp_a2a = (
+            self.shard_config.enable_sequence_parallelism
+            and self.shard_config.sequence_parallelism_mode == "all_to_all"
+        )
+        if sp_a2a:
+            policy[LlamaAttention] = ModulePolicyDescription(
+                attribute_replacement={
+                    "sp_mode": "all_to_all",
+                    "sp_group": self.shard_config.sequence_parallel_process_group,
+                }
+            )
         tp = self.shard_config.tensor_parallel_size
         if self.shard_config.enable_tensor_parallelism and tp > 1:
             model = self.model.model if hasattr(self.model, "model") else self.model
@@ -46,11 +57,13 @@ class LlamaPolicy(Policy):
             inter = first.mlp.gate_up_proj.out_features // 2
             sp_mode = self.shard_config.sequence_parallelism_mode if self.shard_config.enable_sequence_parallelism else None
 
+            attn_attrs = {"num_heads": Hq // tp, "num_kv_heads": Hkv // tp}
+            if sp_a2a:
+                attn_attrs.update(
+                    sp_mode="all_to_all", sp_group=self.shard_config.sequence_parallel_process_group
+                )
             policy[LlamaAttention] = ModulePolicyDescription(
-                attribute_replacement={
-                    "num_heads": Hq // tp,
-                    "num_kv_heads": Hkv // tp,
-                },
+                attribute_replacement=attn_attrs,
                 sub_module_replacement=[
                     SubModuleReplacementDescription(
                         suffix="qkv_proj",
@@ -113,4 +126,9 @@ class LlamaForCausalLMPolicy(LlamaPolicy):
         if self.shard_config.enable_tensor_parallelism and tp > 1 and self.shard_config.parallel_output:
             self.model.tp_group = self.shard_config.tensor_parallel_process_group
             self.model.parallel_logits = True
+        if (
+            self.shard_config.enable_sequence_parallelism
+            and self.shard_config.sequence_parallelism_mode == "all_to_all"
+        ):
+            self.model.sp_group = self.shard_config.sequence_parallel_process_group
         return self.model

@@ -43,7 +43,7 @@ def _pad_to(n: int, align: int) -> int:
 
 
 class _Bucket:
-    __slots__ = ("start", "end", "n_params", "n_done", "work", "master", "exp_avg", "exp_avg_sq")
+    __slots__ = ("start", "end", "n_params", "n_done", "work", "master", "exp_avg", "exp_avg_sq", "offloaded")
 
     def __init__(self, start: int, end: int):
         self.start = start
@@ -54,6 +54,7 @@ class _Bucket:
         self.master: Optional[Tensor] = None
         self.exp_avg: Optional[Tensor] = None
         self.exp_avg_sq: Optional[Tensor] = None
+        self.offloaded: bool = False
 
 
 class _ZeroFP16Mixin(FP16MixedPrecisionMixin):
@@ -87,6 +88,7 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         dp_process_group: Optional[dist.ProcessGroup] = None,
         forced_dtype: Optional[torch.dtype] = None,
         master_weights: bool = True,
+        cpu_offload_frac: float = 0.0,
     ):
         super().__init__(optimizer)
         from ...nn.optimizer.fused_adam import FusedAdam
@@ -101,6 +103,7 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         self.bucket_elems = reduce_bucket_size
         self.overlap = overlap_communication and torch.cuda.is_available()
         self.master_weights = master_weights
+        self.cpu_offload_frac = float(cpu_offload_frac)
         self.require_grad_sync = True
         self._accum_steps_pending = False
 
@@ -171,17 +174,32 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
                         self._param_bucket[p] = b
                         b.n_params += 1
                         break
-            # master shards (this rank's slice of each bucket)
+            # master shards (this rank's slice of each bucket); a prefix
+            # fraction lives in pinned host memory (Gemini-style static
+            # optimizer-state offload — 288 GB HBM usually makes this 0)
+            total_elems = sum(b.end - b.start for b in buckets)
+            offload_budget = int(total_elems * self.cpu_offload_frac)
+            seen = 0
             for b in buckets:
                 blen = b.end - b.start
                 assert blen % self.world == 0
                 shard = flat[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
+                offload = self.master_weights and torch.cuda.is_available() and seen < offload_budget
+                b.offloaded = offload
+                seen += blen
                 if self.master_weights:
-                    b.master = shard.detach().float()
+                    if offload:
+                        b.master = torch.empty(shard.numel(), dtype=torch.float32, pin_memory=True)
+                        b.master.copy_(shard.detach().float().cpu())
+                    else:
+                        b.master = shard.detach().float()
                 else:
                     b.master = shard  # in-dtype "master"
-                b.exp_avg = torch.zeros_like(b.master, dtype=torch.float32)
-                b.exp_avg_sq = torch.zeros_like(b.master, dtype=torch.float32)
+                dev = b.master.device
+                b.exp_avg = torch.zeros(b.master.numel(), dtype=torch.float32, device=dev,
+                                        pin_memory=offload)
+                b.exp_avg_sq = torch.zeros(b.master.numel(), dtype=torch.float32, device=dev,
+                                           pin_memory=offload)
 
             self._flat_params.append(flat)
             self._flat_grads.append(flat_grad)
@@ -334,21 +352,32 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
             if true_norm > self.clip_grad_norm:
                 clip_factor = true_norm / self.clip_grad_norm
 
-        # fused adam over all bucket shards in one multi-tensor batch per group
+        # fused adam over all bucket shards in one multi-tensor batch per group;
+        # offloaded buckets step on the host (grad D2H -> CPU adam -> param H2D)
         for gi, (group, buckets) in enumerate(zip(self.optim.param_groups, self._group_buckets)):
             flat_grad = self._flat_grads[gi]
             flat = self._flat_params[gi]
             grads, masters, mlist, vlist, outs = [], [], [], [], []
+            cpu_jobs = []
             for b in buckets:
                 blen = b.end - b.start
                 gshard = flat_grad[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
                 pshard = flat[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
-                grads.append(gshard)
-                masters.append(b.master)
-                mlist.append(b.exp_avg)
-                vlist.append(b.exp_avg_sq)
-                outs.append(pshard)
-            self._fused_step(group, grads, masters, mlist, vlist, outs, div_scale * clip_factor)
+                if getattr(b, "offloaded", False):
+                    cpu_jobs.append((b, gshard, pshard))
+                else:
+                    grads.append(gshard)
+                    masters.append(b.master)
+                    mlist.append(b.exp_avg)
+                    vlist.append(b.exp_avg_sq)
+                    outs.append(pshard)
+            if grads:
+                self._fused_step(group, grads, masters, mlist, vlist, outs, div_scale * clip_factor)
+            elif cpu_jobs:
+                group.setdefault("step", 0)
+                group["step"] += 1
+            for b, gshard, pshard in cpu_jobs:
+                self._cpu_step(group, b, gshard, pshard, div_scale * clip_factor)
 
         # all-gather updated working params (bucket-wise, async on comm stream)
         if self.world > 1:
@@ -390,6 +419,18 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
                 fused_adam_step_cpu(p, g, m, v, lr, beta1, beta2, eps, wd, group["step"], adamw, bias_corr, div_scale)
                 if self.master_weights:
                     o.copy_(p.to(o.dtype))
+
+    def _cpu_step(self, group, b, gshard, pshard, div_scale):
+        from ...nn.optimizer.fused_adam import fused_adam_step_cpu
+
+        beta1, beta2 = group.get("betas", (0.9, 0.999))
+        g_cpu = gshard.detach().to("cpu", dtype=torch.float32)
+        fused_adam_step_cpu(
+            b.master, g_cpu, b.exp_avg, b.exp_avg_sq, group["lr"], beta1, beta2,
+            group.get("eps", 1e-8), group.get("weight_decay", 0.0), group["step"],
+            getattr(self.optim, "adamw_mode", True), group.get("bias_correction", True), div_scale,
+        )
+        pshard.copy_(b.master.to(pshard.device, dtype=pshard.dtype, non_blocking=True))
 
     # ------------------------------------------------------------ checkpoint
     def state_dict(self):
