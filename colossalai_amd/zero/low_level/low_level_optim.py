@@ -89,6 +89,8 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         forced_dtype: Optional[torch.dtype] = None,
         master_weights: bool = True,
         cpu_offload_frac: float = 0.0,
+        group_dp_pgs: Optional[List[Optional[dist.ProcessGroup]]] = None,
+        group_grad_divisors: Optional[List[int]] = None,
     ):
         super().__init__(optimizer)
         from ...nn.optimizer.fused_adam import FusedAdam
@@ -98,6 +100,18 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         self.dp_pg = dp_process_group
         self.world = dist.get_world_size(self.dp_pg) if dist.is_initialized() else 1
         self.rank = dist.get_rank(self.dp_pg) if dist.is_initialized() else 0
+        # per-param-group process groups (MoE: expert groups sync over a
+        # smaller group than dense params but divide by the same world)
+        n_groups = len(optimizer.param_groups)
+        if group_dp_pgs is None:
+            group_dp_pgs = [self.dp_pg] * n_groups
+        assert len(group_dp_pgs) == n_groups
+        self._g_pg = list(group_dp_pgs)
+        self._g_world = [dist.get_world_size(pg) if dist.is_initialized() else 1 for pg in self._g_pg]
+        self._g_rank = [dist.get_rank(pg) if dist.is_initialized() else 0 for pg in self._g_pg]
+        if group_grad_divisors is None:
+            group_grad_divisors = list(self._g_world)
+        self._g_div = list(group_grad_divisors)
         self.partition_grad = partition_grad
         self.clip_grad_norm = clip_grad_norm
         self.bucket_elems = reduce_bucket_size
@@ -121,7 +135,6 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         else:
             self.mixin = BF16MixedPrecisionMixin()
 
-        align = 64 * self.world
         self._flat_params: List[Tensor] = []
         self._flat_grads: List[Optional[Tensor]] = []
         self._group_buckets: List[List[_Bucket]] = []
@@ -131,6 +144,9 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         self._hook_handles = []
 
         for gi, group in enumerate(optimizer.param_groups):
+            gworld = self._g_world[gi]
+            grank = self._g_rank[gi]
+            align = 64 * gworld
             params = [p for p in group["params"] if p.requires_grad]
             for p in params:
                 assert p.dtype == self.dtype, f"all params must be {self.dtype}, got {p.dtype}"
@@ -182,8 +198,8 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
             seen = 0
             for b in buckets:
                 blen = b.end - b.start
-                assert blen % self.world == 0
-                shard = flat[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
+                assert blen % gworld == 0
+                shard = flat[b.start + grank * blen // gworld : b.start + (grank + 1) * blen // gworld]
                 offload = self.master_weights and torch.cuda.is_available() and seen < offload_budget
                 b.offloaded = offload
                 seen += blen
@@ -224,7 +240,10 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
             self._reduce_bucket(gi, b)
 
     def _reduce_bucket(self, gi: int, b: _Bucket):
-        if self.world == 1:
+        world = self._g_world[gi]
+        rank = self._g_rank[gi]
+        pg = self._g_pg[gi]
+        if world == 1:
             return
         flat_grad = self._flat_grads[gi]
         seg = flat_grad[b.start : b.end]
@@ -232,15 +251,15 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
 
         def _issue():
             if self.partition_grad:
-                shard = seg[self.rank * blen // self.world : (self.rank + 1) * blen // self.world]
+                shard = seg[rank * blen // world : (rank + 1) * blen // world]
                 # reduce_scatter needs a separate output; reuse the shard slice
                 # via an intermediate to keep the flat layout
                 out = torch.empty_like(shard)
                 dist.reduce_scatter_tensor(out, seg.contiguous() if not seg.is_contiguous() else seg,
-                                           group=self.dp_pg)
+                                           group=pg)
                 shard.copy_(out)
             else:
-                dist.all_reduce(seg, group=self.dp_pg)
+                dist.all_reduce(seg, group=pg)
 
         if self.overlap:
             self._comm_stream.wait_stream(torch.cuda.current_stream())
@@ -316,9 +335,10 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         for gi, buckets in enumerate(self._group_buckets):
             flat_grad = self._flat_grads[gi]
             device = flat_grad.device
+            world, rank = self._g_world[gi], self._g_rank[gi]
             for b in buckets:
                 blen = b.end - b.start
-                shard = flat_grad[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
+                shard = flat_grad[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
                 partials.append(torch.linalg.vector_norm(shard.float()) ** 2)
         if not partials:
             return 0.0
@@ -338,31 +358,30 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
 
         self._wait_all_reductions()
 
-        div_scale = self.mixin.get_grad_div_scale()
-        if not self.partition_grad and self.world > 1:
-            div_scale *= self.world  # all_reduce summed
-        elif self.partition_grad and self.world > 1:
-            div_scale *= self.world  # reduce_scatter summed
+        loss_scale = self.mixin.get_grad_div_scale()
 
         # grad clipping: _compute_grad_norm sees grads that are summed over dp
         # and still loss-scaled, so the true norm is summed_norm / div_scale.
         clip_factor = 1.0
         if self.clip_grad_norm > 0:
-            true_norm = self._compute_grad_norm() / div_scale
+            assert len(set(self._g_div)) == 1, "grad clipping with mixed process groups is not supported yet"
+            true_norm = self._compute_grad_norm() / (loss_scale * self._g_div[0])
             if true_norm > self.clip_grad_norm:
                 clip_factor = true_norm / self.clip_grad_norm
 
         # fused adam over all bucket shards in one multi-tensor batch per group;
         # offloaded buckets step on the host (grad D2H -> CPU adam -> param H2D)
         for gi, (group, buckets) in enumerate(zip(self.optim.param_groups, self._group_buckets)):
+            world, rank = self._g_world[gi], self._g_rank[gi]
+            div_scale = loss_scale * self._g_div[gi]  # default divisor == group world (1 if undistributed)
             flat_grad = self._flat_grads[gi]
             flat = self._flat_params[gi]
             grads, masters, mlist, vlist, outs = [], [], [], [], []
             cpu_jobs = []
             for b in buckets:
                 blen = b.end - b.start
-                gshard = flat_grad[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
-                pshard = flat[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
+                gshard = flat_grad[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
+                pshard = flat[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
                 if getattr(b, "offloaded", False):
                     cpu_jobs.append((b, gshard, pshard))
                 else:
@@ -379,18 +398,20 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
             for b, gshard, pshard in cpu_jobs:
                 self._cpu_step(group, b, gshard, pshard, div_scale * clip_factor)
 
-        # all-gather updated working params (bucket-wise, async on comm stream)
-        if self.world > 1:
-            works = []
-            for gi, buckets in enumerate(self._group_buckets):
-                flat = self._flat_params[gi]
-                for b in buckets:
-                    blen = b.end - b.start
-                    seg = flat[b.start : b.end]
-                    shard = flat[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
-                    works.append(dist.all_gather_into_tensor(seg, shard.clone(), group=self.dp_pg, async_op=True))
-            for wk in works:
-                wk.wait()
+        # all-gather updated working params (bucket-wise, async)
+        works = []
+        for gi, buckets in enumerate(self._group_buckets):
+            world, rank, pg = self._g_world[gi], self._g_rank[gi], self._g_pg[gi]
+            if world == 1:
+                continue
+            flat = self._flat_params[gi]
+            for b in buckets:
+                blen = b.end - b.start
+                seg = flat[b.start : b.end]
+                shard = flat[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
+                works.append(dist.all_gather_into_tensor(seg, shard.clone(), group=pg, async_op=True))
+        for wk in works:
+            wk.wait()
         self.zero_grad()
 
     def _fused_step(self, group, grads, masters, mlist, vlist, outs, div_scale):
@@ -465,12 +486,13 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
     def update_master_params(self, model: torch.nn.Module):
         """Re-derive master shards from (newly loaded) working params."""
         for gi, buckets in enumerate(self._group_buckets):
+            world, rank = self._g_world[gi], self._g_rank[gi]
             flat = self._flat_params[gi]
             for b in buckets:
                 blen = b.end - b.start
-                shard = flat[b.start + self.rank * blen // self.world : b.start + (self.rank + 1) * blen // self.world]
+                shard = flat[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
                 if self.master_weights:
-                    b.master.copy_(shard.float())
+                    b.master.copy_(shard.detach().float().to(b.master.device))
 
     @property
     def loss_scale(self) -> float:
