@@ -280,14 +280,26 @@ class LlamaForCausalLM(nn.Module):
         if stage_range is not None and stage_range[1] < len(self.model.layers):
             return {"hidden_states": out}
         hidden = out
-        logits = self.lm_head(hidden)
         loss = None
         if labels is not None:
-            if getattr(self, "_sp_labels_shifted", False):
+            parallel = getattr(self, "tp_group", None) is not None and getattr(self, "parallel_logits", False)
+            pre_shifted = getattr(self, "_sp_labels_shifted", False)
+            if not parallel and getattr(self, "use_fused_loss", True):
+                # chunked fused linear+CE: the full logits are never built
+                from ..ops.fused_ce import fused_linear_cross_entropy
+
+                if pre_shifted:
+                    h_in, lab = hidden, labels
+                else:
+                    h_in, lab = hidden[:, :-1, :], labels[:, 1:]
+                loss = fused_linear_cross_entropy(h_in, self.lm_head.weight, lab)
+                return {"logits": None, "loss": loss}
+            logits = self.lm_head(hidden)
+            if pre_shifted:
                 shift_logits, shift_labels = logits, labels.contiguous()
             else:
                 shift_logits, shift_labels = logits[:, :-1, :], labels[:, 1:].contiguous()
-            if getattr(self, "tp_group", None) is not None and getattr(self, "parallel_logits", False):
+            if parallel:
                 # vocab-parallel CE over the tp group (logits stay sharded)
                 from ..shardformer.layer.loss import dist_cross_entropy
 
@@ -298,6 +310,8 @@ class LlamaForCausalLM(nn.Module):
                     shift_labels.view(-1),
                     ignore_index=-100,
                 )
+            return {"logits": logits, "loss": loss}
+        logits = self.lm_head(hidden)
         return {"logits": logits, "loss": loss}
 
     @property

@@ -14,8 +14,11 @@ def test_forward_backward_cpu():
     torch.manual_seed(0)
     m = LlamaForCausalLM(tiny_cfg())
     x = torch.randint(0, 128, (2, 32))
+    # logits-only path
+    assert m(x)["logits"].shape == (2, 32, 128)
+    # fused-CE training path (logits skipped by design)
     out = m(x, labels=x)
-    assert out["logits"].shape == (2, 32, 128)
+    assert out["logits"] is None and out["loss"] is not None
     out["loss"].backward()
     for n, p in m.named_parameters():
         assert p.grad is not None, n
@@ -50,3 +53,13 @@ def test_configs_table():
         + c.hidden_size
     )
     assert 6.5e9 < n < 7.1e9
+
+
+def test_fused_loss_matches_unfused():
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(tiny_cfg())
+    x = torch.randint(0, 128, (2, 32))
+    loss_fused = m(x, labels=x)["loss"]
+    m.use_fused_loss = False
+    loss_plain = m(x, labels=x)["loss"]
+    torch.testing.assert_close(loss_fused, loss_plain, rtol=1e-5, atol=1e-6)
