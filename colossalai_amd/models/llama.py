@@ -177,6 +177,7 @@ class LlamaModel(nn.Module):
         self.norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
         self.eps = cfg.rms_norm_eps
         self.gradient_checkpointing = cfg.gradient_checkpointing
+        self.ckpt_ratio = 1.0
         self._rope_table = None
 
     def rope_table(self, device) -> torch.Tensor:
@@ -212,9 +213,10 @@ class LlamaModel(nn.Module):
             device = hidden_states.device
         table = self.rope_table(device)
         hidden = rms_norm(residual, self.layers[start].input_layernorm_weight, self.eps)
+        n_ckpt = int(len(self.layers) * getattr(self, "ckpt_ratio", 1.0) + 0.999)
         for i in range(start, end):
             layer = self.layers[i]
-            if self.gradient_checkpointing and self.training:
+            if self.gradient_checkpointing and self.training and i < n_ckpt:
                 out, residual = torch.utils.checkpoint.checkpoint(
                     layer, hidden, residual, table, use_reentrant=False
                 )
@@ -246,8 +248,11 @@ class LlamaForCausalLM(nn.Module):
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
 
-    def gradient_checkpointing_enable(self):
+    def gradient_checkpointing_enable(self, ratio: float = 1.0):
+        """Checkpoint the first ``ratio`` fraction of layers (288 GB HBM often
+        leaves room to keep the tail of the stack un-checkpointed)."""
         self.model.gradient_checkpointing = True
+        self.model.ckpt_ratio = ratio
 
     def gradient_checkpointing_disable(self):
         self.model.gradient_checkpointing = False
