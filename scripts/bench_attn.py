@@ -1,5 +1,11 @@
 """Isolated flash-attention micro-benchmark (TF/s for fwd and bwd kernels)."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import math
 import time
 

@@ -1,6 +1,12 @@
 """Bisect the flash-attention backward mismatch: check lse, delta, dV, dK, dQ
 independently against fp32 references on a minimal config."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import math
 import torch
 
