@@ -72,10 +72,24 @@ DEV_INLINE __bf16 f2b(float f) {
 DEV_INLINE constexpr int crow(int j, int half) { return (j & 3) + 8 * (j >> 2) + 4 * half; }
 
 // Swizzled byte offset inside a row-major LDS tile with ROWB bytes per row.
-// XOR spreads a column access over 8 distinct 16-byte slots (guide §6 G4).
+// XOR spreads a column access across ALL of the row's 16-byte slots
+// (ROWB/16 of them) — guide §6 G4; PMC-verified: an 8-slot mask on 256 B
+// rows left 4-way read / 8-way write conflicts.
 template <int ROWB>
 DEV_INLINE int swz(int row, int byte_in_row) {
-  return row * ROWB + (byte_in_row ^ ((row & 7) << 4));
+  constexpr int CPR_MASK = (ROWB / 16) - 1;
+  return row * ROWB + (byte_in_row ^ ((row & CPR_MASK) << 4));
+}
+
+// Swizzle for TRANSPOSED tiles ([D][rows]): the mask folds d with d>>3 so it
+// varies across lanes both when STORING (lanes sweep ch = d>>3 at fixed
+// d&7) and when READING B-fragments (lanes sweep d&7 at fixed d>>3) —
+// a plain (d&7) mask degenerates to a 16-way conflict on the store side
+// (PMC: conflict cycles ~3x busy). No per-lane data rotation → no
+// runtime-indexed vector access → no scratch (guide rule #20).
+template <int ROWB>
+DEV_INLINE int swzT(int d, int byte_in_row) {
+  return d * ROWB + (byte_in_row ^ (((d ^ (d >> 3)) & 7) << 4));
 }
 
 // ---- cooperative staging helpers (512 threads) ----------------------------
@@ -105,7 +119,7 @@ DEV_INLINE void stage_transposed(char* ldsT, const unsigned short* base, long to
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       const int d = ch * 8 + e;
-      *reinterpret_cast<unsigned short*>(ldsT + swz<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
+      *reinterpret_cast<unsigned short*>(ldsT + swzT<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
     }
   }
 }
@@ -123,7 +137,7 @@ DEV_INLINE void stage_both(char* lds, char* ldsT, const unsigned short* base, lo
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       const int d = ch * 8 + e;
-      *reinterpret_cast<unsigned short*>(ldsT + swz<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
+      *reinterpret_cast<unsigned short*>(ldsT + swzT<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
     }
   }
 }
@@ -162,7 +176,7 @@ struct RegStage {
       const int row = c / CPR, ch = c % CPR;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        *reinterpret_cast<unsigned short*>(ldsT + swz<ROWS * 2>(ch * 8 + e, row * 2)) = bf_raw(r[i][e]);
+        *reinterpret_cast<unsigned short*>(ldsT + swzT<ROWS * 2>(ch * 8 + e, row * 2)) = bf_raw(r[i][e]);
       }
     }
   }
@@ -325,7 +339,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
         // B: lane col = d (nb*32+ln), k-rows = ks*16 + half*8 + [0..7] → V_T[d][k]
-        bf16x8_t vb = ld_lds16(VT, swz<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        bf16x8_t vb = ld_lds16(VT, swzT<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
         __builtin_amdgcn_s_setprio(1);
         oacc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, oacc[nb], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
@@ -552,7 +566,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
         bf16x8_t pa = ld_lds16(P, swz<QB * 2>(ln, (ks * 16 + half * 8) * 2));
 #pragma unroll
         for (int nb = 0; nb < D / 32; ++nb) {
-          bf16x8_t dob = ld_lds16(dOT, swz<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+          bf16x8_t dob = ld_lds16(dOT, swzT<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
           __builtin_amdgcn_s_setprio(1);
           dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[nb], 0, 0, 0);
           __builtin_amdgcn_s_setprio(0);
@@ -592,7 +606,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
         bf16x8_t dsa = ld_lds16(P, swz<QB * 2>(ln, (ks * 16 + half * 8) * 2));
 #pragma unroll
         for (int nb = 0; nb < D / 32; ++nb) {
-          bf16x8_t qtb = ld_lds16(QT, swz<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+          bf16x8_t qtb = ld_lds16(QT, swzT<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
           __builtin_amdgcn_s_setprio(1);
           dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[nb], 0, 0, 0);
           __builtin_amdgcn_s_setprio(0);
@@ -774,7 +788,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
       bf16x8_t dsa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
-        bf16x8_t ktb = ld_lds16(KT, swz<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        bf16x8_t ktb = ld_lds16(KT, swzT<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
         __builtin_amdgcn_s_setprio(1);
         dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, ktb, dq_acc[nb], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
