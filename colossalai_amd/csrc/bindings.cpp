@@ -31,6 +31,10 @@ at::Tensor swiglu_bwd(at::Tensor dout, at::Tensor gate_up);
 // mfma_selftest.hip
 std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor A32, at::Tensor B32);
 
+// decode_attn.hip
+at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, at::Tensor seq_lens,
+                            double scale);
+
 // flash_attn.hip
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale);
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
@@ -50,6 +54,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &cai::swiglu_fwd, "fused SwiGLU forward");
   m.def("swiglu_bwd", &cai::swiglu_bwd, "fused SwiGLU backward");
   m.def("mfma_selftest", &cai::mfma_selftest, "MFMA layout self-test probes");
+  m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache");
   m.def("flash_attn_fwd", &cai::flash_attn_fwd, "flash attention forward (bf16, causal, GQA)");
   m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward");
 }

@@ -1,0 +1,147 @@
+// Single-token (decode) attention over a KV cache, gfx950.
+//
+// Memory-bound: the whole job is streaming K/V rows at HBM speed. One
+// 256-thread block per (batch, q-head); each 16-lane group owns a strided
+// subset of cache positions with its own online-softmax partial (m, l,
+// o[128]) — 16 partials merge via LDS with the standard LSE rescale.
+// K/V loads are 16 B/lane vectorized (guide §6 G13); GQA folds kv-head
+// selection into pointer math.
+//
+// Reference equivalent: extensions/csrc/kernel/cuda/flash_decoding_attention
+// _kernel.cu (contiguous-cache variant; paged cache lands with the paged
+// KV manager).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace cai {
+
+constexpr int DEC_BLOCK = 256;
+constexpr int GROUPS = DEC_BLOCK / 16;  // 16-lane groups
+
+template <int D>
+__global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
+    const unsigned short* __restrict__ Q,   // [B, Hq, D]
+    const unsigned short* __restrict__ KC,  // [B, Smax, Hkv, D]
+    const unsigned short* __restrict__ VC,  // [B, Smax, Hkv, D]
+    unsigned short* __restrict__ O,         // [B, Hq, D]
+    const int* __restrict__ seq_lens,       // [B] (length INCLUDING current token)
+    int B, int Smax, int Hq, int Hkv, float scale) {
+  constexpr int EPL = D / 16;  // elements per lane (8 for D=128)
+  __shared__ float sm_m[GROUPS];
+  __shared__ float sm_l[GROUPS];
+  __shared__ float sm_o[GROUPS][D];
+
+  const int bh = blockIdx.x;
+  const int b = bh / Hq, h = bh % Hq;
+  const int hk = h / (Hq / Hkv);
+  const int S = seq_lens[b];
+  const int g = threadIdx.x / 16;   // group id
+  const int e = threadIdx.x % 16;   // lane-in-group: owns d = e*EPL..+EPL
+
+  const unsigned short* q = Q + ((long)b * Hq + h) * D + e * EPL;
+  const long kv_tok = (long)Hkv * D;
+  const unsigned short* kbase = KC + (long)b * Smax * kv_tok + (long)hk * D + e * EPL;
+  const unsigned short* vbase = VC + (long)b * Smax * kv_tok + (long)hk * D + e * EPL;
+
+  float qf[EPL];
+#pragma unroll
+  for (int j = 0; j < EPL; ++j) qf[j] = bf2f(q[j]);
+
+  float m = -INFINITY, l = 0.0f, o[EPL];
+#pragma unroll
+  for (int j = 0; j < EPL; ++j) o[j] = 0.0f;
+
+  for (int s = g; s < S; s += GROUPS) {
+    const unsigned short* kp = kbase + (long)s * kv_tok;
+    float dot = 0.0f;
+    if constexpr (EPL == 8) {
+      short8 kv = *reinterpret_cast<const short8*>(kp);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f((unsigned short)kv[j]);
+    } else {
+      short4v kv = *reinterpret_cast<const short4v*>(kp);
+#pragma unroll
+      for (int j = 0; j < EPL; ++j) dot += qf[j] * bf2f((unsigned short)kv[j]);
+    }
+    // sum over the 16 lanes of the group
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) dot += __shfl_xor(dot, off);
+    const float sc = dot * scale;
+    const float m_new = fmaxf(m, sc);
+    const float corr = (m == -INFINITY) ? 0.0f : __expf(m - m_new);
+    const float p = __expf(sc - m_new);
+    l = l * corr + p;
+    const unsigned short* vp = vbase + (long)s * kv_tok;
+    if constexpr (EPL == 8) {
+      short8 vv = *reinterpret_cast<const short8*>(vp);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o[j] = o[j] * corr + p * bf2f((unsigned short)vv[j]);
+    } else {
+      short4v vv = *reinterpret_cast<const short4v*>(vp);
+#pragma unroll
+      for (int j = 0; j < EPL; ++j) o[j] = o[j] * corr + p * bf2f((unsigned short)vv[j]);
+    }
+    m = m_new;
+  }
+
+  // merge the GROUPS partials
+  if (e == 0) {
+    sm_m[g] = m;
+    sm_l[g] = l;
+  }
+#pragma unroll
+  for (int j = 0; j < EPL; ++j) sm_o[g][e * EPL + j] = o[j];
+  __syncthreads();
+
+  if (g == 0) {
+    float m_all = -INFINITY;
+#pragma unroll
+    for (int i = 0; i < GROUPS; ++i) m_all = fmaxf(m_all, sm_m[i]);
+    float l_all = 0.0f;
+    float acc[EPL];
+#pragma unroll
+    for (int j = 0; j < EPL; ++j) acc[j] = 0.0f;
+#pragma unroll
+    for (int i = 0; i < GROUPS; ++i) {
+      const float c = (sm_m[i] == -INFINITY) ? 0.0f : __expf(sm_m[i] - m_all);
+      l_all += sm_l[i] * c;
+#pragma unroll
+      for (int j = 0; j < EPL; ++j) acc[j] += sm_o[i][e * EPL + j] * c;
+    }
+    const float inv_l = l_all > 0.0f ? 1.0f / l_all : 0.0f;
+    unsigned short* op = O + ((long)b * Hq + h) * D + e * EPL;
+#pragma unroll
+    for (int j = 0; j < EPL; ++j) op[j] = f2bf(acc[j] * inv_l);
+  }
+}
+
+at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, at::Tensor seq_lens,
+                            double scale) {
+  TORCH_CHECK(q.dim() == 3 && q.scalar_type() == at::kBFloat16 && q.is_contiguous(), "q must be [B,Hq,D] bf16");
+  TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous(), "kv cache must be contiguous");
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
+  const int B = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+  const int Smax = (int)kcache.size(1), Hkv = (int)kcache.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "decode_attention: head dim 64/128");
+  auto out = at::empty_like(q);
+  auto stream = at::hip::getCurrentHIPStream();
+  const dim3 grid(B * Hq);
+  if (D == 128) {
+    hipLaunchKernelGGL((decode_attn_kernel<128>), grid, dim3(DEC_BLOCK), 0, stream.stream(),
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kcache.data_ptr(),
+                       (const unsigned short*)vcache.data_ptr(), (unsigned short*)out.data_ptr(),
+                       seq_lens.data_ptr<int>(), B, Smax, Hq, Hkv, (float)scale);
+  } else {
+    hipLaunchKernelGGL((decode_attn_kernel<64>), grid, dim3(DEC_BLOCK), 0, stream.stream(),
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kcache.data_ptr(),
+                       (const unsigned short*)vcache.data_ptr(), (unsigned short*)out.data_ptr(),
+                       seq_lens.data_ptr<int>(), B, Smax, Hq, Hkv, (float)scale);
+  }
+  HIP_CHECK_LAST();
+  return out;
+}
+
+}  // namespace cai

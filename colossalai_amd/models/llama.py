@@ -136,6 +136,57 @@ class LlamaAttention(nn.Module):
         return self.o_proj(attn.reshape(B, S, -1))
 
 
+    @torch.no_grad()
+    def forward_with_cache(self, hidden, rope_table, kcache, vcache, positions, seq_lens, prefill: bool):
+        """Inference path: RoPE + cache append + (flash prefill | decode) attention."""
+        from ..ops import has_kernels
+        from ..ops.attention import attention_ref
+        from ..ops.rope import apply_rope_ref
+
+        B, S, _ = hidden.shape
+        Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+        qkv = self.qkv_proj(hidden)
+        q = qkv[:, :, : Hq * D].view(B, S, Hq, D)
+        k = qkv[:, :, Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+        v = qkv[:, :, (Hq + Hkv) * D :].view(B, S, Hkv, D)
+        use_hip_path = hidden.is_cuda and has_kernels()
+        if use_hip_path:
+            from ..ops import kernels
+
+            kernels().rope_inplace(q, k, rope_table, positions, False)
+        else:
+            q2, k2 = apply_rope_ref(q, k, rope_table, positions.long(), S, False)
+            q, k = q2, k2
+        if prefill:
+            kcache[:, :S] = k
+            vcache[:, :S] = v
+            if use_hip_path:
+                from ..ops import kernels
+
+                out, _ = kernels().flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), True, self.scale)
+            else:
+                out = attention_ref(q, k, v, causal=True, scale=self.scale)
+        else:
+            idx = (seq_lens - 1).long()
+            kcache[torch.arange(B, device=hidden.device), idx] = k.squeeze(1)
+            vcache[torch.arange(B, device=hidden.device), idx] = v.squeeze(1)
+            if use_hip_path:
+                from ..ops import kernels
+
+                out = kernels().decode_attention(q.squeeze(1).contiguous(), kcache, vcache,
+                                                 seq_lens.int(), self.scale).unsqueeze(1)
+            else:
+                # CPU reference decode: full attention over the cache prefix
+                outs = []
+                for b in range(B):
+                    n = int(seq_lens[b])
+                    ob = attention_ref(q[b : b + 1], kcache[b : b + 1, :n], vcache[b : b + 1, :n],
+                                       causal=False, scale=self.scale)
+                    outs.append(ob)
+                out = torch.cat(outs, dim=0)
+        return self.o_proj(out.reshape(B, S, -1))
+
+
 class LlamaMLP(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
