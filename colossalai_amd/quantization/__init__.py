@@ -1,0 +1,10 @@
+from .fp8 import all_gather_fp8, all_reduce_fp8, all_to_all_single_fp8, cast_from_fp8, cast_to_fp8, reduce_scatter_fp8
+
+__all__ = [
+    "cast_to_fp8",
+    "cast_from_fp8",
+    "all_reduce_fp8",
+    "all_gather_fp8",
+    "reduce_scatter_fp8",
+    "all_to_all_single_fp8",
+]

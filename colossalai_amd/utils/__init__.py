@@ -1,0 +1,5 @@
+from .device import get_current_device, free_port_util
+from .memory import report_memory_usage
+from .timer import MultiTimer, Timer
+
+__all__ = ["Timer", "MultiTimer", "get_current_device", "report_memory_usage", "free_port_util"]
