@@ -100,12 +100,9 @@ class HybridParallelCheckpointIO(GeneralCheckpointIO):
                              use_safetensors: bool = False, use_async: bool = False):
         if isinstance(model, ModelWrapper):
             model = model.unwrap()
-        if self.dp_rank != 0:
-            dist.barrier()
-            return
-        local_sd = _full_state_dict(model, self.tp_group) if self.tp_rank == 0 else {}
-        # merge pipeline stages on pp rank 0
-        if self.pp_size > 1 and self.tp_rank == 0:
+        # TP weight gathers are collective: every rank participates
+        local_sd = _full_state_dict(model, self.tp_group)
+        if self.pp_size > 1:
             gathered = [None] * self.pp_size
             dist.all_gather_object(gathered, local_sd, group=self.pp_group)
             merged = {}
@@ -129,11 +126,8 @@ class HybridParallelCheckpointIO(GeneralCheckpointIO):
         # v1: gather to full on rank 0 then reuse the general sharded writer
         if isinstance(model, ModelWrapper):
             model = model.unwrap()
-        if self.dp_rank != 0:
-            dist.barrier()
-            return
-        local_sd = _full_state_dict(model, self.tp_group) if self.tp_rank == 0 else {}
-        if self.pp_size > 1 and self.tp_rank == 0:
+        local_sd = _full_state_dict(model, self.tp_group)
+        if self.pp_size > 1:
             gathered = [None] * self.pp_size
             dist.all_gather_object(gathered, local_sd, group=self.pp_group)
             merged = {}

@@ -19,11 +19,14 @@ _TABLE_CACHE = {}
 def build_rope_table(max_pos: int, dim: int, theta: float = 10000.0, device="cpu") -> torch.Tensor:
     key = (max_pos, dim, theta, str(device))
     if key not in _TABLE_CACHE:
-        inv_freq = 1.0 / (theta ** (torch.arange(0, dim, 2, dtype=torch.float64) / dim))
-        t = torch.arange(max_pos, dtype=torch.float64)
-        freqs = torch.outer(t, inv_freq)  # [max_pos, dim/2]
-        table = torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().to(device)
-        _TABLE_CACHE[key] = table.contiguous()
+        # never create the cached table as an inference tensor: it is shared
+        # with training graphs that must save it for backward
+        with torch.inference_mode(False):
+            inv_freq = 1.0 / (theta ** (torch.arange(0, dim, 2, dtype=torch.float64) / dim))
+            t = torch.arange(max_pos, dtype=torch.float64)
+            freqs = torch.outer(t, inv_freq)  # [max_pos, dim/2]
+            table = torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().to(device)
+            _TABLE_CACHE[key] = table.contiguous()
     return _TABLE_CACHE[key]
 
 
