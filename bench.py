@@ -24,7 +24,14 @@ def parse_args():
     p.add_argument("--model", type=str, default="llama-7b")
     p.add_argument("--batch", type=int, default=36, help="per-DP-rank batch size (reference: bs/DP=36)")
     p.add_argument("--seq", type=int, default=4096)
-    p.add_argument("--plugin", type=str, default="zero2", choices=["ddp", "zero2", "zero1"])
+    p.add_argument("--plugin", type=str, default="zero2",
+                   choices=["ddp", "zero2", "zero1", "gemini", "hybrid", "moe"])
+    p.add_argument("--tp", type=int, default=1)
+    p.add_argument("--pp", type=int, default=1)
+    p.add_argument("--sp", type=int, default=1)
+    p.add_argument("--ep", type=int, default=1)
+    p.add_argument("--zero", type=int, default=1, help="zero stage inside hybrid/moe plugins")
+    p.add_argument("--microbatches", type=int, default=None)
     p.add_argument("--grad-ckpt", dest="grad_ckpt", action="store_true", default=True)
     p.add_argument("--ckpt-ratio", type=float, default=1.0, help="fraction of layers checkpointed")
     p.add_argument("--no-grad-ckpt", dest="grad_ckpt", action="store_false")
@@ -46,10 +53,17 @@ def main():
     if device == "cpu":
         raise SystemExit("bench.py requires an MI355X GPU")
 
-    cfg = LLAMA_CONFIGS[args.model]
+    if args.model.startswith("mixtral"):
+        from colossalai_amd.models.mixtral import MIXTRAL_CONFIGS, MixtralForCausalLM
+
+        cfg = MIXTRAL_CONFIGS[args.model]
+        model_cls = MixtralForCausalLM
+    else:
+        cfg = LLAMA_CONFIGS[args.model]
+        model_cls = LlamaForCausalLM
     torch.manual_seed(42)
     with torch.device("meta"):
-        model = LlamaForCausalLM(cfg)
+        model = model_cls(cfg)
     model = model.to_empty(device=device).to(torch.bfloat16)
     # cheap random init (init on meta then re-init materialized weights)
     with torch.no_grad():
@@ -72,6 +86,36 @@ def main():
             backward = lambda loss: booster.backward(loss, optimizer)
         else:
             backward = lambda loss: loss.backward()
+    elif args.plugin in ("hybrid", "moe"):
+        from colossalai_amd import Booster
+        from colossalai_amd.booster.plugin import HybridParallelPlugin, MoeHybridParallelPlugin
+        from colossalai_amd.nn import FusedAdam
+
+        kwargs = dict(tp_size=args.tp, pp_size=args.pp, sp_size=args.sp, precision="bf16",
+                      zero_stage=args.zero, num_microbatches=args.microbatches)
+        if args.sp > 1:
+            kwargs.update(enable_sequence_parallelism=True, sequence_parallelism_mode="all_to_all")
+        if args.plugin == "moe":
+            plugin = MoeHybridParallelPlugin(ep_size=args.ep, **kwargs)
+        else:
+            plugin = HybridParallelPlugin(**kwargs)
+        optimizer = FusedAdam(model.parameters(), lr=1e-5, weight_decay=0.1)
+        booster = Booster(plugin=plugin)
+        model, optimizer, *_ = booster.boost(model, optimizer)
+        if args.pp > 1:
+            criterion = lambda out, micro: out["loss"]
+            backward = None  # handled inside execute_pipeline
+        else:
+            backward = lambda loss: booster.backward(loss, optimizer)
+    elif args.plugin == "gemini":
+        from colossalai_amd import Booster
+        from colossalai_amd.booster.plugin import GeminiPlugin
+        from colossalai_amd.nn import HybridAdam
+
+        optimizer = HybridAdam(model.parameters(), lr=1e-5, weight_decay=0.1)
+        booster = Booster(plugin=GeminiPlugin(precision="bf16"))
+        model, optimizer, *_ = booster.boost(model, optimizer)
+        backward = lambda loss: booster.backward(loss, optimizer)
     else:
         from colossalai_amd import Booster
         from colossalai_amd.booster.plugin import LowLevelZeroPlugin
@@ -86,13 +130,21 @@ def main():
     B, S = args.batch, args.seq
     data = torch.randint(0, cfg.vocab_size, (B, S), device=device)
 
-    def step():
-        out = model(data, labels=data)
-        loss = out["loss"]
-        backward(loss)
-        optimizer.step()
-        optimizer.zero_grad()
-        return loss
+    if args.plugin in ("hybrid", "moe") and args.pp > 1:
+        def step():
+            batch = {"input_ids": data, "labels": data}
+            result = booster.execute_pipeline(iter([batch]), model, criterion, optimizer, return_loss=True)
+            optimizer.step()
+            optimizer.zero_grad()
+            return result["loss"]
+    else:
+        def step():
+            out = model(data, labels=data)
+            loss = out["loss"]
+            backward(loss)
+            optimizer.step()
+            optimizer.zero_grad()
+            return loss
 
     for _ in range(args.warmup):
         step()
@@ -136,7 +188,7 @@ def main():
                 "model": args.model,
                 "global_batch": B * world,
                 "seq_len": S,
-                "parallelism": f"{args.plugin}(dp{world})",
+                "parallelism": f"{args.plugin}(dp{world // (args.tp * args.pp * args.sp)} tp{args.tp} pp{args.pp} sp{args.sp})".format(args=args, world=world) if args.plugin in ("hybrid", "moe") else f"{args.plugin}(dp{world})",
                 "grad_ckpt": args.grad_ckpt,
                 "tflops_per_gpu": round(tflops_per_gpu, 1),
                 "params": numel,
