@@ -12,9 +12,8 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     # native models
     "LlamaForCausalLM": ("colossalai_amd.shardformer.policies.llama", "LlamaForCausalLMPolicy"),
     "LlamaModel": ("colossalai_amd.shardformer.policies.llama", "LlamaPolicy"),
-    # HF transformers models
-    "GPT2LMHeadModel": ("colossalai_amd.shardformer.policies.gpt2", "GPT2LMHeadModelPolicy"),
     "MixtralForCausalLM": ("colossalai_amd.shardformer.policies.mixtral", "MixtralForCausalLMPolicy"),
+    "MixtralModel": ("colossalai_amd.shardformer.policies.mixtral", "MixtralPolicy"),
 }
 
 
