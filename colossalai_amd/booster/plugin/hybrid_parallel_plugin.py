@@ -151,11 +151,14 @@ class HybridParallelPlugin(Plugin):
         overlap_communication: bool = True,
         parallel_output: bool = True,
         pp_style: str = "1f1b",
+        num_model_chunks: int = 1,
         **kwargs,
     ):
         assert dist.is_initialized(), "launch colossalai_amd before creating HybridParallelPlugin"
         assert zero_stage in (0, 1, 2)
-        assert pp_style == "1f1b", "only 1f1b pipeline style is implemented so far"
+        assert pp_style in ("1f1b", "interleaved"), f"unsupported pp_style {pp_style}"
+        if pp_style == "interleaved":
+            assert num_model_chunks > 1, "interleaved pipeline needs num_model_chunks > 1"
         world = dist.get_world_size()
         if sp_size is None:
             sp_size = 1
@@ -189,14 +192,26 @@ class HybridParallelPlugin(Plugin):
 
         self.stage_manager = None
         self.scheduler = None
+        self.pp_style = pp_style
+        self.num_model_chunks = num_model_chunks
         if pp_size > 1:
             assert num_microbatches is not None or microbatch_size is not None, (
                 "pipeline parallelism requires num_microbatches or microbatch_size"
             )
-            self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
-            self.scheduler = OneForwardOneBackwardSchedule(
-                self.stage_manager, num_microbatches=num_microbatches, microbatch_size=microbatch_size
-            )
+            if pp_style == "interleaved":
+                from ...pipeline.schedule.interleaved_pp import InterleavedSchedule
+
+                self.stage_manager = PipelineStageManager(
+                    self.pg_mesh, PP_AXIS, enable_interleave=True, num_model_chunks=num_model_chunks
+                )
+                self.scheduler = InterleavedSchedule(
+                    self.stage_manager, num_model_chunks=num_model_chunks, num_microbatches=num_microbatches
+                )
+            else:
+                self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
+                self.scheduler = OneForwardOneBackwardSchedule(
+                    self.stage_manager, num_microbatches=num_microbatches, microbatch_size=microbatch_size
+                )
 
         self.shard_config = ShardConfig(
             tensor_parallel_process_group=self.tp_group if tp_size > 1 else None,
@@ -288,6 +303,28 @@ class HybridParallelPlugin(Plugin):
         inner = model.model if hasattr(model, "model") else model
         assert hasattr(inner, "layers"), "pipeline parallelism needs a .layers decoder stack"
         n_layers = len(inner.layers)
+        if self.pp_style == "interleaved":
+            V, pp = self.num_model_chunks, self.pp_size
+            per = [n_layers // (pp * V)] * (pp * V)
+            for i in range(n_layers % (pp * V)):
+                per[i] += 1
+            starts = [sum(per[:i]) for i in range(pp * V)]
+            held = []
+            model.chunk_ranges = []
+            for c in range(V):
+                vs = c * pp + self.stage_manager.stage
+                rng = (starts[vs], starts[vs] + per[vs])
+                model.chunk_ranges.append(rng)
+                held.extend(range(*rng))
+            model.stage_range = model.chunk_ranges[0]
+            for i in range(n_layers):
+                if i not in held:
+                    inner.layers[i] = _StageStub()
+            if not self.stage_manager.is_first_stage() and hasattr(inner, "embed_tokens"):
+                inner.embed_tokens = _StageStub()
+            if not self.stage_manager.is_last_stage(self.num_model_chunks - 1) and hasattr(model, "lm_head"):
+                model.lm_head = _StageStub()
+            return
         start, end = self.stage_manager.stage_index(n_layers)
         model.stage_range = (start, end)
         for i in range(n_layers):

@@ -313,8 +313,12 @@ class LlamaForCausalLM(nn.Module):
         input_ids: Optional[torch.Tensor] = None,
         labels: Optional[torch.Tensor] = None,
         hidden_states: Optional[torch.Tensor] = None,
+        pp_chunk: Optional[int] = None,
     ):
-        stage_range = getattr(self, "stage_range", None)
+        if pp_chunk is not None:
+            stage_range = self.chunk_ranges[pp_chunk]
+        else:
+            stage_range = getattr(self, "stage_range", None)
         sp_group = getattr(self, "sp_group", None)
         if sp_group is not None and input_ids is not None:
             # Ulysses SP: each rank runs its sequence shard; labels are shifted

@@ -35,9 +35,25 @@ class PipelineP2PCommunication:
         self._send_meta: Dict[int, bool] = {}  # peer -> header already sent
         self._recv_meta: Dict[int, Tuple[torch.Size, torch.dtype]] = {}
         self._device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        self._pending = []  # outstanding (work, tensor) isends
 
     # ------------------------------------------------------------- internals
+    def _isend(self, tensor: torch.Tensor, peer: int) -> None:
+        work = dist.isend(tensor, peer)
+        self._pending.append((work, tensor))
+        if len(self._pending) > 32:
+            w, _ = self._pending.pop(0)
+            w.wait()
+
+    def flush_sends(self) -> None:
+        for w, _ in self._pending:
+            w.wait()
+        self._pending.clear()
+
     def _send_tensor(self, tensor: torch.Tensor, peer: int) -> None:
+        # sends are NON-BLOCKING: interleaved/1F1B warmups legitimately have
+        # both neighbors sending before anyone receives (rendezvous sends
+        # deadlock there). Tensors are kept alive until the work completes.
         tensor = tensor.contiguous()
         if not self._send_meta.get(peer, False):
             header = torch.zeros(_HEADER_LEN, dtype=torch.int64, device=self._device)
@@ -45,9 +61,9 @@ class PipelineP2PCommunication:
             header[1] = _DTYPE_CODES[tensor.dtype]
             for i, d in enumerate(tensor.shape):
                 header[2 + i] = d
-            dist.send(header, peer)
+            self._isend(header, peer)
             self._send_meta[peer] = True
-        dist.send(tensor, peer)
+        self._isend(tensor, peer)
 
     def _recv_tensor(self, peer: int) -> torch.Tensor:
         if peer not in self._recv_meta:

@@ -156,6 +156,7 @@ class OneForwardOneBackwardSchedule(PipelineSchedule):
             if not is_first and input_grad is not None:
                 self.comm.send_backward(input_grad)
 
+        self.comm.flush_sends()
         result = {"loss": accum_loss.squeeze() if accum_loss is not None else None}
         if outputs is not None:
             result["outputs"] = outputs
