@@ -60,6 +60,17 @@ class HybridParallelModule(ModelWrapper):
         finally:
             self.require_grad_sync = old
 
+    def sync_partial_sp_grads(self):
+        """split_gather SP: norm-weight grads are partial over seq shards —
+        sum them over the tp/sp group so every rank holds the full grad
+        (reference: SeqParallelUtils.allreduce_partial_data_grad)."""
+        group = self.tp_group
+        if group is None or dist.get_world_size(group) == 1:
+            return
+        for p in self.module.parameters():
+            if getattr(p, "_sp_partial_grad", False) and p.grad is not None:
+                dist.all_reduce(p.grad, group=group)
+
     def sync_dp_grads(self):
         """ZeRO-0 path: average grads over the dp group (bucket-coalesced)."""
         if self.dp_group is None or dist.get_world_size(self.dp_group) == 1:
@@ -99,11 +110,13 @@ class HybridParallelNaiveOptimizer(MixedPrecisionOptimizer):
     def backward(self, loss, inputs=None, retain_graph=False, **kwargs):
         super().backward(loss, inputs=inputs, retain_graph=retain_graph, **kwargs)
         if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_partial_sp_grads()
             self.model_wrapper.sync_dp_grads()
 
     def backward_by_grad(self, tensor, grad, inputs=None, retain_graph=False):
         super().backward_by_grad(tensor, grad, inputs=inputs, retain_graph=retain_graph)
         if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_partial_sp_grads()
             self.model_wrapper.sync_dp_grads()
 
 
@@ -117,11 +130,13 @@ class HybridParallelFP32Optimizer(OptimizerWrapper):
     def backward(self, loss, inputs=None, retain_graph=False, **kwargs):
         loss.backward(inputs=inputs, retain_graph=retain_graph, **kwargs)
         if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_partial_sp_grads()
             self.model_wrapper.sync_dp_grads()
 
     def backward_by_grad(self, tensor, grad, inputs=None, retain_graph=False):
         torch.autograd.backward(tensor, grad, inputs=inputs, retain_graph=retain_graph)
         if self.model_wrapper.require_grad_sync:
+            self.model_wrapper.sync_partial_sp_grads()
             self.model_wrapper.sync_dp_grads()
 
 

@@ -133,8 +133,9 @@ class LlamaAttention(nn.Module):
             qkv, rope_table, self.num_heads, self.num_kv_heads, self.head_dim,
             causal=True, scale=self.scale,
         )
-        return self.o_proj(attn.reshape(B, S, -1))
-
+        # under split_gather SP the column-linear gathered the sequence, so
+        # flatten with the attention output's own length, not the input's
+        return self.o_proj(attn.reshape(B, attn.shape[1], -1))
 
     @torch.no_grad()
     def forward_with_cache(self, hidden, rope_table, kcache, vcache, positions, seq_lens, prefill: bool):
@@ -258,6 +259,13 @@ class LlamaModel(nn.Module):
             assert input_ids is not None
             residual = self.embed_tokens(input_ids)
             device = input_ids.device
+            if getattr(self, "sp_split_gather_group", None) is not None:
+                # Megatron-style SP: the residual stream lives on seq shards;
+                # column-linears gather / row-linears reduce-scatter around
+                # attention and MLP (wired by the policy on the linears).
+                from ..shardformer.layer import split_forward_gather_backward
+
+                residual = split_forward_gather_backward(residual, 1, self.sp_split_gather_group)
         else:
             assert hidden_states is not None
             residual = hidden_states
@@ -320,6 +328,7 @@ class LlamaForCausalLM(nn.Module):
         else:
             stage_range = getattr(self, "stage_range", None)
         sp_group = getattr(self, "sp_group", None)
+        sp_mode = getattr(self, "sp_mode", None)
         if sp_group is not None and input_ids is not None:
             # Ulysses SP: each rank runs its sequence shard; labels are shifted
             # globally first so the boundary token is not lost.
@@ -330,16 +339,26 @@ class LlamaForCausalLM(nn.Module):
             S = input_ids.shape[1]
             assert S % sp == 0, f"seq len {S} must divide sp size {sp}"
             shard = S // sp
-            if labels is not None:
-                shifted = torch.full_like(labels, -100)
-                shifted[:, :-1] = labels[:, 1:]
-                labels = shifted[:, rank * shard : (rank + 1) * shard]
-                self._sp_labels_shifted = True
-            input_ids = input_ids[:, rank * shard : (rank + 1) * shard]
+            if sp_mode != "split_gather":
+                # Ulysses: inputs and labels are seq-sharded before embedding;
+                # split_gather keeps full inputs (residual split after embed,
+                # output gathered before the loss).
+                if labels is not None:
+                    shifted = torch.full_like(labels, -100)
+                    shifted[:, :-1] = labels[:, 1:]
+                    labels = shifted[:, rank * shard : (rank + 1) * shard]
+                    self._sp_labels_shifted = True
+                input_ids = input_ids[:, rank * shard : (rank + 1) * shard]
         out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range)
         if stage_range is not None and stage_range[1] < len(self.model.layers):
             return {"hidden_states": out}
         hidden = out
+        if sp_group is not None and sp_mode == "split_gather":
+            # gather the sequence so the loss sees full outputs; backward
+            # splits dy back to shards (reference: gather_sp_output)
+            from ..shardformer.layer import gather_forward_split_backward
+
+            hidden = gather_forward_split_backward(hidden, 1, sp_group)
         loss = None
         if labels is not None:
             parallel = getattr(self, "tp_group", None) is not None and getattr(self, "parallel_logits", False)

@@ -37,6 +37,12 @@ class LlamaPolicy(Policy):
             self.shard_config.enable_sequence_parallelism
             and self.shard_config.sequence_parallelism_mode == "all_to_all"
         )
+        sp_sg = (
+            self.shard_config.enable_sequence_parallelism
+            and self.shard_config.sequence_parallelism_mode == "split_gather"
+        )
+        if sp_sg:
+            assert self.shard_config.tensor_parallel_size > 1 or True, "split_gather uses the tp/sp group"
         if sp_a2a:
             policy[LlamaAttention] = ModulePolicyDescription(
                 attribute_replacement={
@@ -126,9 +132,21 @@ class LlamaForCausalLMPolicy(LlamaPolicy):
         if self.shard_config.enable_tensor_parallelism and tp > 1 and self.shard_config.parallel_output:
             self.model.tp_group = self.shard_config.tensor_parallel_process_group
             self.model.parallel_logits = True
-        if (
-            self.shard_config.enable_sequence_parallelism
-            and self.shard_config.sequence_parallelism_mode == "all_to_all"
-        ):
-            self.model.sp_group = self.shard_config.sequence_parallel_process_group
+        if self.shard_config.enable_sequence_parallelism:
+            mode = self.shard_config.sequence_parallelism_mode
+            if mode == "all_to_all":
+                self.model.sp_group = self.shard_config.sequence_parallel_process_group
+                self.model.sp_mode = "all_to_all"
+            elif mode == "split_gather":
+                # split_gather reuses the tp group (reference semantics)
+                group = self.shard_config.tensor_parallel_process_group
+                self.model.sp_group = group
+                self.model.sp_mode = "split_gather"
+                inner = self.model.model if hasattr(self.model, "model") else self.model
+                inner.sp_split_gather_group = group
+                # norm weights see seq-shard activations: mark their grads as
+                # partial so the plugin sums them over the sp group pre-sync
+                for name, param in self.model.named_parameters():
+                    if "layernorm" in name or name.endswith("norm_weight"):
+                        param._sp_partial_grad = True
         return self.model
