@@ -916,6 +916,18 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
   const long qbs = q.stride(0), qts = q.stride(1), kbs = k.stride(0), kts = k.stride(1);
   const long dqbs = dq.stride(0), dqts = dq.stride(1), dkbs = dk.stride(0), dkts = dk.stride(1);
 
+  // dK/dV and dQ are independent: run them on two streams so the two
+  // latency-bound kernels co-occupy the CUs.
+  static hipStream_t side_stream = nullptr;
+  static hipEvent_t ev_fork = nullptr, ev_join = nullptr;
+  if (side_stream == nullptr) {
+    (void)hipStreamCreateWithFlags(&side_stream, hipStreamNonBlocking);
+    (void)hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
+    (void)hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
+  }
+  (void)hipEventRecord(ev_fork, stream.stream());
+  (void)hipStreamWaitEvent(side_stream, ev_fork, 0);
+
 #define LAUNCH_BWD(DD)                                                                              \
   do {                                                                                              \
     constexpr int QB = 64;                                                                          \
@@ -932,7 +944,7 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
     const size_t lds_q = 2 * (KVB * DD * 2) + DD * KVB * 2 + NWB * (QW * KVB * 2)                   \
                          + 2 * NWB * QW * sizeof(float);                                            \
     set_lds_limit((const void*)fa_bwd_dq_kernel<DD>, lds_q);                                        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), grid_q, dim3(NTB), lds_q, stream.stream(),           \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), grid_q, dim3(NTB), lds_q, side_stream,           \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
@@ -945,6 +957,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
   if (D == 128) LAUNCH_BWD(128);
   else LAUNCH_BWD(64);
 #undef LAUNCH_BWD
+  (void)hipEventRecord(ev_join, side_stream);
+  (void)hipStreamWaitEvent(stream.stream(), ev_join, 0);
   return {dq, dk, dv};
 }
 
