@@ -12,6 +12,14 @@ import json
 import os
 import time
 
+# hipBLASLt autotuned GEMM algos (tuned offline on MI355X, committed in-repo);
+# read-only: shapes missing from the file silently use the default algo.
+_TUNE = os.path.join(os.path.dirname(os.path.abspath(__file__)), "profiles", "tunableop", "results.csv")
+if os.path.exists(_TUNE.replace("results.csv", "results0.csv")) and os.environ.get("CAI_TUNABLEOP", "1") == "1":
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _TUNE)
+
 import torch
 import torch.distributed as dist
 
