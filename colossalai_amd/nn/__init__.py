@@ -1,3 +1,3 @@
-from .optimizer import CPUAdam, FusedAdam, HybridAdam
+from .optimizer import Adafactor, CPUAdam, FusedAdam, HybridAdam, Lamb, Lars
 
-__all__ = ["FusedAdam", "HybridAdam", "CPUAdam"]
+__all__ = ["FusedAdam", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor"]

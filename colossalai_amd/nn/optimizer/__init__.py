@@ -1,5 +1,8 @@
+from .adafactor import Adafactor
 from .cpu_adam import CPUAdam
 from .fused_adam import FusedAdam
 from .hybrid_adam import HybridAdam
+from .lamb import Lamb
+from .lars import Lars
 
-__all__ = ["FusedAdam", "HybridAdam", "CPUAdam"]
+__all__ = ["FusedAdam", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor"]
