@@ -39,7 +39,8 @@ def test_lamb():
 
 
 def test_lars():
-    _converges(Lars, lr=1e-2)
+    # LARS scales lr by eeta*||w||/||g|| — needs a larger base lr to move
+    _converges(Lars, lr=2.0)
 
 
 def test_adafactor():
