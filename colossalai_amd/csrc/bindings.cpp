@@ -20,6 +20,12 @@ std::vector<at::Tensor> rmsnorm_fused_add_fwd(at::Tensor input, at::Tensor resid
                                               double eps, bool save_inv_rms);
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor inv_rms);
 
+// layernorm.hip
+std::vector<at::Tensor> layernorm_fwd(at::Tensor input, at::Tensor gamma, c10::optional<at::Tensor> beta,
+                                      double eps, bool save_stats);
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor mean,
+                                      at::Tensor invstd);
+
 // rope.hip
 void rope_inplace(at::Tensor q, c10::optional<at::Tensor> k, at::Tensor table,
                   c10::optional<at::Tensor> positions, bool backward);
@@ -50,6 +56,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &cai::rmsnorm_fwd, "RMSNorm forward");
   m.def("rmsnorm_fused_add_fwd", &cai::rmsnorm_fused_add_fwd, "fused residual-add RMSNorm forward");
   m.def("rmsnorm_bwd", &cai::rmsnorm_bwd, "RMSNorm backward");
+  m.def("layernorm_fwd", &cai::layernorm_fwd, "LayerNorm forward");
+  m.def("layernorm_bwd", &cai::layernorm_bwd, "LayerNorm backward");
   m.def("rope_inplace", &cai::rope_inplace, "in-place rotary embedding (fwd/bwd)");
   m.def("swiglu_fwd", &cai::swiglu_fwd, "fused SwiGLU forward");
   m.def("swiglu_bwd", &cai::swiglu_bwd, "fused SwiGLU backward");

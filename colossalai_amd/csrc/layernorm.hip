@@ -1,0 +1,196 @@
+// Fused LayerNorm forward/backward for gfx950 (bf16 in/out, fp32 stats).
+//
+// Same MI355X design as rmsnorm.hip: one 256-thread block per row
+// (grid-strided), short8-vectorized bf16 traffic, forward saves (mean,
+// inv_std), backward accumulates dgamma/dbeta per block in LDS with
+// thread-owned columns and a single atomic pass.
+//
+// Reference equivalent: extensions/csrc/kernel/cuda/layer_norm_kernel.cu
+// (warp-32 Welford scheme re-derived for 64-wide waves; plain two-pass
+// moments since bf16 rows <= 8192 are L1-resident after the first read).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace cai {
+
+constexpr int LN_BLOCK = 256;
+constexpr int LN_MAX_H = 32 * LN_BLOCK / 8;  // matches rmsnorm (<= 8192 cols)
+
+__global__ __launch_bounds__(LN_BLOCK) void layernorm_fwd_kernel(
+    unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ input,
+    const unsigned short* __restrict__ gamma,  // [H]
+    const unsigned short* __restrict__ beta,   // [H] or null
+    float* __restrict__ mean_out,              // [rows]
+    float* __restrict__ invstd_out,            // [rows]
+    float eps,
+    long rows,
+    int H) {
+  __shared__ float red_smem[LN_BLOCK / WAVE];
+  const int nvec = H / 8;
+
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* x = input + row * H;
+    unsigned short* y = out + row * H;
+
+    float vals[32 * 8 / 8];  // up to 32 short8 slots
+    float sum = 0.0f;
+    int slot = 0;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK, ++slot) {
+      short8 xv = *reinterpret_cast<const short8*>(x + i * 8);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float f = bf2f((unsigned short)xv[k]);
+        vals[slot * 8 + k] = f;
+        sum += f;
+      }
+    }
+    sum = block_reduce_sum(sum, red_smem);
+    const float mu = sum / (float)H;
+    float var = 0.0f;
+    slot = 0;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK, ++slot) {
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float d = vals[slot * 8 + k] - mu;
+        var += d * d;
+      }
+    }
+    var = block_reduce_sum(var, red_smem);
+    const float invstd = rsqrtf(var / (float)H + eps);
+    if (threadIdx.x == 0) {
+      if (mean_out) mean_out[row] = mu;
+      if (invstd_out) invstd_out[row] = invstd;
+    }
+    slot = 0;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK, ++slot) {
+      short8 gv = *reinterpret_cast<const short8*>(gamma + i * 8);
+      short8 bv;
+      if (beta) bv = *reinterpret_cast<const short8*>(beta + i * 8);
+      short8 yv;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float r = (vals[slot * 8 + k] - mu) * invstd * bf2f((unsigned short)gv[k]);
+        if (beta) r += bf2f((unsigned short)bv[k]);
+        yv[k] = (short)f2bf(r);
+      }
+      *reinterpret_cast<short8*>(y + i * 8) = yv;
+    }
+  }
+}
+
+// dx = invstd * (dy*g - mean(dy*g) - xhat * mean(dy*g*xhat))
+__global__ __launch_bounds__(LN_BLOCK) void layernorm_bwd_kernel(
+    unsigned short* __restrict__ dx,
+    float* __restrict__ dgamma,  // [H] fp32 (pre-zeroed)
+    float* __restrict__ dbeta,   // [H] fp32 (pre-zeroed)
+    const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ gamma,
+    const float* __restrict__ mean,
+    const float* __restrict__ invstd,
+    long rows,
+    int H) {
+  __shared__ float red_smem[LN_BLOCK / WAVE];
+  extern __shared__ float dgb_local[];  // [2*H]: dgamma | dbeta
+  const int nvec = H / 8;
+  float* dg_local = dgb_local;
+  float* db_local = dgb_local + H;
+
+  for (int i = threadIdx.x; i < H; i += LN_BLOCK) {
+    dg_local[i] = 0.0f;
+    db_local[i] = 0.0f;
+  }
+  __syncthreads();
+
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* xr = x + row * H;
+    const unsigned short* dyr = dy + row * H;
+    unsigned short* dxr = dx + row * H;
+    const float mu = mean[row];
+    const float istd = invstd[row];
+
+    float xhat_f[32 * 8 / 8], dyg_f[32 * 8 / 8];
+    float s1 = 0.0f, s2 = 0.0f;
+    int slot = 0;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK, ++slot) {
+      short8 xv = *reinterpret_cast<const short8*>(xr + i * 8);
+      short8 dv = *reinterpret_cast<const short8*>(dyr + i * 8);
+      short8 gv = *reinterpret_cast<const short8*>(gamma + i * 8);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float xh = (bf2f((unsigned short)xv[k]) - mu) * istd;
+        const float df = bf2f((unsigned short)dv[k]);
+        const float dg = df * bf2f((unsigned short)gv[k]);
+        xhat_f[slot * 8 + k] = xh;
+        dyg_f[slot * 8 + k] = dg;
+        s1 += dg;
+        s2 += dg * xh;
+        dg_local[i * 8 + k] += df * xh;
+        db_local[i * 8 + k] += df;
+      }
+    }
+    s1 = block_reduce_sum(s1, red_smem) / (float)H;
+    s2 = block_reduce_sum(s2, red_smem) / (float)H;
+
+    slot = 0;
+    for (int i = threadIdx.x; i < nvec; i += LN_BLOCK, ++slot) {
+      short8 ov;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        ov[k] = (short)f2bf(istd * (dyg_f[slot * 8 + k] - s1 - xhat_f[slot * 8 + k] * s2));
+      }
+      *reinterpret_cast<short8*>(dxr + i * 8) = ov;
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < H; i += LN_BLOCK) {
+    atomicAdd(&dgamma[i], dg_local[i]);
+    atomicAdd(&dbeta[i], db_local[i]);
+  }
+}
+
+std::vector<at::Tensor> layernorm_fwd(at::Tensor input, at::Tensor gamma, c10::optional<at::Tensor> beta,
+                                      double eps, bool save_stats) {
+  const int H = (int)input.size(-1);
+  const long rows = input.numel() / H;
+  TORCH_CHECK(input.is_contiguous() && input.scalar_type() == at::kBFloat16, "layernorm: bf16 contiguous");
+  TORCH_CHECK(H % 8 == 0 && H <= LN_MAX_H, "layernorm: hidden %8==0 and <= 8192");
+  auto out = at::empty_like(input);
+  auto mean = save_stats ? at::empty({rows}, input.options().dtype(at::kFloat)) : at::Tensor();
+  auto invstd = save_stats ? at::empty({rows}, input.options().dtype(at::kFloat)) : at::Tensor();
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = capped_grid(rows, 8192);
+  hipLaunchKernelGGL(layernorm_fwd_kernel, dim3(grid), dim3(LN_BLOCK), 0, stream.stream(),
+                     (unsigned short*)out.data_ptr(), (const unsigned short*)input.data_ptr(),
+                     (const unsigned short*)gamma.data_ptr(),
+                     beta.has_value() ? (const unsigned short*)beta->data_ptr() : nullptr,
+                     save_stats ? mean.data_ptr<float>() : nullptr,
+                     save_stats ? invstd.data_ptr<float>() : nullptr, (float)eps, rows, H);
+  HIP_CHECK_LAST();
+  return {out, mean, invstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor gamma, at::Tensor mean,
+                                      at::Tensor invstd) {
+  const int H = (int)x.size(-1);
+  const long rows = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dgamma = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = capped_grid(rows, 2048);
+  const size_t lds = 2 * (size_t)H * sizeof(float);
+  hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(grid), dim3(LN_BLOCK), lds, stream.stream(),
+                     (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                     (const unsigned short*)dy.data_ptr(), (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)gamma.data_ptr(), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), rows, H);
+  HIP_CHECK_LAST();
+  return {dx, dgamma, dbeta};
+}
+
+}  // namespace cai

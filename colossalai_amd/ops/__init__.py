@@ -1,5 +1,6 @@
 from ._kernels import has_kernels, kernels, use_hip
 from .attention import attention_ref, flash_attention, fused_rope_attention
+from .layernorm import layer_norm
 from .norm import fused_add_rms_norm, rms_norm, rms_norm_ref
 from .rope import apply_rope, apply_rope_ref, build_rope_table
 from .swiglu import swiglu, swiglu_ref
@@ -11,6 +12,7 @@ __all__ = [
     "flash_attention",
     "attention_ref",
     "rms_norm",
+    "layer_norm",
     "fused_add_rms_norm",
     "rms_norm_ref",
     "apply_rope",

@@ -14,6 +14,7 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "LlamaModel": ("colossalai_amd.shardformer.policies.llama", "LlamaPolicy"),
     "MixtralForCausalLM": ("colossalai_amd.shardformer.policies.mixtral", "MixtralForCausalLMPolicy"),
     "MixtralModel": ("colossalai_amd.shardformer.policies.mixtral", "MixtralPolicy"),
+    "GPT2LMHeadModel": ("colossalai_amd.shardformer.policies.gpt2", "GPT2LMHeadModelPolicy"),
 }
 
 

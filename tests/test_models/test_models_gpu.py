@@ -94,3 +94,17 @@ def test_gemini_offload_step():
         if l0 is None:
             l0 = out["loss"].item()
     assert out["loss"].item() < l0
+
+
+def test_gpt2_gpu_train_step():
+    from colossalai_amd.models.gpt2 import GPT2Config, GPT2LMHeadModel
+
+    torch.manual_seed(0)
+    m = GPT2LMHeadModel(GPT2Config(vocab_size=512, n_positions=512, n_embd=256, n_layer=2, n_head=4))
+    m = m.to("cuda").bfloat16()
+    x = torch.randint(0, 512, (2, 256), device="cuda")
+    out = m(x, labels=x)
+    assert torch.isfinite(out["loss"])
+    out["loss"].backward()
+    for n, p in m.named_parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all(), n

@@ -247,3 +247,26 @@ def test_fused_rope_attention_autograd():
     loss_ref.backward()
     _bf16_close(out, ref, rtol=3e-2, atol=3e-2, frac=1e-5)
     _bf16_close(qkv_hip.grad, qkv_ref.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+
+
+# ----------------------------------------------------------------- layernorm
+@pytest.mark.parametrize("rows,H", [(512, 4096), (300, 768)])
+def test_layernorm_fwd_bwd(rows, H):
+    torch.manual_seed(11)
+    x = torch.randn(rows, H, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    out, mean, invstd = _C.layernorm_fwd(x, g, b, 1e-5, True)
+    ref = torch.nn.functional.layer_norm(x.float(), (H,), g.float(), b.float(), 1e-5)
+    _bf16_close(out, ref, frac=1e-5)
+
+    dy = torch.randn_like(x)
+    dx, dgamma, dbeta = _C.layernorm_bwd(dy, x, g, mean, invstd)
+    xf = x.float().requires_grad_(True)
+    gf = g.float().requires_grad_(True)
+    bf = b.float().requires_grad_(True)
+    ref2 = torch.nn.functional.layer_norm(xf, (H,), gf, bf, 1e-5)
+    ref2.backward(dy.float())
+    _bf16_close(dx, xf.grad, rtol=4e-2, atol=4e-2, frac=2e-4)
+    torch.testing.assert_close(dgamma.cpu(), gf.grad.cpu(), rtol=2e-2, atol=2e-1)
+    torch.testing.assert_close(dbeta.cpu(), bf.grad.cpu(), rtol=2e-2, atol=2e-1)
