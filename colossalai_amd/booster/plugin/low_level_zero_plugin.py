@@ -115,7 +115,17 @@ class LowLevelZeroPlugin(Plugin):
         return True
 
     def support_lora(self) -> bool:
-        return False
+        return True
+
+    def enable_lora(self, model, pretrained_dir=None, lora_config=None, **kwargs):
+        from ...lora import LoraConfig, apply_lora
+
+        model = apply_lora(model, lora_config if isinstance(lora_config, LoraConfig) else None)
+        if pretrained_dir:
+            from ...checkpoint_io.utils import load_state_dict
+
+            model.load_state_dict(load_state_dict(pretrained_dir), strict=False)
+        return model
 
     def control_checkpoint_io(self) -> bool:
         return True
