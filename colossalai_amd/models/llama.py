@@ -109,6 +109,23 @@ class LlamaAttention(nn.Module):
         B, S, _ = hidden.shape
         qkv = self.qkv_proj(hidden)
         sp_mode = getattr(self, "sp_mode", None)
+        if sp_mode == "ring_attn":
+            # context parallelism: Q stays, K/V blocks travel the xGMI ring
+            import torch.distributed as dist
+
+            from ..ops import apply_rope
+            from ..shardformer.layer.ring_attn import ring_flash_attention
+
+            sp_group = self.sp_group
+            rank = dist.get_rank(sp_group)
+            Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+            q = qkv[:, :, : Hq * D].reshape(B, S, Hq, D).contiguous()
+            k = qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D).contiguous()
+            v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D).contiguous()
+            positions = (torch.arange(S, device=hidden.device) + rank * S).repeat(B).int()
+            q, k = apply_rope(q, k, rope_table, positions)
+            attn = ring_flash_attention(q, k, v, sp_group, causal=True, scale=self.scale)
+            return self.o_proj(attn.reshape(B, S, -1))
         if sp_mode == "all_to_all":
             # Ulysses: scatter heads / gather sequence around attention
             import torch.distributed as dist

@@ -37,6 +37,17 @@ class LlamaPolicy(Policy):
             self.shard_config.enable_sequence_parallelism
             and self.shard_config.sequence_parallelism_mode == "all_to_all"
         )
+        sp_ring = (
+            self.shard_config.enable_sequence_parallelism
+            and self.shard_config.sequence_parallelism_mode == "ring_attn"
+        )
+        if sp_ring:
+            policy[LlamaAttention] = ModulePolicyDescription(
+                attribute_replacement={
+                    "sp_mode": "ring_attn",
+                    "sp_group": self.shard_config.sequence_parallel_process_group,
+                }
+            )
         sp_sg = (
             self.shard_config.enable_sequence_parallelism
             and self.shard_config.sequence_parallelism_mode == "split_gather"
@@ -67,6 +78,10 @@ class LlamaPolicy(Policy):
             if sp_a2a:
                 attn_attrs.update(
                     sp_mode="all_to_all", sp_group=self.shard_config.sequence_parallel_process_group
+                )
+            if sp_ring:
+                attn_attrs.update(
+                    sp_mode="ring_attn", sp_group=self.shard_config.sequence_parallel_process_group
                 )
             policy[LlamaAttention] = ModulePolicyDescription(
                 attribute_replacement=attn_attrs,
@@ -134,9 +149,9 @@ class LlamaForCausalLMPolicy(LlamaPolicy):
             self.model.parallel_logits = True
         if self.shard_config.enable_sequence_parallelism:
             mode = self.shard_config.sequence_parallelism_mode
-            if mode == "all_to_all":
+            if mode in ("all_to_all", "ring_attn"):
                 self.model.sp_group = self.shard_config.sequence_parallel_process_group
-                self.model.sp_mode = "all_to_all"
+                self.model.sp_mode = mode
             elif mode == "split_gather":
                 # split_gather reuses the tp group (reference semantics)
                 group = self.shard_config.tensor_parallel_process_group
