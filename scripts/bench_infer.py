@@ -1,0 +1,32 @@
+"""Inference engine throughput smoke (decode tokens/s on llama-7b)."""
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from colossalai_amd.inference import GenerationConfig, InferenceConfig, LLMEngine
+from colossalai_amd.models import LLAMA_CONFIGS, LlamaForCausalLM
+
+cfg = LLAMA_CONFIGS["llama-7b"]
+with torch.device("meta"):
+    model = LlamaForCausalLM(cfg)
+model = model.to_empty(device="cuda").to(torch.bfloat16)
+with torch.no_grad():
+    for p in model.parameters():
+        p.normal_(0, 0.02)
+
+engine = LLMEngine(model, InferenceConfig(max_batch_size=8, max_input_len=512, max_output_len=256))
+prompts = [[int(x) for x in torch.randint(0, 32000, (128,))] for _ in range(8)]
+# warmup
+engine.generate(prompts, GenerationConfig(max_new_tokens=8))
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+N = 128
+out = engine.generate(prompts, GenerationConfig(max_new_tokens=N))
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+total_new = sum(len(o) - 128 for o in out)
+print(f"llama-7b bf16 bs8: {total_new} tokens in {dt:.2f}s = {total_new/dt:.1f} tok/s ({total_new/dt/8:.1f}/seq)")
