@@ -17,7 +17,7 @@
 namespace cai {
 
 constexpr int LN_BLOCK = 256;
-constexpr int LN_MAX_H = 32 * LN_BLOCK / 8;  // matches rmsnorm (<= 8192 cols)
+constexpr int LN_MAX_H = 32 * LN_BLOCK;  // 32 fp32 vals per thread -> up to 8192 cols
 
 __global__ __launch_bounds__(LN_BLOCK) void layernorm_fwd_kernel(
     unsigned short* __restrict__ out,
