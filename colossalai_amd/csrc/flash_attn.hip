@@ -294,9 +294,14 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
         rowmax = fmaxf(rowmax, p[kb][j]);
       }
     rowmax = fmaxf(rowmax, __shfl_xor(rowmax, 32));
-    const float m_new = fmaxf(m_run, rowmax);
+    // defer-max (guide T13): if this tile's max is within 8 of the running
+    // max, keep the old max (P bounded by e^8, fine in fp32/bf16) and skip
+    // the O-rescale pass entirely when every row defers.
+    const bool defer = (m_run != -INFINITY) && (rowmax - m_run <= 8.0f);
+    const float m_new = defer ? m_run : fmaxf(m_run, rowmax);
     const float msafe = (m_new == -INFINITY) ? 0.0f : m_new;
-    const float corr = (m_run == -INFINITY) ? ((m_new == -INFINITY) ? 1.0f : 0.0f) : __expf(m_run - m_new);
+    const float corr =
+        defer ? 1.0f : ((m_run == -INFINITY) ? ((m_new == -INFINITY) ? 1.0f : 0.0f) : __expf(m_run - m_new));
     float rowsum = 0.0f;
 #pragma unroll
     for (int kb = 0; kb < 2; ++kb)
@@ -322,13 +327,16 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
         *reinterpret_cast<short4v*>(P + swz<KVB * 2>(ln, kcol * 2)) = pk;
       }
 
-    // ---- rescale O accumulator by corr (per q-row, via shfl broadcast)
+    // ---- rescale O accumulator by corr (per q-row, via shfl broadcast);
+    // skipped entirely when every row of the wave deferred (corr == 1)
+    if (!__all(corr == 1.0f)) {
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const int r = crow(j, half);
-      const float c = __shfl(corr, r);  // corr for q-row r lives on lanes r, r+32
+      for (int j = 0; j < 16; ++j) {
+        const int r = crow(j, half);
+        const float c = __shfl(corr, r);  // corr for q-row r lives on lanes r, r+32
 #pragma unroll
-      for (int nb = 0; nb < D / 32; ++nb) oacc[nb][j] *= c;
+        for (int nb = 0; nb < D / 32; ++nb) oacc[nb][j] *= c;
+      }
     }
 
     // ---- O += P · V  (A = P from LDS, B = V_T reads)
