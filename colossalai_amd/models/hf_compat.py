@@ -19,9 +19,12 @@ def hf_to_native_llama(hf_sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor
     out = {}
     packs: Dict[str, Dict[str, torch.Tensor]] = {}
     for k, v in hf_sd.items():
-        if ".self_attn.q_proj.weight" in k or ".self_attn.k_proj.weight" in k or ".self_attn.v_proj.weight" in k:
+        if any(f".self_attn.{p_}_proj.weight" in k for p_ in "qkv"):
             base = k.rsplit(".self_attn.", 1)[0]
             packs.setdefault(base + ".qkv", {})[k.split(".")[-2]] = v
+        elif any(f".self_attn.{p_}_proj.bias" in k for p_ in "qkv"):
+            base = k.rsplit(".self_attn.", 1)[0]
+            packs.setdefault(base + ".qkvbias", {})[k.split(".")[-2]] = v
         elif ".mlp.gate_proj.weight" in k or ".mlp.up_proj.weight" in k:
             base = k.rsplit(".mlp.", 1)[0]
             packs.setdefault(base + ".gateup", {})[k.split(".")[-2]] = v
@@ -36,7 +39,12 @@ def hf_to_native_llama(hf_sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor
         else:
             out[k] = v
     for base, parts in packs.items():
-        if base.endswith(".qkv"):
+        if base.endswith(".qkvbias"):
+            prefix = base[: -len(".qkvbias")]
+            out[f"{prefix}.self_attn.qkv_proj.bias"] = torch.cat(
+                [parts["q_proj"], parts["k_proj"], parts["v_proj"]], dim=0
+            )
+        elif base.endswith(".qkv"):
             prefix = base[: -len(".qkv")]
             out[f"{prefix}.self_attn.qkv_proj.weight"] = torch.cat(
                 [parts["q_proj"], parts["k_proj"], parts["v_proj"]], dim=0

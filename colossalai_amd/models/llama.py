@@ -44,6 +44,7 @@ class LlamaConfig:
     tie_word_embeddings: bool = False
     initializer_range: float = 0.02
     gradient_checkpointing: bool = False
+    attention_bias: bool = False  # Qwen2-style qkv bias
 
     def __post_init__(self):
         if self.num_key_value_heads is None:
@@ -68,6 +69,13 @@ LLAMA_CONFIGS = {
                              max_position_embeddings=8192, rope_theta=500000.0),
     "llama-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
                              num_attention_heads=64, num_key_value_heads=8, max_position_embeddings=4096),
+    # same decoder family: Mistral (GQA) and Qwen2 (GQA + qkv bias)
+    "mistral-7b": LlamaConfig(vocab_size=32000, hidden_size=4096, intermediate_size=14336,
+                              num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
+                              max_position_embeddings=4096, rope_theta=10000.0),
+    "qwen2-7b": LlamaConfig(vocab_size=152064, hidden_size=3584, intermediate_size=18944,
+                            num_hidden_layers=28, num_attention_heads=28, num_key_value_heads=4,
+                            max_position_embeddings=4096, rope_theta=1e6, attention_bias=True),
 }
 
 
@@ -101,7 +109,7 @@ class LlamaAttention(nn.Module):
         self.num_kv_heads = cfg.num_key_value_heads
         self.head_dim = cfg.head_dim
         D, Hq, Hkv = self.head_dim, self.num_heads, self.num_kv_heads
-        self.qkv_proj = nn.Linear(cfg.hidden_size, (Hq + 2 * Hkv) * D, bias=False)
+        self.qkv_proj = nn.Linear(cfg.hidden_size, (Hq + 2 * Hkv) * D, bias=cfg.attention_bias)
         self.o_proj = nn.Linear(Hq * D, cfg.hidden_size, bias=False)
         self.scale = 1.0 / math.sqrt(D)
 
@@ -321,6 +329,8 @@ class LlamaForCausalLM(nn.Module):
         std = self.config.initializer_range
         if isinstance(module, nn.Linear):
             module.weight.data.normal_(0.0, std)
+            if module.bias is not None:
+                module.bias.data.zero_()
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
 
