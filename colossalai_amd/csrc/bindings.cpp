@@ -34,6 +34,10 @@ void rope_inplace(at::Tensor q, c10::optional<at::Tensor> k, at::Tensor table,
 at::Tensor swiglu_fwd(at::Tensor gate_up);
 at::Tensor swiglu_bwd(at::Tensor dout, at::Tensor gate_up);
 
+// scaled_softmax.hip
+at::Tensor scaled_masked_softmax_fwd(at::Tensor x, c10::optional<at::Tensor> mask, double scale, bool causal);
+at::Tensor scaled_masked_softmax_bwd(at::Tensor dy, at::Tensor y, double scale);
+
 // mfma_selftest.hip
 std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor A32, at::Tensor B32);
 
@@ -61,6 +65,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_inplace", &cai::rope_inplace, "in-place rotary embedding (fwd/bwd)");
   m.def("swiglu_fwd", &cai::swiglu_fwd, "fused SwiGLU forward");
   m.def("swiglu_bwd", &cai::swiglu_bwd, "fused SwiGLU backward");
+  m.def("scaled_masked_softmax_fwd", &cai::scaled_masked_softmax_fwd, "fused scale+mask+softmax fwd");
+  m.def("scaled_masked_softmax_bwd", &cai::scaled_masked_softmax_bwd, "fused scale+mask+softmax bwd");
   m.def("mfma_selftest", &cai::mfma_selftest, "MFMA layout self-test probes");
   m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache");
   m.def("flash_attn_fwd", &cai::flash_attn_fwd, "flash attention forward (bf16, causal, GQA)");

@@ -270,3 +270,31 @@ def test_layernorm_fwd_bwd(rows, H):
     _bf16_close(dx, xf.grad, rtol=4e-2, atol=4e-2, frac=2e-4)
     torch.testing.assert_close(dgamma.cpu(), gf.grad.cpu(), rtol=2e-2, atol=2e-1)
     torch.testing.assert_close(dbeta.cpu(), bf.grad.cpu(), rtol=2e-2, atol=2e-1)
+
+
+# --------------------------------------------------- scaled masked softmax
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("with_mask", [False, True])
+def test_scaled_masked_softmax(causal, with_mask):
+    torch.manual_seed(12)
+    B, H, Sq, Sk, scale = 2, 3, 64, 128, 0.3
+    x = torch.randn(B, H, Sq, Sk, device="cuda", dtype=torch.bfloat16)
+    mask = None
+    mref = 0
+    if with_mask:
+        mask = (torch.randn(B, 1, Sq, Sk, device="cuda") * 2).bfloat16()
+        mref = mask.float()
+    y = _C.scaled_masked_softmax_fwd(x, mask, scale, causal)
+    scores = x.float() * scale + mref
+    if causal:
+        cm = torch.triu(torch.ones(Sq, Sk, dtype=torch.bool, device="cuda"), 1)
+        scores = scores.masked_fill(cm, float("-inf"))
+    ref = torch.softmax(scores, dim=-1)
+    _bf16_close(y, ref, frac=1e-5)
+
+    dy = torch.randn_like(x)
+    dx = _C.scaled_masked_softmax_bwd(dy, y, scale)
+    yf = y.float()
+    dot = (dy.float() * yf).sum(-1, keepdim=True)
+    dx_ref = scale * yf * (dy.float() - dot)
+    _bf16_close(dx, dx_ref, frac=1e-4)
