@@ -35,6 +35,9 @@ MIXTRAL_CONFIGS = {
     "mixtral-tiny": MixtralConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
                                   num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64,
                                   num_local_experts=4, num_experts_per_tok=2),
+    "mixtral-small": MixtralConfig(vocab_size=32000, hidden_size=2048, intermediate_size=5632,
+                                   num_hidden_layers=16, num_attention_heads=16, num_key_value_heads=4,
+                                   max_position_embeddings=4096, num_local_experts=8, num_experts_per_tok=2),
     "mixtral-8x7b": MixtralConfig(vocab_size=32000, hidden_size=4096, intermediate_size=14336,
                                   num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
                                   max_position_embeddings=4096, rope_theta=1e6,
