@@ -57,6 +57,11 @@ def main():
         rank, world = dist.get_rank(), dist.get_world_size()
     else:
         rank, world = 0, 1
+        if args.plugin in ("hybrid", "moe"):
+            # mesh-based plugins need an initialized process group even at N=1
+            from colossalai_amd.testing import free_port
+
+            colossalai_amd.launch(0, 1, "127.0.0.1", free_port(), verbose=False)
     device = "cuda" if torch.cuda.is_available() else "cpu"
     if device == "cpu":
         raise SystemExit("bench.py requires an MI355X GPU")
