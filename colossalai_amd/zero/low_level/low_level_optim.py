@@ -369,57 +369,41 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
             if true_norm > self.clip_grad_norm:
                 clip_factor = true_norm / self.clip_grad_norm
 
-        # fused adam over all bucket shards in one multi-tensor batch per group;
-        # offloaded buckets step on the host (grad D2H -> CPU adam -> param H2D)
+        # Per-bucket fused Adam immediately followed by that bucket's ASYNC
+        # all-gather: the gather of bucket i rides under the Adam of bucket
+        # i+1, so update compute and the xGMI traffic pipeline. Offloaded
+        # buckets step on the host (grad D2H -> CPU adam -> param H2D).
+        works = []
         for gi, (group, buckets) in enumerate(zip(self.optim.param_groups, self._group_buckets)):
-            world, rank = self._g_world[gi], self._g_rank[gi]
+            world, rank, pg = self._g_world[gi], self._g_rank[gi], self._g_pg[gi]
             div_scale = loss_scale * self._g_div[gi]  # default divisor == group world (1 if undistributed)
             flat_grad = self._flat_grads[gi]
             flat = self._flat_params[gi]
-            grads, masters, mlist, vlist, outs = [], [], [], [], []
-            cpu_jobs = []
+            group.setdefault("step", 0)
+            group["step"] += 1
             for b in buckets:
                 blen = b.end - b.start
                 gshard = flat_grad[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
                 pshard = flat[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
                 if getattr(b, "offloaded", False):
-                    cpu_jobs.append((b, gshard, pshard))
+                    self._cpu_step(group, b, gshard, pshard, div_scale * clip_factor)
                 else:
-                    grads.append(gshard)
-                    masters.append(b.master)
-                    mlist.append(b.exp_avg)
-                    vlist.append(b.exp_avg_sq)
-                    outs.append(pshard)
-            if grads:
-                self._fused_step(group, grads, masters, mlist, vlist, outs, div_scale * clip_factor)
-            elif cpu_jobs:
-                group.setdefault("step", 0)
-                group["step"] += 1
-            for b, gshard, pshard in cpu_jobs:
-                self._cpu_step(group, b, gshard, pshard, div_scale * clip_factor)
-
-        # all-gather updated working params (bucket-wise, async)
-        works = []
-        for gi, buckets in enumerate(self._group_buckets):
-            world, rank, pg = self._g_world[gi], self._g_rank[gi], self._g_pg[gi]
-            if world == 1:
-                continue
-            flat = self._flat_params[gi]
-            for b in buckets:
-                blen = b.end - b.start
-                seg = flat[b.start : b.end]
-                shard = flat[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
-                works.append(dist.all_gather_into_tensor(seg, shard.clone(), group=pg, async_op=True))
+                    self._fused_step(group, [gshard], [b.master], [b.exp_avg], [b.exp_avg_sq], [pshard],
+                                     div_scale * clip_factor, bump_step=False)
+                if world > 1:
+                    seg = flat[b.start : b.end]
+                    works.append(dist.all_gather_into_tensor(seg, pshard.clone(), group=pg, async_op=True))
         for wk in works:
             wk.wait()
         self.zero_grad()
 
-    def _fused_step(self, group, grads, masters, mlist, vlist, outs, div_scale):
+    def _fused_step(self, group, grads, masters, mlist, vlist, outs, div_scale, bump_step=True):
         from ...nn.optimizer.fused_adam import fused_adam_step_cpu
         from ...ops import has_kernels, kernels
 
-        group.setdefault("step", 0)
-        group["step"] += 1
+        if bump_step:
+            group.setdefault("step", 0)
+            group["step"] += 1
         beta1, beta2 = group.get("betas", (0.9, 0.999))
         lr = group["lr"]
         eps = group.get("eps", 1e-8)
