@@ -12,12 +12,15 @@ chunked flat buffers as the ZeRO engine:
   (gemini_plugin.py:390) and the configuration its Llama benchmarks use.
 - ``offload_optim_frac>0``: that fraction of master/momentum chunks lives in
   pinned host memory; their Adam step runs on CPU (PCIe Gen5 D2H/H2D).
-- ``shard_param_frac=1.0``: full parameter sharding — use TorchFSDPPlugin
-  (torch FSDP over RCCL) until the native chunk-gather manager lands.
+- ``shard_param_frac=1.0``: full parameter sharding via the native
+  chunk-gather manager (``zero.gemini.GeminiDDP`` — storage-resizing
+  all-gather/release at decoder-layer granularity + ``GeminiOptimizer``).
+  bf16/fp32 only (no loss-scale state on this path yet).
 
-Dynamic (auto) placement — chunk eviction driven by runtime memory stats —
-is intentionally deferred: measured HBM headroom on the target workloads
-makes it a no-op on this hardware generation.
+Fractional ``shard_param_frac`` and dynamic (auto) placement — chunk
+eviction driven by runtime memory stats — are intentionally deferred:
+measured HBM headroom on the target workloads makes them no-ops on this
+hardware generation.
 """
 
 from typing import Callable, Iterator, List, Optional, Tuple
@@ -64,14 +67,18 @@ class GeminiPlugin(Plugin):
         **kwargs,
     ):
         assert placement_policy in ("static", "auto"), "placement_policy must be static or auto"
-        if shard_param_frac not in (0.0,):
+        if shard_param_frac not in (0.0, 1.0):
             raise NotImplementedError(
-                "GeminiPlugin currently implements static placement with replicated params "
-                "(shard_param_frac=0.0, the reference's ZeRO-2-equivalent benchmark config). "
-                "For fully sharded parameters use TorchFSDPPlugin."
+                "GeminiPlugin implements shard_param_frac 0.0 (params replicated, ZeRO-2 "
+                "semantics — the reference's benchmark config) and 1.0 (native chunk-sharded "
+                "params, ZeRO-3 semantics); fractional placement is deferred."
             )
         if offload_param_frac > 0.0:
-            raise NotImplementedError("offload_param_frac>0 requires sharded params; use TorchFSDPPlugin")
+            raise NotImplementedError("offload_param_frac>0 is deferred; use offload_optim_frac")
+        if shard_param_frac == 1.0 and precision == "fp16":
+            raise NotImplementedError("shard_param_frac=1.0 supports bf16/fp32 (no loss scaler on this path)")
+        self.shard_param_frac = shard_param_frac
+        self.max_norm = max_norm
         self.precision = precision
         self.offload_optim_frac = offload_optim_frac
         # chunk size: reuse the ZeRO bucket machinery; chunks sized in MiB-elements
@@ -123,6 +130,14 @@ class GeminiPlugin(Plugin):
         dataloader: Optional[DataLoader] = None,
         lr_scheduler: Optional[LRScheduler] = None,
     ) -> Tuple[nn.Module, OptimizerWrapper, Callable, DataLoader, LRScheduler]:
+        if self.shard_param_frac == 1.0:
+            from ...zero.gemini import GeminiDDP, GeminiOptimizer
+
+            if not isinstance(model, ModelWrapper):
+                model = GeminiDDP(model, chunk_size_m=self.chunk_size_m, precision=self.precision)
+            if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
+                optimizer = GeminiOptimizer(optimizer, model, max_norm=self.max_norm)
+            return model, optimizer, criterion, dataloader, lr_scheduler
         if not isinstance(model, ModelWrapper):
             model = LowLevelZeroModel(model, self.precision)
         if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
@@ -130,5 +145,5 @@ class GeminiPlugin(Plugin):
         return model, optimizer, criterion, dataloader, lr_scheduler
 
     def no_sync(self, model: nn.Module, optimizer: OptimizerWrapper = None) -> Iterator[None]:
-        assert isinstance(optimizer, LowLevelZeroOptimizer)
+        assert isinstance(optimizer, LowLevelZeroOptimizer), "no_sync is not supported on the sharded-param path"
         return optimizer.no_sync()

@@ -1,0 +1,309 @@
+"""Chunk-based parameter-sharded data parallelism (native ZeRO-3)
+(reference: colossalai/zero/gemini/gemini_ddp.py:56 + chunk/chunk.py:59 —
+re-designed for MI355X; no tensor subclass, no tracing).
+
+Mechanism (unit-granular, FSDP-shaped):
+
+- The model is split into UNITS — by default the elements of its largest
+  repeated ``ModuleList`` (the decoder layers; ~98 % of weight bytes for
+  the transformer families). Each unit's parameters are packed into flat
+  bf16 CHUNKS (default 64 Mi elements = 128 MB — xGMI-sized all-gathers).
+  Each rank persistently stores only its 1/world shard; the full payload is
+  a resizable storage (``untyped_storage().resize_``) materialized by
+  ``all_gather_into_tensor`` at the unit's pre-forward hook and dropped to
+  zero bytes at its post-forward hook. Parameters are views into the
+  chunk, so autograd-saved weights stay valid across release/regather
+  (same storage object).
+- Parameters OUTSIDE units (embeddings, final norm, lm_head — including
+  tied pairs) go into PERSISTENT chunks: always materialized, but still
+  grad- and optimizer-state-sharded (ZeRO-2 for the residuals, ZeRO-3 for
+  the layers).
+- Backward: a full-backward-pre hook on the unit re-gathers its chunks and
+  materializes their grad buffers (every ``param.grad`` becomes a view);
+  when all of a chunk's grads have accumulated
+  (``post_accumulate_grad`` hooks), the grad buffer is reduce-scattered
+  into the rank's grad shard and both full buffers are released.
+- Optimizer: ``GeminiOptimizer`` steps fp32 master shards against the grad
+  shards with the fused multi-tensor Adam kernel, writing bf16 straight
+  into the param shards; the next forward's gathers publish the update.
+
+With 288 GB HBM3E this path is for 70B+ models / small dp groups; the
+flat-buffer ZeRO-2 engine remains the default below that. Gradient
+accumulation across steps (``no_sync``) is not supported on this path yet.
+"""
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ...interface import ModelWrapper
+
+__all__ = ["GeminiDDP", "Chunk"]
+
+
+def _pad(n: int, align: int) -> int:
+    return (n + align - 1) // align * align
+
+
+class Chunk:
+    def __init__(self, chunk_id: int, numel: int, world: int, rank: int, device, dtype,
+                 persistent: bool = False):
+        self.id = chunk_id
+        self.numel = numel  # padded, divisible by world
+        self.world = world
+        self.rank = rank
+        self.persistent = persistent
+        self.flat = torch.zeros(numel, dtype=dtype, device=device)
+        self.grad_flat = torch.zeros(0, dtype=dtype, device=device)
+        self.shard = torch.zeros(numel // world, dtype=dtype, device=device)
+        self.grad_shard: Optional[torch.Tensor] = None
+        self.params: List[Tuple[nn.Parameter, int]] = []  # (param, offset)
+        self.gathered = True  # starts materialized (init copies weights in)
+        self.grads_done = 0
+
+    # ---------------------------------------------------------- param space
+    def seal(self):
+        """Copy this rank's slice into the persistent shard; release unless
+        the chunk is persistent."""
+        n = self.numel // self.world
+        self.shard.copy_(self.flat[self.rank * n : (self.rank + 1) * n])
+        if not self.persistent:
+            self.release()
+
+    def gather(self, group):
+        if self.gathered:
+            return
+        self.flat.untyped_storage().resize_(self.numel * self.flat.element_size())
+        if self.world > 1:
+            dist.all_gather_into_tensor(self.flat, self.shard, group=group)
+        else:
+            self.flat.copy_(self.shard)
+        self.gathered = True
+
+    def publish_shard(self, group):
+        """Push an updated shard into an already-materialized (persistent)
+        chunk after an optimizer step."""
+        if self.world > 1:
+            dist.all_gather_into_tensor(self.flat, self.shard, group=group)
+        else:
+            self.flat.copy_(self.shard)
+
+    def release(self):
+        if self.persistent or not self.gathered:
+            return
+        self.flat.untyped_storage().resize_(0)
+        self.gathered = False
+
+    # ----------------------------------------------------------- grad space
+    def materialize_grads(self):
+        if self.grad_flat.numel() == 0:
+            self.grad_flat = torch.zeros(self.numel, dtype=self.flat.dtype, device=self.shard.device)
+        for p, off in self.params:
+            if p.grad is None:
+                p.grad = self.grad_flat[off : off + p.numel()].view_as(p)
+
+    def reduce_grads(self, group):
+        n = self.numel // self.world
+        if self.grad_shard is None:
+            self.grad_shard = torch.zeros(n, dtype=self.flat.dtype, device=self.shard.device)
+        if self.world > 1:
+            out = torch.empty_like(self.grad_shard)
+            dist.reduce_scatter_tensor(out, self.grad_flat, group=group)
+            self.grad_shard.add_(out)
+        else:
+            self.grad_shard.add_(self.grad_flat)
+        for p, _ in self.params:
+            p.grad = None
+        self.grad_flat = torch.zeros(0, dtype=self.flat.dtype, device=self.shard.device)
+        self.grads_done = 0
+
+
+def _auto_units(module: nn.Module) -> List[nn.Module]:
+    """Pick the largest same-class ModuleList as the shard units (the
+    decoder-layer stack for every native model family)."""
+    best: List[nn.Module] = []
+    best_numel = 0
+    for m in module.modules():
+        if isinstance(m, nn.ModuleList) and len(m) > 0:
+            total = sum(p.numel() for p in m.parameters())
+            if total > best_numel:
+                best_numel = total
+                best = list(m)
+    return best
+
+
+class GeminiDDP(ModelWrapper):
+    def __init__(
+        self,
+        module: nn.Module,
+        process_group=None,
+        chunk_size_m: int = 64,
+        precision: str = "bf16",
+        units: Optional[List[nn.Module]] = None,
+    ):
+        dtype = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}[precision]
+        module = module.to(dtype)
+        device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        module = module.to(device)
+        super().__init__(module)
+        self.dtype = dtype
+        self.device = device
+        self.group = process_group
+        self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self.rank = dist.get_rank(process_group) if dist.is_initialized() else 0
+        self.chunk_elems = chunk_size_m * 1024 * 1024
+        self._align = 64 * self.world
+
+        self.chunks: List[Chunk] = []
+        self.param_chunk: Dict[int, Chunk] = {}
+        self.unit_chunks: Dict[int, List[Chunk]] = {}  # id(unit module) -> chunks
+
+        units = _auto_units(module) if units is None else units
+        # A unit's DIRECT params (the hoisted per-layer norm weights in the
+        # fused residual chain) are read BETWEEN layer calls — they stay
+        # persistent (H-sized, negligible). Only submodule weights
+        # (attention/MLP — used strictly inside the module call) are
+        # gathered/released at the unit boundary.
+        unit_param_ids = set()
+        unit_params: Dict[int, List[nn.Parameter]] = {}
+        for u in units:
+            direct = {id(p) for p in u.parameters(recurse=False)}
+            ps = [p for p in u.parameters() if p.requires_grad and id(p) not in direct]
+            unit_params[id(u)] = ps
+            unit_param_ids.update(id(p) for p in ps)
+
+        for u in units:
+            self.unit_chunks[id(u)] = self._pack(unit_params[id(u)], persistent=False)
+        # residuals: everything not owned by a unit (dedup handles tied pairs)
+        residual = [p for p in module.parameters() if p.requires_grad and id(p) not in unit_param_ids]
+        self.persistent_chunks = self._pack(residual, persistent=True)
+
+        for u in units:
+            u.register_forward_pre_hook(self._make_fwd_gather(self.unit_chunks[id(u)]))
+            u.register_forward_hook(self._make_fwd_release(self.unit_chunks[id(u)]))
+            u.register_full_backward_pre_hook(self._make_bwd_gather(self.unit_chunks[id(u)]))
+
+        self._hook_handles = []
+        for p in module.parameters():
+            if p.requires_grad:
+                self._hook_handles.append(p.register_post_accumulate_grad_hook(self._grad_hook))
+
+    # ------------------------------------------------------------------ pack
+    def _pack(self, params: List[nn.Parameter], persistent: bool) -> List[Chunk]:
+        made: List[Chunk] = []
+        group: List[nn.Parameter] = []
+        elems = 0
+        seen = set()
+
+        def flush():
+            nonlocal group, elems
+            if not group:
+                return
+            total = max(sum(_pad(p.numel(), self._align) for p in group), self._align)
+            c = Chunk(len(self.chunks), total, self.world, self.rank, self.device, self.dtype,
+                      persistent=persistent)
+            off = 0
+            for p in group:
+                c.flat[off : off + p.numel()].copy_(p.data.reshape(-1))
+                p.data = c.flat[off : off + p.numel()].view_as(p.data)
+                c.params.append((p, off))
+                self.param_chunk[id(p)] = c
+                off += _pad(p.numel(), self._align)
+            c.seal()
+            self.chunks.append(c)
+            made.append(c)
+            group = []
+            elems = 0
+
+        for p in params:
+            if id(p) in seen:
+                continue
+            seen.add(id(p))
+            group.append(p)
+            elems += _pad(p.numel(), self._align)
+            if elems >= self.chunk_elems:
+                flush()
+        flush()
+        return made
+
+    # ------------------------------------------------------------------ hooks
+    def _make_fwd_gather(self, chunks):
+        def hook(module, args):
+            for c in chunks:
+                c.gather(self.group)
+            return None
+        return hook
+
+    def _make_fwd_release(self, chunks):
+        def hook(module, args, output):
+            for c in chunks:
+                c.release()
+            return None
+        return hook
+
+    def _make_bwd_gather(self, chunks):
+        def hook(module, grad_output):
+            for c in chunks:
+                c.gather(self.group)
+                c.materialize_grads()
+            return None
+        return hook
+
+    def _grad_hook(self, p: nn.Parameter):
+        c = self.param_chunk[id(p)]
+        g = p.grad
+        if g is not None and (
+            c.grad_flat.numel() == 0
+            or g.untyped_storage().data_ptr() != c.grad_flat.untyped_storage().data_ptr()
+        ):
+            # autograd allocated a fresh grad (param used outside its hooked
+            # unit): fold it into the chunk's grad buffer
+            p.grad = None
+            c.materialize_grads()
+            p.grad.add_(g)
+        c.grads_done += 1
+        if c.grads_done == len(c.params):
+            c.reduce_grads(self.group)
+            c.release()
+
+    # -------------------------------------------------------------------- api
+    def forward(self, *args, **kwargs):
+        if self.module.training and torch.is_grad_enabled():
+            for c in self.persistent_chunks:
+                c.materialize_grads()
+        args = [a.to(self.dtype) if isinstance(a, torch.Tensor) and a.is_floating_point() else a for a in args]
+        kwargs = {
+            k: (v.to(self.dtype) if isinstance(v, torch.Tensor) and v.is_floating_point() else v)
+            for k, v in kwargs.items()
+        }
+        return self.module(*args, **kwargs)
+
+    def publish_persistent(self):
+        for c in self.persistent_chunks:
+            c.publish_shard(self.group)
+
+    def zero_grad_shards(self):
+        for c in self.chunks:
+            if c.grad_shard is not None:
+                c.grad_shard.zero_()
+
+    def gather_all(self):
+        """Materialize every chunk (checkpoint save / export)."""
+        for c in self.chunks:
+            c.gather(self.group)
+
+    def release_all(self):
+        for c in self.chunks:
+            c.release()
+
+    def state_dict(self, *args, **kwargs):
+        self.gather_all()
+        try:
+            sd = self.module.state_dict(*args, **kwargs)
+            # params are views into chunk payloads that release_all() frees:
+            # detach-copy everything before handing the dict out
+            return {k: v.detach().clone() for k, v in sd.items()}
+        finally:
+            self.release_all()

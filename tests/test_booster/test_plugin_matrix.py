@@ -46,6 +46,7 @@ def _plugins():
         "zero1": lambda: LowLevelZeroPlugin(stage=1, precision="fp32", overlap_communication=False),
         "zero2": lambda: LowLevelZeroPlugin(stage=2, precision="fp32", overlap_communication=False),
         "gemini": lambda: GeminiPlugin(precision="fp16", initial_scale=1.0),
+        "gemini3": lambda: GeminiPlugin(shard_param_frac=1.0, precision="fp32", min_chunk_size_m=1),
         "hybrid_tp2": lambda: HybridParallelPlugin(tp_size=2, pp_size=1, precision="fp32", zero_stage=0),
     }
 
