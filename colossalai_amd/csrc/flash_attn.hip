@@ -404,11 +404,18 @@ __global__ __launch_bounds__(256) void fa_delta_kernel(
     float acc = 0.0f;
     const unsigned short* dop = dO + row * D + e16 * epl;
     const unsigned short* op = Oin + row * D + e16 * epl;
-    for (int k = 0; k < epl; k += 8) {
-      short8 a = *reinterpret_cast<const short8*>(dop + k);
-      short8 b = *reinterpret_cast<const short8*>(op + k);
+    if (epl >= 8) {
+      for (int k = 0; k < epl; k += 8) {
+        short8 a = *reinterpret_cast<const short8*>(dop + k);
+        short8 b = *reinterpret_cast<const short8*>(op + k);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc += bf2f((unsigned short)a[j]) * bf2f((unsigned short)b[j]);
+        for (int j = 0; j < 8; ++j) acc += bf2f((unsigned short)a[j]) * bf2f((unsigned short)b[j]);
+      }
+    } else {  // D=64: 4 elems per lane — load width must match the slice
+      short4v a = *reinterpret_cast<const short4v*>(dop);
+      short4v b = *reinterpret_cast<const short4v*>(op);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc += bf2f((unsigned short)a[j]) * bf2f((unsigned short)b[j]);
     }
 #pragma unroll
     for (int off = 8; off > 0; off >>= 1) acc += __shfl_xor(acc, off);

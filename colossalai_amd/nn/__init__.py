@@ -1,3 +1,3 @@
-from .optimizer import Adafactor, CPUAdam, FusedAdam, HybridAdam, Lamb, Lars
+from .optimizer import CAME, Adafactor, CPUAdam, FusedAdam, GaLoreAdamW, HybridAdam, Lamb, Lars
 
-__all__ = ["FusedAdam", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor"]
+__all__ = ["FusedAdam", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]

@@ -1,5 +1,6 @@
 from .device import get_current_device, free_port_util
 from .memory import report_memory_usage
+from .memory_tracer import MemoryTracer
 from .timer import MultiTimer, Timer
 
-__all__ = ["Timer", "MultiTimer", "get_current_device", "report_memory_usage", "free_port_util"]
+__all__ = ["Timer", "MultiTimer", "get_current_device", "report_memory_usage", "MemoryTracer", "free_port_util"]

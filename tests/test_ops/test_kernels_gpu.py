@@ -196,6 +196,8 @@ def test_flash_attention_fwd(B, S, Hq, Hkv, D, causal):
     (2, 256, 4, 4, 128),
     (1, 512, 8, 2, 128),
     (2, 320, 4, 4, 128),
+    (4, 256, 8, 8, 64),   # D=64 (delta kernel lane-slice width)
+    (1, 512, 8, 2, 64),   # D=64 GQA
 ])
 def test_flash_attention_bwd(B, S, Hq, Hkv, D):
     from colossalai_amd.ops.attention import attention_ref

@@ -60,3 +60,29 @@ def test_fused_adam_matches_torch_adamw():
         o1.step()
         o2.step()
     torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-6)
+
+
+def test_came():
+    from colossalai_amd.nn import CAME
+
+    _converges(CAME, lr=2e-2)
+
+
+def test_galore_adamw():
+    from colossalai_amd.nn import GaLoreAdamW
+
+    # rank 4 on 16x32/32x1 weights: the linears are small, so force the
+    # low-rank path with min_dim=1
+    _converges(GaLoreAdamW, lr=2e-2, rank=4, update_proj_gap=10, galore_scale=1.0, min_dim=1)
+
+
+def test_galore_state_is_lowrank():
+    from colossalai_amd.nn import GaLoreAdamW
+
+    torch.manual_seed(0)
+    p = torch.randn(256, 512, requires_grad=True)
+    opt = GaLoreAdamW([p], lr=1e-3, rank=8, min_dim=1)
+    p.grad = torch.randn_like(p)
+    opt.step()
+    st = opt.state[p]
+    assert st["exp_avg"].shape in ((256, 8), (8, 512)), st["exp_avg"].shape
