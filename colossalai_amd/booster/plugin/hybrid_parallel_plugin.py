@@ -171,9 +171,14 @@ class HybridParallelPlugin(Plugin):
     ):
         assert dist.is_initialized(), "launch colossalai_amd before creating HybridParallelPlugin"
         assert zero_stage in (0, 1, 2)
-        assert pp_style in ("1f1b", "interleaved"), f"unsupported pp_style {pp_style}"
+        assert pp_style in ("1f1b", "interleaved", "zb"), f"unsupported pp_style {pp_style}"
         if pp_style == "interleaved":
             assert num_model_chunks > 1, "interleaved pipeline needs num_model_chunks > 1"
+        if pp_style == "zb":
+            assert zero_stage == 0, (
+                "zero-bubble defers weight grads past AccumulateGrad, which ZeRO's "
+                "bucket hooks rely on — use pp_style='zb' with zero_stage=0"
+            )
         world = dist.get_world_size()
         if sp_size is None:
             sp_size = 1
@@ -221,6 +226,13 @@ class HybridParallelPlugin(Plugin):
                 )
                 self.scheduler = InterleavedSchedule(
                     self.stage_manager, num_model_chunks=num_model_chunks, num_microbatches=num_microbatches
+                )
+            elif pp_style == "zb":
+                from ...pipeline.schedule.zero_bubble import ZeroBubbleSchedule
+
+                self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
+                self.scheduler = ZeroBubbleSchedule(
+                    self.stage_manager, num_microbatches=num_microbatches, microbatch_size=microbatch_size
                 )
             else:
                 self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
@@ -288,6 +300,10 @@ class HybridParallelPlugin(Plugin):
             model = model.to(dtype)
             if torch.cuda.is_available():
                 model = model.to("cuda")
+            if self.pp_style == "zb":
+                from ...ops.zb_linear import convert_to_zb_linears
+
+                convert_to_zb_linears(model)
             model = HybridParallelModule(model, dtype, self.dp_sp_group, self.tp_group, self.sp_group)
 
         if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):

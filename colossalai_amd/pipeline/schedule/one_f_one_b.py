@@ -103,11 +103,7 @@ class OneForwardOneBackwardSchedule(PipelineSchedule):
         def backward_step(input_obj, output_obj, output_grad):
             if forward_only:
                 return None
-            if output_grad is None:  # last stage: output_obj is the loss
-                optimizer.backward(output_obj, retain_graph=False)
-            else:
-                optimizer.backward_by_grad(output_obj, output_grad, retain_graph=False)
-            return input_obj.grad if input_obj is not None else None
+            return self._backward_step(optimizer, input_obj, output_obj, output_grad)
 
         # ---- warmup: forwards only
         for _ in range(num_warmup):
@@ -155,9 +151,26 @@ class OneForwardOneBackwardSchedule(PipelineSchedule):
             input_grad = backward_step(in_obj, out_obj, grad)
             if not is_first and input_grad is not None:
                 self.comm.send_backward(input_grad)
+            self._on_cooldown_backward()
 
         self.comm.flush_sends()
+        if not forward_only:
+            self._finalize_backward()
         result = {"loss": accum_loss.squeeze() if accum_loss is not None else None}
         if outputs is not None:
             result["outputs"] = outputs
         return result
+
+    # ---- hooks for schedule variants (zero-bubble overrides these) -------
+    def _backward_step(self, optimizer, input_obj, output_obj, output_grad):
+        if output_grad is None:  # last stage: output_obj is the loss
+            optimizer.backward(output_obj, retain_graph=False)
+        else:
+            optimizer.backward_by_grad(output_obj, output_grad, retain_graph=False)
+        return input_obj.grad if input_obj is not None else None
+
+    def _on_cooldown_backward(self):
+        pass
+
+    def _finalize_backward(self):
+        pass
