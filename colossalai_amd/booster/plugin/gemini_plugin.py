@@ -37,7 +37,94 @@ from ...zero import LowLevelZeroOptimizer
 from .low_level_zero_plugin import LowLevelZeroCheckpointIO, LowLevelZeroModel, _PRECISION_DTYPE
 from .plugin_base import Plugin
 
-__all__ = ["GeminiPlugin"]
+__all__ = ["GeminiPlugin", "GeminiCheckpointIO"]
+
+
+class GeminiCheckpointIO(LowLevelZeroCheckpointIO):
+    """Model save/load must stay COLLECTIVE for chunk-sharded params: every
+    rank joins the all_gathers inside GeminiDDP.state_dict/load_state_dict;
+    only the master writes files. Optimizer states keep the rank-local
+    shard-file scheme of the ZeRO IO."""
+
+    def load_model(self, model, checkpoint: str, strict: bool = True):
+        # keep the GeminiDDP wrapper: the base class unwraps, which would
+        # bypass the collective gather and read released chunk storage
+        from ...zero.gemini import GeminiDDP
+
+        if isinstance(model, GeminiDDP):
+            from pathlib import Path
+
+            from ...checkpoint_io.checkpoint_io_base import _resolve_single_file, _search_index_file
+
+            index_file_exists, index_file_path = _search_index_file(Path(checkpoint))
+            if index_file_exists:
+                self.load_sharded_model(model, index_file_path, strict)
+            else:
+                self.load_unsharded_model(model, str(_resolve_single_file(Path(checkpoint))), strict)
+            return model
+        return super().load_model(model, checkpoint, strict)
+
+    def save_model(self, model, checkpoint: str, shard: bool = False, gather_dtensor: bool = True,
+                   prefix=None, size_per_shard: int = 1024, use_safetensors: bool = False,
+                   use_async: bool = False):
+        from ...zero.gemini import GeminiDDP
+
+        if isinstance(model, GeminiDDP):
+            if shard:
+                self.save_sharded_model(model, checkpoint, gather_dtensor, prefix, size_per_shard,
+                                        use_safetensors, use_async)
+            else:
+                self.save_unsharded_model(model, checkpoint, gather_dtensor, use_safetensors, use_async)
+            return
+        return super().save_model(model, checkpoint, shard, gather_dtensor, prefix, size_per_shard,
+                                  use_safetensors, use_async)
+
+    def save_unsharded_model(self, model, checkpoint: str, gather_dtensor, use_safetensors, use_async=False):
+        from ...checkpoint_io.utils import save_state_dict
+        from ...zero.gemini import GeminiDDP
+
+        if isinstance(model, GeminiDDP):
+            sd = model.state_dict()  # collective
+            if self.coordinator.is_master():
+                save_state_dict(sd, checkpoint, use_safetensors)
+            return
+        super().save_unsharded_model(model, checkpoint, gather_dtensor, use_safetensors, use_async)
+
+    def save_sharded_model(self, model, checkpoint_path: str, gather_dtensor=False, prefix=None,
+                           max_shard_size=1024, use_safetensors=False, use_async=False):
+        from types import SimpleNamespace
+
+        from ...checkpoint_io import GeneralCheckpointIO
+        from ...zero.gemini import GeminiDDP
+
+        if isinstance(model, GeminiDDP):
+            sd = model.state_dict()  # collective
+            if self.coordinator.is_master():
+                GeneralCheckpointIO.save_sharded_model(
+                    self, SimpleNamespace(state_dict=lambda: sd), checkpoint_path,
+                    gather_dtensor, prefix, max_shard_size, use_safetensors, use_async,
+                )
+            return
+        super().save_sharded_model(model, checkpoint_path, gather_dtensor, prefix,
+                                   max_shard_size, use_safetensors, use_async)
+
+    def load_unsharded_model(self, model, checkpoint: str, strict: bool = True):
+        from ...checkpoint_io.utils import load_state_dict
+        from ...zero.gemini import GeminiDDP
+
+        if isinstance(model, GeminiDDP):
+            model.load_state_dict(load_state_dict(checkpoint), strict=strict)  # collective
+            return
+        super().load_unsharded_model(model, checkpoint, strict)
+
+    def load_sharded_model(self, model, index_file_path: str, strict: bool = False):
+        from ...checkpoint_io import GeneralCheckpointIO
+        from ...zero.gemini import GeminiDDP
+
+        if isinstance(model, GeminiDDP):
+            GeneralCheckpointIO.load_sharded_model(self, model, index_file_path, strict)
+            return
+        super().load_sharded_model(model, index_file_path, strict)
 
 
 class GeminiPlugin(Plugin):
@@ -120,7 +207,7 @@ class GeminiPlugin(Plugin):
         return True
 
     def get_checkpoint_io(self) -> CheckpointIO:
-        return LowLevelZeroCheckpointIO()
+        return GeminiCheckpointIO() if self.shard_param_frac == 1.0 else LowLevelZeroCheckpointIO()
 
     def configure(
         self,

@@ -50,6 +50,16 @@ class LowLevelZeroCheckpointIO(TorchDDPCheckpointIO):
         path = checkpoint if self.coordinator.world_size == 1 else f"{checkpoint}.rank{self.coordinator.rank}"
         _t.save(state, path)
 
+    def load_optimizer(self, optimizer, checkpoint: str):
+        import os
+
+        # rank-suffixed shard files (world>1) don't resolve as a single
+        # file: dispatch straight to the unsharded loader, which knows the
+        # ``<path>.rank<N>`` scheme
+        if os.path.isdir(checkpoint):
+            return super().load_optimizer(optimizer, checkpoint)
+        return self.load_unsharded_optimizer(optimizer, checkpoint)
+
     def load_unsharded_optimizer(self, optimizer, checkpoint: str):
         import os
 

@@ -307,3 +307,18 @@ class GeminiDDP(ModelWrapper):
             return {k: v.detach().clone() for k, v in sd.items()}
         finally:
             self.release_all()
+
+    def load_state_dict(self, state_dict, strict: bool = True):
+        """Collective: every rank materializes, loads, and re-seals its
+        shards (strict=False supports HF-style per-shard loading)."""
+        self.gather_all()
+        try:
+            ret = self.module.load_state_dict(
+                {k: v.to(self.dtype) if torch.is_tensor(v) and v.is_floating_point() else v
+                 for k, v in state_dict.items()},
+                strict=strict,
+            )
+        finally:
+            for c in self.chunks:
+                c.seal()  # copies the slice back to the shard; releases unless persistent
+        return ret
