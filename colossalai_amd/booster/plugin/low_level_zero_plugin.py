@@ -88,6 +88,7 @@ class LowLevelZeroPlugin(Plugin):
         reduce_bucket_size_in_m: int = 32,
         overlap_communication: bool = True,
         master_weights: bool = True,
+        fp8_communication: bool = False,
         verbose: bool = False,
     ):
         assert stage in (1, 2), "LowLevelZeroPlugin supports stage 1 or 2"
@@ -107,6 +108,7 @@ class LowLevelZeroPlugin(Plugin):
             overlap_communication=overlap_communication,
             partition_grad=(stage == 2),
             master_weights=master_weights,
+            fp8_communication=fp8_communication,
         )
 
     def supported_devices(self) -> List[str]:

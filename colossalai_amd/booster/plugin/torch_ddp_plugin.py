@@ -84,7 +84,9 @@ class TorchDDPPlugin(Plugin):
         check_reduction: bool = False,
         gradient_as_bucket_view: bool = False,
         static_graph: bool = False,
+        fp8_communication: bool = False,
     ):
+        self.fp8_communication = fp8_communication
         self.ddp_kwargs = dict(
             broadcast_buffers=broadcast_buffers,
             bucket_cap_mb=bucket_cap_mb,
@@ -141,6 +143,10 @@ class TorchDDPPlugin(Plugin):
         device = "cuda" if torch.cuda.is_available() else "cpu"
         model = model.to(device)
         model = TorchDDPModel(model, **self.ddp_kwargs)
+        if self.fp8_communication:
+            from ...quantization.fp8_hook import fp8_compress_ddp_grad_comm_hook
+
+            model.module.register_comm_hook(None, fp8_compress_ddp_grad_comm_hook)
         if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
             optimizer = OptimizerWrapper(optimizer)
         return model, optimizer, criterion, dataloader, lr_scheduler

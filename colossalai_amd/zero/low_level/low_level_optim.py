@@ -91,6 +91,7 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         cpu_offload_frac: float = 0.0,
         group_dp_pgs: Optional[List[Optional[dist.ProcessGroup]]] = None,
         group_grad_divisors: Optional[List[int]] = None,
+        fp8_communication: bool = False,
     ):
         super().__init__(optimizer)
         from ...nn.optimizer.fused_adam import FusedAdam
@@ -118,6 +119,7 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         self.overlap = overlap_communication and torch.cuda.is_available()
         self.master_weights = master_weights
         self.cpu_offload_frac = float(cpu_offload_frac)
+        self.fp8_communication = fp8_communication
         self.require_grad_sync = True
         self._accum_steps_pending = False
 
@@ -392,7 +394,14 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
                                      div_scale * clip_factor, bump_step=False)
                 if world > 1:
                     seg = flat[b.start : b.end]
-                    works.append(dist.all_gather_into_tensor(seg, pshard.clone(), group=pg, async_op=True))
+                    if self.fp8_communication:
+                        # e4m3 wire format for the post-step weight gather
+                        # (halves xGMI bytes; weights re-quantize next step)
+                        from ...quantization.fp8 import all_gather_fp8
+
+                        seg.copy_(all_gather_fp8(pshard, group=pg).view_as(seg))
+                    else:
+                        works.append(dist.all_gather_into_tensor(seg, pshard.clone(), group=pg, async_op=True))
         for wk in works:
             wk.wait()
         self.zero_grad()
