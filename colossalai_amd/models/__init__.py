@@ -1,3 +1,7 @@
 from .llama import LLAMA_CONFIGS, LlamaConfig, LlamaForCausalLM, llama_flops_per_token
+from .bert import BERT_CONFIGS, BertConfig, BertForMaskedLM, BertForSequenceClassification
+from .opt import OPT_CONFIGS, OPTConfig, OPTForCausalLM
 
-__all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_token"]
+__all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_token",
+           "OPTConfig", "OPTForCausalLM", "OPT_CONFIGS",
+           "BertConfig", "BertForMaskedLM", "BertForSequenceClassification", "BERT_CONFIGS"]

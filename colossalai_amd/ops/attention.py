@@ -23,7 +23,7 @@ __all__ = ["flash_attention", "fused_rope_attention", "attention_ref"]
 
 def attention_ref(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, causal: bool = True, scale: Optional[float] = None,
-    upcast: bool = True,
+    upcast: bool = True, bias: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """fp32 reference attention on [B,S,H,D] bshd tensors (GQA-aware)."""
     B, S, Hq, D = q.shape
@@ -39,6 +39,8 @@ def attention_ref(
         kt = kt.repeat_interleave(rep, dim=1)
         vt = vt.repeat_interleave(rep, dim=1)
     scores = torch.matmul(qt, kt.transpose(-1, -2)) * scale
+    if bias is not None:
+        scores = scores + bias.to(scores.dtype)
     if causal:
         mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device=q.device), diagonal=1)
         scores = scores.masked_fill(mask, float("-inf"))

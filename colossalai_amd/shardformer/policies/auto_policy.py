@@ -15,6 +15,11 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "MixtralForCausalLM": ("colossalai_amd.shardformer.policies.mixtral", "MixtralForCausalLMPolicy"),
     "MixtralModel": ("colossalai_amd.shardformer.policies.mixtral", "MixtralPolicy"),
     "GPT2LMHeadModel": ("colossalai_amd.shardformer.policies.gpt2", "GPT2LMHeadModelPolicy"),
+    "OPTForCausalLM": ("colossalai_amd.shardformer.policies.opt", "OPTForCausalLMPolicy"),
+    "OPTModel": ("colossalai_amd.shardformer.policies.opt", "OPTPolicy"),
+    "BertForMaskedLM": ("colossalai_amd.shardformer.policies.bert", "BertForMaskedLMPolicy"),
+    "BertForSequenceClassification": ("colossalai_amd.shardformer.policies.bert", "BertForSequenceClassificationPolicy"),
+    "BertModel": ("colossalai_amd.shardformer.policies.bert", "BertPolicy"),
 }
 
 
