@@ -44,6 +44,8 @@ std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor
 // decode_attn.hip
 at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, at::Tensor seq_lens,
                             double scale);
+at::Tensor decode_attention_paged(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
+                                  at::Tensor block_tables, at::Tensor seq_lens, double scale);
 
 // flash_attn.hip
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale);
@@ -69,6 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scaled_masked_softmax_bwd", &cai::scaled_masked_softmax_bwd, "fused scale+mask+softmax bwd");
   m.def("mfma_selftest", &cai::mfma_selftest, "MFMA layout self-test probes");
   m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache");
+  m.def("decode_attention_paged", &cai::decode_attention_paged,
+        "single-token attention over a paged (block-table) KV pool");
   m.def("flash_attn_fwd", &cai::flash_attn_fwd, "flash attention forward (bf16, causal, GQA)");
   m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward");
 }

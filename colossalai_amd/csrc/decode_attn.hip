@@ -8,8 +8,8 @@
 // selection into pointer math.
 //
 // Reference equivalent: extensions/csrc/kernel/cuda/flash_decoding_attention
-// _kernel.cu (contiguous-cache variant; paged cache lands with the paged
-// KV manager).
+// _kernel.cu. Both contiguous caches and vLLM-style paged pools (block
+// tables, power-of-two block size folded into shift/mask addressing).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -21,13 +21,19 @@ namespace cai {
 constexpr int DEC_BLOCK = 256;
 constexpr int GROUPS = DEC_BLOCK / 16;  // 16-lane groups
 
-template <int D>
+// PAGED=false: KC/VC are [B, Smax, Hkv, D] contiguous caches (Smax = stride).
+// PAGED=true: KC/VC are [num_blocks, block_size, Hkv, D] pools; logical
+// position s of sequence b lives at physical row
+//   block_tables[b*max_blocks + (s >> lbs)] * block_size + (s & (bs-1)).
+template <int D, bool PAGED>
 __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
     const unsigned short* __restrict__ Q,   // [B, Hq, D]
-    const unsigned short* __restrict__ KC,  // [B, Smax, Hkv, D]
-    const unsigned short* __restrict__ VC,  // [B, Smax, Hkv, D]
+    const unsigned short* __restrict__ KC,
+    const unsigned short* __restrict__ VC,
     unsigned short* __restrict__ O,         // [B, Hq, D]
     const int* __restrict__ seq_lens,       // [B] (length INCLUDING current token)
+    const int* __restrict__ block_tables,   // [B, max_blocks] (PAGED only)
+    int max_blocks, int lbs,                // log2(block_size)
     int B, int Smax, int Hq, int Hkv, float scale) {
   constexpr int EPL = D / 16;  // elements per lane (8 for D=128)
   __shared__ float sm_m[GROUPS];
@@ -43,8 +49,10 @@ __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
 
   const unsigned short* q = Q + ((long)b * Hq + h) * D + e * EPL;
   const long kv_tok = (long)Hkv * D;
-  const unsigned short* kbase = KC + (long)b * Smax * kv_tok + (long)hk * D + e * EPL;
-  const unsigned short* vbase = VC + (long)b * Smax * kv_tok + (long)hk * D + e * EPL;
+  const long seq_off = PAGED ? 0 : (long)b * Smax * kv_tok;
+  const unsigned short* kbase = KC + seq_off + (long)hk * D + e * EPL;
+  const unsigned short* vbase = VC + seq_off + (long)hk * D + e * EPL;
+  const int* bt = PAGED ? block_tables + (long)b * max_blocks : nullptr;
 
   float qf[EPL];
 #pragma unroll
@@ -55,7 +63,8 @@ __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
   for (int j = 0; j < EPL; ++j) o[j] = 0.0f;
 
   for (int s = g; s < S; s += GROUPS) {
-    const unsigned short* kp = kbase + (long)s * kv_tok;
+    const long row = PAGED ? (((long)bt[s >> lbs] << lbs) | (s & ((1 << lbs) - 1))) : (long)s;
+    const unsigned short* kp = kbase + row * kv_tok;
     float dot = 0.0f;
     if constexpr (EPL == 8) {
       short8 kv = *reinterpret_cast<const short8*>(kp);
@@ -74,7 +83,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
     const float corr = (m == -INFINITY) ? 0.0f : __expf(m - m_new);
     const float p = __expf(sc - m_new);
     l = l * corr + p;
-    const unsigned short* vp = vbase + (long)s * kv_tok;
+    const unsigned short* vp = vbase + row * kv_tok;
     if constexpr (EPL == 8) {
       short8 vv = *reinterpret_cast<const short8*>(vp);
 #pragma unroll
@@ -129,17 +138,43 @@ at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, 
   auto out = at::empty_like(q);
   auto stream = at::hip::getCurrentHIPStream();
   const dim3 grid(B * Hq);
-  if (D == 128) {
-    hipLaunchKernelGGL((decode_attn_kernel<128>), grid, dim3(DEC_BLOCK), 0, stream.stream(),
-                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kcache.data_ptr(),
-                       (const unsigned short*)vcache.data_ptr(), (unsigned short*)out.data_ptr(),
-                       seq_lens.data_ptr<int>(), B, Smax, Hq, Hkv, (float)scale);
-  } else {
-    hipLaunchKernelGGL((decode_attn_kernel<64>), grid, dim3(DEC_BLOCK), 0, stream.stream(),
-                       (const unsigned short*)q.data_ptr(), (const unsigned short*)kcache.data_ptr(),
-                       (const unsigned short*)vcache.data_ptr(), (unsigned short*)out.data_ptr(),
-                       seq_lens.data_ptr<int>(), B, Smax, Hq, Hkv, (float)scale);
-  }
+#define LAUNCH_DEC(DD)                                                                             \
+  hipLaunchKernelGGL((decode_attn_kernel<DD, false>), grid, dim3(DEC_BLOCK), 0, stream.stream(),   \
+                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kcache.data_ptr(),\
+                     (const unsigned short*)vcache.data_ptr(), (unsigned short*)out.data_ptr(),    \
+                     seq_lens.data_ptr<int>(), nullptr, 0, 0, B, Smax, Hq, Hkv, (float)scale)
+  if (D == 128) LAUNCH_DEC(128); else LAUNCH_DEC(64);
+#undef LAUNCH_DEC
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// Paged variant: kpool/vpool [num_blocks, block_size, Hkv, D],
+// block_tables [B, max_blocks] int32, block_size a power of two.
+at::Tensor decode_attention_paged(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
+                                  at::Tensor block_tables, at::Tensor seq_lens, double scale) {
+  TORCH_CHECK(q.dim() == 3 && q.scalar_type() == at::kBFloat16 && q.is_contiguous(), "q must be [B,Hq,D] bf16");
+  TORCH_CHECK(kpool.dim() == 4 && kpool.is_contiguous() && vpool.is_contiguous(), "kv pool [NB,BS,Hkv,D]");
+  TORCH_CHECK(block_tables.dim() == 2 && block_tables.scalar_type() == at::kInt && block_tables.is_contiguous());
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt && seq_lens.is_contiguous());
+  const int B = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+  const int BS = (int)kpool.size(1), Hkv = (int)kpool.size(2);
+  const int max_blocks = (int)block_tables.size(1);
+  TORCH_CHECK(D == 64 || D == 128, "decode_attention: head dim 64/128");
+  TORCH_CHECK((BS & (BS - 1)) == 0, "block_size must be a power of two");
+  int lbs = 0;
+  while ((1 << lbs) < BS) ++lbs;
+  auto out = at::empty_like(q);
+  auto stream = at::hip::getCurrentHIPStream();
+  const dim3 grid(B * Hq);
+#define LAUNCH_DECP(DD)                                                                            \
+  hipLaunchKernelGGL((decode_attn_kernel<DD, true>), grid, dim3(DEC_BLOCK), 0, stream.stream(),    \
+                     (const unsigned short*)q.data_ptr(), (const unsigned short*)kpool.data_ptr(), \
+                     (const unsigned short*)vpool.data_ptr(), (unsigned short*)out.data_ptr(),     \
+                     seq_lens.data_ptr<int>(), block_tables.data_ptr<int>(), max_blocks, lbs,      \
+                     B, 0, Hq, Hkv, (float)scale)
+  if (D == 128) LAUNCH_DECP(128); else LAUNCH_DECP(64);
+#undef LAUNCH_DECP
   HIP_CHECK_LAST();
   return out;
 }

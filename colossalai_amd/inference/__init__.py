@@ -1,4 +1,8 @@
 from .config import GenerationConfig, InferenceConfig
 from .engine import LLMEngine
+from .kv_cache import KVCacheManager
+from .paged_engine import ContinuousBatchEngine
+from .request_manager import Request, RequestManager, RequestStatus
 
-__all__ = ["InferenceConfig", "GenerationConfig", "LLMEngine"]
+__all__ = ["InferenceConfig", "GenerationConfig", "LLMEngine", "ContinuousBatchEngine",
+           "KVCacheManager", "RequestManager", "Request", "RequestStatus"]

@@ -212,6 +212,60 @@ class LlamaAttention(nn.Module):
                 out = torch.cat(outs, dim=0)
         return self.o_proj(out.reshape(B, S, -1))
 
+    def forward_with_paged_cache(self, hidden, rope_table, kv, layer_idx, seq_ids, block_tables,
+                                 positions, seq_lens, prefill: bool):
+        """Paged-KV inference path: RoPE + block-pool writes + (flash
+        prefill | paged decode kernel) attention. ``kv`` is a
+        ``KVCacheManager``; ``block_tables`` is the step's [B, max_blocks]
+        int32 tensor (decode only)."""
+        from ..ops import has_kernels
+        from ..ops.attention import attention_ref
+        from ..ops.rope import apply_rope_ref
+
+        B, S, _ = hidden.shape
+        Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+        qkv = self.qkv_proj(hidden)
+        q = qkv[:, :, : Hq * D].view(B, S, Hq, D)
+        k = qkv[:, :, Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+        v = qkv[:, :, (Hq + Hkv) * D :].view(B, S, Hkv, D)
+        use_hip_path = hidden.is_cuda and has_kernels()
+        if use_hip_path:
+            from ..ops import kernels
+
+            kernels().rope_inplace(q, k, rope_table, positions, False)
+        else:
+            q, k = apply_rope_ref(q, k, rope_table, positions.long(), S, False)
+        if prefill:
+            for i, sid in enumerate(seq_ids):
+                n = int(seq_lens[i])
+                kv.write_prefill(layer_idx, sid, k[i, :n], v[i, :n])
+            if use_hip_path:
+                from ..ops import kernels
+
+                out, _ = kernels().flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
+                                                  True, self.scale)
+            else:
+                out = attention_ref(q, k, v, causal=True, scale=self.scale)
+        else:
+            for i, sid in enumerate(seq_ids):
+                kv.write_token(layer_idx, sid, int(seq_lens[i]) - 1, k[i, 0], v[i, 0])
+            if use_hip_path:
+                from ..ops import kernels
+
+                out = kernels().decode_attention_paged(
+                    q.squeeze(1).contiguous(), kv.k_pools[layer_idx], kv.v_pools[layer_idx],
+                    block_tables, seq_lens.int(), self.scale
+                ).unsqueeze(1)
+            else:
+                outs = []
+                for i, sid in enumerate(seq_ids):
+                    n = int(seq_lens[i])
+                    kc, vc = kv.gather_contiguous(layer_idx, sid, n)
+                    outs.append(attention_ref(q[i : i + 1], kc[None], vc[None],
+                                              causal=False, scale=self.scale))
+                out = torch.cat(outs, dim=0)
+        return self.o_proj(out.reshape(B, S, -1))
+
 
 class LlamaMLP(nn.Module):
     def __init__(self, cfg: LlamaConfig):

@@ -300,3 +300,51 @@ def test_scaled_masked_softmax(causal, with_mask):
     dot = (dy.float() * yf).sum(-1, keepdim=True)
     dx_ref = scale * yf * (dy.float() - dot)
     _bf16_close(dx, dx_ref, frac=1e-4)
+
+
+# ------------------------------------------------------------ decode attention
+@pytest.mark.parametrize("D", [64, 128])
+def test_decode_attention_vs_ref(D):
+    from colossalai_amd.ops.attention import attention_ref
+
+    torch.manual_seed(11)
+    B, Smax, Hq, Hkv = 4, 96, 8, 2
+    lens = torch.tensor([17, 96, 1, 40], dtype=torch.int32, device="cuda")
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(B, Smax, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(B, Smax, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    out = _C.decode_attention(q.contiguous(), kc, vc, lens, scale)
+    for b in range(B):
+        n = int(lens[b])
+        ref = attention_ref(q[b : b + 1].unsqueeze(1).float(), kc[b : b + 1, :n].float(),
+                            vc[b : b + 1, :n].float(), causal=False, scale=scale)[0, 0]
+        _bf16_close(out[b], ref, rtol=3e-2, atol=3e-2, frac=1e-4)
+
+
+@pytest.mark.parametrize("D", [64, 128])
+def test_decode_attention_paged_matches_contiguous(D):
+    torch.manual_seed(12)
+    B, Smax, Hq, Hkv, BS = 4, 96, 8, 2, 16
+    lens = torch.tensor([17, 96, 1, 40], dtype=torch.int32, device="cuda")
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(B, Smax, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(B, Smax, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    ref = _C.decode_attention(q.contiguous(), kc, vc, lens, scale)
+
+    # scatter the same logical KV into a shuffled block pool
+    nb_per = Smax // BS
+    total = B * nb_per
+    perm = torch.randperm(total)
+    kpool = torch.zeros(total, BS, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    vpool = torch.zeros_like(kpool)
+    bt = torch.zeros(B, nb_per, dtype=torch.int32, device="cuda")
+    for b in range(B):
+        for j in range(nb_per):
+            blk = int(perm[b * nb_per + j])
+            kpool[blk] = kc[b, j * BS : (j + 1) * BS]
+            vpool[blk] = vc[b, j * BS : (j + 1) * BS]
+            bt[b, j] = blk
+    out = _C.decode_attention_paged(q.contiguous(), kpool, vpool, bt, lens, scale)
+    torch.testing.assert_close(out, ref, rtol=0.0, atol=0.0)
