@@ -167,6 +167,7 @@ class HybridParallelPlugin(Plugin):
         parallel_output: bool = True,
         pp_style: str = "1f1b",
         num_model_chunks: int = 1,
+        sp_zigzag: bool = False,
         **kwargs,
     ):
         assert dist.is_initialized(), "launch colossalai_amd before creating HybridParallelPlugin"
@@ -247,6 +248,7 @@ class HybridParallelPlugin(Plugin):
             enable_tensor_parallelism=tp_size > 1,
             enable_sequence_parallelism=enable_sequence_parallelism,
             sequence_parallelism_mode=sequence_parallelism_mode,
+            sp_zigzag=sp_zigzag,
             enable_flash_attention=enable_flash_attention,
             enable_fused_normalization=enable_fused_normalization,
             parallel_output=parallel_output,

@@ -130,9 +130,17 @@ class LlamaAttention(nn.Module):
             q = qkv[:, :, : Hq * D].reshape(B, S, Hq, D).contiguous()
             k = qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D).contiguous()
             v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D).contiguous()
-            positions = (torch.arange(S, device=hidden.device) + rank * S).repeat(B).int()
+            zigzag = getattr(self, "sp_zigzag", False)
+            if zigzag:
+                # shard = chunks (rank, 2sp-1-rank) of the global sequence
+                sp = dist.get_world_size(sp_group)
+                C = S // 2
+                ar = torch.arange(C, device=hidden.device)
+                positions = torch.cat([ar + rank * C, ar + (2 * sp - 1 - rank) * C]).repeat(B).int()
+            else:
+                positions = (torch.arange(S, device=hidden.device) + rank * S).repeat(B).int()
             q, k = apply_rope(q, k, rope_table, positions)
-            attn = ring_flash_attention(q, k, v, sp_group, causal=True, scale=self.scale)
+            attn = ring_flash_attention(q, k, v, sp_group, causal=True, scale=self.scale, zigzag=zigzag)
             return self.o_proj(attn.reshape(B, S, -1))
         if sp_mode == "all_to_all":
             # Ulysses: scatter heads / gather sequence around attention
@@ -421,15 +429,24 @@ class LlamaForCausalLM(nn.Module):
             assert S % sp == 0, f"seq len {S} must divide sp size {sp}"
             shard = S // sp
             if sp_mode != "split_gather":
-                # Ulysses: inputs and labels are seq-sharded before embedding;
-                # split_gather keeps full inputs (residual split after embed,
-                # output gathered before the loss).
+                # Ulysses/ring: inputs and labels are seq-sharded before the
+                # embedding; split_gather keeps full inputs (residual split
+                # after embed, output gathered before the loss). Zigzag ring
+                # shards are chunks (r, 2sp-1-r) — positionwise loss terms are
+                # layout-invariant, so labels just follow the same split.
+                zigzag = getattr(self, "sp_zigzag", False)
+                if zigzag:
+                    from ..shardformer.layer.ring_attn import zigzag_split
+
+                    assert S % (2 * sp) == 0, f"zigzag ring needs seq {S} % {2 * sp} == 0"
                 if labels is not None:
                     shifted = torch.full_like(labels, -100)
                     shifted[:, :-1] = labels[:, 1:]
-                    labels = shifted[:, rank * shard : (rank + 1) * shard]
+                    labels = (zigzag_split(shifted, sp, rank) if zigzag
+                              else shifted[:, rank * shard : (rank + 1) * shard])
                     self._sp_labels_shifted = True
-                input_ids = input_ids[:, rank * shard : (rank + 1) * shard]
+                input_ids = (zigzag_split(input_ids, sp, rank) if zigzag
+                             else input_ids[:, rank * shard : (rank + 1) * shard])
         out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range)
         if stage_range is not None and stage_range[1] < len(self.model.layers):
             return {"hidden_states": out}

@@ -9,9 +9,21 @@ forwarding, so gradients arrive home after a full cycle. Block math reuses
 the flash-attention HIP kernels (the per-block backward consumes the GLOBAL
 lse/out, FA2-style, so contributions sum exactly).
 
-Round-1 scope: contiguous (non-zigzag) shards — the diagonal-causal load
-imbalance (~2x on the last rank) is accepted; zigzag balancing is a later
-optimization. GQA supported on the HIP path.
+Two shard layouts:
+
+- contiguous: rank r owns sequence block r. Simple, but causal masking
+  makes the last rank do ~sp× the work of the first.
+- zigzag (``zigzag=True``): the sequence splits into 2·sp chunks and rank
+  r owns chunks (r, 2sp−1−r) concatenated. Within a shard all positions
+  of the first chunk precede the second, so the DIAGONAL block is plain
+  causal attention over the concatenated shard; KV arriving from an
+  earlier rank s<r contributes only its first half (full attention), and
+  from a later rank s>r contributes fully but only to the shard's second
+  half of Q — three dense flash calls on slices, no custom masks, and
+  every rank does identical work. Use ``zigzag_split``/``zigzag_gather``
+  to lay out the batch.
+
+GQA supported on the HIP path.
 """
 
 import math
@@ -22,7 +34,23 @@ import torch.distributed as dist
 
 from ...ops import has_kernels
 
-__all__ = ["ring_flash_attention", "RingComm"]
+__all__ = ["ring_flash_attention", "RingComm", "zigzag_split", "zigzag_gather"]
+
+
+def zigzag_split(t: torch.Tensor, world: int, rank: int, dim: int = 1) -> torch.Tensor:
+    """Rank r's zigzag shard: chunks (r, 2·world−1−r) of 2·world chunks."""
+    chunks = t.chunk(2 * world, dim=dim)
+    return torch.cat([chunks[rank], chunks[2 * world - 1 - rank]], dim=dim).contiguous()
+
+
+def zigzag_gather(shards: list, world: int, dim: int = 1) -> torch.Tensor:
+    """Inverse of zigzag_split given every rank's shard (oracle/tests)."""
+    chunks = [None] * (2 * world)
+    for r, sh in enumerate(shards):
+        a, b = sh.chunk(2, dim=dim)
+        chunks[r] = a
+        chunks[2 * world - 1 - r] = b
+    return torch.cat(chunks, dim=dim)
 
 
 class RingComm:
@@ -119,33 +147,50 @@ def _merge(out_a, lse_a, out_b, lse_b):
 
 class _RingFlashAttention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, group, causal, scale):
+    def forward(ctx, q, k, v, group, causal, scale, zigzag):
         comm = RingComm(group)
         rank, world = comm.rank, comm.world
-        out, lse = None, None
+        B, S, Hq, D = q.shape
+        S2 = S // 2
+        # -inf lse + zero out make partial-row merges uniform
+        out = torch.zeros(B, S, Hq, D, dtype=q.dtype, device=q.device)
+        lse = torch.full((B, Hq, S), float("-inf"), dtype=torch.float32, device=q.device)
         cur_k, cur_v = k, v
         for step in range(world):
             src = (rank - step) % world
-            contributes = (not causal) or (src <= rank)
-            if contributes:
-                blk_causal = causal and (src == rank)
-                o_blk, l_blk = _block_fwd(q, cur_k, cur_v, blk_causal, scale)
-                if out is None:
-                    out, lse = o_blk, l_blk.float()
-                else:
+            if not causal:
+                o_blk, l_blk = _block_fwd(q, cur_k, cur_v, False, scale)
+                out, lse = _merge(out, lse, o_blk, l_blk.float())
+            elif not zigzag:
+                if src <= rank:
+                    o_blk, l_blk = _block_fwd(q, cur_k, cur_v, src == rank, scale)
                     out, lse = _merge(out, lse, o_blk, l_blk.float())
+            else:
+                if src == rank:  # diagonal: plain causal over the concat shard
+                    o_blk, l_blk = _block_fwd(q, cur_k, cur_v, True, scale)
+                    out, lse = _merge(out, lse, o_blk, l_blk.float())
+                elif src < rank:  # earlier src: only its first chunk is visible
+                    o_blk, l_blk = _block_fwd(q, cur_k[:, :S2], cur_v[:, :S2], False, scale)
+                    out, lse = _merge(out, lse, o_blk, l_blk.float())
+                else:  # later src: visible only to the shard's second half of Q
+                    o_blk, l_blk = _block_fwd(q[:, S2:].contiguous(), cur_k, cur_v, False, scale)
+                    o_new, l_new = _merge(out[:, S2:], lse[:, :, S2:], o_blk, l_blk.float())
+                    out = torch.cat([out[:, :S2], o_new], dim=1)
+                    lse = torch.cat([lse[:, :, :S2], l_new], dim=2)
             if step + 1 < world:
                 cur_k, cur_v = comm.send_recv([cur_k, cur_v])
         ctx.save_for_backward(q, k, v, out, lse)
-        ctx.group, ctx.causal, ctx.scale = group, causal, scale
+        ctx.group, ctx.causal, ctx.scale, ctx.zigzag = group, causal, scale, zigzag
         return out
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
         group, causal, scale = ctx.group, ctx.causal, ctx.scale
+        zigzag = ctx.zigzag
         comm = RingComm(group)
         rank, world = comm.rank, comm.world
+        S2 = q.shape[1] // 2
         dout = dout.contiguous()
         dq = torch.zeros_like(q, dtype=torch.float32)
         cur_k, cur_v = k, v
@@ -154,23 +199,43 @@ class _RingFlashAttention(torch.autograd.Function):
         # after `world` ring steps the (k, dk) pair returns to its owner
         for step in range(world):
             src = (rank - step) % world
-            contributes = (not causal) or (src <= rank)
-            if contributes:
+            if not causal or (not zigzag and src <= rank):
                 blk_causal = causal and (src == rank)
                 dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k, cur_v, out, lse, blk_causal, scale)
                 dq += dq_b.float()
                 cur_dk += dk_b.float()
                 cur_dv += dv_b.float()
+            elif zigzag:
+                if src == rank:
+                    dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k, cur_v, out, lse, True, scale)
+                    dq += dq_b.float()
+                    cur_dk += dk_b.float()
+                    cur_dv += dv_b.float()
+                elif src < rank:
+                    dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k[:, :S2].contiguous(),
+                                                  cur_v[:, :S2].contiguous(), out, lse, False, scale)
+                    dq += dq_b.float()
+                    cur_dk[:, :S2] += dk_b.float()
+                    cur_dv[:, :S2] += dv_b.float()
+                else:
+                    dq_b, dk_b, dv_b = _block_bwd(
+                        dout[:, S2:].contiguous(), q[:, S2:].contiguous(), cur_k, cur_v,
+                        out[:, S2:].contiguous(), lse[:, :, S2:].contiguous(), False, scale)
+                    dq[:, S2:] += dq_b.float()
+                    cur_dk += dk_b.float()
+                    cur_dv += dv_b.float()
             cur_k, cur_v, cur_dk, cur_dv = comm.send_recv([cur_k, cur_v, cur_dk, cur_dv])
         # one full cycle: cur_dk/cur_dv now hold this rank's own grads
-        return dq.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype), None, None, None
+        return dq.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype), None, None, None, None
 
 
 def ring_flash_attention(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, group, causal: bool = True,
-    scale: Optional[float] = None,
+    scale: Optional[float] = None, zigzag: bool = False,
 ) -> torch.Tensor:
-    """q/k/v [B, S/sp, H, D] contiguous sequence shards -> local out shard."""
+    """q/k/v [B, S/sp, H, D] sequence shards -> local out shard.
+    ``zigzag=True`` expects shards laid out by ``zigzag_split`` and
+    balances causal work exactly across the ring."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    return _RingFlashAttention.apply(q, k, v, group, causal, scale)
+    return _RingFlashAttention.apply(q, k, v, group, causal, scale, zigzag)

@@ -46,6 +46,7 @@ class LlamaPolicy(Policy):
                 attribute_replacement={
                     "sp_mode": "ring_attn",
                     "sp_group": self.shard_config.sequence_parallel_process_group,
+                    "sp_zigzag": self.shard_config.sp_zigzag,
                 }
             )
         sp_sg = (
@@ -81,7 +82,8 @@ class LlamaPolicy(Policy):
                 )
             if sp_ring:
                 attn_attrs.update(
-                    sp_mode="ring_attn", sp_group=self.shard_config.sequence_parallel_process_group
+                    sp_mode="ring_attn", sp_group=self.shard_config.sequence_parallel_process_group,
+                    sp_zigzag=self.shard_config.sp_zigzag,
                 )
             policy[LlamaAttention] = ModulePolicyDescription(
                 attribute_replacement=attn_attrs,
@@ -152,6 +154,7 @@ class LlamaForCausalLMPolicy(LlamaPolicy):
             if mode in ("all_to_all", "ring_attn"):
                 self.model.sp_group = self.shard_config.sequence_parallel_process_group
                 self.model.sp_mode = mode
+                self.model.sp_zigzag = self.shard_config.sp_zigzag and mode == "ring_attn"
             elif mode == "split_gather":
                 # split_gather reuses the tp group (reference semantics)
                 group = self.shard_config.tensor_parallel_process_group

@@ -19,6 +19,7 @@ class ShardConfig:
     enable_tensor_parallelism: bool = True
     enable_sequence_parallelism: bool = False
     sequence_parallelism_mode: Optional[str] = None
+    sp_zigzag: bool = False  # zigzag-balanced ring_attn shards
     enable_flash_attention: bool = True
     enable_fused_normalization: bool = True
     enable_jit_fused: bool = False

@@ -1,0 +1,33 @@
+"""API server over the continuous-batching engine (in-process TestClient)."""
+
+import pytest
+import torch
+
+fastapi = pytest.importorskip("fastapi")
+
+from colossalai_amd.inference import ContinuousBatchEngine, InferenceConfig
+from colossalai_amd.inference.server import create_app
+from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
+
+def test_generate_endpoint():
+    from starlette.testclient import TestClient
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=128)
+    model = LlamaForCausalLM(cfg).eval()
+    engine = ContinuousBatchEngine(model, InferenceConfig(max_batch_size=2, max_input_len=32,
+                                                          max_output_len=16), block_size=4)
+    client = TestClient(create_app(engine))
+
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+
+    r = client.post("/generate", json={"prompt_ids": [5, 17, 42], "max_new_tokens": 6})
+    assert r.status_code == 200
+    out = r.json()["output_ids"]
+    assert out[:3] == [5, 17, 42] and len(out) == 9
+
+    r = client.post("/generate", json={"max_new_tokens": 4})
+    assert r.status_code == 400  # no ids and no tokenizer
