@@ -57,12 +57,20 @@ class ContinuousBatchEngine:
 
         m = self.model.model
         table = m.rope_table(self.device)
+        slot_rows = None
+        if not prefill:
+            # pool row of each sequence's current token, shared by all layers
+            bs = self.kv.block_size
+            rows = [self.kv.table(s)[(int(seq_lens[i]) - 1) // bs] * bs + (int(seq_lens[i]) - 1) % bs
+                    for i, s in enumerate(seq_ids)]
+            slot_rows = torch.tensor(rows, dtype=torch.long, device=self.device)
         residual = m.embed_tokens(input_ids)
         hidden = rms_norm(residual, m.layers[0].input_layernorm_weight, m.eps)
         n = len(m.layers)
         for i, layer in enumerate(m.layers):
             attn_out = layer.self_attn.forward_with_paged_cache(
-                hidden, table, self.kv, i, seq_ids, block_tables, positions, seq_lens, prefill
+                hidden, table, self.kv, i, seq_ids, block_tables, positions, seq_lens, prefill,
+                slot_rows=slot_rows,
             )
             hidden, residual = fused_add_rms_norm(attn_out, residual,
                                                   layer.post_attention_layernorm_weight, m.eps)

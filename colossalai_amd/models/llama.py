@@ -221,7 +221,7 @@ class LlamaAttention(nn.Module):
         return self.o_proj(out.reshape(B, S, -1))
 
     def forward_with_paged_cache(self, hidden, rope_table, kv, layer_idx, seq_ids, block_tables,
-                                 positions, seq_lens, prefill: bool):
+                                 positions, seq_lens, prefill: bool, slot_rows=None):
         """Paged-KV inference path: RoPE + block-pool writes + (flash
         prefill | paged decode kernel) attention. ``kv`` is a
         ``KVCacheManager``; ``block_tables`` is the step's [B, max_blocks]
@@ -255,8 +255,14 @@ class LlamaAttention(nn.Module):
             else:
                 out = attention_ref(q, k, v, causal=True, scale=self.scale)
         else:
-            for i, sid in enumerate(seq_ids):
-                kv.write_token(layer_idx, sid, int(seq_lens[i]) - 1, k[i, 0], v[i, 0])
+            if slot_rows is not None:
+                # one scatter per layer: slot_rows [B] = pool row of each
+                # sequence's current token (precomputed once per step)
+                kv.k_pools[layer_idx].view(-1, Hkv, D)[slot_rows] = k[:, 0]
+                kv.v_pools[layer_idx].view(-1, Hkv, D)[slot_rows] = v[:, 0]
+            else:
+                for i, sid in enumerate(seq_ids):
+                    kv.write_token(layer_idx, sid, int(seq_lens[i]) - 1, k[i, 0], v[i, 0])
             if use_hip_path:
                 from ..ops import kernels
 
