@@ -305,6 +305,93 @@ void multi_tensor_adam(
 #undef DISPATCH_ADAM
 }
 
+// --------------------------------------------------------------------- sgd
+template <typename GT, typename PT, bool HAS_OUT, bool MOMENTUM, bool NESTEROV>
+__global__ __launch_bounds__(ADAM_BLOCK) void multi_tensor_sgd_kernel(
+    TensorListMeta meta, long chunk_size, float lr, float momentum, float dampening,
+    float weight_decay, float inv_div_scale) {
+  const int tensor_id = meta.block_to_tensor[blockIdx.x];
+  const long chunk_id = meta.block_to_chunk[blockIdx.x];
+  const long offset = chunk_id * chunk_size;
+  const long n = min(chunk_size, meta.sizes[tensor_id] - offset);
+
+  const GT* g = reinterpret_cast<const GT*>(meta.addresses[0][tensor_id]) + offset;
+  PT* p = reinterpret_cast<PT*>(meta.addresses[1][tensor_id]) + offset;
+  float* buf = MOMENTUM ? reinterpret_cast<float*>(meta.addresses[2][tensor_id]) + offset : nullptr;
+  unsigned short* out =
+      HAS_OUT ? reinterpret_cast<unsigned short*>(meta.addresses[MOMENTUM ? 3 : 2][tensor_id]) + offset
+              : nullptr;
+
+  for (long i = threadIdx.x; i < n; i += blockDim.x) {
+    float gk = VecIO<GT>::load1(g + i) * inv_div_scale;
+    float pk = VecIO<PT>::load1(p + i);
+    gk += weight_decay * pk;
+    float upd = gk;
+    if (MOMENTUM) {
+      float bk = momentum * buf[i] + (1.0f - dampening) * gk;
+      buf[i] = bk;
+      upd = NESTEROV ? gk + momentum * bk : bk;
+    }
+    pk -= lr * upd;
+    VecIO<PT>::store1(p + i, pk);
+    if (HAS_OUT) VecIO<unsigned short>::store1(out + i, pk);
+  }
+}
+
+void multi_tensor_sgd(
+    std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> params,
+    std::vector<at::Tensor> momentum_bufs,  // empty when momentum == 0
+    std::vector<at::Tensor> param_outs,     // empty, or bf16 working copies
+    double lr, double momentum, double dampening, double weight_decay,
+    bool nesterov, double div_scale, long chunk_size) {
+  TORCH_CHECK(!grads.empty(), "multi_tensor_sgd: empty tensor list");
+  const bool has_mom = !momentum_bufs.empty();
+  const bool has_out = !param_outs.empty();
+  const float inv_div_scale = (float)(1.0 / div_scale);
+  auto stream = at::hip::getCurrentHIPStream();
+  const bool g_bf16 = is_bf16(grads[0]);
+  const bool p_bf16 = is_bf16(params[0]);
+
+  std::vector<std::vector<at::Tensor>> lists = {grads, params};
+  if (has_mom) lists.push_back(momentum_bufs);
+  if (has_out) lists.push_back(param_outs);
+
+  auto run = [&](auto gt, auto pt, auto out_c, auto mom_c, auto nest_c) {
+    using GT = decltype(gt);
+    using PT = decltype(pt);
+    multi_tensor_apply(lists, chunk_size, [&](const TensorListMeta& meta, int nblocks) {
+      hipLaunchKernelGGL((multi_tensor_sgd_kernel<GT, PT, decltype(out_c)::value,
+                                                  decltype(mom_c)::value, decltype(nest_c)::value>),
+                         dim3(nblocks), dim3(ADAM_BLOCK), 0, stream.stream(), meta, chunk_size,
+                         (float)lr, (float)momentum, (float)dampening, (float)weight_decay,
+                         inv_div_scale);
+      HIP_CHECK_LAST();
+    });
+  };
+  using TrueT = std::integral_constant<bool, true>;
+  using FalseT = std::integral_constant<bool, false>;
+  using US = unsigned short;
+#define DISPATCH_SGD(GT, PT)                                                        \
+  do {                                                                              \
+    if (has_mom && nesterov) {                                                      \
+      if (has_out) run(GT{}, PT{}, TrueT{}, TrueT{}, TrueT{});                      \
+      else run(GT{}, PT{}, FalseT{}, TrueT{}, TrueT{});                             \
+    } else if (has_mom) {                                                           \
+      if (has_out) run(GT{}, PT{}, TrueT{}, TrueT{}, FalseT{});                     \
+      else run(GT{}, PT{}, FalseT{}, TrueT{}, FalseT{});                            \
+    } else {                                                                        \
+      if (has_out) run(GT{}, PT{}, TrueT{}, FalseT{}, FalseT{});                    \
+      else run(GT{}, PT{}, FalseT{}, FalseT{}, FalseT{});                           \
+    }                                                                               \
+  } while (0)
+  if (g_bf16 && p_bf16) DISPATCH_SGD(US, US);
+  else if (g_bf16) DISPATCH_SGD(US, float);
+  else if (p_bf16) DISPATCH_SGD(float, US);
+  else DISPATCH_SGD(float, float);
+#undef DISPATCH_SGD
+}
+
 void multi_tensor_scale(std::vector<at::Tensor> inputs, std::vector<at::Tensor> outputs, double scale,
                         long chunk_size) {
   auto stream = at::hip::getCurrentHIPStream();

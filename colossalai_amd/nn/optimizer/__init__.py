@@ -2,9 +2,11 @@ from .adafactor import Adafactor
 from .came import CAME
 from .cpu_adam import CPUAdam
 from .fused_adam import FusedAdam
+from .fused_sgd import FusedSGD
 from .galore import GaLoreAdamW
 from .hybrid_adam import HybridAdam
 from .lamb import Lamb
 from .lars import Lars
 
-__all__ = ["FusedAdam", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
+__all__ = ["FusedAdam",
+    "FusedSGD", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]

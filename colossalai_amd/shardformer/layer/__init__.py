@@ -10,8 +10,9 @@ from ._operation import (
 )
 from .embedding import VocabParallelEmbedding1D
 from .linear import Linear1D_Col, Linear1D_Row
-from .loss import DistCrossEntropy, dist_cross_entropy
+from .loss import DistCrossEntropy, DistLogProb, dist_cross_entropy, dist_log_prob
 from .parallel_module import ParallelModule
+from .randomizer import Randomizer
 
 __all__ = [
     "Linear1D_Col",
@@ -20,6 +21,9 @@ __all__ = [
     "DistCrossEntropy",
     "dist_cross_entropy",
     "ParallelModule",
+    "Randomizer",
+    "DistLogProb",
+    "dist_log_prob",
     "linear_with_async_comm",
     "reduce_forward",
     "reduce_backward",

@@ -348,3 +348,23 @@ def test_decode_attention_paged_matches_contiguous(D):
             bt[b, j] = blk
     out = _C.decode_attention_paged(q.contiguous(), kpool, vpool, bt, lens, scale)
     torch.testing.assert_close(out, ref, rtol=0.0, atol=0.0)
+
+
+def test_multi_tensor_sgd():
+    torch.manual_seed(13)
+    shapes = [(1000,), (333,), (512, 64)]
+    lr, mom, damp, wd = 1e-2, 0.9, 0.0, 0.1
+    gs, ps, bufs, refs = [], [], [], []
+    for sh in shapes:
+        g = torch.randn(sh, device="cuda", dtype=torch.bfloat16)
+        p = torch.randn(sh, device="cuda", dtype=torch.float32)
+        b = torch.randn(sh, device="cuda").abs() * 0.1
+        refs.append((g.clone(), p.clone(), b.clone()))
+        gs.append(g); ps.append(p); bufs.append(b)
+    _C.multi_tensor_sgd(gs, ps, bufs, [], lr, mom, damp, wd, True, 1.0, 65536)
+    for (g0, p0, b0), p, b in zip(refs, ps, bufs):
+        gf = g0.float() + wd * p0
+        br = mom * b0 + gf
+        pr = p0 - lr * (gf + mom * br)
+        torch.testing.assert_close(b.cpu(), br.cpu(), rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(p.cpu(), pr.cpu(), rtol=1e-5, atol=1e-6)

@@ -86,3 +86,26 @@ def test_galore_state_is_lowrank():
     opt.step()
     st = opt.state[p]
     assert st["exp_avg"].shape in ((256, 8), (8, 512)), st["exp_avg"].shape
+
+
+def test_fused_sgd():
+    from colossalai_amd.nn import FusedSGD
+
+    _converges(FusedSGD, lr=0.5, momentum=0.9)
+
+
+def test_fused_sgd_matches_torch():
+    from colossalai_amd.nn import FusedSGD
+
+    torch.manual_seed(0)
+    p1 = torch.randn(100, requires_grad=True)
+    p2 = p1.detach().clone().requires_grad_(True)
+    o1 = FusedSGD([p1], lr=1e-2, momentum=0.9, weight_decay=0.1, nesterov=True)
+    o2 = torch.optim.SGD([p2], lr=1e-2, momentum=0.9, weight_decay=0.1, nesterov=True)
+    for step in range(5):
+        g = torch.randn(100)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        o1.step()
+        o2.step()
+        torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-6), step
