@@ -1,0 +1,164 @@
+"""Speculative decoding (reference: colossalai/inference/core/llm_engine.py
+:301-388 speculative + GLIDE path — greedy draft-verify rebuilt on the
+native cache kernels).
+
+A small draft model proposes ``gamma`` tokens autoregressively; the target
+model scores all of them in ONE chunk forward against its KV cache —
+attention of the chunk = LSE-merge of (non-causal chunk×prefix) and
+(causal chunk×chunk), the same block merge the ring attention uses, so no
+special kernel is needed. Greedy acceptance: keep draft tokens while they
+match the target argmax; the first mismatch is replaced by the target's
+token (one guaranteed token per round, up to gamma+1).
+
+Cache rollback is free: sequence length bookkeeping moves back and stale
+cache rows are overwritten by position on the next write.
+"""
+
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..models.llama import LlamaForCausalLM
+from .config import GenerationConfig, InferenceConfig
+
+__all__ = ["SpeculativeEngine"]
+
+
+def _chunk_attend(attn, hidden, table, kcache, vcache, prefix_len: int, positions):
+    """Chunk of new tokens vs cache prefix + itself (causal)."""
+    from ..ops import has_kernels
+    from ..shardformer.layer.ring_attn import _block_fwd, _merge
+
+    B, S, _ = hidden.shape
+    Hq, Hkv, D = attn.num_heads, attn.num_kv_heads, attn.head_dim
+    qkv = attn.qkv_proj(hidden)
+    q = qkv[:, :, : Hq * D].view(B, S, Hq, D)
+    k = qkv[:, :, Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+    v = qkv[:, :, (Hq + Hkv) * D :].view(B, S, Hkv, D)
+    if hidden.is_cuda and has_kernels():
+        from ..ops import kernels
+
+        kernels().rope_inplace(q, k, table, positions, False)
+    else:
+        from ..ops.rope import apply_rope_ref
+
+        q, k = apply_rope_ref(q, k, table, positions.long(), S, False)
+    kcache[:, prefix_len : prefix_len + S] = k
+    vcache[:, prefix_len : prefix_len + S] = v
+    out, lse = _block_fwd(q.contiguous(), k.contiguous(), v.contiguous(), True, attn.scale)
+    if prefix_len > 0:
+        o_pre, l_pre = _block_fwd(q.contiguous(), kcache[:, :prefix_len].contiguous(),
+                                  vcache[:, :prefix_len].contiguous(), False, attn.scale)
+        out, lse = _merge(out, lse.float(), o_pre, l_pre.float())
+    return attn.o_proj(out.reshape(B, S, Hq * D))
+
+
+class _CachedModel:
+    """Target/draft wrapper: KV caches + chunk forward returning all-position logits."""
+
+    def __init__(self, model: LlamaForCausalLM, max_seq_len: int):
+        self.model = model.eval()
+        self.device = next(model.parameters()).device
+        self.dtype = next(model.parameters()).dtype
+        cfg = model.config
+        self.caches = [
+            (torch.zeros(1, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+                         device=self.device, dtype=self.dtype),
+             torch.zeros(1, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+                         device=self.device, dtype=self.dtype))
+            for _ in range(cfg.num_hidden_layers)
+        ]
+        self.len = 0
+
+    @torch.inference_mode()
+    def forward_chunk(self, token_ids: List[int]) -> torch.Tensor:
+        """Append token_ids at the current length; -> logits [S, V]."""
+        from ..ops import fused_add_rms_norm, rms_norm
+
+        m = self.model.model
+        ids = torch.tensor([token_ids], device=self.device)
+        S = ids.shape[1]
+        positions = (torch.arange(S, device=self.device) + self.len).int()
+        table = m.rope_table(self.device)
+        residual = m.embed_tokens(ids)
+        hidden = rms_norm(residual, m.layers[0].input_layernorm_weight, m.eps)
+        n = len(m.layers)
+        for i, layer in enumerate(m.layers):
+            kc, vc = self.caches[i]
+            attn_out = _chunk_attend(layer.self_attn, hidden, table, kc, vc, self.len, positions)
+            hidden, residual = fused_add_rms_norm(attn_out, residual,
+                                                  layer.post_attention_layernorm_weight, m.eps)
+            mlp_out = layer.mlp(hidden)
+            next_w = m.layers[i + 1].input_layernorm_weight if i + 1 < n else m.norm_weight
+            hidden, residual = fused_add_rms_norm(mlp_out, residual, next_w, m.eps)
+        self.len += S
+        return self.model.lm_head(hidden[0]).float()
+
+    def rollback(self, new_len: int):
+        self.len = new_len
+
+
+class SpeculativeEngine:
+    """Greedy speculative decoding, batch size 1."""
+
+    def __init__(self, target: LlamaForCausalLM, draft: LlamaForCausalLM,
+                 config: Optional[InferenceConfig] = None, gamma: int = 4):
+        self.config = config or InferenceConfig()
+        self.gamma = gamma
+        self.target = _CachedModel(target, self.config.max_seq_len)
+        self.draft = _CachedModel(draft, self.config.max_seq_len)
+        self.accepted = 0
+        self.proposed = 0
+
+    @torch.inference_mode()
+    def generate(self, prompt: List[int], generation_config: Optional[GenerationConfig] = None) -> List[int]:
+        gen = generation_config or GenerationConfig()
+        self.target.rollback(0)
+        self.draft.rollback(0)
+        seq = list(prompt)
+        t_logits = self.target.forward_chunk(seq)
+        self.draft.forward_chunk(seq)
+        next_tok = int(t_logits[-1].argmax())
+        seq.append(next_tok)
+
+        while len(seq) - len(prompt) < gen.max_new_tokens and len(seq) < self.config.max_seq_len - self.gamma - 1:
+            # ---- draft proposes gamma tokens from the last accepted token
+            proposal = []
+            tok = next_tok
+            for _ in range(self.gamma):
+                d_logits = self.draft.forward_chunk([tok])
+                tok = int(d_logits[-1].argmax())
+                proposal.append(tok)
+            # ---- target verifies [next_tok, proposal...] in one chunk
+            t_logits = self.target.forward_chunk([next_tok] + proposal)
+            self.proposed += len(proposal)
+            n_acc = 0
+            for j, p in enumerate(proposal):
+                want = int(t_logits[j].argmax())
+                if p == want:
+                    n_acc += 1
+                else:
+                    break
+            self.accepted += n_acc
+            accepted_tokens = proposal[:n_acc]
+            # first rejected (or bonus) token comes from the target
+            correction = int(t_logits[n_acc].argmax())
+            seq.extend(accepted_tokens)
+            seq.append(correction)
+            next_tok = correction
+            # ---- resync both caches to cover exactly seq[:-1]
+            needed = len(seq) - 1
+            self.target.rollback(needed)
+            if self.draft.len > needed:
+                self.draft.rollback(needed)
+            elif self.draft.len < needed:
+                # every draft token accepted: the last proposal was never
+                # fed to the draft — replay the gap
+                self.draft.forward_chunk(seq[self.draft.len : needed])
+            if self.config.eos_token_id is not None and correction == self.config.eos_token_id:
+                break
+        return seq[: len(prompt) + gen.max_new_tokens]
+
+    @property
+    def acceptance_rate(self) -> float:
+        return self.accepted / max(self.proposed, 1)

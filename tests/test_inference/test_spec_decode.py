@@ -1,0 +1,49 @@
+"""Speculative decoding: output must equal plain greedy decoding of the
+TARGET model exactly (lossless property of greedy spec decode)."""
+
+import pytest
+import torch
+
+from colossalai_amd.inference import GenerationConfig, InferenceConfig, SpeculativeEngine
+from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
+
+def _cfg(layers, hidden):
+    return LlamaConfig(vocab_size=128, hidden_size=hidden, intermediate_size=hidden * 2,
+                       num_hidden_layers=layers, num_attention_heads=4, num_key_value_heads=2,
+                       max_position_embeddings=128)
+
+
+def _oracle(model, prompt, n_new):
+    seq = list(prompt)
+    for _ in range(n_new):
+        logits = model(torch.tensor([seq]))["logits"][0, -1]
+        seq.append(int(logits.argmax()))
+    return seq
+
+
+@pytest.mark.parametrize("gamma", [1, 3, 5])
+def test_spec_decode_matches_target_greedy(gamma):
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(_cfg(3, 64)).eval()
+    draft = LlamaForCausalLM(_cfg(1, 32)).eval()
+    eng = SpeculativeEngine(target, draft,
+                            InferenceConfig(max_batch_size=1, max_input_len=32, max_output_len=32),
+                            gamma=gamma)
+    for prompt in ([5, 17, 42, 7], [99]):
+        out = eng.generate(prompt, GenerationConfig(max_new_tokens=16))
+        ref = _oracle(target, prompt, 16)
+        assert out == ref, f"gamma={gamma}: spec {out} vs greedy {ref}"
+    assert eng.proposed > 0
+
+
+def test_spec_decode_self_draft_accepts_everything():
+    """Draft == target: every proposal must be accepted."""
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(_cfg(2, 64)).eval()
+    eng = SpeculativeEngine(target, target,
+                            InferenceConfig(max_batch_size=1, max_input_len=32, max_output_len=32),
+                            gamma=4)
+    out = eng.generate([5, 17, 42], GenerationConfig(max_new_tokens=12))
+    assert out == _oracle(target, [5, 17, 42], 12)
+    assert eng.acceptance_rate == 1.0, eng.acceptance_rate
