@@ -51,6 +51,10 @@ at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, 
 at::Tensor decode_attention_paged(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
                                   at::Tensor block_tables, at::Tensor seq_lens, double scale);
 
+// moe.hip
+at::Tensor moe_combine_fwd(at::Tensor y, at::Tensor inv, at::Tensor topw);
+std::vector<at::Tensor> moe_combine_bwd(at::Tensor dout, at::Tensor y, at::Tensor inv, at::Tensor topw);
+
 // flash_attn.hip
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale);
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
@@ -78,6 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache");
   m.def("decode_attention_paged", &cai::decode_attention_paged,
         "single-token attention over a paged (block-table) KV pool");
+  m.def("moe_combine_fwd", &cai::moe_combine_fwd, "fused MoE un-permute + weighted top-k sum");
+  m.def("moe_combine_bwd", &cai::moe_combine_bwd, "MoE combine backward (dy + routing-weight grads)");
   m.def("flash_attn_fwd", &cai::flash_attn_fwd, "flash attention forward (bf16, causal, GQA)");
   m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward");
 }

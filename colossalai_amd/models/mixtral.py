@@ -87,8 +87,7 @@ class MixtralSparseMoeBlock(nn.Module):
         logits = self.gate(x).float()
         probs = F.softmax(logits, dim=-1)
         topw, topi = torch.topk(probs, self.top_k, dim=-1)
-        topw = topw / topw.sum(-1, keepdim=True)
-        topw = topw.to(hidden.dtype)
+        topw = topw / topw.sum(-1, keepdim=True)  # fp32 routing weights
 
         flat_expert = topi.reshape(-1)  # [T*k]
         order = torch.argsort(flat_expert, stable=True)
@@ -126,11 +125,12 @@ class MixtralSparseMoeBlock(nn.Module):
             local_eids = sorted_expert
             y = self._experts_forward(x_disp, local_eids)
 
-        # un-sort dispatch slots and combine with routing weights
-        slot_out = torch.empty_like(y)
-        slot_out[order] = y
-        slot_out = slot_out.reshape(T, self.top_k, H)
-        combined = (slot_out * topw.unsqueeze(-1)).sum(dim=1)
+        # fused un-permute + routing-weight combine (ops/moe.py -> csrc/moe.hip)
+        from ..ops import moe_combine
+
+        inv_perm = torch.empty_like(order)
+        inv_perm[order] = torch.arange(order.numel(), device=x.device)
+        combined = moe_combine(y, inv_perm, topw)
         return combined.reshape(B, S, H).to(hidden.dtype)
 
 

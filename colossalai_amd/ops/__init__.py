@@ -3,6 +3,7 @@ from .attention import attention_ref, flash_attention, fused_rope_attention
 from .layernorm import layer_norm
 from .norm import fused_add_rms_norm, rms_norm, rms_norm_ref
 from .rope import apply_rope, apply_rope_ref, build_rope_table
+from .moe import moe_combine
 from .swiglu import swiglu, swiglu_ref
 
 __all__ = [
@@ -18,6 +19,7 @@ __all__ = [
     "apply_rope",
     "apply_rope_ref",
     "build_rope_table",
+    "moe_combine",
     "swiglu",
     "swiglu_ref",
 ]

@@ -368,3 +368,25 @@ def test_multi_tensor_sgd():
         pr = p0 - lr * (gf + mom * br)
         torch.testing.assert_close(b.cpu(), br.cpu(), rtol=1e-5, atol=1e-6)
         torch.testing.assert_close(p.cpu(), pr.cpu(), rtol=1e-5, atol=1e-6)
+
+
+def test_moe_combine():
+    from colossalai_amd.ops.moe import _MoeCombine
+
+    torch.manual_seed(14)
+    T, k, H = 64, 2, 256
+    N = T * k
+    y = torch.randn(N, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    inv = torch.randperm(N, device="cuda").int()
+    w = torch.rand(T, k, device="cuda", requires_grad=True)
+    out = _MoeCombine.apply(y, inv, w)
+    # fp32 reference
+    yr = y.detach().float().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    ref = (yr[inv.long()].view(T, k, H) * wr.unsqueeze(-1)).sum(1)
+    _bf16_close(out, ref, rtol=2e-2, atol=2e-2)
+    d = torch.randn(T, H, device="cuda", dtype=torch.bfloat16)
+    out.backward(d)
+    ref.backward(d.float())
+    _bf16_close(y.grad, yr.grad, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(w.grad.cpu(), wr.grad.cpu(), rtol=2e-2, atol=2e-1)
