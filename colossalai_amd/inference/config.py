@@ -16,6 +16,9 @@ class InferenceConfig:
     dtype: torch.dtype = torch.bfloat16
     pad_token_id: int = 0
     eos_token_id: Optional[int] = None
+    # capture the decode step in a hipGraph (one graph per batch size):
+    # removes per-layer launch gaps in the latency-bound decode loop
+    use_hip_graph: bool = False
 
     @property
     def max_seq_len(self) -> int:

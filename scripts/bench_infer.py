@@ -60,3 +60,15 @@ torch.cuda.synchronize()
 dt = time.perf_counter() - t0
 total_new = sum(len(t) - 64 for t in done.values())
 print(f"llama-7b bf16 16-req ragged (8 slots): {total_new} tokens in {dt:.2f}s = {total_new/dt:.1f} tok/s")
+
+# ---- hipGraph-captured decode loop
+gengine = LLMEngine(model, InferenceConfig(max_batch_size=8, max_input_len=512, max_output_len=256,
+                                           use_hip_graph=True))
+gengine.generate(prompts, GenerationConfig(max_new_tokens=8))  # warmup + capture
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+out = gengine.generate(prompts, GenerationConfig(max_new_tokens=N))
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+total_new = sum(len(o) - 128 for o in out)
+print(f"llama-7b bf16 bs8 hipGraph: {total_new} tokens in {dt:.2f}s = {total_new/dt:.1f} tok/s")

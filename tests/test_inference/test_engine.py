@@ -53,3 +53,18 @@ def test_engine_gpu():
     # decode kernel numerics: compare against bf16 CPU full recompute for a few tokens
     ref = _oracle_generate(model_cpu.bfloat16(), prompts[0], 4)
     assert out[0][: len(prompts[0]) + 2] == ref[: len(prompts[0]) + 2], (out[0], ref)
+
+
+@pytest.mark.gpu
+def test_engine_hip_graph_matches_eager():
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=512, hidden_size=256, intermediate_size=512, num_hidden_layers=2,
+                      num_attention_heads=2, num_key_value_heads=2, max_position_embeddings=512)
+    model = LlamaForCausalLM(cfg).to("cuda").bfloat16().eval()
+    prompts = [[5, 17, 42, 7, 100, 250], [99, 3, 4]]
+    ref = LLMEngine(model, InferenceConfig(max_batch_size=2, max_input_len=64, max_output_len=32)
+                    ).generate(prompts, GenerationConfig(max_new_tokens=12))
+    out = LLMEngine(model, InferenceConfig(max_batch_size=2, max_input_len=64, max_output_len=32,
+                                           use_hip_graph=True)
+                    ).generate(prompts, GenerationConfig(max_new_tokens=12))
+    assert out == ref, f"graph {out} vs eager {ref}"
