@@ -1,6 +1,7 @@
 from .adafactor import Adafactor
 from .came import CAME
 from .cpu_adam import CPUAdam
+from .distributed_lamb import DistributedLamb
 from .fused_adam import FusedAdam
 from .fused_sgd import FusedSGD
 from .galore import GaLoreAdamW
@@ -9,4 +10,4 @@ from .lamb import Lamb
 from .lars import Lars
 
 __all__ = ["FusedAdam",
-    "FusedSGD", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
+    "FusedSGD", "DistributedLamb", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]

@@ -104,11 +104,14 @@ class Linear1D_Col(ParallelModule):
         layer.seq_parallel_dim = kwargs.get("seq_parallel_dim", 1)
         layer.split_sizes = kwargs.get("split_sizes", None)
         layer.weight = Parameter(_shard_rows(module.weight.data, process_group, layer.split_sizes))
+        layer.weight.tp_sharded = True  # distributed optimizers: norms need the tp all-reduce
         layer.bias = (
             Parameter(_shard_rows(module.bias.data.unsqueeze(-1), process_group, layer.split_sizes).squeeze(-1))
             if module.bias is not None
             else None
         )
+        if layer.bias is not None:
+            layer.bias.tp_sharded = True
         return layer
 
     def gather_weight(self) -> torch.Tensor:
@@ -171,6 +174,7 @@ class Linear1D_Row(ParallelModule):
         else:
             w = module.weight.data.chunk(layer.world, dim=1)[rank].contiguous()
         layer.weight = Parameter(w)
+        layer.weight.tp_sharded = True
         # bias applied once (after reduce), kept replicated
         layer.bias = Parameter(module.bias.data.clone()) if module.bias is not None else None
         return layer
