@@ -107,3 +107,9 @@ class MoeHybridParallelPlugin(HybridParallelPlugin):
                 **({} if self.precision != "fp16" else self.amp_kwargs),
             )
         return model, optimizer, criterion, dataloader, lr_scheduler
+
+    def get_checkpoint_io(self):
+        from ...checkpoint_io.moe_checkpoint import MoECheckpointIO
+
+        return MoECheckpointIO(self.dp_group, self.pp_group, self.tp_group, self.ep_group,
+                               self.sp_size)

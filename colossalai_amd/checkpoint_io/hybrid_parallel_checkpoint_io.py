@@ -96,12 +96,21 @@ class HybridParallelCheckpointIO(GeneralCheckpointIO):
         self.pp_size = dist.get_world_size(pp_group) if pp_group is not None else 1
         self.global_rank = dist.get_rank()
 
+    def _local_state_dict(self, model) -> dict:
+        """Collective per-rank full-weight state dict (TP gathered); MoE IO
+        extends this with the EP expert gather."""
+        return _full_state_dict(model, self.tp_group)
+
+    def _pre_load(self, model, full_sd: dict) -> dict:
+        """Hook: adapt a full checkpoint to this rank before sharded load."""
+        return full_sd
+
     def save_unsharded_model(self, model, checkpoint: str, gather_dtensor: bool = True,
                              use_safetensors: bool = False, use_async: bool = False):
         if isinstance(model, ModelWrapper):
             model = model.unwrap()
         # TP weight gathers are collective: every rank participates
-        local_sd = _full_state_dict(model, self.tp_group)
+        local_sd = self._local_state_dict(model)
         if self.pp_size > 1:
             gathered = [None] * self.pp_size
             dist.all_gather_object(gathered, local_sd, group=self.pp_group)
@@ -116,8 +125,8 @@ class HybridParallelCheckpointIO(GeneralCheckpointIO):
     def load_unsharded_model(self, model, checkpoint: str, strict: bool = False):
         if isinstance(model, ModelWrapper):
             model = model.unwrap()
-        full_sd = load_state_dict(checkpoint)
-        _load_into_sharded(model, dict(full_sd))
+        full_sd = self._pre_load(model, dict(load_state_dict(checkpoint)))
+        _load_into_sharded(model, full_sd)
         if dist.is_initialized():
             dist.barrier()
 
@@ -126,7 +135,7 @@ class HybridParallelCheckpointIO(GeneralCheckpointIO):
         # v1: gather to full on rank 0 then reuse the general sharded writer
         if isinstance(model, ModelWrapper):
             model = model.unwrap()
-        local_sd = _full_state_dict(model, self.tp_group)
+        local_sd = self._local_state_dict(model)
         if self.pp_size > 1:
             gathered = [None] * self.pp_size
             dist.all_gather_object(gathered, local_sd, group=self.pp_group)
