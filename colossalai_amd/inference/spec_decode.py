@@ -124,25 +124,50 @@ class SpeculativeEngine:
         while len(seq) - len(prompt) < gen.max_new_tokens and len(seq) < self.config.max_seq_len - self.gamma - 1:
             # ---- draft proposes gamma tokens from the last accepted token
             proposal = []
+            q_probs = []  # draft prob of each proposed token (sampling mode)
             tok = next_tok
             for _ in range(self.gamma):
                 d_logits = self.draft.forward_chunk([tok])
-                tok = int(d_logits[-1].argmax())
+                if gen.do_sample:
+                    q = torch.softmax(d_logits[-1] / max(gen.temperature, 1e-5), dim=-1)
+                    tok = int(torch.multinomial(q, 1))
+                    q_probs.append(q)
+                else:
+                    tok = int(d_logits[-1].argmax())
                 proposal.append(tok)
             # ---- target verifies [next_tok, proposal...] in one chunk
             t_logits = self.target.forward_chunk([next_tok] + proposal)
             self.proposed += len(proposal)
-            n_acc = 0
-            for j, p in enumerate(proposal):
-                want = int(t_logits[j].argmax())
-                if p == want:
-                    n_acc += 1
-                else:
+            if gen.do_sample:
+                # Leviathan et al. acceptance: keep x_j with prob
+                # min(1, p(x_j)/q(x_j)); on rejection resample from
+                # max(0, p - q) normalized
+                n_acc = 0
+                correction = None
+                for j, x in enumerate(proposal):
+                    p = torch.softmax(t_logits[j] / max(gen.temperature, 1e-5), dim=-1)
+                    q = q_probs[j]
+                    ratio = float(p[x] / q[x].clamp_min(1e-20))
+                    if float(torch.rand(())) < ratio:
+                        n_acc += 1
+                        continue
+                    resid = (p - q).clamp_min(0)
+                    s = resid.sum()
+                    correction = int(torch.multinomial(resid / s if s > 0 else p, 1))
                     break
+                if correction is None:  # all accepted: bonus from the target
+                    p = torch.softmax(t_logits[self.gamma] / max(gen.temperature, 1e-5), dim=-1)
+                    correction = int(torch.multinomial(p, 1))
+            else:
+                n_acc = 0
+                for j, x in enumerate(proposal):
+                    if x == int(t_logits[j].argmax()):
+                        n_acc += 1
+                    else:
+                        break
+                correction = int(t_logits[n_acc].argmax())
             self.accepted += n_acc
             accepted_tokens = proposal[:n_acc]
-            # first rejected (or bonus) token comes from the target
-            correction = int(t_logits[n_acc].argmax())
             seq.extend(accepted_tokens)
             seq.append(correction)
             next_tok = correction

