@@ -20,6 +20,7 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "BertForMaskedLM": ("colossalai_amd.shardformer.policies.bert", "BertForMaskedLMPolicy"),
     "BertForSequenceClassification": ("colossalai_amd.shardformer.policies.bert", "BertForSequenceClassificationPolicy"),
     "BertModel": ("colossalai_amd.shardformer.policies.bert", "BertPolicy"),
+    "T5ForConditionalGeneration": ("colossalai_amd.shardformer.policies.t5", "T5ForConditionalGenerationPolicy"),
 }
 
 

@@ -64,3 +64,34 @@ def test_tp_opt():
 @rerun_if_address_is_in_use()
 def test_tp_bert():
     spawn(run_tp_bert, 2)
+
+
+def run_tp_t5(rank, world_size, port):
+    from colossalai_amd.models.t5 import T5Config, T5ForConditionalGeneration
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = T5Config(vocab_size=256, d_model=64, d_kv=16, d_ff=128, num_layers=2,
+                   num_decoder_layers=2, num_heads=4, relative_attention_num_buckets=8,
+                   relative_attention_max_distance=32)
+    ref = T5ForConditionalGeneration(cfg)
+    model = copy.deepcopy(ref)
+    model, _ = ShardFormer(ShardConfig(tensor_parallel_process_group=dist.group.WORLD)).optimize(model)
+    assert model.encoder.block[0].self_attn.num_heads == 2
+    assert model.encoder.block[0].self_attn.relative_attention_bias.weight.shape[1] == 2
+
+    x = torch.randint(0, 256, (2, 12))
+    y = torch.randint(0, 256, (2, 8))
+    out = model(input_ids=x, labels=y)
+    out_ref = ref(input_ids=x, labels=y)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.encoder.block[0].self_ln_weight.grad,
+                       ref.encoder.block[0].self_ln_weight.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_tp_t5():
+    spawn(run_tp_t5, 2)
