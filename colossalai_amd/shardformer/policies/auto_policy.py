@@ -21,6 +21,7 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "BertForSequenceClassification": ("colossalai_amd.shardformer.policies.bert", "BertForSequenceClassificationPolicy"),
     "BertModel": ("colossalai_amd.shardformer.policies.bert", "BertPolicy"),
     "T5ForConditionalGeneration": ("colossalai_amd.shardformer.policies.t5", "T5ForConditionalGenerationPolicy"),
+    "ViTForImageClassification": ("colossalai_amd.shardformer.policies.vit", "ViTForImageClassificationPolicy"),
 }
 
 

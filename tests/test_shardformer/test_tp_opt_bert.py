@@ -95,3 +95,32 @@ def run_tp_t5(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_tp_t5():
     spawn(run_tp_t5, 2)
+
+
+def run_tp_vit(rank, world_size, port):
+    from colossalai_amd.models.vit import ViTConfig, ViTForImageClassification
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = ViTConfig(image_size=32, patch_size=8, hidden_size=64, num_hidden_layers=2,
+                    num_attention_heads=4, intermediate_size=128, num_labels=5)
+    ref = ViTForImageClassification(cfg)
+    model = copy.deepcopy(ref)
+    model, _ = ShardFormer(ShardConfig(tensor_parallel_process_group=dist.group.WORLD)).optimize(model)
+    assert model.vit.layers[0].attention.num_heads == 2
+
+    x = torch.randn(2, 3, 32, 32)
+    y = torch.randint(0, 5, (2,))
+    out = model(x, labels=y)
+    out_ref = ref(x, labels=y)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.vit.layers[0].ln1_weight.grad,
+                       ref.vit.layers[0].ln1_weight.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_tp_vit():
+    spawn(run_tp_vit, 2)
