@@ -62,6 +62,7 @@ class Chunk:
         self.params: List[Tuple[nn.Parameter, int]] = []  # (param, offset)
         self.gathered = True  # starts materialized (init copies weights in)
         self.grads_done = 0
+        self.pending_event = None  # hipEvent of an in-flight prefetch gather
 
     # ---------------------------------------------------------- param space
     def seal(self):
@@ -74,6 +75,12 @@ class Chunk:
 
     def gather(self, group):
         if self.gathered:
+            return
+        if self.pending_event is not None:
+            # a prefetch issued this gather on the comm stream: just join
+            torch.cuda.current_stream().wait_event(self.pending_event)
+            self.pending_event = None
+            self.gathered = True
             return
         self.flat.untyped_storage().resize_(self.numel * self.flat.element_size())
         if self.world > 1:
@@ -91,7 +98,14 @@ class Chunk:
             self.flat.copy_(self.shard)
 
     def release(self):
-        if self.persistent or not self.gathered:
+        if self.persistent:
+            return
+        if self.pending_event is not None:
+            # never free a payload the comm stream is still writing
+            torch.cuda.current_stream().wait_event(self.pending_event)
+            self.pending_event = None
+            self.gathered = True
+        if not self.gathered:
             return
         self.flat.untyped_storage().resize_(0)
         self.gathered = False
@@ -180,10 +194,16 @@ class GeminiDDP(ModelWrapper):
         residual = [p for p in module.parameters() if p.requires_grad and id(p) not in unit_param_ids]
         self.persistent_chunks = self._pack(residual, persistent=True)
 
-        for u in units:
-            u.register_forward_pre_hook(self._make_fwd_gather(self.unit_chunks[id(u)]))
+        # xGMI/compute overlap: unit i's pre-hooks issue unit i±1's
+        # all-gather on a dedicated comm stream under the current unit's GEMMs
+        self._units = list(units)
+        self._prefetch_stream = torch.cuda.Stream() if device.type == "cuda" else None
+        for i, u in enumerate(units):
+            nxt = self.unit_chunks[id(units[i + 1])] if i + 1 < len(units) else []
+            prv = self.unit_chunks[id(units[i - 1])] if i > 0 else []
+            u.register_forward_pre_hook(self._make_fwd_gather(self.unit_chunks[id(u)], nxt))
             u.register_forward_hook(self._make_fwd_release(self.unit_chunks[id(u)]))
-            u.register_full_backward_pre_hook(self._make_bwd_gather(self.unit_chunks[id(u)]))
+            u.register_full_backward_pre_hook(self._make_bwd_gather(self.unit_chunks[id(u)], prv))
 
         self._hook_handles = []
         for p in module.parameters():
@@ -229,10 +249,29 @@ class GeminiDDP(ModelWrapper):
         return made
 
     # ------------------------------------------------------------------ hooks
-    def _make_fwd_gather(self, chunks):
+    def _prefetch(self, chunks):
+        if self._prefetch_stream is None:
+            return
+        todo = [c for c in chunks if not c.gathered and c.pending_event is None]
+        if not todo:
+            return
+        self._prefetch_stream.wait_stream(torch.cuda.current_stream())
+        for c in todo:
+            c.flat.untyped_storage().resize_(c.numel * c.flat.element_size())
+            with torch.cuda.stream(self._prefetch_stream):
+                if self.world > 1:
+                    dist.all_gather_into_tensor(c.flat, c.shard, group=self.group)
+                else:
+                    c.flat.copy_(c.shard)
+                ev = torch.cuda.Event()
+                ev.record(self._prefetch_stream)
+            c.pending_event = ev
+
+    def _make_fwd_gather(self, chunks, next_chunks):
         def hook(module, args):
             for c in chunks:
                 c.gather(self.group)
+            self._prefetch(next_chunks)
             return None
         return hook
 
@@ -243,11 +282,12 @@ class GeminiDDP(ModelWrapper):
             return None
         return hook
 
-    def _make_bwd_gather(self, chunks):
+    def _make_bwd_gather(self, chunks, prev_chunks):
         def hook(module, grad_output):
             for c in chunks:
                 c.gather(self.group)
                 c.materialize_grads()
+            self._prefetch(prev_chunks)
             return None
         return hook
 
