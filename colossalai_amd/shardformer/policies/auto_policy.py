@@ -22,6 +22,7 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "BertModel": ("colossalai_amd.shardformer.policies.bert", "BertPolicy"),
     "T5ForConditionalGeneration": ("colossalai_amd.shardformer.policies.t5", "T5ForConditionalGenerationPolicy"),
     "ViTForImageClassification": ("colossalai_amd.shardformer.policies.vit", "ViTForImageClassificationPolicy"),
+    "FalconForCausalLM": ("colossalai_amd.shardformer.policies.falcon", "FalconForCausalLMPolicy"),
 }
 
 
