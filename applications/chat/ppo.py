@@ -19,7 +19,7 @@ from colossalai_amd import Booster
 
 from .dpo import sequence_log_probs  # noqa: F401  (re-exported convenience)
 
-__all__ = ["PPOTrainer", "ValueCritic"]
+__all__ = ["PPOTrainer", "ValueCritic", "GRPOTrainer"]
 
 
 class ValueCritic(nn.Module):
@@ -137,4 +137,58 @@ class PPOTrainer:
             self.critic_opt.step()
             stats = {"policy_loss": float(policy_loss), "value_loss": float(value_loss),
                      "reward": exp["mean_reward"]}
+        return stats
+
+
+class GRPOTrainer(PPOTrainer):
+    """Group Relative Policy Optimization (reference:
+    applications/ColossalChat GRPO trainer; Shao et al., DeepSeekMath).
+
+    Critic-free: for each prompt, sample ``group_size`` continuations and
+    use the group-normalized reward as a per-sequence advantage (broadcast
+    over its tokens). Keeps PPO's clipped surrogate and the reference-KL
+    shaping; drops the value network entirely."""
+
+    def __init__(self, actor, reward_fn, actor_optimizer, booster=None,
+                 group_size: int = 4, clip_eps: float = 0.2, kl_coef: float = 0.02,
+                 ppo_epochs: int = 2):
+        booster = booster or Booster()
+        criterion = lambda out, batch: out
+        self.actor, self.actor_opt, *_ = booster.boost(actor, actor_optimizer, criterion)
+        self.booster = booster
+        self.ref = copy.deepcopy(actor).eval()
+        for p in self.ref.parameters():
+            p.requires_grad_(False)
+        self.reward_fn = reward_fn
+        self.group_size = group_size
+        self.clip_eps = clip_eps
+        self.kl_coef = kl_coef
+        self.ppo_epochs = ppo_epochs
+
+    def train_step(self, prompts: torch.Tensor, max_new_tokens: int = 8):
+        B, P = prompts.shape
+        G = self.group_size
+        rep = prompts.repeat_interleave(G, dim=0)  # [B*G, P]
+        with torch.no_grad():
+            seq = self._rollout(rep, max_new_tokens)
+            old_logp = _token_logprobs(self.actor(input_ids=seq)["logits"], seq)[:, P - 1 :]
+            ref_logp = _token_logprobs(self.ref(input_ids=seq)["logits"], seq)[:, P - 1 :]
+            reward = self.reward_fn(seq, P).float().view(B, G)
+            adv = (reward - reward.mean(dim=1, keepdim=True)) / (reward.std(dim=1, keepdim=True) + 1e-6)
+            adv = adv.view(B * G, 1)  # broadcast over the sequence's tokens
+
+        stats = {}
+        for _ in range(self.ppo_epochs):
+            logp = _token_logprobs(self.actor(input_ids=seq)["logits"], seq)[:, P - 1 :]
+            ratio = (logp - old_logp).exp()
+            s1 = ratio * adv
+            s2 = ratio.clamp(1 - self.clip_eps, 1 + self.clip_eps) * adv
+            kl = (old_logp - logp)  # k1 estimator vs the sampling policy
+            ref_kl = (logp - ref_logp)
+            loss = (-torch.min(s1, s2) + self.kl_coef * ref_kl).mean()
+            self.actor_opt.zero_grad()
+            self.booster.backward(loss, self.actor_opt)
+            self.actor_opt.step()
+            stats = {"policy_loss": float(loss), "reward": float(reward.mean()),
+                     "kl": float(kl.mean())}
         return stats

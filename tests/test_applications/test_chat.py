@@ -89,3 +89,29 @@ def test_ppo_trainer_shifts_policy_toward_reward():
     after = low_frac()
     assert after > max(before + 0.2, 0.8), (
         f"policy did not shift: {before:.2f} -> {after:.2f} (rewards {rewards})")
+
+
+def test_grpo_trainer_shifts_policy():
+    from applications.chat import GRPOTrainer
+
+    torch.manual_seed(1)
+    actor = LlamaForCausalLM(_tiny())
+
+    def reward_fn(seq, prompt_len):
+        gen = seq[:, prompt_len:]
+        return (gen < 32).float().mean(dim=1) * 2 - 1
+
+    trainer = GRPOTrainer(actor, reward_fn, torch.optim.AdamW(actor.parameters(), lr=5e-3),
+                          group_size=4, kl_coef=0.0, ppo_epochs=2)
+    prompts = torch.randint(0, 64, (8, 6))
+
+    def low_frac():
+        with torch.no_grad():
+            fr = [float((trainer._rollout(prompts, 8)[:, 6:] < 32).float().mean()) for _ in range(4)]
+        return sum(fr) / len(fr)
+
+    before = low_frac()
+    for _ in range(20):
+        trainer.train_step(prompts, max_new_tokens=8)
+    after = low_frac()
+    assert after > max(before + 0.2, 0.8), f"GRPO did not shift: {before:.2f} -> {after:.2f}"
