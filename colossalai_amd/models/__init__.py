@@ -4,10 +4,12 @@ from .opt import OPT_CONFIGS, OPTConfig, OPTForCausalLM
 from .t5 import T5_CONFIGS, T5Config, T5ForConditionalGeneration
 from .falcon import FALCON_CONFIGS, FalconConfig, FalconForCausalLM
 from .vit import VIT_CONFIGS, ViTConfig, ViTForImageClassification
+from .whisper import WHISPER_CONFIGS, WhisperConfig, WhisperForConditionalGeneration
 
 __all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_token",
            "OPTConfig", "OPTForCausalLM", "OPT_CONFIGS",
            "BertConfig", "BertForMaskedLM", "BertForSequenceClassification", "BERT_CONFIGS",
            "T5Config", "T5ForConditionalGeneration", "T5_CONFIGS",
            "ViTConfig", "ViTForImageClassification", "VIT_CONFIGS",
-           "FalconConfig", "FalconForCausalLM", "FALCON_CONFIGS"]
+           "FalconConfig", "FalconForCausalLM", "FALCON_CONFIGS",
+           "WhisperConfig", "WhisperForConditionalGeneration", "WHISPER_CONFIGS"]
