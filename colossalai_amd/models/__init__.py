@@ -2,6 +2,7 @@ from .llama import LLAMA_CONFIGS, LlamaConfig, LlamaForCausalLM, llama_flops_per
 from .bert import BERT_CONFIGS, BertConfig, BertForMaskedLM, BertForSequenceClassification
 from .opt import OPT_CONFIGS, OPTConfig, OPTForCausalLM
 from .t5 import T5_CONFIGS, T5Config, T5ForConditionalGeneration
+from .bloom import BLOOM_CONFIGS, BloomConfig, BloomForCausalLM
 from .falcon import FALCON_CONFIGS, FalconConfig, FalconForCausalLM
 from .vit import VIT_CONFIGS, ViTConfig, ViTForImageClassification
 from .whisper import WHISPER_CONFIGS, WhisperConfig, WhisperForConditionalGeneration
@@ -12,4 +13,5 @@ __all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_
            "T5Config", "T5ForConditionalGeneration", "T5_CONFIGS",
            "ViTConfig", "ViTForImageClassification", "VIT_CONFIGS",
            "FalconConfig", "FalconForCausalLM", "FALCON_CONFIGS",
-           "WhisperConfig", "WhisperForConditionalGeneration", "WHISPER_CONFIGS"]
+           "WhisperConfig", "WhisperForConditionalGeneration", "WHISPER_CONFIGS",
+           "BloomConfig", "BloomForCausalLM", "BLOOM_CONFIGS"]
