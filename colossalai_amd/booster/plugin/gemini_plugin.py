@@ -232,5 +232,10 @@ class GeminiPlugin(Plugin):
         return model, optimizer, criterion, dataloader, lr_scheduler
 
     def no_sync(self, model: nn.Module, optimizer: OptimizerWrapper = None) -> Iterator[None]:
-        assert isinstance(optimizer, LowLevelZeroOptimizer), "no_sync is not supported on the sharded-param path"
+        if self.shard_param_frac == 1.0:
+            from ...zero.gemini import GeminiDDP
+
+            assert isinstance(model, GeminiDDP)
+            return model.no_sync()
+        assert isinstance(optimizer, LowLevelZeroOptimizer)
         return optimizer.no_sync()

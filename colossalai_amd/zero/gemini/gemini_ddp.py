@@ -28,8 +28,9 @@ Mechanism (unit-granular, FSDP-shaped):
   into the param shards; the next forward's gathers publish the update.
 
 With 288 GB HBM3E this path is for 70B+ models / small dp groups; the
-flat-buffer ZeRO-2 engine remains the default below that. Gradient
-accumulation across steps (``no_sync``) is not supported on this path yet.
+flat-buffer ZeRO-2 engine remains the default below that. ``no_sync``
+accumulates grads in the (unsharded) chunk grad buffers and defers the
+reduce-scatter to the sync step.
 """
 
 from typing import Dict, List, Optional, Tuple
@@ -205,6 +206,7 @@ class GeminiDDP(ModelWrapper):
             u.register_forward_hook(self._make_fwd_release(self.unit_chunks[id(u)]))
             u.register_full_backward_pre_hook(self._make_bwd_gather(self.unit_chunks[id(u)], prv))
 
+        self.require_grad_sync = True
         self._hook_handles = []
         for p in module.parameters():
             if p.requires_grad:
@@ -305,10 +307,32 @@ class GeminiDDP(ModelWrapper):
             p.grad.add_(g)
         c.grads_done += 1
         if c.grads_done == len(c.params):
-            c.reduce_grads(self.group)
+            if self.require_grad_sync:
+                c.reduce_grads(self.group)
+            else:
+                # grad accumulation: keep grad_flat materialized so the next
+                # backward accumulates into the same views; params still
+                # release (re-gathered per unit as usual)
+                c.grads_done = 0
             c.release()
 
     # -------------------------------------------------------------------- api
+    def no_sync(self):
+        """Gradient accumulation: defer the reduce-scatter; grad buffers
+        stay materialized (unsharded) across the accumulation window."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def ctx():
+            old = self.require_grad_sync
+            self.require_grad_sync = False
+            try:
+                yield
+            finally:
+                self.require_grad_sync = old
+
+        return ctx()
+
     def forward(self, *args, **kwargs):
         if self.module.training and torch.is_grad_enabled():
             for c in self.persistent_chunks:

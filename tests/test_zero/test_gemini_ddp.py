@@ -93,3 +93,41 @@ def test_gemini_ddp_oracle():
 @rerun_if_address_is_in_use()
 def test_gemini_plugin_sharded():
     spawn(run_gemini_plugin, 2)
+
+
+def run_gemini_no_sync(rank, world_size, port):
+    """Grad accumulation: no_sync backward + sync backward + step must
+    match the oracle that sums both micro-grads."""
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(_cfg())
+    ref = copy.deepcopy(model).float()
+    gm = GeminiDDP(copy.deepcopy(model), precision="fp32", chunk_size_m=1)
+    opt = GeminiOptimizer(FusedAdam(gm.parameters(), lr=1e-2), gm)
+    ref_opt = FusedAdam(ref.parameters(), lr=1e-2)
+
+    torch.manual_seed(50)
+    micros = [[torch.randint(0, 128, (2, 16)) for _ in range(world_size)] for _ in range(2)]
+
+    with gm.no_sync():
+        opt.backward(gm(input_ids=micros[0][rank], labels=micros[0][rank])["loss"])
+    opt.backward(gm(input_ids=micros[1][rank], labels=micros[1][rank])["loss"])
+    opt.step()
+    opt.zero_grad()
+
+    ref_opt.zero_grad()
+    for mb in micros:
+        for x in mb:
+            (ref(input_ids=x, labels=x)["loss"] / world_size).backward()
+    ref_opt.step()
+
+    sd = gm.state_dict()
+    ref_sd = ref.state_dict()
+    for k, v in sd.items():
+        assert torch.allclose(v.float(), ref_sd[k].float(), atol=1e-5, rtol=1e-5), f"{k} diverged"
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_gemini_no_sync():
+    spawn(run_gemini_no_sync, 2)
