@@ -63,6 +63,10 @@ class MixtralSparseMoeBlock(nn.Module):
         self.expert_start = 0
         self.num_local_experts = E
 
+    def _route(self, probs: torch.Tensor):
+        topw, topi = torch.topk(probs, self.top_k, dim=-1)
+        return topw / topw.sum(-1, keepdim=True), topi
+
     def _experts_forward(self, x: torch.Tensor, expert_ids: torch.Tensor) -> torch.Tensor:
         """x [N, H] grouped so rows of the same LOCAL expert are contiguous;
         expert_ids [N] gives each row's local expert."""
@@ -86,8 +90,7 @@ class MixtralSparseMoeBlock(nn.Module):
         T = x.shape[0]
         logits = self.gate(x).float()
         probs = F.softmax(logits, dim=-1)
-        topw, topi = torch.topk(probs, self.top_k, dim=-1)
-        topw = topw / topw.sum(-1, keepdim=True)  # fp32 routing weights
+        topw, topi = self._route(probs)  # fp32 routing weights
 
         flat_expert = topi.reshape(-1)  # [T*k]
         order = torch.argsort(flat_expert, stable=True)

@@ -23,6 +23,7 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "T5ForConditionalGeneration": ("colossalai_amd.shardformer.policies.t5", "T5ForConditionalGenerationPolicy"),
     "ViTForImageClassification": ("colossalai_amd.shardformer.policies.vit", "ViTForImageClassificationPolicy"),
     "FalconForCausalLM": ("colossalai_amd.shardformer.policies.falcon", "FalconForCausalLMPolicy"),
+    "DeepseekForCausalLM": ("colossalai_amd.shardformer.policies.deepseek", "DeepseekForCausalLMPolicy"),
 }
 
 
