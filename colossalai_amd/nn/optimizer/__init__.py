@@ -11,3 +11,19 @@ from .lars import Lars
 
 __all__ = ["FusedAdam",
     "FusedSGD", "DistributedLamb", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
+
+
+def cast_to_distributed(optimizer):
+    """Swap a plain optimizer for its TP/ZeRO-aware variant when one exists
+    (reference: colossalai/nn/optimizer/__init__.py cast_to_distributed).
+    Returns the original optimizer unchanged otherwise; call
+    ``setup_distributed(...)`` on the result before stepping."""
+    mapping = {Lamb: DistributedLamb}
+    cls = mapping.get(type(optimizer))
+    if cls is None:
+        return optimizer
+    # rebuild from the same param groups + defaults
+    groups = [dict(g) for g in optimizer.param_groups]
+    kwargs = {k: v for k, v in optimizer.defaults.items() if k in ("lr", "betas", "eps", "weight_decay")}
+    return cls([{"params": g["params"], **{k: g.get(k, kwargs.get(k)) for k in kwargs}} for g in groups],
+               **kwargs)
