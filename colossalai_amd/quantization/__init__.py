@@ -1,4 +1,5 @@
 from .fp8 import all_gather_fp8, all_reduce_fp8, all_to_all_single_fp8, cast_from_fp8, cast_to_fp8, reduce_scatter_fp8
+from .fp8_linear import Fp8Linear, fp8_linear
 
 __all__ = [
     "cast_to_fp8",
@@ -7,4 +8,6 @@ __all__ = [
     "all_gather_fp8",
     "reduce_scatter_fp8",
     "all_to_all_single_fp8",
+    "fp8_linear",
+    "Fp8Linear",
 ]

@@ -110,3 +110,20 @@ def _run_plugins(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_fp8_plugin_flags():
     spawn(_run_plugins, 2)
+
+
+def test_fp8_linear_cpu_wiring():
+    """CPU path: plain linear fallback with working autograd."""
+    from colossalai_amd.quantization import Fp8Linear, fp8_linear
+
+    torch.manual_seed(0)
+    x = torch.randn(4, 8, requires_grad=True)
+    lin = torch.nn.Linear(8, 6)
+    out = fp8_linear(x, lin.weight, lin.bias)
+    ref = torch.nn.functional.linear(x, lin.weight, lin.bias)
+    torch.testing.assert_close(out, ref)
+    out.sum().backward()
+    assert x.grad is not None and lin.weight.grad is not None
+
+    f = Fp8Linear.from_linear(lin)
+    torch.testing.assert_close(f(x.detach()), ref.detach())

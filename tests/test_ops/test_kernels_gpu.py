@@ -390,3 +390,23 @@ def test_moe_combine():
     ref.backward(d.float())
     _bf16_close(y.grad, yr.grad, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(w.grad.cpu(), wr.grad.cpu(), rtol=2e-2, atol=2e-1)
+
+
+def test_fp8_linear_gpu():
+    """e4m3 _scaled_mm forward vs bf16 linear (per-tensor scales)."""
+    from colossalai_amd.quantization import fp8_linear
+
+    torch.manual_seed(15)
+    x = torch.randn(64, 256, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = fp8_linear(x, w)
+    ref = torch.nn.functional.linear(x.float(), w.float())
+    # e4m3 has ~2 decimal digits: compare loosely, elementwise fraction
+    _bf16_close(out, ref, rtol=8e-2, atol=8e-1, frac=2e-3)
+    d = torch.randn_like(out)
+    out.backward(d)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    torch.nn.functional.linear(xr, wr).backward(d.float())
+    _bf16_close(x.grad, xr.grad, rtol=3e-2, atol=3e-2, frac=1e-4)
+    _bf16_close(w.grad, wr.grad, rtol=3e-2, atol=3e-2, frac=1e-4)
