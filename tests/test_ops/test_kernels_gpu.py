@@ -401,8 +401,9 @@ def test_fp8_linear_gpu():
     w = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16, requires_grad=True)
     out = fp8_linear(x, w)
     ref = torch.nn.functional.linear(x.float(), w.float())
-    # e4m3 has ~2 decimal digits: compare loosely, elementwise fraction
-    _bf16_close(out, ref, rtol=8e-2, atol=8e-1, frac=2e-3)
+    # e4m3 carries ~3 mantissa bits: judge by relative RMS, not elementwise
+    rms = (out.float() - ref).pow(2).mean().sqrt() / ref.pow(2).mean().sqrt()
+    assert rms < 0.05, f"fp8 forward rel-RMS {rms:.4f}"
     d = torch.randn_like(out)
     out.backward(d)
     xr = x.detach().float().requires_grad_(True)
