@@ -262,10 +262,11 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     // waves entirely above the diagonal produce nothing (barriers stay uniform)
     const bool active = !(causal && k0 > qw + QW - 1) && (qw < S);
     if (active) {
-    // ---- S^T = K · Q^T : two 32-k blocks. C: col = q (ln), row = k (crow).
-    float p[2][16];
+    // ---- S^T = K · Q^T : KVB/32 32-k blocks. C: col = q (ln), row = k (crow).
+    constexpr int KB = KVB / 32;
+    float p[KB][16];
 #pragma unroll
-    for (int kb = 0; kb < 2; ++kb) {
+    for (int kb = 0; kb < KB; ++kb) {
       f32x16 acc;
 #pragma unroll
       for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
@@ -285,7 +286,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     const int q_abs = qw + ln;
     float rowmax = -INFINITY;
 #pragma unroll
-    for (int kb = 0; kb < 2; ++kb)
+    for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const int k_abs = k0 + kb * 32 + crow(j, half);
@@ -304,7 +305,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
         defer ? 1.0f : ((m_run == -INFINITY) ? ((m_new == -INFINITY) ? 1.0f : 0.0f) : __expf(m_run - m_new));
     float rowsum = 0.0f;
 #pragma unroll
-    for (int kb = 0; kb < 2; ++kb)
+    for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         p[kb][j] = __expf(p[kb][j] - msafe);
@@ -317,7 +318,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     // ---- stage P (bf16) into swizzled per-wave LDS: P[q=ln][k]
     // C-regs j=0..3 within a group are 4 consecutive k values → 8 B packed.
 #pragma unroll
-    for (int kb = 0; kb < 2; ++kb)
+    for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
       for (int jj = 0; jj < 4; ++jj) {
         const int kcol = kb * 32 + 8 * jj + 4 * half;
@@ -761,9 +762,10 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
     if (active) {
 
     // ---- S[q][kv] = Q · K^T : C col = kv (ln within kb), row = q (crow).
-    float p[2][16], dp[2][16];
+    constexpr int KB = KVB / 32;
+    float p[KB][16], dp[KB][16];
 #pragma unroll
-    for (int kb = 0; kb < 2; ++kb) {
+    for (int kb = 0; kb < KB; ++kb) {
       f32x16 acc, acc2;
 #pragma unroll
       for (int j = 0; j < 16; ++j) { acc[j] = 0.0f; acc2[j] = 0.0f; }
@@ -782,7 +784,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
 
     // ---- dS = scale * P ⊙ (dP - delta[q]) with P = exp(scale*s - lse[q])
 #pragma unroll
-    for (int kb = 0; kb < 2; ++kb)
+    for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const int r = crow(j, half);
