@@ -33,7 +33,7 @@ def parse_args():
     p.add_argument("--batch", type=int, default=36, help="per-DP-rank batch size (reference: bs/DP=36)")
     p.add_argument("--seq", type=int, default=4096)
     p.add_argument("--plugin", type=str, default="zero2",
-                   choices=["ddp", "zero2", "zero1", "gemini", "hybrid", "moe"])
+                   choices=["ddp", "zero2", "zero1", "gemini", "gemini3", "hybrid", "moe"])
     p.add_argument("--tp", type=int, default=1)
     p.add_argument("--pp", type=int, default=1)
     p.add_argument("--sp", type=int, default=1)
@@ -120,13 +120,17 @@ def main():
             backward = None  # handled inside execute_pipeline
         else:
             backward = lambda loss: booster.backward(loss, optimizer)
-    elif args.plugin == "gemini":
+    elif args.plugin in ("gemini", "gemini3"):
         from colossalai_amd import Booster
         from colossalai_amd.booster.plugin import GeminiPlugin
-        from colossalai_amd.nn import HybridAdam
+        from colossalai_amd.nn import FusedAdam, HybridAdam
 
-        optimizer = HybridAdam(model.parameters(), lr=1e-5, weight_decay=0.1)
-        booster = Booster(plugin=GeminiPlugin(precision="bf16"))
+        if args.plugin == "gemini3":  # native chunk-sharded params (ZeRO-3)
+            optimizer = FusedAdam(model.parameters(), lr=1e-5, weight_decay=0.1)
+            booster = Booster(plugin=GeminiPlugin(precision="bf16", shard_param_frac=1.0))
+        else:
+            optimizer = HybridAdam(model.parameters(), lr=1e-5, weight_decay=0.1)
+            booster = Booster(plugin=GeminiPlugin(precision="bf16"))
         model, optimizer, *_ = booster.boost(model, optimizer)
         backward = lambda loss: booster.backward(loss, optimizer)
     else:
