@@ -30,7 +30,24 @@ def _models():
                                                  attention_bias=True)),
         "gpt2": GPT2LMHeadModel(GPT2Config(vocab_size=128, n_positions=64, n_embd=64, n_layer=2, n_head=4)),
         "mixtral": MixtralForCausalLM(MIXTRAL_CONFIGS["mixtral-tiny"]),
+        "opt": _opt(),
+        "falcon": _falcon(),
     }
+
+
+def _opt():
+    from colossalai_amd.models.opt import OPTConfig, OPTForCausalLM
+
+    return OPTForCausalLM(OPTConfig(vocab_size=128, hidden_size=64, ffn_dim=128,
+                                    num_hidden_layers=2, num_attention_heads=4,
+                                    max_position_embeddings=64))
+
+
+def _falcon():
+    from colossalai_amd.models.falcon import FalconConfig, FalconForCausalLM
+
+    return FalconForCausalLM(FalconConfig(vocab_size=128, hidden_size=64, num_hidden_layers=2,
+                                          num_attention_heads=4, max_position_embeddings=64))
 
 
 def _plugins():
@@ -56,7 +73,7 @@ def run_matrix(rank, world_size, port):
     x = torch.randint(0, 128, (4, 16))
     for pname, make_plugin in _plugins().items():
         for mname, model in _models().items():
-            if pname == "hybrid_tp2" and mname == "mixtral":
+            if pname == "hybrid_tp2" and mname in ("mixtral", "falcon"):
                 continue  # tiny mixtral: 4 experts, head split covered by ep tests
             model = copy.deepcopy(model)
             booster = Booster(plugin=make_plugin())
