@@ -108,3 +108,62 @@ def test_gpt2_gpu_train_step():
     out["loss"].backward()
     for n, p in m.named_parameters():
         assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+
+@pytest.mark.gpu
+def test_new_families_gpu_fwd_bwd():
+    """Every round-1 model family steps on the HIP paths (bf16, head_dim 64):
+    finite loss + finite grads."""
+    import torch
+
+    from colossalai_amd.models.bert import BertConfig, BertForMaskedLM
+    from colossalai_amd.models.bloom import BloomConfig, BloomForCausalLM
+    from colossalai_amd.models.falcon import FalconConfig, FalconForCausalLM
+    from colossalai_amd.models.opt import OPTConfig, OPTForCausalLM
+    from colossalai_amd.models.t5 import T5Config, T5ForConditionalGeneration
+    from colossalai_amd.models.vit import ViTConfig, ViTForImageClassification
+    from colossalai_amd.models.whisper import WhisperConfig, WhisperForConditionalGeneration
+
+    torch.manual_seed(0)
+    x = torch.randint(0, 512, (2, 128), device="cuda")
+    y = torch.randint(0, 512, (2, 32), device="cuda")
+
+    cases = [
+        (OPTForCausalLM(OPTConfig(vocab_size=512, hidden_size=256, ffn_dim=512,
+                                  num_hidden_layers=2, num_attention_heads=4,
+                                  max_position_embeddings=256)),
+         lambda m: m(x, labels=x)),
+        (BertForMaskedLM(BertConfig(vocab_size=512, hidden_size=256, num_hidden_layers=2,
+                                    num_attention_heads=4, intermediate_size=512,
+                                    max_position_embeddings=256)),
+         lambda m: m(x, labels=x)),
+        (FalconForCausalLM(FalconConfig(vocab_size=512, hidden_size=256, num_hidden_layers=2,
+                                        num_attention_heads=4, max_position_embeddings=256)),
+         lambda m: m(x, labels=x)),
+        (BloomForCausalLM(BloomConfig(vocab_size=512, hidden_size=256, num_hidden_layers=2,
+                                      num_attention_heads=4)),
+         lambda m: m(x, labels=x)),
+        (T5ForConditionalGeneration(T5Config(vocab_size=512, d_model=256, d_kv=64, d_ff=512,
+                                             num_layers=2, num_decoder_layers=2, num_heads=4)),
+         lambda m: m(x, labels=y)),
+        (ViTForImageClassification(ViTConfig(image_size=64, patch_size=8, hidden_size=256,
+                                             num_hidden_layers=2, num_attention_heads=4,
+                                             intermediate_size=512, num_labels=7)),
+         lambda m: m(torch.randn(2, 3, 64, 64, device="cuda", dtype=torch.bfloat16),
+                     labels=torch.randint(0, 7, (2,), device="cuda"))),
+        (WhisperForConditionalGeneration(WhisperConfig(vocab_size=512, num_mel_bins=32,
+                                                       d_model=256, encoder_layers=2,
+                                                       decoder_layers=2, num_heads=4, d_ff=512,
+                                                       max_source_positions=64,
+                                                       max_target_positions=64,
+                                                       decoder_start_token_id=1, pad_token_id=0)),
+         lambda m: m(torch.randn(2, 32, 128, device="cuda", dtype=torch.bfloat16), labels=y)),
+    ]
+    for model, run in cases:
+        name = type(model).__name__
+        model = model.to("cuda").bfloat16()
+        out = run(model)
+        assert out["loss"] is not None and torch.isfinite(out["loss"]), name
+        out["loss"].backward()
+        for n, p in model.named_parameters():
+            assert p.grad is None or torch.isfinite(p.grad).all(), f"{name}.{n}"
