@@ -28,6 +28,10 @@ def hf_to_native_llama(hf_sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor
         elif ".mlp.gate_proj.weight" in k or ".mlp.up_proj.weight" in k:
             base = k.rsplit(".mlp.", 1)[0]
             packs.setdefault(base + ".gateup", {})[k.split(".")[-2]] = v
+        elif k.endswith(".self_attn.q_norm.weight"):
+            out[k.replace(".self_attn.q_norm.weight", ".self_attn.q_norm_weight")] = v
+        elif k.endswith(".self_attn.k_norm.weight"):
+            out[k.replace(".self_attn.k_norm.weight", ".self_attn.k_norm_weight")] = v
         elif k.endswith("input_layernorm.weight"):
             out[k.replace("input_layernorm.weight", "input_layernorm_weight")] = v
         elif k.endswith("post_attention_layernorm.weight"):

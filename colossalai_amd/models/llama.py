@@ -45,6 +45,8 @@ class LlamaConfig:
     initializer_range: float = 0.02
     gradient_checkpointing: bool = False
     attention_bias: bool = False  # Qwen2-style qkv bias
+    qk_norm: bool = False  # Qwen3-style per-head RMSNorm on q/k before RoPE
+    head_dim_override: Optional[int] = None  # Qwen3 decouples head_dim from hidden/heads
 
     def __post_init__(self):
         if self.num_key_value_heads is None:
@@ -52,6 +54,8 @@ class LlamaConfig:
 
     @property
     def head_dim(self) -> int:
+        if self.head_dim_override is not None:
+            return self.head_dim_override
         return self.hidden_size // self.num_attention_heads
 
 
@@ -76,6 +80,11 @@ LLAMA_CONFIGS = {
     "qwen2-7b": LlamaConfig(vocab_size=152064, hidden_size=3584, intermediate_size=18944,
                             num_hidden_layers=28, num_attention_heads=28, num_key_value_heads=4,
                             max_position_embeddings=4096, rope_theta=1e6, attention_bias=True),
+    # Qwen3: qk-norm + decoupled head_dim
+    "qwen3-8b": LlamaConfig(vocab_size=151936, hidden_size=4096, intermediate_size=12288,
+                            num_hidden_layers=36, num_attention_heads=32, num_key_value_heads=8,
+                            head_dim_override=128, max_position_embeddings=8192, rope_theta=1e6,
+                            rms_norm_eps=1e-6, qk_norm=True),
 }
 
 
@@ -112,6 +121,11 @@ class LlamaAttention(nn.Module):
         self.qkv_proj = nn.Linear(cfg.hidden_size, (Hq + 2 * Hkv) * D, bias=cfg.attention_bias)
         self.o_proj = nn.Linear(Hq * D, cfg.hidden_size, bias=False)
         self.scale = 1.0 / math.sqrt(D)
+        self.qk_norm = cfg.qk_norm
+        if cfg.qk_norm:
+            self.q_norm_weight = nn.Parameter(torch.ones(D))
+            self.k_norm_weight = nn.Parameter(torch.ones(D))
+            self.norm_eps = cfg.rms_norm_eps
 
     def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
         B, S, _ = hidden.shape
@@ -162,6 +176,19 @@ class LlamaAttention(nn.Module):
             attn = flash_attention(q, k, v.contiguous(), causal=True, scale=self.scale)
             attn = all_to_all_comm(attn, sp_group, scatter_dim=1, gather_dim=2)  # back to [B, S, Hq, D]
             return self.o_proj(attn.reshape(B, S, -1))
+        if self.qk_norm:
+            # Qwen3: per-head RMSNorm on q/k before RoPE — unpack, norm,
+            # rope, flash (the packed fused path skips the norm)
+            from ..ops import apply_rope, flash_attention, rms_norm
+
+            Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+            q = rms_norm(qkv[:, :, : Hq * D].reshape(B, S, Hq, D), self.q_norm_weight, self.norm_eps)
+            k = rms_norm(qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D),
+                         self.k_norm_weight, self.norm_eps)
+            v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
+            q, k = apply_rope(q.contiguous(), k.contiguous(), rope_table)
+            attn = flash_attention(q, k, v.contiguous(), causal=True, scale=self.scale)
+            return self.o_proj(attn.reshape(B, S, Hq * D))
         attn = fused_rope_attention(
             qkv, rope_table, self.num_heads, self.num_kv_heads, self.head_dim,
             causal=True, scale=self.scale,
