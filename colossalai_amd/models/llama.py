@@ -182,8 +182,9 @@ class LlamaAttention(nn.Module):
             from ..ops import apply_rope, flash_attention, rms_norm
 
             Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
-            q = rms_norm(qkv[:, :, : Hq * D].reshape(B, S, Hq, D), self.q_norm_weight, self.norm_eps)
-            k = rms_norm(qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D),
+            q = rms_norm(qkv[:, :, : Hq * D].reshape(B, S, Hq, D).contiguous(),
+                         self.q_norm_weight, self.norm_eps)
+            k = rms_norm(qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D).contiguous(),
                          self.k_norm_weight, self.norm_eps)
             v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
             q, k = apply_rope(q.contiguous(), k.contiguous(), rope_table)

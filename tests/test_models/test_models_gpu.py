@@ -128,7 +128,14 @@ def test_new_families_gpu_fwd_bwd():
     x = torch.randint(0, 512, (2, 128), device="cuda")
     y = torch.randint(0, 512, (2, 32), device="cuda")
 
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
     cases = [
+        (LlamaForCausalLM(LlamaConfig(vocab_size=512, hidden_size=256, intermediate_size=512,
+                                      num_hidden_layers=2, num_attention_heads=4,
+                                      num_key_value_heads=2, head_dim_override=64,
+                                      max_position_embeddings=256, qk_norm=True)),
+         lambda m: m(x, labels=x)),  # Qwen3-style qk-norm path
         (OPTForCausalLM(OPTConfig(vocab_size=512, hidden_size=256, ffn_dim=512,
                                   num_hidden_layers=2, num_attention_heads=4,
                                   max_position_embeddings=256)),
