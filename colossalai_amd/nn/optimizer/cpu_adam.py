@@ -12,7 +12,23 @@ from torch.optim import Optimizer
 from ...ops import has_kernels, kernels
 from .fused_adam import DEFAULT_CHUNK, fused_adam_step_cpu
 
+try:  # native omp-simd CPU kernel (csrc_cpu/cpu_adam.cpp, ~16x the torch path)
+    from ... import _C_cpu
+except ImportError:  # pragma: no cover - source checkout without build
+    _C_cpu = None
+
 __all__ = ["CPUAdam"]
+
+
+def native_cpu_adam_available() -> bool:
+    return _C_cpu is not None
+
+
+def native_cpu_adam_step(p, g, m, v, out, lr, beta1, beta2, eps, step, adamw, bias_corr,
+                         weight_decay, div_scale):
+    _C_cpu.cpu_adam_step(p.reshape(-1), g.reshape(-1).float(), m.reshape(-1), v.reshape(-1),
+                         out, lr, beta1, beta2, eps, step, adamw, bias_corr, weight_decay,
+                         div_scale)
 
 
 class CPUAdam(Optimizer):
@@ -52,6 +68,13 @@ class CPUAdam(Optimizer):
                         [p.grad], [p], [state["exp_avg"]], [state["exp_avg_sq"]], [],
                         group["lr"], beta1, beta2, group["eps"], group["step"], self.adamw_mode,
                         group["bias_correction"], group["weight_decay"], div_scale, DEFAULT_CHUNK,
+                    )
+                elif _C_cpu is not None and p.dtype == torch.float32 and p.is_contiguous():
+                    empty = torch.empty(0, dtype=torch.bfloat16)
+                    native_cpu_adam_step(
+                        p.data, p.grad, state["exp_avg"], state["exp_avg_sq"], empty,
+                        group["lr"], beta1, beta2, group["eps"], group["step"], self.adamw_mode,
+                        group["bias_correction"], group["weight_decay"], div_scale,
                     )
                 else:
                     fused_adam_step_cpu(

@@ -435,15 +435,24 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
                     o.copy_(p.to(o.dtype))
 
     def _cpu_step(self, group, b, gshard, pshard, div_scale):
+        from ...nn.optimizer.cpu_adam import native_cpu_adam_available, native_cpu_adam_step
         from ...nn.optimizer.fused_adam import fused_adam_step_cpu
 
         beta1, beta2 = group.get("betas", (0.9, 0.999))
         g_cpu = gshard.detach().to("cpu", dtype=torch.float32)
-        fused_adam_step_cpu(
-            b.master, g_cpu, b.exp_avg, b.exp_avg_sq, group["lr"], beta1, beta2,
-            group.get("eps", 1e-8), group.get("weight_decay", 0.0), group["step"],
-            getattr(self.optim, "adamw_mode", True), group.get("bias_correction", True), div_scale,
-        )
+        if native_cpu_adam_available():
+            empty = torch.empty(0, dtype=torch.bfloat16)
+            native_cpu_adam_step(
+                b.master, g_cpu, b.exp_avg, b.exp_avg_sq, empty, group["lr"], beta1, beta2,
+                group.get("eps", 1e-8), group["step"], getattr(self.optim, "adamw_mode", True),
+                group.get("bias_correction", True), group.get("weight_decay", 0.0), div_scale,
+            )
+        else:
+            fused_adam_step_cpu(
+                b.master, g_cpu, b.exp_avg, b.exp_avg_sq, group["lr"], beta1, beta2,
+                group.get("eps", 1e-8), group.get("weight_decay", 0.0), group["step"],
+                getattr(self.optim, "adamw_mode", True), group.get("bias_correction", True), div_scale,
+            )
         pshard.copy_(b.master.to(pshard.device, dtype=pshard.dtype, non_blocking=True))
 
     # ------------------------------------------------------------ checkpoint

@@ -109,3 +109,24 @@ def test_fused_sgd_matches_torch():
         o1.step()
         o2.step()
         torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-6), step
+
+
+def test_native_cpu_adam_matches_reference():
+    from colossalai_amd.nn.optimizer.cpu_adam import native_cpu_adam_available, native_cpu_adam_step
+    from colossalai_amd.nn.optimizer.fused_adam import fused_adam_step_cpu
+
+    if not native_cpu_adam_available():
+        import pytest
+
+        pytest.skip("native CPU adam extension not built")
+    torch.manual_seed(0)
+    n = 10007
+    p = torch.randn(n); g = torch.randn(n); m = torch.rand(n) * 0.1; v = torch.rand(n) * 0.01
+    p2, g2, m2, v2 = p.clone(), g.clone(), m.clone(), v.clone()
+    out = torch.zeros(n, dtype=torch.bfloat16)
+    native_cpu_adam_step(p, g, m, v, out, 1e-2, 0.9, 0.95, 1e-8, 7, True, True, 0.1, 2.0)
+    fused_adam_step_cpu(p2, g2, m2, v2, 1e-2, 0.9, 0.95, 1e-8, 0.1, 7, True, True, 2.0)
+    torch.testing.assert_close(p, p2, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(m, m2, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(v, v2, rtol=1e-5, atol=1e-6)
+    assert torch.equal(out, p.bfloat16())
