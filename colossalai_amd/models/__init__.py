@@ -4,6 +4,7 @@ from .opt import OPT_CONFIGS, OPTConfig, OPTForCausalLM
 from .t5 import T5_CONFIGS, T5Config, T5ForConditionalGeneration
 from .bloom import BLOOM_CONFIGS, BloomConfig, BloomForCausalLM
 from .deepseek import DEEPSEEK_CONFIGS, DeepseekConfig, DeepseekForCausalLM
+from .gptj import GPTJ_CONFIGS, GPTJConfig, GPTJForCausalLM
 from .falcon import FALCON_CONFIGS, FalconConfig, FalconForCausalLM
 from .vit import VIT_CONFIGS, ViTConfig, ViTForImageClassification
 from .whisper import WHISPER_CONFIGS, WhisperConfig, WhisperForConditionalGeneration
@@ -16,4 +17,5 @@ __all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_
            "FalconConfig", "FalconForCausalLM", "FALCON_CONFIGS",
            "WhisperConfig", "WhisperForConditionalGeneration", "WHISPER_CONFIGS",
            "BloomConfig", "BloomForCausalLM", "BLOOM_CONFIGS",
-           "DeepseekConfig", "DeepseekForCausalLM", "DEEPSEEK_CONFIGS"]
+           "DeepseekConfig", "DeepseekForCausalLM", "DEEPSEEK_CONFIGS",
+           "GPTJConfig", "GPTJForCausalLM", "GPTJ_CONFIGS"]
