@@ -3,7 +3,7 @@ from .engine import LLMEngine
 from .kv_cache import KVCacheManager
 from .paged_engine import ContinuousBatchEngine
 from .request_manager import Request, RequestManager, RequestStatus
-from .spec_decode import SpeculativeEngine
+from .spec_decode import BatchedSpeculativeEngine, SpeculativeEngine
 
 __all__ = ["InferenceConfig", "GenerationConfig", "LLMEngine", "ContinuousBatchEngine",
-           "KVCacheManager", "SpeculativeEngine", "RequestManager", "Request", "RequestStatus"]
+           "KVCacheManager", "SpeculativeEngine", "BatchedSpeculativeEngine", "RequestManager", "Request", "RequestStatus"]

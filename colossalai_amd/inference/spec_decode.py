@@ -21,7 +21,7 @@ import torch
 from ..models.llama import LlamaForCausalLM
 from .config import GenerationConfig, InferenceConfig
 
-__all__ = ["SpeculativeEngine"]
+__all__ = ["SpeculativeEngine", "BatchedSpeculativeEngine"]
 
 
 def _chunk_attend(attn, hidden, table, kcache, vcache, prefix_len: int, positions):
@@ -183,6 +183,193 @@ class SpeculativeEngine:
             if self.config.eos_token_id is not None and correction == self.config.eos_token_id:
                 break
         return seq[: len(prompt) + gen.max_new_tokens]
+
+    @property
+    def acceptance_rate(self) -> float:
+        return self.accepted / max(self.proposed, 1)
+
+
+class _BatchedCachedModel:
+    """Batch-of-sequences cache wrapper with per-sequence lengths.
+
+    Single-token chunks run the batched decode path
+    (``forward_with_cache``, native decode kernel on GPU); multi-token
+    chunks (prefill / verification) run causal flash over the chunk and
+    LSE-merge a per-sequence non-causal block against the cache prefix."""
+
+    def __init__(self, model: LlamaForCausalLM, batch: int, max_seq_len: int):
+        self.model = model.eval()
+        self.device = next(model.parameters()).device
+        self.dtype = next(model.parameters()).dtype
+        cfg = model.config
+        self.caches = [
+            (torch.zeros(batch, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+                         device=self.device, dtype=self.dtype),
+             torch.zeros(batch, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+                         device=self.device, dtype=self.dtype))
+            for _ in range(cfg.num_hidden_layers)
+        ]
+        self.lens = torch.zeros(batch, dtype=torch.int32, device=self.device)
+
+    @torch.inference_mode()
+    def forward_chunk(self, ids: torch.Tensor, valid: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """ids [B, n] appended at per-seq self.lens; valid [B] caps how many
+        of the n tokens are real (prefill right-padding). -> logits
+        [B, n, V]. Advances lens by ``valid`` (or n)."""
+        from ..ops import fused_add_rms_norm, rms_norm
+        from ..shardformer.layer.ring_attn import _block_fwd, _merge
+
+        B, n = ids.shape
+        m = self.model.model
+        table = m.rope_table(self.device)
+        n_valid = valid if valid is not None else torch.full((B,), n, dtype=torch.int32,
+                                                            device=self.device)
+        if n == 1:
+            new_lens = (self.lens + 1).int()
+            positions = (new_lens - 1).int()
+            residual = m.embed_tokens(ids)
+            hidden = rms_norm(residual, m.layers[0].input_layernorm_weight, m.eps)
+            nl = len(m.layers)
+            for i, layer in enumerate(m.layers):
+                kc, vc = self.caches[i]
+                attn = layer.self_attn.forward_with_cache(hidden, table, kc, vc, positions,
+                                                          new_lens, prefill=False)
+                hidden, residual = fused_add_rms_norm(attn, residual,
+                                                      layer.post_attention_layernorm_weight, m.eps)
+                mlp_out = layer.mlp(hidden)
+                next_w = m.layers[i + 1].input_layernorm_weight if i + 1 < nl else m.norm_weight
+                hidden, residual = fused_add_rms_norm(mlp_out, residual, next_w, m.eps)
+            self.lens = new_lens
+            return self.model.lm_head(hidden).float()
+
+        # multi-token chunk with per-seq prefix lengths
+        positions = (self.lens[:, None] + torch.arange(n, device=self.device)[None]).reshape(-1).int()
+        residual = m.embed_tokens(ids)
+        hidden = rms_norm(residual, m.layers[0].input_layernorm_weight, m.eps)
+        nl = len(m.layers)
+        for i, layer in enumerate(m.layers):
+            attn_mod = layer.self_attn
+            Hq, Hkv, D = attn_mod.num_heads, attn_mod.num_kv_heads, attn_mod.head_dim
+            qkv = attn_mod.qkv_proj(hidden)
+            q = qkv[:, :, : Hq * D].view(B, n, Hq, D)
+            k = qkv[:, :, Hq * D : (Hq + Hkv) * D].view(B, n, Hkv, D)
+            v = qkv[:, :, (Hq + Hkv) * D :].view(B, n, Hkv, D)
+            from ..ops import has_kernels
+
+            if hidden.is_cuda and has_kernels():
+                from ..ops import kernels
+
+                kernels().rope_inplace(q, k, table, positions, False)
+            else:
+                from ..ops.rope import apply_rope_ref
+
+                q, k = apply_rope_ref(q, k, table, positions.long(), n, False)
+            kc, vc = self.caches[i]
+            for b in range(B):
+                lo = int(self.lens[b])
+                nb = int(n_valid[b])
+                kc[b, lo : lo + nb] = k[b, :nb]
+                vc[b, lo : lo + nb] = v[b, :nb]
+            out, lse = _block_fwd(q.contiguous(), k.contiguous(), v.contiguous(), True,
+                                  attn_mod.scale)
+            lse = lse.float()
+            merged = []
+            for b in range(B):
+                lo = int(self.lens[b])
+                if lo == 0:
+                    merged.append(out[b : b + 1])
+                    continue
+                o_pre, l_pre = _block_fwd(q[b : b + 1].contiguous(), kc[b : b + 1, :lo].contiguous(),
+                                          vc[b : b + 1, :lo].contiguous(), False, attn_mod.scale)
+                o_m, _ = _merge(out[b : b + 1], lse[b : b + 1], o_pre, l_pre.float())
+                merged.append(o_m)
+            attn = torch.cat(merged, dim=0).reshape(B, n, Hq * D)
+            attn = attn_mod.o_proj(attn)
+            hidden, residual = fused_add_rms_norm(attn, residual,
+                                                  layer.post_attention_layernorm_weight, m.eps)
+            mlp_out = layer.mlp(hidden)
+            next_w = m.layers[i + 1].input_layernorm_weight if i + 1 < nl else m.norm_weight
+            hidden, residual = fused_add_rms_norm(mlp_out, residual, next_w, m.eps)
+        self.lens = (self.lens + n_valid).int()
+        return self.model.lm_head(hidden).float()
+
+    def rollback(self, new_lens: torch.Tensor):
+        self.lens = new_lens.int().to(self.device)
+
+
+class BatchedSpeculativeEngine:
+    """Greedy speculative decoding over a batch of sequences: the draft
+    proposes gamma tokens for every sequence; one batched chunk forward
+    verifies all of them; acceptance/rollback are per-sequence length
+    bookkeeping (cache rows overwrite by position)."""
+
+    def __init__(self, target: LlamaForCausalLM, draft: LlamaForCausalLM,
+                 config: Optional[InferenceConfig] = None, gamma: int = 4):
+        self.config = config or InferenceConfig()
+        self.gamma = gamma
+        self._target_model = target
+        self._draft_model = draft
+        self.accepted = 0
+        self.proposed = 0
+
+    @torch.inference_mode()
+    def generate(self, prompts, generation_config: Optional[GenerationConfig] = None):
+        gen = generation_config or GenerationConfig()
+        B = len(prompts)
+        dev = next(self._target_model.parameters()).device
+        target = _BatchedCachedModel(self._target_model, B, self.config.max_seq_len)
+        draft = _BatchedCachedModel(self._draft_model, B, self.config.max_seq_len)
+
+        plens = torch.tensor([len(p) for p in prompts], dtype=torch.int32, device=dev)
+        S = int(plens.max())
+        padded = torch.full((B, S), self.config.pad_token_id, dtype=torch.long, device=dev)
+        for b, p in enumerate(prompts):
+            padded[b, : len(p)] = torch.tensor(p, device=dev)
+        t_logits = target.forward_chunk(padded, valid=plens)
+        draft.forward_chunk(padded, valid=plens)
+        gather = (plens.long() - 1).view(B, 1, 1).expand(B, 1, t_logits.shape[-1])
+        next_tok = t_logits.gather(1, gather)[:, 0].argmax(-1)  # [B]
+
+        seqs = [list(p) + [int(next_tok[b])] for b, p in enumerate(prompts)]
+        done = [False] * B
+        G = self.gamma
+        while not all(done):
+            if max(len(s) for s in seqs) + G + 2 > self.config.max_seq_len:
+                break
+            # ---- draft proposes G tokens for every sequence (batched)
+            proposals = torch.zeros(B, G, dtype=torch.long, device=dev)
+            tok = next_tok.view(B, 1)
+            for j in range(G):
+                d_logits = draft.forward_chunk(tok)
+                tok = d_logits[:, -1].argmax(-1, keepdim=True)
+                proposals[:, j] = tok[:, 0]
+            draft.forward_chunk(tok)  # feed the last proposal: keeps draft cache uniform
+            # ---- verify [next_tok | proposals] in one batched chunk
+            chunk = torch.cat([next_tok.view(B, 1), proposals], dim=1)
+            t_logits = target.forward_chunk(chunk)
+            want = t_logits.argmax(-1)  # [B, G+1]: target's next token after each prefix
+            n_acc = torch.zeros(B, dtype=torch.long)
+            for b in range(B):
+                j = 0
+                while j < G and int(proposals[b, j]) == int(want[b, j]):
+                    j += 1
+                n_acc[b] = j
+            corrections = want[torch.arange(B), n_acc]  # first mismatch (or bonus)
+            for b in range(B):
+                if done[b]:
+                    continue
+                self.proposed += G
+                self.accepted += int(n_acc[b])
+                budget = gen.max_new_tokens - (len(seqs[b]) - len(prompts[b]))
+                new = [int(x) for x in proposals[b, : int(n_acc[b])]] + [int(corrections[b])]
+                seqs[b].extend(new[:budget])
+                if len(seqs[b]) - len(prompts[b]) >= gen.max_new_tokens:
+                    done[b] = True
+            next_tok = torch.tensor([s[-1] for s in seqs], device=dev)
+            new_lens = torch.tensor([len(s) - 1 for s in seqs], dtype=torch.int32, device=dev)
+            target.rollback(new_lens)
+            draft.rollback(new_lens)
+        return [s[: len(p) + gen.max_new_tokens] for s, p in zip(seqs, prompts)]
 
     @property
     def acceptance_rate(self) -> float:

@@ -78,3 +78,35 @@ def test_spec_decode_sampled_runs_with_real_draft():
     out = eng.generate([5, 17, 42], GenerationConfig(max_new_tokens=16, do_sample=True, temperature=1.0))
     assert len(out) == 19
     assert 0.0 <= eng.acceptance_rate <= 1.0
+
+
+def test_batched_spec_decode_matches_per_sequence_greedy():
+    from colossalai_amd.inference import BatchedSpeculativeEngine
+
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(_cfg(3, 64)).eval()
+    draft = LlamaForCausalLM(_cfg(1, 32)).eval()
+    eng = BatchedSpeculativeEngine(target, draft,
+                                   InferenceConfig(max_batch_size=4, max_input_len=32,
+                                                   max_output_len=32), gamma=3)
+    prompts = [[5, 17, 42, 7], [99], [1, 2, 3], [88, 6]]
+    outs = eng.generate(prompts, GenerationConfig(max_new_tokens=12))
+    for p, o in zip(prompts, outs):
+        ref = _oracle(target, p, 12)
+        assert o == ref, f"batched spec {o} vs greedy {ref}"
+    assert eng.proposed > 0 and 0.0 <= eng.acceptance_rate <= 1.0
+
+
+def test_batched_spec_decode_self_draft():
+    from colossalai_amd.inference import BatchedSpeculativeEngine
+
+    torch.manual_seed(0)
+    target = LlamaForCausalLM(_cfg(2, 64)).eval()
+    eng = BatchedSpeculativeEngine(target, target,
+                                   InferenceConfig(max_batch_size=2, max_input_len=32,
+                                                   max_output_len=16), gamma=4)
+    prompts = [[5, 17, 42], [9, 8]]
+    outs = eng.generate(prompts, GenerationConfig(max_new_tokens=10))
+    for p, o in zip(prompts, outs):
+        assert o == _oracle(target, p, 10)
+    assert eng.acceptance_rate == 1.0
