@@ -41,6 +41,43 @@ def create_app(engine: ContinuousBatchEngine, tokenizer=None):
         return {"status": "ok", "running": len(engine.rm.running), "waiting": len(engine.rm.waiting),
                 "free_kv_blocks": engine.kv.free_blocks}
 
+    class CompletionRequest(BaseModel):
+        model: str = "colossalai_amd"
+        prompt: Optional[str] = None
+        prompt_ids: Optional[List[int]] = None
+        max_tokens: int = 64
+        temperature: float = 1.0
+        top_p: float = 1.0
+        n: int = 1
+
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        """OpenAI-compatible completions endpoint (token ids accepted via
+        the prompt_ids extension when no tokenizer is attached)."""
+        if req.prompt_ids is None:
+            if req.prompt is None or tokenizer is None:
+                raise HTTPException(400, "provide prompt_ids, or prompt with a tokenizer attached")
+            ids = tokenizer(req.prompt)["input_ids"]
+        else:
+            ids = req.prompt_ids
+        gen = GenerationConfig(max_new_tokens=req.max_tokens, do_sample=req.temperature > 0,
+                               temperature=max(req.temperature, 1e-5), top_p=req.top_p)
+        choices = []
+        for i in range(req.n):
+            out = engine.generate([ids], gen)[0]
+            new = out[len(ids):]
+            choices.append({
+                "index": i,
+                "text": tokenizer.decode(new) if tokenizer is not None else None,
+                "token_ids": new,
+                "finish_reason": "length" if len(new) >= req.max_tokens else "stop",
+            })
+        return {"id": f"cmpl-{id(choices) & 0xFFFFFF:x}", "object": "text_completion",
+                "model": req.model, "choices": choices,
+                "usage": {"prompt_tokens": len(ids),
+                          "completion_tokens": sum(len(c["token_ids"]) for c in choices),
+                          "total_tokens": len(ids) + sum(len(c["token_ids"]) for c in choices)}}
+
     @app.post("/generate", response_model=GenerateResponse)
     def generate(req: GenerateRequest):
         if req.prompt_ids is None:

@@ -31,3 +31,23 @@ def test_generate_endpoint():
 
     r = client.post("/generate", json={"max_new_tokens": 4})
     assert r.status_code == 400  # no ids and no tokenizer
+
+
+def test_openai_completions_endpoint():
+    from starlette.testclient import TestClient
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=128)
+    model = LlamaForCausalLM(cfg).eval()
+    engine = ContinuousBatchEngine(model, InferenceConfig(max_batch_size=2, max_input_len=32,
+                                                          max_output_len=16), block_size=4)
+    client = TestClient(create_app(engine))
+    r = client.post("/v1/completions", json={"prompt_ids": [5, 17, 42], "max_tokens": 6,
+                                             "temperature": 0.0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "text_completion"
+    assert len(body["choices"]) == 1
+    assert len(body["choices"][0]["token_ids"]) == 6
+    assert body["usage"]["total_tokens"] == 9
