@@ -3,6 +3,7 @@ from .bert import BERT_CONFIGS, BertConfig, BertForMaskedLM, BertForSequenceClas
 from .opt import OPT_CONFIGS, OPTConfig, OPTForCausalLM
 from .t5 import T5_CONFIGS, T5Config, T5ForConditionalGeneration
 from .bloom import BLOOM_CONFIGS, BloomConfig, BloomForCausalLM
+from .cohere import COHERE_CONFIGS, CohereConfig, CohereForCausalLM
 from .deepseek import DEEPSEEK_CONFIGS, DeepseekConfig, DeepseekForCausalLM
 from .gptj import GPTJ_CONFIGS, GPTJConfig, GPTJForCausalLM
 from .falcon import FALCON_CONFIGS, FalconConfig, FalconForCausalLM
@@ -18,4 +19,5 @@ __all__ = ["LlamaConfig", "LlamaForCausalLM", "LLAMA_CONFIGS", "llama_flops_per_
            "WhisperConfig", "WhisperForConditionalGeneration", "WHISPER_CONFIGS",
            "BloomConfig", "BloomForCausalLM", "BLOOM_CONFIGS",
            "DeepseekConfig", "DeepseekForCausalLM", "DEEPSEEK_CONFIGS",
-           "GPTJConfig", "GPTJForCausalLM", "GPTJ_CONFIGS"]
+           "GPTJConfig", "GPTJForCausalLM", "GPTJ_CONFIGS",
+           "CohereConfig", "CohereForCausalLM", "COHERE_CONFIGS"]
