@@ -162,12 +162,16 @@ class GeminiPlugin(Plugin):
             )
         if offload_param_frac > 0.0:
             raise NotImplementedError("offload_param_frac>0 is deferred; use offload_optim_frac")
-        if shard_param_frac == 1.0 and precision == "fp16":
-            raise NotImplementedError("shard_param_frac=1.0 supports bf16/fp32 (no loss scaler on this path)")
+        self._scaler_kwargs = dict(initial_scale=initial_scale, min_scale=min_scale,
+                                   growth_factor=growth_factor, backoff_factor=backoff_factor,
+                                   growth_interval=growth_interval, hysteresis=hysteresis,
+                                   max_scale=max_scale)
         self.shard_param_frac = shard_param_frac
         self.max_norm = max_norm
         self.precision = precision
+        self.placement_policy = placement_policy
         self.offload_optim_frac = offload_optim_frac
+        self._memory_ratio = kwargs.get("memory_ratio", 0.9)  # HBM budget for auto placement
         # chunk size: reuse the ZeRO bucket machinery; chunks sized in MiB-elements
         self.chunk_size_m = max(int(min_chunk_size_m), 1)
         self.zero_kwargs = dict(
@@ -209,6 +213,25 @@ class GeminiPlugin(Plugin):
     def get_checkpoint_io(self) -> CheckpointIO:
         return GeminiCheckpointIO() if self.shard_param_frac == 1.0 else LowLevelZeroCheckpointIO()
 
+    @staticmethod
+    def auto_offload_frac(param_numel: int, capacity_bytes: int, memory_ratio: float = 0.9,
+                          activation_reserve: float = 0.35) -> float:
+        """Auto placement (reference: zero/gemini/placement_policy.py:128
+        AutoPlacementPolicy — re-derived for 288 GB HBM3E): choose how much
+        optimizer state must live in pinned host memory so that
+        params(bf16) + grads(bf16) + resident fp32 states (master+m+v = 12
+        B/elem) fit under memory_ratio·capacity with an activation reserve.
+        Returns 0.0 when everything fits (the common case on MI355X) —
+        runtime chunk eviction would only add traffic then."""
+        budget = capacity_bytes * memory_ratio * (1.0 - activation_reserve)
+        fixed = param_numel * 4  # bf16 params + bf16 grads
+        state_bytes = param_numel * 12
+        if fixed + state_bytes <= budget:
+            return 0.0
+        if fixed >= budget:
+            return 1.0
+        return min(1.0, max(0.0, 1.0 - (budget - fixed) / state_bytes))
+
     def configure(
         self,
         model: nn.Module,
@@ -217,13 +240,21 @@ class GeminiPlugin(Plugin):
         dataloader: Optional[DataLoader] = None,
         lr_scheduler: Optional[LRScheduler] = None,
     ) -> Tuple[nn.Module, OptimizerWrapper, Callable, DataLoader, LRScheduler]:
+        if self.placement_policy == "auto" and torch.cuda.is_available():
+            numel = sum(p.numel() for p in model.parameters()) if not isinstance(model, ModelWrapper) else 0
+            if numel:
+                cap = torch.cuda.get_device_properties(0).total_memory
+                frac = self.auto_offload_frac(numel, cap, self._memory_ratio)
+                self.offload_optim_frac = max(self.offload_optim_frac, frac)
+                self.zero_kwargs["cpu_offload_frac"] = self.offload_optim_frac
         if self.shard_param_frac == 1.0:
             from ...zero.gemini import GeminiDDP, GeminiOptimizer
 
             if not isinstance(model, ModelWrapper):
                 model = GeminiDDP(model, chunk_size_m=self.chunk_size_m, precision=self.precision)
             if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
-                optimizer = GeminiOptimizer(optimizer, model, max_norm=self.max_norm)
+                kw = self._scaler_kwargs if self.precision == "fp16" else {}
+                optimizer = GeminiOptimizer(optimizer, model, max_norm=self.max_norm, **kw)
             return model, optimizer, criterion, dataloader, lr_scheduler
         if not isinstance(model, ModelWrapper):
             model = LowLevelZeroModel(model, self.precision)

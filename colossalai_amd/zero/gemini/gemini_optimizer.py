@@ -26,11 +26,30 @@ class GeminiOptimizer(OptimizerWrapper):
         optim: Optimizer,
         model: GeminiDDP,
         max_norm: float = 0.0,
+        **scaler_kwargs,
     ):
         super().__init__(optim)
         self.model = model
         self.max_norm = max_norm
         self.world = model.world
+        self.mixin = None
+        if model.dtype == torch.float16:
+            from ...amp.mixed_precision_mixin import FP16MixedPrecisionMixin
+
+            class _GeminiFP16Mixin(FP16MixedPrecisionMixin):
+                def __init__(m, outer, **kw):
+                    super().__init__(**kw)
+                    m._outer = outer
+
+                def check_local_overflow(m) -> bool:
+                    for c in m._outer.model.chunks:
+                        if c.grad_shard is not None and not torch.isfinite(c.grad_shard.sum()):
+                            return True
+                        if c.grad_flat.numel() and not torch.isfinite(c.grad_flat.sum()):
+                            return True
+                    return False
+
+            self.mixin = _GeminiFP16Mixin(self, **scaler_kwargs)
         # masters + adam state, one triple per chunk shard (fp32)
         self.masters: List[torch.Tensor] = []
         self.exp_avg: List[torch.Tensor] = []
@@ -45,6 +64,8 @@ class GeminiOptimizer(OptimizerWrapper):
         )
 
     def backward(self, loss: torch.Tensor, inputs=None, retain_graph: bool = False, **kwargs):
+        if self.mixin is not None:
+            loss = self.mixin.pre_backward(loss)
         loss.backward(inputs=inputs, retain_graph=retain_graph, **kwargs)
 
     def _grad_norm(self) -> torch.Tensor:
@@ -62,6 +83,15 @@ class GeminiOptimizer(OptimizerWrapper):
         from ...nn.optimizer.fused_adam import fused_adam_step_cpu
         from ...ops import has_kernels, kernels
 
+        if self.mixin is not None and self.mixin.should_skip_step():
+            # overflowed fp16 step: drop grads, shrink the scale, move on
+            self.zero_grad()
+            for c in self.model.chunks:
+                c.grad_flat = torch.zeros(0, dtype=c.flat.dtype, device=c.shard.device)
+                c.grads_done = 0
+                for p, _ in c.params:
+                    p.grad = None
+            return
         group = self.optim.param_groups[0]
         group.setdefault("step", 0)
         group["step"] += 1
@@ -73,6 +103,8 @@ class GeminiOptimizer(OptimizerWrapper):
         adamw = getattr(self.optim, "adamw_mode", True)
 
         div_scale = float(self.world)
+        if self.mixin is not None:
+            div_scale *= float(self.mixin.get_grad_div_scale())
         if self.max_norm > 0:
             norm = self._grad_norm()
             clip = (norm / self.max_norm).clamp(min=1.0)

@@ -91,3 +91,23 @@ def run_matrix(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_plugin_model_matrix():
     spawn(run_matrix, 2)
+
+
+def test_gemini_auto_offload_frac():
+    """Auto placement heuristic: 0 when states fit, rises smoothly, 1 when
+    even params+grads blow the budget."""
+    from colossalai_amd.booster.plugin import GeminiPlugin
+
+    GiB = 1 << 30
+    cap = 288 * GiB
+    # llama-7b on MI355X: everything fits -> no offload
+    assert GeminiPlugin.auto_offload_frac(6_738_000_000, cap) == 0.0
+    # 70B on one 288 GB GPU: states cannot all stay resident
+    f = GeminiPlugin.auto_offload_frac(70_000_000_000, cap)
+    assert 0.0 < f <= 1.0
+    # absurd model: full offload
+    assert GeminiPlugin.auto_offload_frac(10**12, cap) == 1.0
+    # monotonic in model size
+    sizes = [5e9, 2e10, 5e10, 1e11, 3e11]
+    fr = [GeminiPlugin.auto_offload_frac(int(s), cap) for s in sizes]
+    assert fr == sorted(fr)

@@ -131,3 +131,47 @@ def run_gemini_no_sync(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_gemini_no_sync():
     spawn(run_gemini_no_sync, 2)
+
+
+def run_gemini_fp16(rank, world_size, port):
+    """fp16 Gemini: scaled backward, overflow skip (scale shrinks, params
+    unchanged), then a normal step makes progress."""
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(_cfg())
+    gm = GeminiDDP(copy.deepcopy(model), precision="fp16", chunk_size_m=1)
+    opt = GeminiOptimizer(FusedAdam(gm.parameters(), lr=1e-3), gm, initial_scale=2.0**8, hysteresis=1)
+    assert opt.mixin is not None
+    x = torch.randint(0, 128, (2, 16))
+
+    before = gm.state_dict()
+    # force an overflow: backward, then poison one grad shard
+    out = gm(input_ids=x, labels=x)
+    opt.backward(out["loss"])
+    for c in gm.chunks:
+        if c.grad_shard is not None:
+            c.grad_shard[0] = float("inf")
+            break
+    scale0 = float(opt.mixin.loss_scale)
+    opt.step()
+    opt.zero_grad()
+    assert float(opt.mixin.loss_scale) < scale0, "scale did not back off"
+    after = gm.state_dict()
+    for k in before:
+        assert torch.equal(before[k], after[k]), f"{k} changed on a skipped step"
+
+    # normal steps still learn
+    losses = []
+    for _ in range(3):
+        out = gm(input_ids=x, labels=x)
+        opt.backward(out["loss"])
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(out["loss"]))
+    assert losses[-1] < losses[0], losses
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_gemini_fp16_scaler():
+    spawn(run_gemini_fp16, 2)
