@@ -47,6 +47,8 @@ class LlamaConfig:
     attention_bias: bool = False  # Qwen2-style qkv bias
     qk_norm: bool = False  # Qwen3-style per-head RMSNorm on q/k before RoPE
     head_dim_override: Optional[int] = None  # Qwen3 decouples head_dim from hidden/heads
+    # GLM-4-style rope: rotate-every-two on the first (factor*head_dim) dims
+    partial_interleaved_rotary_factor: float = 0.0
 
     def __post_init__(self):
         if self.num_key_value_heads is None:
@@ -80,6 +82,11 @@ LLAMA_CONFIGS = {
     "qwen2-7b": LlamaConfig(vocab_size=152064, hidden_size=3584, intermediate_size=18944,
                             num_hidden_layers=28, num_attention_heads=28, num_key_value_heads=4,
                             max_position_embeddings=4096, rope_theta=1e6, attention_bias=True),
+    # GLM-4: qkv bias + partial interleaved rotary
+    "glm4-9b": LlamaConfig(vocab_size=151552, hidden_size=4096, intermediate_size=13696,
+                           num_hidden_layers=40, num_attention_heads=32, num_key_value_heads=2,
+                           max_position_embeddings=8192, attention_bias=True,
+                           partial_interleaved_rotary_factor=0.5, rms_norm_eps=1.5625e-07),
     # Qwen3: qk-norm + decoupled head_dim
     "qwen3-8b": LlamaConfig(vocab_size=151936, hidden_size=4096, intermediate_size=12288,
                             num_hidden_layers=36, num_attention_heads=32, num_key_value_heads=8,
@@ -122,6 +129,8 @@ class LlamaAttention(nn.Module):
         self.o_proj = nn.Linear(Hq * D, cfg.hidden_size, bias=False)
         self.scale = 1.0 / math.sqrt(D)
         self.qk_norm = cfg.qk_norm
+        self.pir_factor = cfg.partial_interleaved_rotary_factor
+        self.rope_theta = cfg.rope_theta
         if cfg.qk_norm:
             self.q_norm_weight = nn.Parameter(torch.ones(D))
             self.k_norm_weight = nn.Parameter(torch.ones(D))
@@ -176,6 +185,25 @@ class LlamaAttention(nn.Module):
             attn = flash_attention(q, k, v.contiguous(), causal=True, scale=self.scale)
             attn = all_to_all_comm(attn, sp_group, scatter_dim=1, gather_dim=2)  # back to [B, S, Hq, D]
             return self.o_proj(attn.reshape(B, S, -1))
+        if self.pir_factor > 0:
+            # GLM-4: partial interleaved rotary — unpack, rope, flash
+            from ..ops import flash_attention
+            from .gptj import _gptj_rope
+
+            Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+            r = int(D * self.pir_factor)
+            q = _gptj_rope(qkv[:, :, : Hq * D].reshape(B, S, Hq, D), r, self.rope_theta)
+            k = _gptj_rope(qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D), r,
+                           self.rope_theta)
+            v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
+            if D in (64, 128) and hidden.dtype == torch.bfloat16:
+                attn = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                                       causal=True, scale=self.scale)
+            else:
+                from ..ops.attention import attention_ref
+
+                attn = attention_ref(q, k, v, causal=True, scale=self.scale, upcast=False)
+            return self.o_proj(attn.reshape(B, S, Hq * D))
         if self.qk_norm:
             # Qwen3: per-head RMSNorm on q/k before RoPE — unpack, norm,
             # rope, flash (the packed fused path skips the norm)
