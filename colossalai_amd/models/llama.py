@@ -140,6 +140,11 @@ class LlamaAttention(nn.Module):
         B, S, _ = hidden.shape
         qkv = self.qkv_proj(hidden)
         sp_mode = getattr(self, "sp_mode", None)
+        if sp_mode is not None:
+            assert not self.qk_norm and self.pir_factor == 0, (
+                "qk-norm / partial-rotary variants are not wired into the SP "
+                "attention branches yet — run these models without SP"
+            )
         if sp_mode == "ring_attn":
             # context parallelism: Q stays, K/V blocks travel the xGMI ring
             import torch.distributed as dist
@@ -229,6 +234,10 @@ class LlamaAttention(nn.Module):
     @torch.no_grad()
     def forward_with_cache(self, hidden, rope_table, kcache, vcache, positions, seq_lens, prefill: bool):
         """Inference path: RoPE + cache append + (flash prefill | decode) attention."""
+        assert not getattr(self, "qk_norm", False) and getattr(self, "pir_factor", 0) == 0, (
+            "qk-norm / partial-rotary variants are not wired into the KV-cache "
+            "path yet — inference engines support the rotate-half families"
+        )
         from ..ops import has_kernels
         from ..ops.attention import attention_ref
         from ..ops.rope import apply_rope_ref
