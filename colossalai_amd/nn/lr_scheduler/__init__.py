@@ -10,6 +10,8 @@ from typing import List
 from torch.optim.lr_scheduler import _LRScheduler
 
 __all__ = [
+    "OneCycleLR",
+    "DelayedCosineAnnealingLR",
     "LinearWarmupLR",
     "CosineAnnealingLR",
     "CosineAnnealingWarmupLR",
@@ -141,3 +143,46 @@ class MultiStepWarmupLR(_LRScheduler):
             return [lr * (self.last_epoch + 1) / max(self.warmup_steps, 1) for lr in self.base_lrs]
         n = sum(1 for m in self.milestones if m <= self.last_epoch)
         return [lr * self.gamma**n for lr in self.base_lrs]
+
+class OneCycleLR(_LRScheduler):
+    """One-cycle policy (reference: colossalai/nn/lr_scheduler/onecycle.py —
+    thin wrapper over torch's implementation with the warmup-style API)."""
+
+    def __init__(self, optimizer, total_steps: int, pct_start: float = 0.3,
+                 div_factor: float = 25.0, final_div_factor: float = 1e4, last_epoch: int = -1):
+        import torch.optim.lr_scheduler as tls
+
+        max_lrs = [g["lr"] for g in optimizer.param_groups]
+        self._inner = tls.OneCycleLR(optimizer, max_lr=max_lrs, total_steps=total_steps,
+                                     pct_start=pct_start, div_factor=div_factor,
+                                     final_div_factor=final_div_factor, last_epoch=last_epoch)
+        self.optimizer = optimizer
+
+    def step(self, epoch=None):
+        self._inner.step(epoch)
+
+    def get_last_lr(self):
+        return self._inner.get_last_lr()
+
+    def state_dict(self):
+        return self._inner.state_dict()
+
+    def load_state_dict(self, sd):
+        self._inner.load_state_dict(sd)
+
+
+class DelayedCosineAnnealingLR(WarmupScheduler):
+    """Hold the base LR for ``delay_steps``, then cosine-anneal (reference:
+    colossalai/nn/lr_scheduler/delayed.py DelayerScheduler + cosine)."""
+
+    def __init__(self, optimizer, total_steps: int, delay_steps: int, last_epoch: int = -1):
+        import torch.optim.lr_scheduler as tls
+
+        base = tls.CosineAnnealingLR(optimizer, max(total_steps - delay_steps, 1))
+        super().__init__(optimizer, delay_steps, base, last_epoch=last_epoch)
+
+    def get_lr(self):
+        # during the delay: hold base lr (WarmupScheduler ramps; override)
+        if self.last_epoch < self.warmup_steps:
+            return self.base_lrs
+        return super().get_lr()
