@@ -101,3 +101,39 @@ def test_tp2_zero1():
 
 if __name__ == "__main__":
     test_pp2()
+
+
+def _run_bf16_pp(rank, world_size, port):
+    """pp2 in bf16: HybridParallelNaiveOptimizer (bf16 master path) through
+    the pipeline schedule — finite losses, steps run, params update."""
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(_tiny())
+    plugin = HybridParallelPlugin(tp_size=1, pp_size=2, precision="bf16",
+                                  num_microbatches=2, zero_stage=0)
+    booster = Booster(plugin=plugin)
+    optimizer = torch.optim.AdamW(model.parameters(), lr=1e-2)
+    criterion = lambda out, micro: out["loss"]
+    model_b, optimizer_b, criterion, _, _ = booster.boost(model, optimizer, criterion)
+
+    x = torch.randint(0, 128, (4, 16))
+    batch = {"input_ids": x, "labels": x}
+    start, end = model_b.module.stage_range
+    before = model_b.module.model.layers[start].input_layernorm_weight.detach().clone()
+    losses = []
+    for _ in range(2):
+        result = booster.execute_pipeline(iter([batch]), model_b, criterion, optimizer_b,
+                                          return_loss=True)
+        if plugin.stage_manager.is_last_stage():
+            assert result["loss"] is not None and torch.isfinite(result["loss"])
+            losses.append(float(result["loss"]))
+        optimizer_b.step()
+        optimizer_b.zero_grad()
+    after = model_b.module.model.layers[start].input_layernorm_weight.detach()
+    assert not torch.equal(before, after), "params did not update"
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_pp2_bf16():
+    spawn(_run_bf16_pp, 2)
