@@ -17,7 +17,48 @@ from torch.optim import Optimizer
 from ..interface.optimizer import OptimizerWrapper
 from .mixed_precision_mixin import BF16MixedPrecisionMixin, FP16MixedPrecisionMixin, MixedPrecisionMixin
 
-__all__ = ["MixedPrecisionOptimizer"]
+__all__ = ["MixedPrecisionOptimizer", "compute_global_grad_norm"]
+
+
+def compute_global_grad_norm(
+    working_grad_pairs: List[Tuple[Tensor, Tensor]],
+    tp_pg: Optional[dist.ProcessGroup] = None,
+    pp_pg: Optional[dist.ProcessGroup] = None,
+) -> float:
+    """Global L2 grad norm across tp and pp.
+
+    tp-sharded params (``p.tp_sharded``) contribute their shard's norm summed
+    over the tp group; replicated params are counted once; pipeline stages
+    hold disjoint params so the pp reduction is a plain sum; tied embed/head
+    duplicates carry ``_grad_norm_skip`` and are counted on one stage only
+    (reference: colossalai/booster/plugin/hybrid_parallel_plugin.py:406-451).
+    """
+    sq_sharded = 0.0
+    sq_replicated = 0.0
+    device = None
+    for working, g in working_grad_pairs:
+        if getattr(working, "_grad_norm_skip", False):
+            continue
+        device = g.device
+        part = torch.linalg.vector_norm(g, 2.0, dtype=torch.float32) ** 2
+        if getattr(working, "tp_sharded", False):
+            sq_sharded = sq_sharded + part
+        else:
+            sq_replicated = sq_replicated + part
+    tp_world = dist.get_world_size(tp_pg) if tp_pg is not None else 1
+    pp_world = dist.get_world_size(pp_pg) if pp_pg is not None else 1
+    if tp_world == 1 and pp_world == 1:
+        total = sq_sharded + sq_replicated
+        return float(total) ** 0.5 if isinstance(total, float) else float(total.sqrt())
+    if device is None:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+    sq_sharded = torch.as_tensor(sq_sharded, dtype=torch.float32, device=device)
+    if tp_world > 1:
+        dist.all_reduce(sq_sharded, group=tp_pg)
+    total = sq_sharded + torch.as_tensor(sq_replicated, dtype=torch.float32, device=device)
+    if pp_world > 1:
+        dist.all_reduce(total, group=pp_pg)
+    return float(total.sqrt())
 
 
 class _NaiveFP16Mixin(FP16MixedPrecisionMixin):
@@ -45,8 +86,15 @@ class MixedPrecisionOptimizer(OptimizerWrapper):
         hysteresis: int = 2,
         max_scale: float = 2**32,
         max_norm: float = 0.0,
+        tp_process_group: Optional[dist.ProcessGroup] = None,
+        pp_process_group: Optional[dist.ProcessGroup] = None,
     ):
         super().__init__(optim)
+        # under tp/pp the clip norm is GLOBAL: tp-sharded grads are summed over
+        # the tp group, stage-local grads over the pp group
+        # (reference: colossalai/booster/plugin/hybrid_parallel_plugin.py:406-451)
+        self.tp_pg = tp_process_group
+        self.pp_pg = pp_process_group
         # Working params from the optimizer's groups; swap in fp32 masters.
         working_params: List[Parameter] = []
         for group in self.optim.param_groups:
@@ -110,13 +158,11 @@ class MixedPrecisionOptimizer(OptimizerWrapper):
                         p.grad.mul_(1.0 / div_scale)
 
     def _compute_grad_norm(self, param_gradient_pairs: List[Tuple[Tensor, Tensor]], norm_type: float = 2.0) -> float:
-        if len(param_gradient_pairs) == 0:
+        if len(param_gradient_pairs) == 0 and self.pp_pg is None:
             return 0.0
-        grads = [g for _, g in param_gradient_pairs]
-        norm = torch.linalg.vector_norm(
-            torch.stack([torch.linalg.vector_norm(g, norm_type, dtype=torch.float32) for g in grads]), norm_type
-        )
-        return norm.item()
+        assert norm_type == 2.0, "only L2 clipping is supported"
+        pairs = [(self.master_to_working.get(m, m), g) for m, g in param_gradient_pairs]
+        return compute_global_grad_norm(pairs, self.tp_pg, self.pp_pg)
 
     def step(self, *args, **kwargs):
         if self.mixin.should_skip_step():
@@ -130,7 +176,7 @@ class MixedPrecisionOptimizer(OptimizerWrapper):
                     master.grad = working.grad.to(master.dtype)
                     working.grad = None
         total_norm = 0.0
-        if self.max_norm > 0.0 or self.mixin.get_grad_div_scale() != 1.0:
+        if self.max_norm > 0.0:
             pairs = [
                 (m, m.grad) for g in self.optim.param_groups for m in g["params"] if m.grad is not None
             ]

@@ -35,12 +35,18 @@ _PRECISION_DTYPE = {"fp16": torch.float16, "bf16": torch.bfloat16, "fp32": torch
 
 
 class HybridParallelModule(ModelWrapper):
-    def __init__(self, module: nn.Module, dtype: torch.dtype, dp_group, tp_group, sp_group):
+    def __init__(self, module: nn.Module, dtype: torch.dtype, dp_group, tp_group, sp_group,
+                 embed_group=None, tied_param: Optional[torch.Tensor] = None):
         super().__init__(module)
         self.dtype = dtype
         self.dp_group = dp_group
         self.tp_group = tp_group
         self.sp_group = sp_group
+        # tied embed/lm_head under pp>1: first and last stage each hold a copy;
+        # grads are summed over the 2-rank embed group every step
+        # (reference: hybrid_parallel_plugin.py:131 sync_shared_params)
+        self.embed_group = embed_group
+        self.tied_param = tied_param
         self.require_grad_sync = True
 
     def forward(self, *args, **kwargs):
@@ -59,6 +65,16 @@ class HybridParallelModule(ModelWrapper):
             yield
         finally:
             self.require_grad_sync = old
+
+    def sync_shared_params(self):
+        """Sum grads of the tied embed/lm_head weight across the pipeline's
+        first and last stage so both copies take the same update."""
+        if self.tied_param is None or self.embed_group is None:
+            return
+        if dist.get_world_size(self.embed_group) == 1:
+            return
+        if self.tied_param.grad is not None:
+            dist.all_reduce(self.tied_param.grad, group=self.embed_group)
 
     def sync_partial_sp_grads(self):
         """split_gather SP: norm-weight grads are partial over seq shards —
@@ -123,9 +139,25 @@ class HybridParallelNaiveOptimizer(MixedPrecisionOptimizer):
 class HybridParallelFP32Optimizer(OptimizerWrapper):
     """ZeRO-0 fp32: plain optimizer + dp×sp grad averaging after backward."""
 
-    def __init__(self, optim: Optimizer, model: HybridParallelModule):
+    def __init__(self, optim: Optimizer, model: HybridParallelModule, max_norm: float = 0.0,
+                 tp_process_group=None, pp_process_group=None):
         super().__init__(optim)
         self.model_wrapper = model
+        self.max_norm = max_norm
+        self.tp_pg = tp_process_group
+        self.pp_pg = pp_process_group
+
+    def step(self, *args, **kwargs):
+        if self.max_norm > 0.0:
+            from ...amp.mixed_precision_optimizer import compute_global_grad_norm
+
+            pairs = [(p, p.grad) for g in self.optim.param_groups for p in g["params"] if p.grad is not None]
+            total_norm = compute_global_grad_norm(pairs, self.tp_pg, self.pp_pg)
+            if total_norm > self.max_norm:
+                scale = self.max_norm / total_norm
+                for _, g in pairs:
+                    g.mul_(scale)
+        return self.optim.step(*args, **kwargs)
 
     def backward(self, loss, inputs=None, retain_graph=False, **kwargs):
         loss.backward(inputs=inputs, retain_graph=retain_graph, **kwargs)
@@ -210,6 +242,11 @@ class HybridParallelPlugin(Plugin):
         self.sp_group = self.pg_mesh.get_group_along_axis(SP_AXIS)
         # grads of dp×sp-replicated params sync over the flattened group
         self.dp_sp_group = self.pg_mesh.get_group_along_axis([DP_AXIS, SP_AXIS])
+        # tied embed/lm_head sync group: first + last pipeline stage
+        # (None on middle stages; the groups are still created collectively)
+        self.embed_group = None
+        if pp_size > 1:
+            self.embed_group = self.pg_mesh.get_group_along_axis(PP_AXIS, indices_at_axis=[0, pp_size - 1])
 
         self.stage_manager = None
         self.scheduler = None
@@ -292,7 +329,11 @@ class HybridParallelPlugin(Plugin):
     ) -> Tuple[nn.Module, OptimizerWrapper, Callable, DataLoader, LRScheduler]:
         dtype = _PRECISION_DTYPE[self.precision]
 
+        pre_shard_names: Optional[dict] = None
         if not isinstance(model, ModelWrapper):
+            # record param identities before surgery so multi-group optimizers
+            # can be re-pointed at the sharded replacements by name
+            pre_shard_names = {id(p): n for n, p in model.named_parameters()}
             # pipeline layer assignment BEFORE sharding (policies may use it)
             if self.stage_manager is not None:
                 self._assign_pipeline_stage(model)
@@ -306,22 +347,39 @@ class HybridParallelPlugin(Plugin):
                 from ...ops.zb_linear import convert_to_zb_linears
 
                 convert_to_zb_linears(model)
-            model = HybridParallelModule(model, dtype, self.dp_sp_group, self.tp_group, self.sp_group)
+            tied = getattr(model, "_tied_embed_param", None)
+            if tied is not None and self.embed_group is not None and dist.get_world_size(self.embed_group) > 1:
+                # both copies must start identical: broadcast from the first stage
+                src = min(self.pg_mesh.get_ranks_in_group(self.embed_group))
+                dist.broadcast(tied.data, src=src, group=self.embed_group)
+            model = HybridParallelModule(
+                model, dtype, self.dp_sp_group, self.tp_group, self.sp_group,
+                embed_group=self.embed_group, tied_param=tied,
+            )
 
         if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
-            self._rebuild_param_groups(optimizer, model.module)
+            self._rebuild_param_groups(optimizer, model.module, pre_shard_names)
             if self.zero_stage == 0:
                 if self.precision == "fp32":
-                    optimizer = HybridParallelFP32Optimizer(optimizer, model)
+                    optimizer = HybridParallelFP32Optimizer(
+                        optimizer, model, max_norm=self.max_norm,
+                        tp_process_group=self.tp_group if self.tp_size > 1 else None,
+                        pp_process_group=self.pp_group if self.pp_size > 1 else None,
+                    )
                 else:
                     optimizer = HybridParallelNaiveOptimizer(
-                        optimizer, model, precision=self.precision, max_norm=self.max_norm, **self.amp_kwargs
+                        optimizer, model, precision=self.precision, max_norm=self.max_norm,
+                        tp_process_group=self.tp_group if self.tp_size > 1 else None,
+                        pp_process_group=self.pp_group if self.pp_size > 1 else None,
+                        **self.amp_kwargs,
                     )
             else:
                 optimizer = LowLevelZeroOptimizer(
                     optimizer,
                     dp_process_group=self.dp_sp_group,
                     forced_dtype=dtype if self.precision != "fp32" else None,
+                    tp_process_group=self.tp_group if self.tp_size > 1 else None,
+                    pp_process_group=self.pp_group if self.pp_size > 1 else None,
                     **self.zero_kwargs,
                     **({} if self.precision != "fp16" else self.amp_kwargs),
                 )
@@ -336,6 +394,26 @@ class HybridParallelPlugin(Plugin):
         inner = model.model if hasattr(model, "model") else model
         assert hasattr(inner, "layers"), "pipeline parallelism needs a .layers decoder stack"
         n_layers = len(inner.layers)
+        # tied embed/lm_head detection BEFORE stubbing: first stage keeps the
+        # weight via embed_tokens, last via lm_head; grads sync over embed_group
+        head = getattr(model, "lm_head", None)
+        embed = getattr(inner, "embed_tokens", None)
+        is_tied = (
+            head is not None and embed is not None
+            and getattr(head, "weight", None) is not None and head.weight is embed.weight
+        )
+        model._tied_embed_param = None
+        if is_tied:
+            last_stage = (
+                self.stage_manager.is_last_stage(self.num_model_chunks - 1)
+                if self.pp_style == "interleaved" else self.stage_manager.is_last_stage()
+            )
+            if self.stage_manager.is_first_stage():
+                model._tied_embed_param = embed.weight
+            elif last_stage:
+                model._tied_embed_param = head.weight
+                # full grad lives on both stages after sync — count it once
+                head.weight._grad_norm_skip = True
         if self.pp_style == "interleaved":
             V, pp = self.num_model_chunks, self.pp_size
             per = [n_layers // (pp * V)] * (pp * V)
@@ -369,18 +447,43 @@ class HybridParallelPlugin(Plugin):
             model.lm_head = _StageStub()
 
     @staticmethod
-    def _rebuild_param_groups(optimizer: Optimizer, model: nn.Module) -> None:
+    def _rebuild_param_groups(optimizer: Optimizer, model: nn.Module,
+                              pre_shard_names: Optional[dict] = None) -> None:
         """Sharding/stage-release replaced parameter objects; re-point the
-        optimizer's param groups at the live model parameters."""
+        optimizer's param groups at the live model parameters.
+
+        Multi-group optimizers (e.g. weight-decay splits) are re-pointed by
+        matching each old param's NAME (recorded before surgery) to the
+        sharded replacement; params released to other pipeline stages are
+        dropped from their group."""
         live = list(model.parameters())
         current = [p for g in optimizer.param_groups for p in g["params"]]
         if len(current) == len(live) and all(a is b for a, b in zip(current, live)):
             return
-        assert len(optimizer.param_groups) == 1, (
-            "after sharding, only single-param-group optimizers can be re-pointed automatically; "
-            "construct the optimizer after booster.boost for multiple groups"
+        if len(optimizer.param_groups) == 1:
+            optimizer.param_groups[0]["params"] = live
+            optimizer.state.clear()
+            return
+        assert pre_shard_names is not None, (
+            "multi-group optimizer re-pointing needs the pre-shard name map; "
+            "construct the optimizer after booster.boost, or pass the bare model"
         )
-        optimizer.param_groups[0]["params"] = live
+        live_by_name = dict(model.named_parameters())
+        seen = set()
+        for group in optimizer.param_groups:
+            new_params = []
+            for p in group["params"]:
+                name = pre_shard_names.get(id(p))
+                if name is None:
+                    raise ValueError(
+                        "optimizer contains a parameter that was not in the model passed to boost()"
+                    )
+                new_p = live_by_name.get(name)
+                if new_p is None or id(new_p) in seen:
+                    continue  # released to another pipeline stage / tied duplicate
+                seen.add(id(new_p))
+                new_params.append(new_p)
+            group["params"] = new_params
         optimizer.state.clear()
 
     # ---------------------------------------------------------------- runtime
@@ -402,7 +505,10 @@ class HybridParallelPlugin(Plugin):
             result = self.scheduler.forward_backward_step(
                 model, data_iter, criterion, optimizer, return_loss, return_outputs
             )
-        # grad sync after all microbatches
+        # grad sync after all microbatches: tied embed/head first (so the dp
+        # reduction sees the summed grad), then sp partials, then dp
+        model.sync_shared_params()
+        model.sync_partial_sp_grads()
         if isinstance(optimizer, LowLevelZeroOptimizer):
             optimizer.sync_dp_grads()
         else:

@@ -76,7 +76,10 @@ class GeminiOptimizer(OptimizerWrapper):
                 sq += (c.grad_shard.float() / self.world).pow(2).sum()
         if self.world > 1:
             dist.all_reduce(sq, group=self.model.group)
-        return sq.sqrt()
+        # grads still carry the fp16 loss scale — divide it out so the norm
+        # compared against max_norm is the true gradient norm
+        scale = float(self.mixin.get_grad_div_scale()) if self.mixin is not None else 1.0
+        return sq.sqrt() / scale
 
     @torch.no_grad()
     def step(self, closure=None):

@@ -92,6 +92,8 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         group_dp_pgs: Optional[List[Optional[dist.ProcessGroup]]] = None,
         group_grad_divisors: Optional[List[int]] = None,
         fp8_communication: bool = False,
+        tp_process_group: Optional[dist.ProcessGroup] = None,
+        pp_process_group: Optional[dist.ProcessGroup] = None,
     ):
         super().__init__(optimizer)
         from ...nn.optimizer.fused_adam import FusedAdam
@@ -120,6 +122,11 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         self.master_weights = master_weights
         self.cpu_offload_frac = float(cpu_offload_frac)
         self.fp8_communication = fp8_communication
+        # tp/pp groups are used only for the global grad-clip norm
+        # (reference reduces the squared norm over tp_pg and pp_pg:
+        # colossalai/booster/plugin/hybrid_parallel_plugin.py:610-658)
+        self.tp_pg = tp_process_group
+        self.pp_pg = pp_process_group
         self.require_grad_sync = True
         self._accum_steps_pending = False
 
@@ -331,34 +338,70 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
 
     # ------------------------------------------------------------------ norm
     def _compute_grad_norm(self, norm_type: float = 2.0) -> float:
-        sq = 0.0
-        device = None
-        partials = []
-        for gi, buckets in enumerate(self._group_buckets):
-            flat_grad = self._flat_grads[gi]
-            device = flat_grad.device
+        """L2 norm of the (still scaled, dp-summed) gradients, global across
+        dp, tp and pp. Each rank norms its own dp-shard of the flat grad
+        buffers, so the dp all-reduce alone gives the full-model norm when
+        tp=pp=1. Under tp, a shard mixes tp-sharded and tp-replicated param
+        grads: sharded contributions are summed over the tp group while
+        replicated ones (identical on every tp rank after grad sync) are
+        counted once. pp stages hold disjoint params — plain sum — except
+        tied embed/head duplicates, which carry ``_grad_norm_skip``."""
+        tp_world = dist.get_world_size(self.tp_pg) if self.tp_pg is not None else 1
+        pp_world = dist.get_world_size(self.pp_pg) if self.pp_pg is not None else 1
+        device = self._flat_grads[0].device if self._flat_grads else torch.device("cpu")
+
+        if tp_world == 1 and pp_world == 1:
+            partials = []
+            for gi, buckets in enumerate(self._group_buckets):
+                flat_grad = self._flat_grads[gi]
+                world, rank = self._g_world[gi], self._g_rank[gi]
+                for b in buckets:
+                    blen = b.end - b.start
+                    shard = flat_grad[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
+                    partials.append(torch.linalg.vector_norm(shard.float()) ** 2)
+            if not partials:
+                return 0.0
+            total = torch.stack(partials).sum()
+            if self.world > 1:
+                dist.all_reduce(total, group=self.dp_pg)
+            return float(total.sqrt())
+
+        # tp/pp-aware path: walk params, intersect each with this rank's
+        # bucket shard (padding regions are zero and excluded either way)
+        sq = torch.zeros(2, dtype=torch.float32, device=device)  # [sharded, replicated]
+        for p, (gi, off, numel) in self._param_slice.items():
+            if getattr(p, "_grad_norm_skip", False):
+                continue
+            b = self._param_bucket[p]
             world, rank = self._g_world[gi], self._g_rank[gi]
-            for b in buckets:
-                blen = b.end - b.start
-                shard = flat_grad[b.start + rank * blen // world : b.start + (rank + 1) * blen // world]
-                partials.append(torch.linalg.vector_norm(shard.float()) ** 2)
-        if not partials:
-            return 0.0
-        total = torch.stack(partials).sum()
+            blen = b.end - b.start
+            lo = max(off, b.start + rank * blen // world)
+            hi = min(off + numel, b.start + (rank + 1) * blen // world)
+            if lo >= hi:
+                continue
+            part = torch.linalg.vector_norm(self._flat_grads[gi][lo:hi].float()) ** 2
+            sq[0 if getattr(p, "tp_sharded", False) else 1] += part
         if self.world > 1:
-            if not self.partition_grad:
-                pass  # all ranks have identical full grads; shard-local partials already partition them
-            dist.all_reduce(total, group=self.dp_pg)
+            dist.all_reduce(sq, group=self.dp_pg)
+        if tp_world > 1:
+            sharded = sq[0].clone()
+            dist.all_reduce(sharded, group=self.tp_pg)
+        else:
+            sharded = sq[0]
+        total = sharded + sq[1]
+        if pp_world > 1:
+            dist.all_reduce(total, group=self.pp_pg)
         return float(total.sqrt())
 
     # ------------------------------------------------------------------ step
     def step(self, closure=None):
         assert closure is None
+        # Reductions must land BEFORE the overflow check: check_local_overflow
+        # reads _flat_grads, which the comm stream may still be writing.
+        self._wait_all_reductions()
         if self.mixin.should_skip_step():
             self.zero_grad()
             return
-
-        self._wait_all_reductions()
 
         loss_scale = self.mixin.get_grad_div_scale()
 
