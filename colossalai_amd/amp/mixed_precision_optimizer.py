@@ -193,6 +193,47 @@ class MixedPrecisionOptimizer(OptimizerWrapper):
             if p in self.working_to_master:
                 self.working_to_master[p].data.copy_(p.data.float())
 
+    # ------------------------------------------------------------ checkpoint
+    def get_param_states(self, names: Dict[int, Tensor]) -> Dict[str, dict]:
+        """Per-param optimizer states keyed by param NAME (topology
+        independent): fp32 master + the inner optimizer's tensor state."""
+        out: Dict[str, dict] = {}
+        for gi, group in enumerate(self.optim.param_groups):
+            for master in group["params"]:
+                working = self.master_to_working.get(master, master)
+                name = names.get(id(working))
+                if name is None:
+                    continue
+                st: dict = {"_group": gi, "master": master.detach().float().cpu().clone()}
+                inner = self.optim.state.get(master, {})
+                for k, v in inner.items():
+                    st[k] = v.cpu().clone() if isinstance(v, Tensor) else v
+                out[name] = st
+        return out
+
+    def set_param_states(self, states: Dict[str, dict], names: Dict[int, Tensor]) -> None:
+        for group in self.optim.param_groups:
+            for master in group["params"]:
+                working = self.master_to_working.get(master, master)
+                name = names.get(id(working))
+                if name is None or name not in states:
+                    continue
+                st = dict(states[name])
+                st.pop("_group", None)
+                m = st.pop("master", None)
+                if m is not None:
+                    master.data.copy_(m.to(master.device, dtype=master.dtype))
+                    working.data.copy_(master.data.to(working.dtype))
+                inner = {}
+                for k, v in st.items():
+                    if isinstance(v, Tensor) and v.shape == master.shape:
+                        inner[k] = v.to(master.device, dtype=v.dtype)
+                    elif k == "step":
+                        inner[k] = torch.tensor(float(v)) if not isinstance(v, Tensor) else v
+                    else:
+                        inner[k] = v
+                self.optim.state[master] = inner
+
     def get_working_to_master_map(self):
         return {id(k): v for k, v in self.working_to_master.items()}
 

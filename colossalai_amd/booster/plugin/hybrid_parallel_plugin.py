@@ -159,6 +159,39 @@ class HybridParallelFP32Optimizer(OptimizerWrapper):
                     g.mul_(scale)
         return self.optim.step(*args, **kwargs)
 
+    # ------------------------------------------------------------ checkpoint
+    def get_param_states(self, names):
+        out = {}
+        for gi, group in enumerate(self.optim.param_groups):
+            for p in group["params"]:
+                name = names.get(id(p))
+                if name is None:
+                    continue
+                st = {"_group": gi}
+                for k, v in self.optim.state.get(p, {}).items():
+                    st[k] = v.cpu().clone() if isinstance(v, torch.Tensor) else v
+                out[name] = st
+        return out
+
+    def set_param_states(self, states, names):
+        for group in self.optim.param_groups:
+            for p in group["params"]:
+                name = names.get(id(p))
+                if name is None or name not in states:
+                    continue
+                st = dict(states[name])
+                st.pop("_group", None)
+                st.pop("master", None)
+                inner = {}
+                for k, v in st.items():
+                    if isinstance(v, torch.Tensor) and v.shape == p.shape:
+                        inner[k] = v.to(p.device)
+                    elif k == "step" and not isinstance(v, torch.Tensor):
+                        inner[k] = torch.tensor(float(v))
+                    else:
+                        inner[k] = v
+                self.optim.state[p] = inner
+
     def backward(self, loss, inputs=None, retain_graph=False, **kwargs):
         loss.backward(inputs=inputs, retain_graph=retain_graph, **kwargs)
         if self.model_wrapper.require_grad_sync:
@@ -383,6 +416,14 @@ class HybridParallelPlugin(Plugin):
                     **self.zero_kwargs,
                     **({} if self.precision != "fp16" else self.amp_kwargs),
                 )
+        if optimizer is not None and isinstance(optimizer, OptimizerWrapper):
+            # topology-independent optimizer checkpoints need param names +
+            # TP shard metadata (checkpoint_io/hybrid_parallel_checkpoint_io.py)
+            from ...checkpoint_io.param_meta import build_tp_shard_map
+
+            optimizer.ckpt_param_names = {id(p): n for n, p in model.module.named_parameters()}
+            optimizer.ckpt_local_params = dict(model.module.named_parameters())
+            optimizer.ckpt_tp_map = build_tp_shard_map(model.module)
         return model, optimizer, criterion, dataloader, lr_scheduler
 
     def _assign_pipeline_stage(self, model: nn.Module) -> None:

@@ -168,18 +168,170 @@ class HybridParallelCheckpointIO(GeneralCheckpointIO):
         if dist.is_initialized():
             dist.barrier()
 
+    # ------------------------------------------------------------- optimizer
+    #
+    # Topology-independent format (reference:
+    # hybrid_parallel_checkpoint_io.py:469 save_sharded_optimizer,
+    # :1017 gather_from_sharded_optimizer_state,
+    # :1082 shard_from_complete_optimizer_state):
+    #   state:  {param_name: {"master": fp32 full, "exp_avg": ..., "step": int,
+    #                         "_group": int}}  — tensors at FULL (unsharded)
+    #           shapes, so any (dp, tp, pp) topology can reload them.
+    #   param_groups: hyperparams per group (no "params").
+    # Save gathers ZeRO shards over dp (inside get_param_states) and TP
+    # shards over tp (via the plugin-attached tp shard map); pp stages hold
+    # disjoint params and are merged at the file level.
+
+    def _gather_global_states(self, optimizer) -> dict:
+        """name -> full-shape state for THIS pp stage (collective on dp+tp)."""
+        names = getattr(optimizer, "ckpt_param_names", None)
+        tp_map = getattr(optimizer, "ckpt_tp_map", None)
+        assert names is not None and hasattr(optimizer, "get_param_states"), (
+            "optimizer was not boosted by HybridParallelPlugin (no checkpoint metadata)"
+        )
+        states = optimizer.get_param_states(names)
+        local_params = getattr(optimizer, "ckpt_local_params", {})
+        for name, st in states.items():
+            info = tp_map.get(name) if tp_map else None
+            if info is None:
+                continue
+            local_shape = local_params[name].shape if name in local_params else None
+            for k, v in list(st.items()):
+                if isinstance(v, torch.Tensor) and local_shape is not None and v.shape == local_shape:
+                    st[k] = info.gather(v.to(local_params[name].device)).cpu()
+        return states
+
+    def _scatter_global_states(self, optimizer, states: dict) -> None:
+        """Slice full-shape states to the current topology and load them."""
+        names = getattr(optimizer, "ckpt_param_names", None)
+        tp_map = getattr(optimizer, "ckpt_tp_map", None)
+        local_params = getattr(optimizer, "ckpt_local_params", {})
+        assert names is not None and hasattr(optimizer, "set_param_states")
+        tp_world = dist.get_world_size(self.tp_group) if self.tp_group is not None else 1
+        tp_rank = self.tp_rank
+        sliced = {}
+        for name, st in states.items():
+            if name not in local_params:
+                continue  # another pp stage's param
+            p = local_params[name]
+            info = tp_map.get(name) if tp_map else None
+            new_st = {}
+            for k, v in st.items():
+                if isinstance(v, torch.Tensor) and info is not None and v.dim() == p.dim() and v.shape != p.shape:
+                    new_st[k] = info.shard(v, tp_world, tp_rank)
+                else:
+                    new_st[k] = v
+            sliced[name] = new_st
+        optimizer.set_param_states(sliced, names)
+
+    def _hyper_param_groups(self, optimizer) -> list:
+        inner = optimizer.optim if hasattr(optimizer, "optim") else optimizer
+        return [{k: v for k, v in g.items() if k != "params"} for g in inner.param_groups]
+
+    def _load_hyper_param_groups(self, optimizer, groups: list) -> None:
+        inner = optimizer.optim if hasattr(optimizer, "optim") else optimizer
+        if len(groups) != len(inner.param_groups):
+            return  # group structure changed; keep current hyperparams
+        for g, gs in zip(inner.param_groups, groups):
+            g.update({k: v for k, v in gs.items() if k != "params"})
+
     def save_unsharded_optimizer(self, optimizer, checkpoint: str, gather_dtensor: bool = True, use_async: bool = False):
-        # rank-local shard states (ZeRO/hybrid): one file per rank
-        state = optimizer.state_dict()
-        path = f"{checkpoint}.rank{self.global_rank}" if dist.get_world_size() > 1 else checkpoint
-        torch.save(state, path)
+        states = self._gather_global_states(optimizer)
+        if self.pp_size > 1:
+            # merge pp stages (disjoint name sets) onto pp rank 0
+            gathered = [None] * self.pp_size
+            dist.all_gather_object(gathered, states, group=self.pp_group)
+            merged = {}
+            for sd in gathered:
+                merged.update(sd)
+            states = merged
+        if self.global_rank == 0:
+            torch.save({"state": states, "param_groups": self._hyper_param_groups(optimizer)}, checkpoint)
         dist.barrier()
 
     def load_unsharded_optimizer(self, optimizer, checkpoint: str):
         import os
 
-        path = f"{checkpoint}.rank{self.global_rank}" if dist.get_world_size() > 1 else checkpoint
-        if not os.path.exists(path):
-            path = checkpoint
-        optimizer.load_state_dict(torch.load(path, weights_only=False))
+        ckpt = torch.load(checkpoint, weights_only=False)
+        if "state" not in ckpt or not isinstance(ckpt.get("state"), dict):
+            raise RuntimeError(f"{checkpoint} is not a hybrid optimizer checkpoint")
+        # legacy rank-file checkpoints are refused loudly rather than mis-loaded
+        first = next(iter(ckpt["state"].values()), None)
+        if first is not None and not isinstance(first, dict):
+            raise RuntimeError("unrecognized optimizer checkpoint format")
+        self._scatter_global_states(optimizer, ckpt["state"])
+        self._load_hyper_param_groups(optimizer, ckpt.get("param_groups", []))
+        if dist.is_initialized():
+            dist.barrier()
+
+    def save_sharded_optimizer(self, optimizer, checkpoint: str, gather_dtensor: bool = False,
+                               prefix: str = None, size_per_shard: int = 1024, use_async: bool = False):
+        """Directory format: per-pp-stage shard files + one merged index —
+        no whole-model gather on any single rank."""
+        import os
+
+        from .index_file import CheckpointIndexFile
+        from .utils import get_optimizer_base_filenames
+
+        os.makedirs(checkpoint, exist_ok=True)
+        states = self._gather_global_states(optimizer)
+        states_name, group_name, save_index_file = get_optimizer_base_filenames(prefix)
+        root, ext = os.path.splitext(states_name)
+
+        weight_map = {}
+        if self.dp_rank == 0 and self.tp_rank == 0:
+            # chunk this stage's states by size
+            budget = size_per_shard * 1024 * 1024
+            blocks, cur, cur_size = [], {}, 0
+            for name, st in states.items():
+                sz = sum(v.numel() * v.element_size() for v in st.values() if isinstance(v, torch.Tensor))
+                if cur and cur_size + sz > budget:
+                    blocks.append(cur)
+                    cur, cur_size = {}, 0
+                cur[name] = st
+                cur_size += sz
+            if cur:
+                blocks.append(cur)
+            for idx, block in enumerate(blocks):
+                shard_file = f"{root}-stage{self.pp_rank:02d}-{idx + 1:05d}{ext}"
+                torch.save(block, os.path.join(checkpoint, shard_file))
+                for name in block:
+                    weight_map[name] = shard_file
+        if self.pp_size > 1:
+            maps = [None] * self.pp_size
+            dist.all_gather_object(maps, weight_map, group=self.pp_group)
+            merged = {}
+            for m in maps:
+                merged.update(m)
+            weight_map = merged
+        if self.global_rank == 0:
+            torch.save(self._hyper_param_groups(optimizer), os.path.join(checkpoint, group_name))
+            index_file = CheckpointIndexFile(checkpoint)
+            index_file.append_meta_data("param_groups", group_name)
+            for name, f in weight_map.items():
+                index_file.append_weight_map(name, f)
+            index_file.write_index_file(save_index_file)
         dist.barrier()
+
+    def load_sharded_optimizer(self, optimizer, index_file_path: str):
+        from pathlib import Path
+
+        from .index_file import CheckpointIndexFile
+
+        index = CheckpointIndexFile.from_file(str(index_file_path))
+        root = Path(index_file_path).parent
+        local_params = getattr(optimizer, "ckpt_local_params", {})
+        # only read the shard files containing params this rank holds
+        needed = set()
+        for name, f in index.weight_map.items():
+            if name in local_params:
+                needed.add(f)
+        states = {}
+        for shard_file in sorted(needed):
+            states.update(torch.load(str(root / shard_file), weights_only=False))
+        self._scatter_global_states(optimizer, states)
+        group_file = index.metadata.get("param_groups")
+        if group_file is not None:
+            self._load_hyper_param_groups(optimizer, torch.load(str(root / group_file), weights_only=False))
+        if dist.is_initialized():
+            dist.barrier()

@@ -1,21 +1,30 @@
 // Flash attention forward + backward for gfx950 (CDNA4), bf16, causal, GQA.
 //
 // MI355X-native design (not a CUDA port):
-//  - mfma_f32_32x32x16_bf16 tiles; 64-wide waves; 8 waves per workgroup.
-//  - forward: each wave owns 32 q-rows (block tile = 256 q-rows); K/V tiles
-//    of 64 rows are staged in LDS (K row-major XOR-swizzled, V transposed)
-//    and shared by all waves. QK^T is computed OPERAND-SWAPPED
-//    (S^T = K·Q^T) so the online-softmax row reduction is lane-local:
-//    the MFMA C-layout puts one q-row on lanes (r, r+32) — row max/sum is
-//    an in-register reduce + one shfl_xor(32) (guide §B "fused attention").
-//  - P is restaged per-wave through a 4 KB swizzled LDS tile to convert the
-//    C-layout into the A-fragment layout for the P·V MFMAs.
-//  - all LDS tiles carry the ((row&7)<<4) byte-XOR swizzle: a row-major
-//    [*][128] bf16 tile read column-wise is otherwise a 32-way bank
-//    conflict, measured as ~50% of attention kernel time (guide §6 G4).
-//  - backward is FA2-style: delta = rowsum(dO⊙O) preprocess; one kernel
-//    recomputes P^T per kv-tile and accumulates dK/dV (GQA group summed
-//    in-register); a second kernel recomputes P per q-tile for dQ.
+//  - mfma_f32_32x32x16_bf16 tiles; 64-wide waves; 8 waves per workgroup
+//    forward, 4-wave blocks backward.
+//  - ALL operand tiles live in LDS in ONE layout: "subtiled row-major"
+//    [D/16 col-groups][ROWS][16 elems] with 32 B row pitch (+16 B pad per
+//    group to stagger banks). Row-slice fragment reads (8 consecutive d at
+//    fixed row) are single 16 B vector loads; COLUMN fragment reads (the
+//    transposed consumption that previously needed separate [D][ROWS]
+//    tiles built with 64 scalar ds_write_b16 per thread per tile) use
+//    gfx950's ds_read_b64_tr_b16 hardware weave (guide T10).
+//    HW-verified semantics (tr16_probe, mfma_selftest.hip): each lane reads
+//    64 bits at its own address; within a 16-lane group, out[l][j] =
+//    pool[(l&15) + 16 j] where pool is the group's reads in lane order.
+//    With lane addressing row=(m>>2), byte=(m&3)*8 this delivers
+//    tile[row0+j][col m] per lane — a free 4x16 transpose with any row pitch.
+//  - forward: each wave owns 32 q rows; K/V tiles of 64 staged subtiled;
+//    QK^T operand-swapped (S^T = K.Q^T) so softmax is lane-local; defer-max
+//    rescale (T13); async-split staging (T14): next tile's global loads
+//    issue before this tile's MFMAs.
+//  - backward FA2-split: delta preprocess; dK/dV kernel (per kv-tile, GQA
+//    group in registers); dQ kernel (per q-tile). P/dS restage between the
+//    MFMA C-layout and A-fragment layout goes through a per-wave 2 KB
+//    subtiled tile: packed 8 B stores (4 consecutive rows of the C-layout
+//    share a lane) + tr-read A-fragments. Score tiles are processed in
+//    32-wide halves, halving live p/dp registers.
 //
 // Shapes: q [B,S,Hq,D], k/v [B,S,Hkv,D] ("bshd" — no transposes in the
 // model), D in {64,128}. lse/delta are [B,Hq,S] fp32.
@@ -30,6 +39,7 @@
 namespace cai {
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4v_t __attribute__((ext_vector_type(4)));
 
 namespace fa {
 
@@ -37,10 +47,8 @@ constexpr int KVB = 64;     // kv rows per tile
 constexpr int QW = 32;      // q rows per wave (fwd / dq); kv rows per wave (dkdv)
 constexpr int NW = 8;       // waves per block (forward)
 constexpr int NT = 512;     // threads per block (forward)
-// Backward kernels hold dK+dV (or dQ) accumulators plus operand fragments in
-// registers (~210 VGPR measured): 4-wave blocks keep the allocator
-// unconstrained (8-wave blocks force a 256-VGPR cap -> spills, measured 21%
-// slower end-to-end).
+// Backward kernels carry dK+dV (or dQ) accumulators plus K/V (or Q/dO)
+// fragments in registers (~380-450 VGPR+AGPR): 1 wave/SIMD — 4-wave blocks.
 constexpr int NWB = 4;
 constexpr int NTB = 256;
 
@@ -71,75 +79,42 @@ DEV_INLINE __bf16 f2b(float f) {
 // C-layout row for mfma_f32_32x32x16 register j (guide §3, HW-verified).
 DEV_INLINE constexpr int crow(int j, int half) { return (j & 3) + 8 * (j >> 2) + 4 * half; }
 
-// Swizzled byte offset inside a row-major LDS tile with ROWB bytes per row.
-// XOR spreads a column access across ALL of the row's 16-byte slots
-// (ROWB/16 of them) — guide §6 G4; PMC-verified: an 8-slot mask on 256 B
-// rows left 4-way read / 8-way write conflicts.
-template <int ROWB>
-DEV_INLINE int swz(int row, int byte_in_row) {
-  constexpr int CPR_MASK = (ROWB / 16) - 1;
-  return row * ROWB + (byte_in_row ^ ((row & CPR_MASK) << 4));
+// ---- subtiled row-major LDS tiles --------------------------------------
+// [COLS/16 groups][ROWS][16 bf16], 32 B row pitch, 16 B pad between groups
+// (group base bank shift of 4 keeps multi-group accesses staggered).
+template <int ROWS>
+__host__ __device__ constexpr int sub_pitch() { return ROWS * 32 + 16; }
+
+template <int ROWS>
+DEV_INLINE int sub_off(int row, int col) {
+  return (col >> 4) * sub_pitch<ROWS>() + row * 32 + (col & 15) * 2;
 }
 
-// Swizzle for TRANSPOSED tiles ([D][rows]): the mask folds d with d>>3 so it
-// varies across lanes both when STORING (lanes sweep ch = d>>3 at fixed
-// d&7) and when READING B-fragments (lanes sweep d&7 at fixed d>>3) —
-// a plain (d&7) mask degenerates to a 16-way conflict on the store side
-// (PMC: conflict cycles ~3x busy). No per-lane data rotation → no
-// runtime-indexed vector access → no scratch (guide rule #20).
-template <int ROWB>
-DEV_INLINE int swzT(int d, int byte_in_row) {
-  return d * ROWB + (byte_in_row ^ (((d ^ (d >> 3)) & 7) << 4));
+template <int ROWS, int COLS>
+__host__ __device__ constexpr int sub_bytes() { return (COLS / 16) * sub_pitch<ROWS>(); }
+
+// ds_read_b64_tr_b16: per-lane 64-bit read + 16-lane weave (see header).
+DEV_INLINE bf16x4v_t ld_tr16(const char* lds, int byte) {
+  typedef __attribute__((address_space(3))) bf16x4v_t* lds_v4p;
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_v4p)(lds + byte));
 }
 
-// ---- cooperative staging helpers (512 threads) ----------------------------
-
-// Row-major [ROWS][D] bf16 tile, swizzled. Source rows clamped to < S.
-template <int D, int ROWS>
-DEV_INLINE void stage_rowmajor(char* lds, const unsigned short* base, long tok_stride, int row0, int S) {
-  constexpr int CPR = D / 8;  // 16B chunks per row
-  constexpr int TOTAL = ROWS * CPR;
-  for (int c = threadIdx.x; c < TOTAL; c += blockDim.x) {
-    const int row = c / CPR, ch = c % CPR;
-    const int grow = min(row0 + row, S - 1);
-    bf16x8_t v = ld_g16(base + (long)grow * tok_stride + ch * 8);
-    st_lds16(lds, swz<D * 2>(row, ch * 16), v);
-  }
-}
-
-// Transposed [D][ROWS] bf16 tile, swizzled: ldsT[d][row] = src[row][d].
-template <int D, int ROWS>
-DEV_INLINE void stage_transposed(char* ldsT, const unsigned short* base, long tok_stride, int row0, int S) {
-  constexpr int CPR = D / 8;
-  constexpr int TOTAL = ROWS * CPR;
-  for (int c = threadIdx.x; c < TOTAL; c += blockDim.x) {
-    const int row = c / CPR, ch = c % CPR;
-    const int grow = min(row0 + row, S - 1);
-    bf16x8_t v = ld_g16(base + (long)grow * tok_stride + ch * 8);
-#pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      const int d = ch * 8 + e;
-      *reinterpret_cast<unsigned short*>(ldsT + swzT<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
-    }
-  }
-}
-
-// Both layouts from one global read (used for Q/dO in the dkdv kernel).
-template <int D, int ROWS>
-DEV_INLINE void stage_both(char* lds, char* ldsT, const unsigned short* base, long tok_stride, int row0, int S) {
-  constexpr int CPR = D / 8;
-  constexpr int TOTAL = ROWS * CPR;
-  for (int c = threadIdx.x; c < TOTAL; c += blockDim.x) {
-    const int row = c / CPR, ch = c % CPR;
-    const int grow = min(row0 + row, S - 1);
-    bf16x8_t v = ld_g16(base + (long)grow * tok_stride + ch * 8);
-    st_lds16(lds, swz<D * 2>(row, ch * 16), v);
-#pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      const int d = ch * 8 + e;
-      *reinterpret_cast<unsigned short*>(ldsT + swzT<ROWS * 2>(d, row * 2)) = bf_raw(v[e]);
-    }
-  }
+// Full 8-element column fragment from a subtiled tile: lane l receives
+// tile[k_base + (l>>5)*8 + e][cg_base*16 + (l&31)] for e = 0..7 — exactly
+// the A/B fragment of mfma_32x32x16 with the tile's ROW axis as the MFMA
+// k-dim and its COLUMN axis as the lane-owned m/n dim.
+template <int ROWS>
+DEV_INLINE bf16x8_t ld_frag_tr(const char* tile, int lane, int k_base, int cg_base) {
+  const int m = lane & 15;
+  const int cg = cg_base + ((lane >> 4) & 1);
+  const int row = k_base + ((lane >> 5) * 8) + (m >> 2);
+  const int base = cg * sub_pitch<ROWS>() + (m & 3) * 8;
+  bf16x4v_t lo = ld_tr16(tile, base + row * 32);
+  bf16x4v_t hi = ld_tr16(tile, base + (row + 4) * 32);
+  bf16x8_t r;
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+  r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+  return r;
 }
 
 // Register-staged tile for async-split staging (guide T14 / G15): global
@@ -162,31 +137,29 @@ struct RegStage {
       r[i] = ld_g16(base + (long)grow * tok_stride + ch * 8);
     }
   }
-  DEV_INLINE void store_rowmajor(char* lds) const {
+  DEV_INLINE void store_subtiled(char* lds) const {
 #pragma unroll
     for (int i = 0; i < NCH; ++i) {
       const int c = (int)threadIdx.x + i * NT_;
-      st_lds16(lds, swz<D * 2>(c / CPR, (c % CPR) * 16), r[i]);
-    }
-  }
-  DEV_INLINE void store_transposed(char* ldsT) const {
-#pragma unroll
-    for (int i = 0; i < NCH; ++i) {
-      const int c = (int)threadIdx.x + i * NT_;
-      const int row = c / CPR, ch = c % CPR;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        *reinterpret_cast<unsigned short*>(ldsT + swzT<ROWS * 2>(ch * 8 + e, row * 2)) = bf_raw(r[i][e]);
-      }
+      st_lds16(lds, sub_off<ROWS>(c / CPR, (c % CPR) * 8), r[i]);
     }
   }
 };
+
+// Swizzled byte offset inside a row-major LDS tile with ROWB bytes per row
+// (forward P tile only). XOR spreads a column access across the row's
+// 16-byte slots — guide §6 G4.
+template <int ROWB>
+DEV_INLINE int swz(int row, int byte_in_row) {
+  constexpr int CPR_MASK = (ROWB / 16) - 1;
+  return row * ROWB + (byte_in_row ^ ((row & CPR_MASK) << 4));
+}
 
 }  // namespace fa
 
 // ============================================================ forward kernel
 //
-// grid: (ceil(S/256), B*Hq). LDS: K[KVB][D] + V_T[D][KVB] + P[NW][32][KVB].
+// grid: (ceil(S/256), B*Hq). LDS: K_sub + V_sub + P[NW][32][KVB].
 template <int D>
 __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     const unsigned short* __restrict__ Q,
@@ -198,11 +171,10 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     int B, int S, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
   extern __shared__ char smem[];
-  constexpr int KB_BYTES = KVB * D * 2;
-  constexpr int VT_BYTES = D * KVB * 2;
+  constexpr int KB_BYTES = sub_bytes<KVB, D>();
   char* Klds = smem;
-  char* VT = smem + KB_BYTES;
-  char* Pw = smem + KB_BYTES + VT_BYTES;  // + wave*QW*KVB*2
+  char* Vlds = smem + KB_BYTES;
+  char* Pw = smem + 2 * KB_BYTES;  // + wave*QW*KVB*2
 
   const int lane = threadIdx.x & 63;
   const int w = threadIdx.x >> 6;
@@ -248,8 +220,8 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   RegStage<D, KVB, NT> kstage, vstage;
   kstage.load(k_base, kts, 0, S);
   vstage.load(v_base, kts, 0, S);
-  kstage.store_rowmajor(Klds);
-  vstage.store_transposed(VT);
+  kstage.store_subtiled(Klds);
+  vstage.store_subtiled(Vlds);
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
@@ -273,7 +245,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
 #pragma unroll
       for (int sl = 0; sl < DSL; ++sl) {
         // A-operand: K rows. lane: row = kb*32+ln, cols d = sl*16 + half*8 + [0..7]
-        bf16x8_t kf = ld_lds16(Klds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        bf16x8_t kf = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
         __builtin_amdgcn_s_setprio(1);
         acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[sl], acc, 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
@@ -340,15 +312,15 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
       }
     }
 
-    // ---- O += P · V  (A = P from LDS, B = V_T reads)
+    // ---- O += P · V  (A = P from LDS, B = V column-fragments via tr-read)
 #pragma unroll
     for (int ks = 0; ks < KVB / 16; ++ks) {
       // A: lane row = q (ln), k-cols = ks*16 + half*8 + [0..7]
       bf16x8_t pa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
-        // B: lane col = d (nb*32+ln), k-rows = ks*16 + half*8 + [0..7] → V_T[d][k]
-        bf16x8_t vb = ld_lds16(VT, swzT<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
+        // B: lane col = d (nb*32+ln), k-rows kv = ks*16 + half*8 + [0..7]
+        bf16x8_t vb = ld_frag_tr<KVB>(Vlds, lane, ks * 16, nb * 2);
         __builtin_amdgcn_s_setprio(1);
         oacc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, oacc[nb], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
@@ -356,10 +328,10 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     }
     }  // active
 
-    __syncthreads();  // all waves done reading Klds/VT for tile t
+    __syncthreads();  // all waves done reading Klds/Vlds for tile t
     if (has_next) {
-      kstage.store_rowmajor(Klds);
-      vstage.store_transposed(VT);
+      kstage.store_subtiled(Klds);
+      vstage.store_subtiled(Vlds);
     }
     __syncthreads();
   }
@@ -432,9 +404,10 @@ __global__ __launch_bounds__(256) void fa_delta_kernel(
 
 // ================================================= backward dK/dV kernel
 //
-// grid: (ceil(S/256) kv-tiles, B*Hkv). Each wave owns 32 kv rows; loops over
-// the GQA group's q-heads and that head's q-tiles (64 q rows staged in LDS
-// in both layouts). dK/dV accumulate in registers across the whole loop.
+// grid: (ceil(S/128) kv-tiles, B*Hkv). Each wave owns 32 kv rows; loops over
+// the GQA group's q-heads and that head's q-tiles (64 q rows staged subtiled
+// in LDS). dK/dV accumulate in registers across the whole loop. Score tiles
+// are processed in 32-q halves through a per-wave 2 KB P tile.
 template <int D>
 __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     const unsigned short* __restrict__ Q,
@@ -450,13 +423,12 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
   using namespace fa;
   constexpr int QB = 64;  // q rows per staged tile
   extern __shared__ char smem[];
-  constexpr int TB = QB * D * 2;  // 16 KB (D=128)
+  constexpr int TB = sub_bytes<QB, D>();
+  constexpr int PB = sub_bytes<32, 32>();  // per-wave P tile: [q 32][kv 32]
   char* Qlds = smem;
-  char* QT = smem + TB;
-  char* dOlds = smem + 2 * TB;
-  char* dOT = smem + 3 * TB;
-  char* Pw = smem + 4 * TB;                      // NWB * 32*QB*2
-  float* lse_lds = reinterpret_cast<float*>(smem + 4 * TB + NWB * (QW * QB * 2));
+  char* dOlds = smem + TB;
+  char* Pw = smem + 2 * TB;  // + wave*PB
+  float* lse_lds = reinterpret_cast<float*>(smem + 2 * TB + NWB * PB);
   float* dta_lds = lse_lds + QB;
 
   const int lane = threadIdx.x & 63;
@@ -474,7 +446,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
   const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
 
   constexpr int DSL = D / 16;
-  char* P = Pw + w * (QW * QB * 2);
+  char* P = Pw + w * PB;
 
   // ---- K, V fragments (A-operands): lane row = kv (ln), d = sl*16+half*8+[0..7]
   const int kv_my = min(kvw + ln, S - 1);
@@ -514,10 +486,8 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     qstage.load(q_base, qts, qt_first * QB, S);
     dostage.load(do_base, do_stride, qt_first * QB, S);
     __syncthreads();  // previous head's compute done before overwriting LDS
-    qstage.store_rowmajor(Qlds);
-    qstage.store_transposed(QT);
-    dostage.store_rowmajor(dOlds);
-    dostage.store_transposed(dOT);
+    qstage.store_subtiled(Qlds);
+    dostage.store_subtiled(dOlds);
     if (threadIdx.x < QB) {
       const int qr = min(qt_first * QB + (int)threadIdx.x, S - 1);
       lse_lds[threadIdx.x] = lse_base[qr];
@@ -535,107 +505,105 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
       const bool active = !(causal && qt0 + QB - 1 < kvw) && (kvw < S);
       if (active) {
 
-      // ---- S'^T[kv][q] = K · Q^T : C col = q (ln within qb), row = kv.
-      float p[2][16], dp[2][16];
+      // 32-q halves: live score state is p[16]+dp[16] instead of [2][16].
 #pragma unroll
       for (int qb = 0; qb < 2; ++qb) {
+        // ---- S'^T[kv][q] = K · Q^T : C col = q (ln within half), row = kv.
         f32x16 acc;
 #pragma unroll
         for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
-          // B: lane col = q (qb*32+ln), rows d → Qlds[q][d]
-          bf16x8_t qb_frag = ld_lds16(Qlds, swz<D * 2>(qb * 32 + ln, (sl * 16 + half * 8) * 2));
+          // B: lane col = q (qb*32+ln), k-rows d → row-slice read
+          bf16x8_t qb_frag = ld_lds16(Qlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
           __builtin_amdgcn_s_setprio(1);
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[sl], qb_frag, acc, 0, 0, 0);
           __builtin_amdgcn_s_setprio(0);
         }
-#pragma unroll
-        for (int j = 0; j < 16; ++j) p[qb][j] = acc[j];
-      }
 
-      // ---- P' = exp(scale*s - lse[q]), masked
+        // ---- P' = exp(scale*s - lse[q]), masked
+        float p[16];
+        {
+          const int q_abs = qt0 + qb * 32 + ln;
+          const float lse = lse_lds[qb * 32 + ln];
 #pragma unroll
-      for (int qb = 0; qb < 2; ++qb) {
-        const int q_abs = qt0 + qb * 32 + ln;
-        const float lse = lse_lds[qb * 32 + ln];
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int k_abs = kvw + crow(j, half);
-          const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
-          p[qb][j] = valid ? __expf(p[qb][j] * scale - lse) : 0.0f;
-        }
-      }
-
-      // ---- stage P' to LDS: P[kv local][q] bf16 swizzled (scalar stores)
-#pragma unroll
-      for (int qb = 0; qb < 2; ++qb)
-#pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int r = crow(j, half);
-          *reinterpret_cast<unsigned short*>(P + swz<QB * 2>(r, (qb * 32 + ln) * 2)) = f2bf(p[qb][j]);
+          for (int j = 0; j < 16; ++j) {
+            const int k_abs = kvw + crow(j, half);
+            const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+            p[j] = valid ? __expf(acc[j] * scale - lse) : 0.0f;
+          }
         }
 
-      // ---- dV += P' · dO   (A = P' from LDS, B = dO_T reads)
+        // ---- stage P' into the per-wave subtiled [q 32][kv 32] tile:
+        // C-regs j in a 4-group are 4 consecutive kv rows → packed 8 B store
 #pragma unroll
-      for (int ks = 0; ks < QB / 16; ++ks) {
-        bf16x8_t pa = ld_lds16(P, swz<QB * 2>(ln, (ks * 16 + half * 8) * 2));
+        for (int jj = 0; jj < 4; ++jj) {
+          const int kv_loc = 8 * jj + 4 * half;
+          short4v pk;
 #pragma unroll
-        for (int nb = 0; nb < D / 32; ++nb) {
-          bf16x8_t dob = ld_lds16(dOT, swzT<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
-          __builtin_amdgcn_s_setprio(1);
-          dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[nb], 0, 0, 0);
-          __builtin_amdgcn_s_setprio(0);
+          for (int e = 0; e < 4; ++e) pk[e] = (short)f2bf(p[jj * 4 + e]);
+          *reinterpret_cast<short4v*>(P + sub_off<32>(ln, kv_loc)) = pk;
         }
-      }
 
-      // ---- dP'[kv][q] = V · dO^T  (A = V regs, B = dOlds[q][d] reads)
+        // ---- dV += P' · dO  (A = P' via tr-read: lane row = kv, k = q-local;
+        //      B = dO column-fragments, k = q-global)
 #pragma unroll
-      for (int qb = 0; qb < 2; ++qb) {
-        f32x16 acc;
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          bf16x8_t pa = ld_frag_tr<32>(P, lane, ks2 * 16, 0);
 #pragma unroll
-        for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+          for (int nb = 0; nb < D / 32; ++nb) {
+            bf16x8_t dob = ld_frag_tr<QB>(dOlds, lane, qb * 32 + ks2 * 16, nb * 2);
+            __builtin_amdgcn_s_setprio(1);
+            dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[nb], 0, 0, 0);
+            __builtin_amdgcn_s_setprio(0);
+          }
+        }
+
+        // ---- dP'[kv][q] = V · dO^T  (A = V regs, B = dO row-slice reads)
+        f32x16 acc2;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) acc2[j] = 0.0f;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
-          bf16x8_t dob = ld_lds16(dOlds, swz<D * 2>(qb * 32 + ln, (sl * 16 + half * 8) * 2));
-          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[sl], dob, acc, 0, 0, 0);
+          bf16x8_t dob = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
+          acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[sl], dob, acc2, 0, 0, 0);
         }
-#pragma unroll
-        for (int j = 0; j < 16; ++j) dp[qb][j] = acc[j];
-      }
 
-      // ---- dS' = scale * P' ⊙ (dP' - delta[q]) → overwrite P LDS
+        // ---- dS' = scale * P' ⊙ (dP' - delta[q]) → overwrite P tile
+        {
+          const float dta = dta_lds[qb * 32 + ln];
 #pragma unroll
-      for (int qb = 0; qb < 2; ++qb) {
-        const float dta = dta_lds[qb * 32 + ln];
+          for (int jj = 0; jj < 4; ++jj) {
+            const int kv_loc = 8 * jj + 4 * half;
+            short4v dk4;
 #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int r = crow(j, half);
-          const float ds = scale * p[qb][j] * (dp[qb][j] - dta);
-          *reinterpret_cast<unsigned short*>(P + swz<QB * 2>(r, (qb * 32 + ln) * 2)) = f2bf(ds);
+            for (int e = 0; e < 4; ++e) {
+              const float ds = scale * p[jj * 4 + e] * (acc2[jj * 4 + e] - dta);
+              dk4[e] = (short)f2bf(ds);
+            }
+            *reinterpret_cast<short4v*>(P + sub_off<32>(ln, kv_loc)) = dk4;
+          }
         }
-      }
 
-      // ---- dK += dS' · Q   (A = dS' from LDS, B = Q_T reads)
+        // ---- dK += dS' · Q   (A = dS' tr-read, B = Q column-fragments)
 #pragma unroll
-      for (int ks = 0; ks < QB / 16; ++ks) {
-        bf16x8_t dsa = ld_lds16(P, swz<QB * 2>(ln, (ks * 16 + half * 8) * 2));
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          bf16x8_t dsa = ld_frag_tr<32>(P, lane, ks2 * 16, 0);
 #pragma unroll
-        for (int nb = 0; nb < D / 32; ++nb) {
-          bf16x8_t qtb = ld_lds16(QT, swzT<QB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
-          __builtin_amdgcn_s_setprio(1);
-          dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[nb], 0, 0, 0);
-          __builtin_amdgcn_s_setprio(0);
+          for (int nb = 0; nb < D / 32; ++nb) {
+            bf16x8_t qtb = ld_frag_tr<QB>(Qlds, lane, qb * 32 + ks2 * 16, nb * 2);
+            __builtin_amdgcn_s_setprio(1);
+            dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[nb], 0, 0, 0);
+            __builtin_amdgcn_s_setprio(0);
+          }
         }
-      }
+      }  // qb halves
       }  // active
 
       __syncthreads();  // all waves done with this q-tile's LDS
       if (has_next) {
-        qstage.store_rowmajor(Qlds);
-        qstage.store_transposed(QT);
-        dostage.store_rowmajor(dOlds);
-        dostage.store_transposed(dOT);
+        qstage.store_subtiled(Qlds);
+        dostage.store_subtiled(dOlds);
         if (threadIdx.x < QB) {
           const int qr = min(qt0 + QB + (int)threadIdx.x, S - 1);
           lse_lds[threadIdx.x] = lse_base[qr];
@@ -666,10 +634,13 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
 
 // ===================================================== backward dQ kernel
 //
-// grid: (ceil(S/256) q-tiles, B*Hq). Each wave owns 32 q rows; kv tiles of
-// 64 staged in LDS (K both layouts, V row-major).
+// grid: (ceil(S/128) q-tiles, B*Hq). Each wave owns 32 q rows; kv tiles of
+// 64 staged subtiled in LDS (K serves both row-slice and column fragments,
+// V row-slice only). dS restages through a per-wave [kv 32][q 32] tile.
+// __launch_bounds__ min-waves/SIMD = 2 caps the allocation at 256 registers:
+// with the tr-read design the kernel fits (vs 441 before), doubling occupancy.
 template <int D>
-__global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
+__global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     const unsigned short* __restrict__ Q,
     const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V,
@@ -681,13 +652,12 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
     int B, int S, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
   extern __shared__ char smem[];
-  constexpr int KB_BYTES = KVB * D * 2;
-  constexpr int KT_BYTES = D * KVB * 2;
+  constexpr int KB_BYTES = sub_bytes<KVB, D>();
+  constexpr int PB = sub_bytes<32, 32>();  // per-wave dS tile: [kv 32][q 32]
   char* Klds = smem;
-  char* KT = smem + KB_BYTES;
-  char* Vlds = smem + KB_BYTES + KT_BYTES;
-  char* Pw = smem + 2 * KB_BYTES + KT_BYTES;  // NWB * 32*KVB*2
-  float* lse_lds = reinterpret_cast<float*>(Pw + NWB * (QW * KVB * 2));
+  char* Vlds = smem + KB_BYTES;
+  char* Pw = smem + 2 * KB_BYTES;  // + wave*PB
+  float* lse_lds = reinterpret_cast<float*>(smem + 2 * KB_BYTES + NWB * PB);
   float* dta_lds = lse_lds + NWB * QW;
 
   const int lane = threadIdx.x & 63;
@@ -708,7 +678,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
   const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
 
   constexpr int DSL = D / 16;
-  char* P = Pw + w * (QW * KVB * 2);
+  char* P = Pw + w * PB;
 
   // lse/delta for the block's q rows
   {
@@ -746,9 +716,8 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
   RegStage<D, KVB, NTB> kstage, vstage;
   kstage.load(k_base, kts, 0, S);
   vstage.load(v_base, kts, 0, S);
-  kstage.store_rowmajor(Klds);
-  kstage.store_transposed(KT);
-  vstage.store_rowmajor(Vlds);
+  kstage.store_subtiled(Klds);
+  vstage.store_subtiled(Vlds);
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
@@ -761,63 +730,65 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dq_kernel(
     const bool active = !(causal && k0 > qw + QW - 1) && (qw < S);
     if (active) {
 
-    // ---- S[q][kv] = Q · K^T : C col = kv (ln within kb), row = q (crow).
-    constexpr int KB = KVB / 32;
-    float p[KB][16], dp[KB][16];
+    // 32-kv halves: live score state is p[16]+dp[16] instead of [2][16].
 #pragma unroll
-    for (int kb = 0; kb < KB; ++kb) {
+    for (int kb = 0; kb < 2; ++kb) {
+      // ---- S[q][kv] = Q · K^T and dP[q][kv] = dO · V^T (shared loop)
       f32x16 acc, acc2;
 #pragma unroll
       for (int j = 0; j < 16; ++j) { acc[j] = 0.0f; acc2[j] = 0.0f; }
 #pragma unroll
       for (int sl = 0; sl < DSL; ++sl) {
-        // B = K^T: lane col = kv (kb*32+ln), rows d → Klds[kv][d]
-        bf16x8_t kb_frag = ld_lds16(Klds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        // B = K^T: lane col = kv (kb*32+ln), rows d → row-slice read
+        bf16x8_t kb_frag = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
         acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa[sl], kb_frag, acc, 0, 0, 0);
-        // B = V^T: same pattern from Vlds → dP = dO · V^T
-        bf16x8_t vb_frag = ld_lds16(Vlds, swz<D * 2>(kb * 32 + ln, (sl * 16 + half * 8) * 2));
+        // B = V^T: same pattern → dP = dO · V^T
+        bf16x8_t vb_frag = ld_lds16(Vlds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
         acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa[sl], vb_frag, acc2, 0, 0, 0);
       }
-#pragma unroll
-      for (int j = 0; j < 16; ++j) { p[kb][j] = acc[j]; dp[kb][j] = acc2[j]; }
-    }
 
-    // ---- dS = scale * P ⊙ (dP - delta[q]) with P = exp(scale*s - lse[q])
+      // ---- dS = scale * P ⊙ (dP - delta[q]) with P = exp(scale*s - lse[q]);
+      // C-regs j in a 4-group are 4 consecutive q rows → packed 8 B store
+      // into the per-wave [kv 32][q 32] subtiled tile.
 #pragma unroll
-    for (int kb = 0; kb < KB; ++kb)
+      for (int jj = 0; jj < 4; ++jj) {
+        const int q_loc0 = 8 * jj + 4 * half;
+        short4v dk4;
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int r = crow(j, half);
-        const int q_abs = qw + r;
-        const int k_abs = k0 + kb * 32 + ln;
-        const float lse = lse_lds[w * QW + r];
-        const float dta = dta_lds[w * QW + r];
-        const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
-        const float pv = valid ? __expf(p[kb][j] * scale - lse) : 0.0f;
-        const float ds = scale * pv * (dp[kb][j] - dta);
-        // stage dS to P LDS: [q local][kv] bf16 swizzled
-        *reinterpret_cast<unsigned short*>(P + swz<KVB * 2>(r, (kb * 32 + ln) * 2)) = f2bf(ds);
+        for (int e = 0; e < 4; ++e) {
+          const int r = q_loc0 + e;
+          const int q_abs = qw + r;
+          const int k_abs = k0 + kb * 32 + ln;
+          const float lse = lse_lds[w * QW + r];
+          const float dta = dta_lds[w * QW + r];
+          const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+          const float pv = valid ? __expf(acc[jj * 4 + e] * scale - lse) : 0.0f;
+          const float ds = scale * pv * (acc2[jj * 4 + e] - dta);
+          dk4[e] = (short)f2bf(ds);
+        }
+        *reinterpret_cast<short4v*>(P + sub_off<32>(ln, q_loc0)) = dk4;
       }
 
-    // ---- dQ += dS · K   (A = dS from LDS, B = K_T reads)
+      // ---- dQ += dS · K   (A = dS tr-read: lane row = q, k = kv-local;
+      //      B = K column-fragments, k = kv-global)
 #pragma unroll
-    for (int ks = 0; ks < KVB / 16; ++ks) {
-      bf16x8_t dsa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        bf16x8_t dsa = ld_frag_tr<32>(P, lane, ks2 * 16, 0);
 #pragma unroll
-      for (int nb = 0; nb < D / 32; ++nb) {
-        bf16x8_t ktb = ld_lds16(KT, swzT<KVB * 2>(nb * 32 + ln, (ks * 16 + half * 8) * 2));
-        __builtin_amdgcn_s_setprio(1);
-        dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, ktb, dq_acc[nb], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+        for (int nb = 0; nb < D / 32; ++nb) {
+          bf16x8_t ktb = ld_frag_tr<KVB>(Klds, lane, kb * 32 + ks2 * 16, nb * 2);
+          __builtin_amdgcn_s_setprio(1);
+          dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, ktb, dq_acc[nb], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+        }
       }
-    }
+    }  // kb halves
     }  // active
 
     __syncthreads();
     if (has_next) {
-      kstage.store_rowmajor(Klds);
-      kstage.store_transposed(KT);
-      vstage.store_rowmajor(Vlds);
+      kstage.store_subtiled(Klds);
+      vstage.store_subtiled(Vlds);
     }
     __syncthreads();
   }
@@ -883,7 +854,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 
 #define LAUNCH_FWD(DD)                                                                            \
   do {                                                                                            \
-    const size_t lds = KVB * DD * 2 + DD * KVB * 2 + NW * (QW * KVB * 2);                         \
+    const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
     set_lds_limit((const void*)fa_fwd_kernel<DD>, lds);                                           \
     hipLaunchKernelGGL((fa_fwd_kernel<DD>), grid, dim3(NT), lds, stream.stream(),                 \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
@@ -948,7 +919,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
 #define LAUNCH_BWD(DD)                                                                              \
   do {                                                                                              \
     constexpr int QB = 64;                                                                          \
-    const size_t lds_kv = 4 * (QB * DD * 2) + NWB * (QW * QB * 2) + 2 * QB * sizeof(float);         \
+    const size_t lds_kv = 2 * sub_bytes<QB, DD>() + NWB * sub_bytes<32, 32>()                       \
+                          + 2 * QB * sizeof(float);                                                 \
     set_lds_limit((const void*)fa_bwd_dkdv_kernel<DD>, lds_kv);                                     \
     hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD>), grid_kv, dim3(NTB), lds_kv, stream.stream(),       \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
@@ -958,7 +930,7 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
                        qbs, qts, kbs, kts, dkbs, dkts,                                              \
                        B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
     HIP_CHECK_LAST();                                                                               \
-    const size_t lds_q = 2 * (KVB * DD * 2) + DD * KVB * 2 + NWB * (QW * KVB * 2)                   \
+    const size_t lds_q = 2 * sub_bytes<KVB, DD>() + NWB * sub_bytes<32, 32>()                       \
                          + 2 * NWB * QW * sizeof(float);                                            \
     set_lds_limit((const void*)fa_bwd_dq_kernel<DD>, lds_q);                                        \
     hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), grid_q, dim3(NTB), lds_q, side_stream,           \

@@ -76,6 +76,28 @@ __global__ void permlane32_probe(int* out0, int* out1) {
   out1[lane] = pair[1];
 }
 
+// ds_read_b64_tr_b16 semantics probe. LDS holds u16 value == element index.
+// Each lane supplies addr = base + 8*lane bytes (element 4*lane); the output
+// records which elements land in lane l's 4 result slots, distinguishing
+//  (a) per-lane strided gather: lane l gets {4l, 4l+16, 4l+32, 4l+48}
+//  (b) 16-lane weave:           lane l gets {(l&15)+16j+(l>>4)*64}
+typedef __bf16 bf16x4v_t __attribute__((ext_vector_type(4)));
+
+__global__ void tr16_probe(int* out) {  // out[64*4]
+  __shared__ unsigned short buf[256];
+  const int lane = threadIdx.x & 63;
+  for (int i = lane; i < 256; i += 64) buf[i] = (unsigned short)i;
+  __syncthreads();
+  typedef __attribute__((address_space(3))) bf16x4v_t* lds_v4p;
+  bf16x4v_t v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_v4p)&buf[lane * 4]);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    union { __bf16 b; unsigned short u; } c;
+    c.b = v[j];
+    out[lane * 4 + j] = (int)c.u;
+  }
+}
+
 std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor A32, at::Tensor B32) {
   TORCH_CHECK(A16.sizes() == at::IntArrayRef({16, 32}) && B16.sizes() == at::IntArrayRef({32, 16}));
   TORCH_CHECK(A32.sizes() == at::IntArrayRef({32, 16}) && B32.sizes() == at::IntArrayRef({16, 32}));
@@ -84,6 +106,7 @@ std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor
   auto C32 = at::zeros({32, 32}, opts);
   auto p0 = at::zeros({64}, opts.dtype(at::kInt));
   auto p1 = at::zeros({64}, opts.dtype(at::kInt));
+  auto tr = at::zeros({64, 4}, opts.dtype(at::kInt));
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(mfma_16x16x32_probe, dim3(1), dim3(64), 0, stream.stream(),
                      A16.data_ptr<float>(), B16.data_ptr<float>(), C16.data_ptr<float>());
@@ -91,8 +114,9 @@ std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor
                      A32.data_ptr<float>(), B32.data_ptr<float>(), C32.data_ptr<float>());
   hipLaunchKernelGGL(permlane32_probe, dim3(1), dim3(64), 0, stream.stream(),
                      p0.data_ptr<int>(), p1.data_ptr<int>());
+  hipLaunchKernelGGL(tr16_probe, dim3(1), dim3(64), 0, stream.stream(), tr.data_ptr<int>());
   HIP_CHECK_LAST();
-  return {C16, C32, p0, p1};
+  return {C16, C32, p0, p1, tr};
 }
 
 }  // namespace cai

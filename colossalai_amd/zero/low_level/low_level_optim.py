@@ -499,6 +499,78 @@ class LowLevelZeroOptimizer(OptimizerWrapper):
         pshard.copy_(b.master.to(pshard.device, dtype=pshard.dtype, non_blocking=True))
 
     # ------------------------------------------------------------ checkpoint
+    def get_param_states(self, names: Dict[int, str]) -> Dict[str, dict]:
+        """Topology-independent per-param optimizer states: for every locally
+        held param, the fp32 master/exp_avg/exp_avg_sq at the param's FULL
+        local (tp-shard) shape, reassembled from the rank-partitioned flat
+        buckets via one all-gather per bucket over the ZeRO group.
+        COLLECTIVE over each param group's dp group.
+        (reference: hybrid_parallel_checkpoint_io.py:1017
+        gather_from_sharded_optimizer_state)"""
+        out: Dict[str, dict] = {}
+        for gi, buckets in enumerate(self._group_buckets):
+            pg, world = self._g_pg[gi], self._g_world[gi]
+            step = self.optim.param_groups[gi].get("step", 0)
+            device = self._flat_params[gi].device
+            by_bucket: Dict[int, list] = {}
+            for p, (g2, off, numel) in self._param_slice.items():
+                if g2 == gi:
+                    by_bucket.setdefault(id(self._param_bucket[p]), []).append((p, off, numel))
+            for b in buckets:
+                plist = by_bucket.get(id(b), [])
+                if not plist:
+                    continue
+                full: Dict[str, Tensor] = {}
+                shards = [("exp_avg", b.exp_avg), ("exp_avg_sq", b.exp_avg_sq)]
+                if self.master_weights:
+                    shards.insert(0, ("master", b.master))
+                for key, shard in shards:
+                    shard_f = shard.detach().float().to(device)
+                    if world > 1:
+                        parts = [torch.empty_like(shard_f) for _ in range(world)]
+                        dist.all_gather(parts, shard_f.contiguous(), group=pg)
+                        full[key] = torch.cat(parts)
+                    else:
+                        full[key] = shard_f
+                for p, off, numel in plist:
+                    lo = off - b.start
+                    st: dict = {"step": int(step)}
+                    for key, flat in full.items():
+                        st[key] = flat[lo : lo + numel].view(p.shape).cpu().clone()
+                    st["_group"] = gi
+                    out[names[id(p)]] = st
+        return out
+
+    def set_param_states(self, states: Dict[str, dict], names: Dict[int, str]) -> None:
+        """Inverse of get_param_states for the CURRENT topology: slice each
+        param's full-local state into this rank's flat shard. Local only.
+        (reference: hybrid_parallel_checkpoint_io.py:1082
+        shard_from_complete_optimizer_state)"""
+        for p, (gi, off, numel) in self._param_slice.items():
+            name = names.get(id(p))
+            if name is None or name not in states:
+                continue
+            st = states[name]
+            b = self._param_bucket[p]
+            world, rank = self._g_world[gi], self._g_rank[gi]
+            blen = b.end - b.start
+            shard_lo = rank * blen // world
+            shard_hi = (rank + 1) * blen // world
+            p_lo, p_hi = off - b.start, off - b.start + numel
+            lo, hi = max(p_lo, shard_lo), min(p_hi, shard_hi)
+            if lo >= hi:
+                continue
+            for key, dst in (("master", b.master), ("exp_avg", b.exp_avg), ("exp_avg_sq", b.exp_avg_sq)):
+                if key == "master" and not self.master_weights:
+                    continue
+                src = st.get(key)
+                if src is None:
+                    continue
+                piece = src.reshape(-1)[lo - p_lo : hi - p_lo].to(dst.device, dtype=dst.dtype)
+                dst[lo - shard_lo : hi - shard_lo].copy_(piece)
+            if "step" in st:
+                self.optim.param_groups[gi]["step"] = int(st["step"])
+
     def state_dict(self):
         """Rank-local shard state (flat). Full gather lives in checkpoint_io."""
         state = {"param_groups": [{k: v for k, v in g.items() if k != "params"} for g in self.optim.param_groups]}

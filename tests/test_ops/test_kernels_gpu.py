@@ -30,7 +30,7 @@ def test_mfma_layouts():
     B16 = torch.randn(32, 16, device="cuda")
     A32 = torch.randn(32, 16, device="cuda")
     B32 = torch.randn(16, 32, device="cuda")
-    C16, C32, p0, p1 = _C.mfma_selftest(A16, B16, A32, B32)
+    C16, C32, p0, p1, tr = _C.mfma_selftest(A16, B16, A32, B32)
     # bf16 inputs -> compare against bf16-rounded matmul
     ref16 = (A16.bfloat16().float() @ B16.bfloat16().float())
     ref32 = (A32.bfloat16().float() @ B32.bfloat16().float())
@@ -40,6 +40,16 @@ def test_mfma_layouts():
     p0, p1 = p0.cpu(), p1.cpu()
     print("permlane32_swap out0:", p0.tolist())
     print("permlane32_swap out1:", p1.tolist())
+    # ds_read_b64_tr_b16 semantics: LDS[i]=i, lane l addr = elem 4l.
+    # gather hypothesis: lane l -> {4l+16j}; weave: lane l -> {(l&15)+16j+(l>>4)*64}
+    tr = tr.cpu()
+    lanes = torch.arange(64)
+    gather = (4 * lanes).unsqueeze(1) + 16 * torch.arange(4).unsqueeze(0)
+    weave = ((lanes & 15) + (lanes >> 4) * 64).unsqueeze(1) + 16 * torch.arange(4).unsqueeze(0)
+    is_gather = torch.equal(tr, gather)
+    is_weave = torch.equal(tr, weave)
+    print("tr16 semantics:", "gather" if is_gather else ("weave" if is_weave else f"OTHER: {tr.tolist()}"))
+    assert is_gather or is_weave, f"unrecognized ds_read_b64_tr_b16 mapping: {tr.tolist()}"
 
 
 # ------------------------------------------------------------------- rmsnorm
