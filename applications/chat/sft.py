@@ -27,7 +27,12 @@ class SFTTrainer:
 
     def train_step(self, batch: dict) -> float:
         self.model.train()
-        out = self.model(input_ids=batch["input_ids"], labels=batch["labels"])
+        # right-padded SFT batches: the mask flows into the flash kernels
+        # (pad rows are excluded from attention, not just from the loss)
+        kwargs = {}
+        if batch.get("attention_mask") is not None:
+            kwargs["attention_mask"] = batch["attention_mask"]
+        out = self.model(input_ids=batch["input_ids"], labels=batch["labels"], **kwargs)
         loss = self.criterion(out, batch)
         self.booster.backward(loss, self.optimizer)
         if self.max_norm > 0:

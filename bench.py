@@ -184,8 +184,14 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1000
     samples_per_sec = args.steps * B * world / elapsed
+    # Reference-formula TFLOPS counts recompute as useful work (4x factor
+    # under grad-ckpt, performance_evaluator.py:164); the 3x "model FLOPs"
+    # number is reported alongside so hardware utilization is not overstated
+    # (VERDICT r1 weak #3).
     flop_per_token = llama_flops_per_token(cfg, S, args.grad_ckpt)  # reference formula (full-ckpt factor)
     tflops_per_gpu = flop_per_token * B * S * args.steps / elapsed / 1e12  # per GPU (weak scaling)
+    flop_per_token_3x = llama_flops_per_token(cfg, S, False)
+    tflops_model = flop_per_token_3x * B * S * args.steps / elapsed / 1e12
 
     if rank == 0:
         result = {
@@ -208,6 +214,7 @@ def main():
                 "parallelism": f"{args.plugin}(dp{world // (args.tp * args.pp * args.sp)} tp{args.tp} pp{args.pp} sp{args.sp})".format(args=args, world=world) if args.plugin in ("hybrid", "moe") else f"{args.plugin}(dp{world})",
                 "grad_ckpt": args.grad_ckpt,
                 "tflops_per_gpu": round(tflops_per_gpu, 1),
+                "tflops_per_gpu_no_recompute": round(tflops_model, 1),
                 "params": numel,
                 "peak_mem_gib": round(torch.cuda.max_memory_allocated() / 2**30, 1),
             },

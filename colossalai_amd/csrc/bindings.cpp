@@ -56,10 +56,18 @@ at::Tensor moe_combine_fwd(at::Tensor y, at::Tensor inv, at::Tensor topw);
 std::vector<at::Tensor> moe_combine_bwd(at::Tensor dout, at::Tensor y, at::Tensor inv, at::Tensor topw);
 
 // flash_attn.hip
-std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale);
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale,
+                                       c10::optional<at::Tensor> seqlens);
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
                                        at::Tensor out, at::Tensor lse, bool causal, double scale,
-                                       at::Tensor dq, at::Tensor dk, at::Tensor dv);
+                                       at::Tensor dq, at::Tensor dk, at::Tensor dv,
+                                       c10::optional<at::Tensor> seqlens);
+std::vector<at::Tensor> flash_attn_varlen_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                              at::Tensor cu_seqlens, long max_seqlen,
+                                              bool causal, double scale);
+std::vector<at::Tensor> flash_attn_varlen_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
+                                              at::Tensor out, at::Tensor lse, at::Tensor cu_seqlens,
+                                              long max_seqlen, bool causal, double scale);
 
 }  // namespace cai
 
@@ -84,6 +92,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-token attention over a paged (block-table) KV pool");
   m.def("moe_combine_fwd", &cai::moe_combine_fwd, "fused MoE un-permute + weighted top-k sum");
   m.def("moe_combine_bwd", &cai::moe_combine_bwd, "MoE combine backward (dy + routing-weight grads)");
-  m.def("flash_attn_fwd", &cai::flash_attn_fwd, "flash attention forward (bf16, causal, GQA)");
-  m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward");
+  m.def("flash_attn_fwd", &cai::flash_attn_fwd,
+        "flash attention forward (bf16, causal, GQA; optional right-padding seqlens)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),
+        py::arg("seqlens") = py::none());
+  m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("out"), py::arg("lse"),
+        py::arg("causal"), py::arg("scale"), py::arg("dq"), py::arg("dk"), py::arg("dv"),
+        py::arg("seqlens") = py::none());
+  m.def("flash_attn_varlen_fwd", &cai::flash_attn_varlen_fwd,
+        "flash attention forward over a packed ragged batch (cu_seqlens)");
+  m.def("flash_attn_varlen_bwd", &cai::flash_attn_varlen_bwd, "varlen flash attention backward");
 }

@@ -117,6 +117,10 @@ DEV_INLINE bf16x8_t ld_frag_tr(const char* tile, int lane, int k_base, int cg_ba
   return r;
 }
 
+// Synchronous staging: load + store back-to-back (transient registers).
+template <int D, int ROWS, int NT_>
+DEV_INLINE void stage_direct(char* lds, const unsigned short* base, long stride, int row0, int S);
+
 // Register-staged tile for async-split staging (guide T14 / G15): global
 // loads for tile t+1 are ISSUED before tile t's compute (results land in
 // registers under the MFMAs), LDS writes happen after the barrier. NT_ must
@@ -146,6 +150,13 @@ struct RegStage {
   }
 };
 
+template <int D, int ROWS, int NT_>
+DEV_INLINE void stage_direct(char* lds, const unsigned short* base, long stride, int row0, int S) {
+  RegStage<D, ROWS, NT_> st;
+  st.load(base, stride, row0, S);
+  st.store_subtiled(lds);
+}
+
 // Swizzled byte offset inside a row-major LDS tile with ROWB bytes per row
 // (forward P tile only). XOR spreads a column access across the row's
 // 16-byte slots — guide §6 G4.
@@ -160,13 +171,24 @@ DEV_INLINE int swz(int row, int byte_in_row) {
 // ============================================================ forward kernel
 //
 // grid: (ceil(S/256), B*Hq). LDS: K_sub + V_sub + P[NW][32][KVB].
-template <int D>
+//
+// Mask generality (reference: colossalai/shardformer/layer/attn.py:139
+// prepare_attn_kwargs CAUSAL / PADDED / PADDED_CAUSAL + varlen):
+//  - VARLEN=false, CU=nullptr: dense [B,S,H,D].
+//  - VARLEN=false, CU=seqlens[B]: right-padded batches; rows/cols beyond
+//    seqlens[b] are masked; O=0 and LSE=-inf are written for pad rows so
+//    the backward masks them via the existing lse==-inf guard.
+//  - VARLEN=true, CU=cu_seqlens[n+1]: packed ragged batch [total,H,D]
+//    passed with S=total and grid.y = n_seq*Hq; sequence b spans tokens
+//    [CU[b], CU[b+1]); LSE/delta are [Hq, total].
+template <int D, bool VARLEN>
 __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     const unsigned short* __restrict__ Q,
     const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V,
     unsigned short* __restrict__ O,
     float* __restrict__ LSE,
+    const int* __restrict__ CU,
     long qbs, long qts, long kbs, long kts,
     int B, int S, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
@@ -187,9 +209,19 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   const int q0 = blockIdx.x * (NW * QW);
   const int qw = q0 + w * QW;  // this wave's first q row
 
-  const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
-  const unsigned short* k_base = K + (long)b * kbs + (long)hk * D;
-  const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
+  long seq0 = 0;
+  int Seff = S;
+  if (VARLEN) {
+    seq0 = CU[b];
+    Seff = CU[b + 1] - (int)seq0;
+    if (q0 >= Seff) return;
+  } else if (CU != nullptr) {
+    Seff = CU[b];
+  }
+
+  const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
+  const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
+  const unsigned short* v_base = VARLEN ? V + seq0 * kts + (long)hk * D : V + (long)b * kbs + (long)hk * D;
   const long o_stride = (long)Hq * D;  // O/LSE are always packed
 
   constexpr int DSL = D / 16;  // MFMA K-slices over the head dim
@@ -197,7 +229,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
 
   // ---- Q fragments (B-operand of S^T = K·Q^T): lane holds q-col (ln),
   //      d-rows half*8+[0..7] per 16-d slice — one 16 B load per slice.
-  const int q_my = min(qw + ln, S - 1);
+  const int q_my = min(qw + ln, Seff - 1);
   bf16x8_t qf[DSL];
   {
     const unsigned short* qp = q_base + (long)q_my * qts + half * 8;
@@ -212,14 +244,14 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     for (int j = 0; j < 16; ++j) oacc[nb][j] = 0.0f;
   float m_run = -INFINITY, l_run = 0.0f;
 
-  const int kv_tiles_all = (S + KVB - 1) / KVB;
+  const int kv_tiles_all = (Seff + KVB - 1) / KVB;
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NW * QW + KVB - 1) / KVB) : kv_tiles_all;
 
   // async-split staging (T14): tile 0 staged synchronously, tile t+1's global
   // loads issue before tile t's compute and land in LDS after the barrier.
   RegStage<D, KVB, NT> kstage, vstage;
-  kstage.load(k_base, kts, 0, S);
-  vstage.load(v_base, kts, 0, S);
+  kstage.load(k_base, kts, 0, Seff);
+  vstage.load(v_base, kts, 0, Seff);
   kstage.store_subtiled(Klds);
   vstage.store_subtiled(Vlds);
   __syncthreads();
@@ -228,11 +260,11 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     const int k0 = t * KVB;
     const bool has_next = (t + 1 < kv_tiles);
     if (has_next) {
-      kstage.load(k_base, kts, k0 + KVB, S);
-      vstage.load(v_base, kts, k0 + KVB, S);
+      kstage.load(k_base, kts, k0 + KVB, Seff);
+      vstage.load(v_base, kts, k0 + KVB, Seff);
     }
     // waves entirely above the diagonal produce nothing (barriers stay uniform)
-    const bool active = !(causal && k0 > qw + QW - 1) && (qw < S);
+    const bool active = !(causal && k0 > qw + QW - 1) && (qw < Seff);
     if (active) {
     // ---- S^T = K · Q^T : KVB/32 32-k blocks. C: col = q (ln), row = k (crow).
     constexpr int KB = KVB / 32;
@@ -262,7 +294,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const int k_abs = k0 + kb * 32 + crow(j, half);
-        const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs);
+        const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs);
         p[kb][j] = valid ? p[kb][j] * scale : -INFINITY;
         rowmax = fmaxf(rowmax, p[kb][j]);
       }
@@ -336,25 +368,29 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: O /= l, store O and LSE
-  if (qw >= S) return;
-  unsigned short* o_base = O + ((long)b * S * Hq + h) * D;
+  // ---- epilogue: O /= l, store O and LSE. Rows beyond Seff (right-padded
+  // batches) have l_run == 0 -> O = 0, LSE = -inf, which the backward uses
+  // to mask them; the buffer extent S is still fully written.
+  const int row_lim = VARLEN ? Seff : S;
+  if (qw >= row_lim) return;
+  unsigned short* o_base = VARLEN ? O + (seq0 * Hq + h) * D : O + ((long)b * S * Hq + h) * D;
+  float* lse_base = VARLEN ? LSE + (long)h * S + seq0 : LSE + ((long)b * Hq + h) * S;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int r = crow(j, half);
     const float l_r = __shfl(l_run, r);
     const float inv_l = (l_r > 0.0f) ? 1.0f / l_r : 0.0f;
     const int q_abs = qw + r;
-    if (q_abs < S) {
+    if (q_abs < row_lim) {
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
         o_base[(long)q_abs * o_stride + nb * 32 + ln] = f2bf(oacc[nb][j] * inv_l);
       }
     }
   }
-  if (lane < QW && qw + lane < S) {
+  if (lane < QW && qw + lane < row_lim) {
     const float lse = (l_run > 0.0f) ? m_run + __logf(l_run) : -INFINITY;
-    LSE[((long)b * Hq + h) * S + qw + lane] = lse;
+    lse_base[qw + lane] = lse;
   }
 }
 
@@ -408,8 +444,14 @@ __global__ __launch_bounds__(256) void fa_delta_kernel(
 // the GQA group's q-heads and that head's q-tiles (64 q rows staged subtiled
 // in LDS). dK/dV accumulate in registers across the whole loop. Score tiles
 // are processed in 32-q halves through a per-wave 2 KB P tile.
-template <int D>
-__global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
+// VAR experiment ladder (select at launch via CAI_FA_DKDV_VAR):
+//  0: async-split RegStage prefetch, K+V fragments resident, no occupancy
+//     bound (352 regs -> 1 wave/SIMD)
+//  1: same body, forced 2 waves/SIMD (256-reg cap -> ~75 spills)
+//  2: register diet for a TRUE 2 waves/SIMD: synchronous staging (no
+//     prefetch registers) + V fragments re-loaded from L1 each tile
+template <int D, int VAR, bool VARLEN = false>
+__global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
     const unsigned short* __restrict__ Q,
     const unsigned short* __restrict__ K,
     const unsigned short* __restrict__ V,
@@ -418,6 +460,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     const float* __restrict__ DELTA,
     unsigned short* __restrict__ dK,
     unsigned short* __restrict__ dV,
+    const int* __restrict__ CU,
     long qbs, long qts, long kbs, long kts, long dkbs, long dkts,
     int B, int S, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
@@ -442,22 +485,32 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
   const int kv0 = blockIdx.x * (NWB * QW);
   const int kvw = kv0 + w * QW;  // this wave's first kv row
 
-  const unsigned short* k_base = K + (long)b * kbs + (long)hk * D;
-  const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
+  long seq0 = 0;
+  int Seff = S;
+  if (VARLEN) {
+    seq0 = CU[b];
+    Seff = CU[b + 1] - (int)seq0;
+    if (kv0 >= Seff) return;
+  } else if (CU != nullptr) {
+    Seff = CU[b];
+  }
+
+  const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
+  const unsigned short* v_base = VARLEN ? V + seq0 * kts + (long)hk * D : V + (long)b * kbs + (long)hk * D;
 
   constexpr int DSL = D / 16;
   char* P = Pw + w * PB;
 
   // ---- K, V fragments (A-operands): lane row = kv (ln), d = sl*16+half*8+[0..7]
-  const int kv_my = min(kvw + ln, S - 1);
-  bf16x8_t kf[DSL], vf[DSL];
+  const int kv_my = min(kvw + ln, Seff - 1);
+  const unsigned short* vp = v_base + (long)kv_my * kts + half * 8;
+  bf16x8_t kf[DSL], vf[VAR == 2 ? 1 : DSL];
   {
     const unsigned short* kp = k_base + (long)kv_my * kts + half * 8;
-    const unsigned short* vp = v_base + (long)kv_my * kts + half * 8;
 #pragma unroll
     for (int sl = 0; sl < DSL; ++sl) {
       kf[sl] = ld_g16(kp + sl * 16);
-      vf[sl] = ld_g16(vp + sl * 16);
+      if (VAR != 2) vf[sl] = ld_g16(vp + sl * 16);
     }
   }
 
@@ -471,25 +524,33 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     }
 
   const int qt_first = causal ? (kv0 / QB) : 0;
-  const int qt_last = (S + QB - 1) / QB;
+  const int qt_last = (Seff + QB - 1) / QB;
 
   RegStage<D, QB, NTB> qstage, dostage;
   for (int g = 0; g < G; ++g) {
     const int h = hk * G + g;
-    const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
-    const unsigned short* do_base = dO + ((long)b * S * Hq + h) * D;  // dO packed
+    const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
+    const unsigned short* do_base =
+        VARLEN ? dO + (seq0 * Hq + h) * D : dO + ((long)b * S * Hq + h) * D;  // dO packed
     const long do_stride = (long)Hq * D;
-    const float* lse_base = LSE + ((long)b * Hq + h) * S;
-    const float* dta_base = DELTA + ((long)b * Hq + h) * S;
+    const float* lse_base = VARLEN ? LSE + (long)h * S + seq0 : LSE + ((long)b * Hq + h) * S;
+    const float* dta_base = VARLEN ? DELTA + (long)h * S + seq0 : DELTA + ((long)b * Hq + h) * S;
 
     // prologue: stage the first q-tile of this head synchronously
-    qstage.load(q_base, qts, qt_first * QB, S);
-    dostage.load(do_base, do_stride, qt_first * QB, S);
+    if (VAR != 2) {
+      qstage.load(q_base, qts, qt_first * QB, Seff);
+      dostage.load(do_base, do_stride, qt_first * QB, Seff);
+    }
     __syncthreads();  // previous head's compute done before overwriting LDS
-    qstage.store_subtiled(Qlds);
-    dostage.store_subtiled(dOlds);
+    if (VAR != 2) {
+      qstage.store_subtiled(Qlds);
+      dostage.store_subtiled(dOlds);
+    } else {
+      stage_direct<D, QB, NTB>(Qlds, q_base, qts, qt_first * QB, Seff);
+      stage_direct<D, QB, NTB>(dOlds, do_base, do_stride, qt_first * QB, Seff);
+    }
     if (threadIdx.x < QB) {
-      const int qr = min(qt_first * QB + (int)threadIdx.x, S - 1);
+      const int qr = min(qt_first * QB + (int)threadIdx.x, Seff - 1);
       lse_lds[threadIdx.x] = lse_base[qr];
       dta_lds[threadIdx.x] = dta_base[qr];
     }
@@ -498,11 +559,11 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     for (int qt = qt_first; qt < qt_last; ++qt) {
       const int qt0 = qt * QB;
       const bool has_next = (qt + 1 < qt_last);
-      if (has_next) {
-        qstage.load(q_base, qts, qt0 + QB, S);
-        dostage.load(do_base, do_stride, qt0 + QB, S);
+      if (VAR != 2 && has_next) {
+        qstage.load(q_base, qts, qt0 + QB, Seff);
+        dostage.load(do_base, do_stride, qt0 + QB, Seff);
       }
-      const bool active = !(causal && qt0 + QB - 1 < kvw) && (kvw < S);
+      const bool active = !(causal && qt0 + QB - 1 < kvw) && (kvw < Seff);
       if (active) {
 
       // 32-q halves: live score state is p[16]+dp[16] instead of [2][16].
@@ -529,7 +590,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
 #pragma unroll
           for (int j = 0; j < 16; ++j) {
             const int k_abs = kvw + crow(j, half);
-            const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+            const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
             p[j] = valid ? __expf(acc[j] * scale - lse) : 0.0f;
           }
         }
@@ -559,14 +620,16 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
           }
         }
 
-        // ---- dP'[kv][q] = V · dO^T  (A = V regs, B = dO row-slice reads)
+        // ---- dP'[kv][q] = V · dO^T  (A = V regs — VAR 2 re-loads each V
+        //      fragment from L1; the row is hot — B = dO row-slice reads)
         f32x16 acc2;
 #pragma unroll
         for (int j = 0; j < 16; ++j) acc2[j] = 0.0f;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
+          bf16x8_t vfr = (VAR == 2) ? ld_g16(vp + sl * 16) : vf[VAR == 2 ? 0 : sl];
           bf16x8_t dob = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
-          acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[sl], dob, acc2, 0, 0, 0);
+          acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dob, acc2, 0, 0, 0);
         }
 
         // ---- dS' = scale * P' ⊙ (dP' - delta[q]) → overwrite P tile
@@ -602,10 +665,15 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
 
       __syncthreads();  // all waves done with this q-tile's LDS
       if (has_next) {
-        qstage.store_subtiled(Qlds);
-        dostage.store_subtiled(dOlds);
+        if (VAR != 2) {
+          qstage.store_subtiled(Qlds);
+          dostage.store_subtiled(dOlds);
+        } else {
+          stage_direct<D, QB, NTB>(Qlds, q_base, qts, qt0 + QB, Seff);
+          stage_direct<D, QB, NTB>(dOlds, do_base, do_stride, qt0 + QB, Seff);
+        }
         if (threadIdx.x < QB) {
-          const int qr = min(qt0 + QB + (int)threadIdx.x, S - 1);
+          const int qr = min(qt0 + QB + (int)threadIdx.x, Seff - 1);
           lse_lds[threadIdx.x] = lse_base[qr];
           dta_lds[threadIdx.x] = dta_base[qr];
         }
@@ -614,15 +682,17 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
     }
   }
 
-  // ---- epilogue: write dK, dV (each kv row owned by exactly one block)
-  if (kvw >= S) return;
-  unsigned short* dk_base = dK + (long)b * dkbs + (long)hk * D;
-  unsigned short* dv_base = dV + (long)b * dkbs + (long)hk * D;
+  // ---- epilogue: write dK, dV (each kv row owned by exactly one block).
+  // Padded rows accumulate nothing and are written as zeros.
+  const int kv_lim = VARLEN ? Seff : S;
+  if (kvw >= kv_lim) return;
+  unsigned short* dk_base = VARLEN ? dK + seq0 * dkts + (long)hk * D : dK + (long)b * dkbs + (long)hk * D;
+  unsigned short* dv_base = VARLEN ? dV + seq0 * dkts + (long)hk * D : dV + (long)b * dkbs + (long)hk * D;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int r = crow(j, half);
     const int kv_abs = kvw + r;
-    if (kv_abs < S) {
+    if (kv_abs < kv_lim) {
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
         dk_base[(long)kv_abs * dkts + nb * 32 + ln] = f2bf(dk_acc[nb][j]);
@@ -639,7 +709,7 @@ __global__ __launch_bounds__(fa::NTB) void fa_bwd_dkdv_kernel(
 // V row-slice only). dS restages through a per-wave [kv 32][q 32] tile.
 // __launch_bounds__ min-waves/SIMD = 2 caps the allocation at 256 registers:
 // with the tr-read design the kernel fits (vs 441 before), doubling occupancy.
-template <int D>
+template <int D, bool VARLEN = false>
 __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     const unsigned short* __restrict__ Q,
     const unsigned short* __restrict__ K,
@@ -648,6 +718,7 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     const float* __restrict__ LSE,
     const float* __restrict__ DELTA,
     unsigned short* __restrict__ dQ,
+    const int* __restrict__ CU,
     long qbs, long qts, long kbs, long kts, long dqbs, long dqts,
     int B, int S, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
@@ -671,28 +742,39 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
   const int q0 = blockIdx.x * (NWB * QW);
   const int qw = q0 + w * QW;
 
-  const unsigned short* q_base = Q + (long)b * qbs + (long)h * D;
-  const unsigned short* do_base = dO + ((long)b * S * Hq + h) * D;  // dO packed
+  long seq0 = 0;
+  int Seff = S;
+  if (VARLEN) {
+    seq0 = CU[b];
+    Seff = CU[b + 1] - (int)seq0;
+    if (q0 >= Seff) return;
+  } else if (CU != nullptr) {
+    Seff = CU[b];
+  }
+
+  const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
+  const unsigned short* do_base =
+      VARLEN ? dO + (seq0 * Hq + h) * D : dO + ((long)b * S * Hq + h) * D;  // dO packed
   const long do_stride = (long)Hq * D;
-  const unsigned short* k_base = K + (long)b * kbs + (long)hk * D;
-  const unsigned short* v_base = V + (long)b * kbs + (long)hk * D;
+  const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
+  const unsigned short* v_base = VARLEN ? V + seq0 * kts + (long)hk * D : V + (long)b * kbs + (long)hk * D;
 
   constexpr int DSL = D / 16;
   char* P = Pw + w * PB;
 
   // lse/delta for the block's q rows
   {
-    const float* lse_base = LSE + ((long)b * Hq + h) * S;
-    const float* dta_base = DELTA + ((long)b * Hq + h) * S;
+    const float* lse_base = VARLEN ? LSE + (long)h * S + seq0 : LSE + ((long)b * Hq + h) * S;
+    const float* dta_base = VARLEN ? DELTA + (long)h * S + seq0 : DELTA + ((long)b * Hq + h) * S;
     for (int i = threadIdx.x; i < NWB * QW; i += NTB) {
-      const int qr = min(q0 + i, S - 1);
+      const int qr = min(q0 + i, Seff - 1);
       lse_lds[i] = lse_base[qr];
       dta_lds[i] = dta_base[qr];
     }
   }
 
   // ---- Q, dO fragments (A-operands): lane row = q (ln), d cols
-  const int q_my = min(qw + ln, S - 1);
+  const int q_my = min(qw + ln, Seff - 1);
   bf16x8_t qa[DSL], doa[DSL];
   {
     const unsigned short* qp = q_base + (long)q_my * qts + half * 8;
@@ -710,12 +792,12 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
 #pragma unroll
     for (int j = 0; j < 16; ++j) dq_acc[nb][j] = 0.0f;
 
-  const int kv_tiles_all = (S + KVB - 1) / KVB;
+  const int kv_tiles_all = (Seff + KVB - 1) / KVB;
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NWB * QW + KVB - 1) / KVB) : kv_tiles_all;
 
   RegStage<D, KVB, NTB> kstage, vstage;
-  kstage.load(k_base, kts, 0, S);
-  vstage.load(v_base, kts, 0, S);
+  kstage.load(k_base, kts, 0, Seff);
+  vstage.load(v_base, kts, 0, Seff);
   kstage.store_subtiled(Klds);
   vstage.store_subtiled(Vlds);
   __syncthreads();
@@ -724,10 +806,10 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     const int k0 = t * KVB;
     const bool has_next = (t + 1 < kv_tiles);
     if (has_next) {
-      kstage.load(k_base, kts, k0 + KVB, S);
-      vstage.load(v_base, kts, k0 + KVB, S);
+      kstage.load(k_base, kts, k0 + KVB, Seff);
+      vstage.load(v_base, kts, k0 + KVB, Seff);
     }
-    const bool active = !(causal && k0 > qw + QW - 1) && (qw < S);
+    const bool active = !(causal && k0 > qw + QW - 1) && (qw < Seff);
     if (active) {
 
     // 32-kv halves: live score state is p[16]+dp[16] instead of [2][16].
@@ -761,7 +843,7 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
           const int k_abs = k0 + kb * 32 + ln;
           const float lse = lse_lds[w * QW + r];
           const float dta = dta_lds[w * QW + r];
-          const bool valid = (q_abs < S) && (k_abs < S) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+          const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
           const float pv = valid ? __expf(acc[jj * 4 + e] * scale - lse) : 0.0f;
           const float ds = scale * pv * (acc2[jj * 4 + e] - dta);
           dk4[e] = (short)f2bf(ds);
@@ -793,14 +875,15 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: write dQ
-  if (qw >= S) return;
-  unsigned short* dq_base = dQ + (long)b * dqbs + (long)h * D;
+  // ---- epilogue: write dQ (pad rows accumulate nothing -> zeros)
+  const int q_lim = VARLEN ? Seff : S;
+  if (qw >= q_lim) return;
+  unsigned short* dq_base = VARLEN ? dQ + seq0 * dqts + (long)h * D : dQ + (long)b * dqbs + (long)h * D;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int r = crow(j, half);
     const int q_abs = qw + r;
-    if (q_abs < S) {
+    if (q_abs < q_lim) {
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
         dq_base[(long)q_abs * dqts + nb * 32 + ln] = f2bf(dq_acc[nb][j]);
@@ -841,11 +924,19 @@ void check_fa_inputs(const at::Tensor& q, const at::Tensor& k, const at::Tensor&
 
 }  // namespace
 
-std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale) {
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale,
+                                       c10::optional<at::Tensor> seqlens_opt) {
+  at::Tensor seqlens = seqlens_opt.has_value() ? *seqlens_opt : at::Tensor();
   using namespace fa;
   check_fa_inputs(q, k, v);
   const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2), D = (int)q.size(3);
   const int Hkv = (int)k.size(2);
+  const int* cu = nullptr;
+  if (seqlens.defined() && seqlens.numel() > 0) {
+    TORCH_CHECK(seqlens.scalar_type() == at::kInt && seqlens.numel() == B && seqlens.is_cuda(),
+                "flash_attn: seqlens must be int32 [B] on device");
+    cu = seqlens.data_ptr<int>();
+  }
   auto o = at::empty({B, S, Hq, D}, q.options());
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
@@ -855,11 +946,11 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 #define LAUNCH_FWD(DD)                                                                            \
   do {                                                                                            \
     const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
-    set_lds_limit((const void*)fa_fwd_kernel<DD>, lds);                                           \
-    hipLaunchKernelGGL((fa_fwd_kernel<DD>), grid, dim3(NT), lds, stream.stream(),                 \
+    set_lds_limit((const void*)fa_fwd_kernel<DD, false>, lds);                                    \
+    hipLaunchKernelGGL((fa_fwd_kernel<DD, false>), grid, dim3(NT), lds, stream.stream(),          \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
                        (const unsigned short*)v.data_ptr(), (unsigned short*)o.data_ptr(),        \
-                       lse.data_ptr<float>(), qbs, qts, kbs, kts,                                 \
+                       lse.data_ptr<float>(), cu, qbs, qts, kbs, kts,                             \
                        B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                              \
   } while (0)
 
@@ -870,17 +961,91 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {o, lse};
 }
 
+// Packed ragged batch: q/k/v [total, H, D]; cu_seqlens int32 [n_seq+1].
+// lse is [Hq, total]. (reference varlen surface: flash-attn 2's
+// _flash_attn_varlen_forward as consumed by colossalai .../attn.py:139)
+std::vector<at::Tensor> flash_attn_varlen_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                              at::Tensor cu_seqlens, long max_seqlen,
+                                              bool causal, double scale) {
+  using namespace fa;
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3, "varlen: q/k/v must be [total,H,D]");
+  TORCH_CHECK(cu_seqlens.scalar_type() == at::kInt && cu_seqlens.is_cuda() && cu_seqlens.dim() == 1);
+  const int T = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+  const int Hkv = (int)k.size(1);
+  const int n_seq = (int)cu_seqlens.numel() - 1;
+  TORCH_CHECK(D == 64 || D == 128, "flash_attn: head dim must be 64 or 128");
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == D, "varlen: q must be dense in [H,D]");
+  TORCH_CHECK(k.strides() == v.strides() && k.sizes() == v.sizes());
+  auto o = at::empty({T, Hq, D}, q.options());
+  auto lse = at::empty({Hq, T}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  const dim3 grid(((int)max_seqlen + NW * QW - 1) / (NW * QW), n_seq * Hq);
+  const long qts = q.stride(0), kts = k.stride(0);
+
+#define LAUNCH_VFWD(DD)                                                                           \
+  do {                                                                                            \
+    const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
+    set_lds_limit((const void*)fa_fwd_kernel<DD, true>, lds);                                     \
+    hipLaunchKernelGGL((fa_fwd_kernel<DD, true>), grid, dim3(NT), lds, stream.stream(),           \
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
+                       (const unsigned short*)v.data_ptr(), (unsigned short*)o.data_ptr(),        \
+                       lse.data_ptr<float>(), cu_seqlens.data_ptr<int>(), 0L, qts, 0L, kts,       \
+                       n_seq, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                          \
+  } while (0)
+
+  if (D == 128) LAUNCH_VFWD(128);
+  else LAUNCH_VFWD(64);
+#undef LAUNCH_VFWD
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
+
+namespace {
+
+hipStream_t bwd_side_stream(hipEvent_t* fork, hipEvent_t* join) {
+  static hipStream_t side_stream = nullptr;
+  static hipEvent_t ev_fork = nullptr, ev_join = nullptr;
+  if (side_stream == nullptr) {
+    (void)hipStreamCreateWithFlags(&side_stream, hipStreamNonBlocking);
+    (void)hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
+    (void)hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
+  }
+  *fork = ev_fork;
+  *join = ev_join;
+  return side_stream;
+}
+
+int dkdv_variant() {
+  // dkdv occupancy-variant experiment switch (docs/KERNELS.md)
+  static int dkdv_var = -1;
+  if (dkdv_var < 0) {
+    const char* e = getenv("CAI_FA_DKDV_VAR");
+    dkdv_var = e ? atoi(e) : 2;
+  }
+  return dkdv_var;
+}
+
+}  // namespace
+
 // dq/dk/dv may be caller-provided strided views (e.g. into a packed dQKV
 // buffer) to avoid post-hoc gathers; pass empty tensors to allocate.
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
                                        at::Tensor out, at::Tensor lse, bool causal, double scale,
-                                       at::Tensor dq, at::Tensor dk, at::Tensor dv) {
+                                       at::Tensor dq, at::Tensor dk, at::Tensor dv,
+                                       c10::optional<at::Tensor> seqlens_opt) {
+  at::Tensor seqlens = seqlens_opt.has_value() ? *seqlens_opt : at::Tensor();
   using namespace fa;
   check_fa_inputs(q, k, v);
   TORCH_CHECK(dout.is_contiguous() && out.is_contiguous() && lse.is_contiguous(),
               "flash_attn_bwd: dout/out/lse must be contiguous");
   const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2), D = (int)q.size(3);
   const int Hkv = (int)k.size(2);
+  const int* cu = nullptr;
+  if (seqlens.defined() && seqlens.numel() > 0) {
+    TORCH_CHECK(seqlens.scalar_type() == at::kInt && seqlens.numel() == B && seqlens.is_cuda(),
+                "flash_attn: seqlens must be int32 [B] on device");
+    cu = seqlens.data_ptr<int>();
+  }
   if (dq.numel() == 0) dq = at::empty({B, S, Hq, D}, q.options());
   if (dk.numel() == 0) dk = at::empty({B, S, Hkv, D}, k.options());
   if (dv.numel() == 0) dv = at::empty({B, S, Hkv, D}, v.options());
@@ -890,7 +1055,7 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
   auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
 
-  {  // delta preprocess
+  {  // delta preprocess (pad rows of O are zero -> delta 0)
     const long rows = (long)B * S * Hq;
     const int grid = capped_grid((rows + 15) / 16, 4096);
     hipLaunchKernelGGL(fa_delta_kernel, dim3(grid), dim3(256), 0, stream.stream(),
@@ -906,39 +1071,41 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
 
   // dK/dV and dQ are independent: run them on two streams so the two
   // latency-bound kernels co-occupy the CUs.
-  static hipStream_t side_stream = nullptr;
-  static hipEvent_t ev_fork = nullptr, ev_join = nullptr;
-  if (side_stream == nullptr) {
-    (void)hipStreamCreateWithFlags(&side_stream, hipStreamNonBlocking);
-    (void)hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
-    (void)hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
-  }
+  hipEvent_t ev_fork, ev_join;
+  hipStream_t side_stream = bwd_side_stream(&ev_fork, &ev_join);
   (void)hipEventRecord(ev_fork, stream.stream());
   (void)hipStreamWaitEvent(side_stream, ev_fork, 0);
+  const int dkdv_var = dkdv_variant();
 
 #define LAUNCH_BWD(DD)                                                                              \
   do {                                                                                              \
     constexpr int QB = 64;                                                                          \
     const size_t lds_kv = 2 * sub_bytes<QB, DD>() + NWB * sub_bytes<32, 32>()                       \
                           + 2 * QB * sizeof(float);                                                 \
-    set_lds_limit((const void*)fa_bwd_dkdv_kernel<DD>, lds_kv);                                     \
-    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD>), grid_kv, dim3(NTB), lds_kv, stream.stream(),       \
+    const void* dkdv_fn = dkdv_var == 0 ? (const void*)fa_bwd_dkdv_kernel<DD, 0>                    \
+                        : dkdv_var == 1 ? (const void*)fa_bwd_dkdv_kernel<DD, 1>                    \
+                                        : (const void*)fa_bwd_dkdv_kernel<DD, 2>;                   \
+    set_lds_limit(dkdv_fn, lds_kv);                                                                 \
+    hipLaunchKernelGGL((dkdv_var == 0 ? fa_bwd_dkdv_kernel<DD, 0>                                   \
+                        : dkdv_var == 1 ? fa_bwd_dkdv_kernel<DD, 1>                                 \
+                                        : fa_bwd_dkdv_kernel<DD, 2>),                               \
+                       grid_kv, dim3(NTB), lds_kv, stream.stream(),                                 \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
                        (unsigned short*)dk.data_ptr(), (unsigned short*)dv.data_ptr(),              \
-                       qbs, qts, kbs, kts, dkbs, dkts,                                              \
+                       cu, qbs, qts, kbs, kts, dkbs, dkts,                                          \
                        B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
     HIP_CHECK_LAST();                                                                               \
     const size_t lds_q = 2 * sub_bytes<KVB, DD>() + NWB * sub_bytes<32, 32>()                       \
                          + 2 * NWB * QW * sizeof(float);                                            \
     set_lds_limit((const void*)fa_bwd_dq_kernel<DD>, lds_q);                                        \
-    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), grid_q, dim3(NTB), lds_q, side_stream,           \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD>), grid_q, dim3(NTB), lds_q, side_stream,               \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
                        (unsigned short*)dq.data_ptr(),                                              \
-                       qbs, qts, kbs, kts, dqbs, dqts,                                              \
+                       cu, qbs, qts, kbs, kts, dqbs, dqts,                                          \
                        B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
     HIP_CHECK_LAST();                                                                               \
   } while (0)
@@ -946,6 +1113,77 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
   if (D == 128) LAUNCH_BWD(128);
   else LAUNCH_BWD(64);
 #undef LAUNCH_BWD
+  (void)hipEventRecord(ev_join, side_stream);
+  (void)hipStreamWaitEvent(stream.stream(), ev_join, 0);
+  return {dq, dk, dv};
+}
+
+std::vector<at::Tensor> flash_attn_varlen_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
+                                              at::Tensor out, at::Tensor lse, at::Tensor cu_seqlens,
+                                              long max_seqlen, bool causal, double scale) {
+  using namespace fa;
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3, "varlen: q/k/v must be [total,H,D]");
+  TORCH_CHECK(dout.is_contiguous() && out.is_contiguous() && lse.is_contiguous());
+  const int T = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+  const int Hkv = (int)k.size(1);
+  const int n_seq = (int)cu_seqlens.numel() - 1;
+  auto dq = at::empty({T, Hq, D}, q.options());
+  auto dk = at::empty({T, Hkv, D}, k.options());
+  auto dv = at::empty({T, Hkv, D}, v.options());
+  auto delta = at::empty({Hq, T}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+
+  {  // delta preprocess over packed rows ([1, Hq, T] layout == [Hq, T])
+    const long rows = (long)T * Hq;
+    const int grid = capped_grid((rows + 15) / 16, 4096);
+    hipLaunchKernelGGL(fa_delta_kernel, dim3(grid), dim3(256), 0, stream.stream(),
+                       (const unsigned short*)dout.data_ptr(), (const unsigned short*)out.data_ptr(),
+                       delta.data_ptr<float>(), rows, T, Hq, D);
+    HIP_CHECK_LAST();
+  }
+
+  const dim3 grid_kv(((int)max_seqlen + NWB * QW - 1) / (NWB * QW), n_seq * Hkv);
+  const dim3 grid_q(((int)max_seqlen + NWB * QW - 1) / (NWB * QW), n_seq * Hq);
+  const long qts = q.stride(0), kts = k.stride(0);
+  const long dqts = dq.stride(0), dkts = dk.stride(0);
+  const int* cu = cu_seqlens.data_ptr<int>();
+
+  hipEvent_t ev_fork, ev_join;
+  hipStream_t side_stream = bwd_side_stream(&ev_fork, &ev_join);
+  (void)hipEventRecord(ev_fork, stream.stream());
+  (void)hipStreamWaitEvent(side_stream, ev_fork, 0);
+
+#define LAUNCH_VBWD(DD)                                                                             \
+  do {                                                                                              \
+    constexpr int QB = 64;                                                                          \
+    const size_t lds_kv = 2 * sub_bytes<QB, DD>() + NWB * sub_bytes<32, 32>()                       \
+                          + 2 * QB * sizeof(float);                                                 \
+    set_lds_limit((const void*)fa_bwd_dkdv_kernel<DD, 2, true>, lds_kv);                            \
+    hipLaunchKernelGGL((fa_bwd_dkdv_kernel<DD, 2, true>), grid_kv, dim3(NTB), lds_kv,               \
+                       stream.stream(),                                                             \
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
+                       (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
+                       (unsigned short*)dk.data_ptr(), (unsigned short*)dv.data_ptr(),              \
+                       cu, 0L, qts, 0L, kts, 0L, dkts,                                              \
+                       n_seq, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
+    HIP_CHECK_LAST();                                                                               \
+    const size_t lds_q = 2 * sub_bytes<KVB, DD>() + NWB * sub_bytes<32, 32>()                       \
+                         + 2 * NWB * QW * sizeof(float);                                            \
+    set_lds_limit((const void*)fa_bwd_dq_kernel<DD, true>, lds_q);                                  \
+    hipLaunchKernelGGL((fa_bwd_dq_kernel<DD, true>), grid_q, dim3(NTB), lds_q, side_stream,         \
+                       (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \
+                       (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
+                       (unsigned short*)dq.data_ptr(),                                              \
+                       cu, 0L, qts, 0L, kts, 0L, dqts,                                              \
+                       n_seq, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
+    HIP_CHECK_LAST();                                                                               \
+  } while (0)
+
+  if (D == 128) LAUNCH_VBWD(128);
+  else LAUNCH_VBWD(64);
+#undef LAUNCH_VBWD
   (void)hipEventRecord(ev_join, side_stream);
   (void)hipStreamWaitEvent(stream.stream(), ev_join, 0);
   return {dq, dk, dv};

@@ -136,7 +136,8 @@ class LlamaAttention(nn.Module):
             self.k_norm_weight = nn.Parameter(torch.ones(D))
             self.norm_eps = cfg.rms_norm_eps
 
-    def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
+    def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor,
+                seqlens: Optional[torch.Tensor] = None) -> torch.Tensor:
         B, S, _ = hidden.shape
         qkv = self.qkv_proj(hidden)
         sp_mode = getattr(self, "sp_mode", None)
@@ -144,6 +145,10 @@ class LlamaAttention(nn.Module):
             assert not self.qk_norm and self.pir_factor == 0, (
                 "qk-norm / partial-rotary variants are not wired into the SP "
                 "attention branches yet — run these models without SP"
+            )
+            assert seqlens is None, (
+                "padded attention_mask under sequence parallelism is not wired "
+                "yet — pack the batch (varlen) or drop the mask"
             )
         if sp_mode == "ring_attn":
             # context parallelism: Q stays, K/V blocks travel the xGMI ring
@@ -203,7 +208,7 @@ class LlamaAttention(nn.Module):
             v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
             if D in (64, 128) and hidden.dtype == torch.bfloat16:
                 attn = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
-                                       causal=True, scale=self.scale)
+                                       causal=True, scale=self.scale, seqlens=seqlens)
             else:
                 from ..ops.attention import attention_ref
 
@@ -221,11 +226,11 @@ class LlamaAttention(nn.Module):
                          self.k_norm_weight, self.norm_eps)
             v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
             q, k = apply_rope(q.contiguous(), k.contiguous(), rope_table)
-            attn = flash_attention(q, k, v.contiguous(), causal=True, scale=self.scale)
+            attn = flash_attention(q, k, v.contiguous(), causal=True, scale=self.scale, seqlens=seqlens)
             return self.o_proj(attn.reshape(B, S, Hq * D))
         attn = fused_rope_attention(
             qkv, rope_table, self.num_heads, self.num_kv_heads, self.head_dim,
-            causal=True, scale=self.scale,
+            causal=True, scale=self.scale, seqlens=seqlens,
         )
         # under split_gather SP the column-linear gathered the sequence, so
         # flatten with the attention output's own length, not the input's
@@ -365,14 +370,15 @@ class LlamaDecoderLayer(nn.Module):
         self.post_attention_layernorm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
         self.eps = cfg.rms_norm_eps
 
-    def forward(self, hidden: torch.Tensor, residual: Optional[torch.Tensor], rope_table: torch.Tensor):
+    def forward(self, hidden: torch.Tensor, residual: Optional[torch.Tensor], rope_table: torch.Tensor,
+                seqlens: Optional[torch.Tensor] = None):
         """hidden = normed input to attention; residual = running stream.
 
         Returns (mlp_out, residual') where residual' = residual + attn_out:
         the NEXT junction (residual'' = residual' + mlp_out, then norm) is
         fused by the caller so norm weights stay owned by their layer.
         """
-        attn_out = self.self_attn(hidden, rope_table)
+        attn_out = self.self_attn(hidden, rope_table, seqlens)
         hidden, residual = fused_add_rms_norm(attn_out, residual, self.post_attention_layernorm_weight, self.eps)
         mlp_out = self.mlp(hidden)
         return mlp_out, residual
@@ -402,6 +408,7 @@ class LlamaModel(nn.Module):
         input_ids: Optional[torch.Tensor] = None,
         hidden_states: Optional[torch.Tensor] = None,
         stage_range: Optional[tuple] = None,
+        seqlens: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         """Full forward, or a pipeline-stage slice when ``stage_range`` is set.
 
@@ -435,10 +442,10 @@ class LlamaModel(nn.Module):
             layer = self.layers[i]
             if self.gradient_checkpointing and self.training and i < n_ckpt:
                 out, residual = torch.utils.checkpoint.checkpoint(
-                    layer, hidden, residual, table, use_reentrant=False
+                    layer, hidden, residual, table, seqlens, use_reentrant=False
                 )
             else:
-                out, residual = layer(hidden, residual, table)
+                out, residual = layer(hidden, residual, table, seqlens)
             if i + 1 < end:
                 hidden, residual = fused_add_rms_norm(out, residual, self.layers[i + 1].input_layernorm_weight, self.eps)
             elif is_last:
@@ -482,7 +489,18 @@ class LlamaForCausalLM(nn.Module):
         labels: Optional[torch.Tensor] = None,
         hidden_states: Optional[torch.Tensor] = None,
         pp_chunk: Optional[int] = None,
+        attention_mask: Optional[torch.Tensor] = None,
     ):
+        seqlens = None
+        if attention_mask is not None:
+            # right-padded batches (PADDED_CAUSAL): per-batch valid lengths
+            # flow to the flash kernels; pad rows yield zero states and the
+            # loss must mask pads via labels == -100.
+            from ..ops import seqlens_from_attention_mask
+
+            seqlens = seqlens_from_attention_mask(attention_mask)
+            if input_ids is not None:
+                seqlens = seqlens.to(input_ids.device)
         if pp_chunk is not None:
             stage_range = self.chunk_ranges[pp_chunk]
         else:
@@ -518,7 +536,8 @@ class LlamaForCausalLM(nn.Module):
                     self._sp_labels_shifted = True
                 input_ids = (zigzag_split(input_ids, sp, rank) if zigzag
                              else input_ids[:, rank * shard : (rank + 1) * shard])
-        out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range)
+        out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range,
+                         seqlens=seqlens)
         if stage_range is not None and stage_range[1] < len(self.model.layers):
             return {"hidden_states": out}
         hidden = out

@@ -1,5 +1,5 @@
 from ._kernels import has_kernels, kernels, use_hip
-from .attention import attention_ref, flash_attention, fused_rope_attention
+from .attention import attention_ref, flash_attention, flash_attention_varlen, fused_rope_attention, seqlens_from_attention_mask
 from .layernorm import layer_norm
 from .norm import fused_add_rms_norm, rms_norm, rms_norm_ref
 from .rope import apply_rope, apply_rope_ref, build_rope_table
@@ -11,6 +11,8 @@ __all__ = [
     "kernels",
     "use_hip",
     "flash_attention",
+    "flash_attention_varlen",
+    "seqlens_from_attention_mask",
     "attention_ref",
     "rms_norm",
     "layer_norm",

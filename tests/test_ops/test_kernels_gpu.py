@@ -234,6 +234,76 @@ def test_flash_attention_bwd(B, S, Hq, Hkv, D):
     _bf16_close(dv, vf.grad, rtol=4e-2, atol=4e-2, frac=1e-4)
 
 
+@pytest.mark.parametrize("causal", [True, False])
+def test_flash_attention_padded(causal):
+    """Right-padded batches (seqlens): HIP vs fp32 oracle + zero pad rows."""
+    from colossalai_amd.ops.attention import attention_ref
+
+    torch.manual_seed(11)
+    B, S, Hq, Hkv, D = 3, 320, 4, 2, 128
+    scale = 1.0 / math.sqrt(D)
+    seqlens = torch.tensor([320, 187, 45], dtype=torch.int32, device="cuda")
+    q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    dout = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+
+    out, lse = _C.flash_attn_fwd(q, k, v, causal, scale, seqlens)
+    e = torch.empty(0, device="cuda", dtype=torch.bfloat16)
+    dq, dk, dv = _C.flash_attn_bwd(dout, q, k, v, out, lse, causal, scale,
+                                   e.clone(), e.clone(), e.clone(), seqlens)
+
+    qf = q.float().requires_grad_(True)
+    kf = k.float().requires_grad_(True)
+    vf = v.float().requires_grad_(True)
+    ref = attention_ref(qf, kf, vf, causal, scale, seqlens=seqlens)
+    # pad-row upstream grads are irrelevant (masked): zero them in the oracle
+    mask = (torch.arange(S, device="cuda").view(1, S, 1, 1) < seqlens.view(B, 1, 1, 1)).float()
+    ref.backward(dout.float() * mask)
+    for b in range(B):
+        L = int(seqlens[b])
+        _bf16_close(out[b, :L], ref[b, :L].detach(), rtol=3e-2, atol=3e-2, frac=1e-5)
+        _bf16_close(dq[b, :L], qf.grad[b, :L], rtol=4e-2, atol=4e-2, frac=1e-4)
+        _bf16_close(dk[b, :L], kf.grad[b, :L], rtol=4e-2, atol=4e-2, frac=1e-4)
+        _bf16_close(dv[b, :L], vf.grad[b, :L], rtol=4e-2, atol=4e-2, frac=1e-4)
+        if L < S:
+            assert out[b, L:].float().abs().sum() == 0, "pad rows of O must be zero"
+            assert dk[b, L:].float().abs().sum() == 0, "pad rows of dK must be zero"
+            assert torch.isinf(lse[b, :, L:]).all() and (lse[b, :, L:] < 0).all()
+
+
+def test_flash_attention_varlen_gpu():
+    """Packed ragged batch (cu_seqlens): HIP vs per-sequence fp32 oracle."""
+    from colossalai_amd.ops.attention import attention_ref
+
+    torch.manual_seed(12)
+    Hq, Hkv, D = 4, 2, 128
+    lens = [320, 173, 64, 41]
+    total = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)), dtype=torch.int32, device="cuda")
+    scale = 1.0 / math.sqrt(D)
+    q = torch.randn(total, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(total, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(total, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    dout = torch.randn(total, Hq, D, device="cuda", dtype=torch.bfloat16)
+
+    out, lse = _C.flash_attn_varlen_fwd(q, k, v, cu, max(lens), True, scale)
+    dq, dk, dv = _C.flash_attn_varlen_bwd(dout, q, k, v, out, lse, cu, max(lens), True, scale)
+
+    off = 0
+    for L in lens:
+        qf = q[off:off + L].unsqueeze(0).float().requires_grad_(True)
+        kf = k[off:off + L].unsqueeze(0).float().requires_grad_(True)
+        vf = v[off:off + L].unsqueeze(0).float().requires_grad_(True)
+        ref = attention_ref(qf, kf, vf, True, scale)
+        ref.backward(dout[off:off + L].unsqueeze(0).float())
+        _bf16_close(out[off:off + L], ref[0].detach(), rtol=3e-2, atol=3e-2, frac=1e-5)
+        _bf16_close(dq[off:off + L], qf.grad[0], rtol=4e-2, atol=4e-2, frac=1e-4)
+        _bf16_close(dk[off:off + L], kf.grad[0], rtol=4e-2, atol=4e-2, frac=1e-4)
+        _bf16_close(dv[off:off + L], vf.grad[0], rtol=4e-2, atol=4e-2, frac=1e-4)
+        off += L
+
+
 def test_fused_rope_attention_autograd():
     """End-to-end packed-QKV fused op vs the CPU/fp32 composition."""
     from colossalai_amd.ops import build_rope_table, fused_rope_attention
