@@ -55,6 +55,11 @@ at::Tensor decode_attention_paged(at::Tensor q, at::Tensor kpool, at::Tensor vpo
 at::Tensor moe_combine_fwd(at::Tensor y, at::Tensor inv, at::Tensor topw);
 std::vector<at::Tensor> moe_combine_bwd(at::Tensor dout, at::Tensor y, at::Tensor inv, at::Tensor topw);
 
+// grouped_gemm.hip
+at::Tensor grouped_gemm_fwd(at::Tensor x, at::Tensor w, std::vector<long> offsets);
+at::Tensor grouped_gemm_dgrad(at::Tensor dy, at::Tensor w, std::vector<long> offsets);
+at::Tensor grouped_gemm_wgrad(at::Tensor dy, at::Tensor x, std::vector<long> offsets);
+
 // flash_attn.hip
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale,
                                        c10::optional<at::Tensor> seqlens);
@@ -92,6 +97,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "single-token attention over a paged (block-table) KV pool");
   m.def("moe_combine_fwd", &cai::moe_combine_fwd, "fused MoE un-permute + weighted top-k sum");
   m.def("moe_combine_bwd", &cai::moe_combine_bwd, "MoE combine backward (dy + routing-weight grads)");
+  m.def("grouped_gemm_fwd", &cai::grouped_gemm_fwd, "grouped per-expert GEMM: y = x @ w[g]^T");
+  m.def("grouped_gemm_dgrad", &cai::grouped_gemm_dgrad, "grouped GEMM data grad: dx = dy @ w[g]");
+  m.def("grouped_gemm_wgrad", &cai::grouped_gemm_wgrad, "grouped GEMM weight grad: dw[g] = dy_g^T @ x_g");
   m.def("flash_attn_fwd", &cai::flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA; optional right-padding seqlens)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),

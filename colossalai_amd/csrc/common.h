@@ -99,6 +99,61 @@ DEV_INLINE float block_reduce_max(float v, float* smem) {
   return v;
 }
 
+// ------------------------------------------- subtiled LDS tiles + tr-read
+// Shared by flash_attn.hip / grouped_gemm.hip. "Subtiled row-major":
+// [COLS/16 groups][ROWS][16 bf16], 32 B row pitch, 16 B pad between groups.
+// Row-slice fragment reads (8 consecutive cols at fixed row) are single
+// 16 B vector loads; column fragments use ds_read_b64_tr_b16 (HW-verified
+// weave semantics: per-lane 64-bit reads, 16-lane redistribution
+// out[l][j] = pool[(l&15)+16j] — see flash_attn.hip header / tr16_probe).
+typedef __bf16 bf16x8_v __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4_v __attribute__((ext_vector_type(4)));
+
+template <int ROWS>
+__host__ __device__ constexpr int sub_pitch() { return ROWS * 32 + 16; }
+
+template <int ROWS>
+DEV_INLINE int sub_off(int row, int col) {
+  return (col >> 4) * sub_pitch<ROWS>() + row * 32 + (col & 15) * 2;
+}
+
+template <int ROWS, int COLS>
+__host__ __device__ constexpr int sub_bytes() { return (COLS / 16) * sub_pitch<ROWS>(); }
+
+DEV_INLINE bf16x8_v ld_g16b(const unsigned short* p) {
+  return *reinterpret_cast<const bf16x8_v*>(p);
+}
+
+DEV_INLINE void st_lds16b(char* lds, int byte, bf16x8_v v) {
+  *reinterpret_cast<bf16x8_v*>(lds + byte) = v;
+}
+
+DEV_INLINE bf16x8_v ld_lds16b(const char* lds, int byte) {
+  return *reinterpret_cast<const bf16x8_v*>(lds + byte);
+}
+
+DEV_INLINE bf16x4_v ld_tr16(const char* lds, int byte) {
+  typedef __attribute__((address_space(3))) bf16x4_v* lds_v4p;
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_v4p)(lds + byte));
+}
+
+// Full 8-element column fragment from a subtiled tile: lane l receives
+// tile[k_base + (l>>5)*8 + e][cg_base*16 + (l&31)] for e = 0..7 — the A/B
+// fragment of mfma_32x32x16 with the tile's ROW axis as the MFMA k-dim.
+template <int ROWS>
+DEV_INLINE bf16x8_v ld_frag_tr(const char* tile, int lane, int k_base, int cg_base) {
+  const int m = lane & 15;
+  const int cg = cg_base + ((lane >> 4) & 1);
+  const int row = k_base + ((lane >> 5) * 8) + (m >> 2);
+  const int base = cg * sub_pitch<ROWS>() + (m & 3) * 8;
+  bf16x4_v lo = ld_tr16(tile, base + row * 32);
+  bf16x4_v hi = ld_tr16(tile, base + (row + 4) * 32);
+  bf16x8_v r;
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+  r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+  return r;
+}
+
 // ------------------------------------------------------- grid sizing helper
 // Memory-bound kernels: cap grid and stride (guide §6 G11).
 inline int capped_grid(long total_blocks, int cap = 2048) {
