@@ -42,13 +42,13 @@ constexpr int TN = 128;  // output cols per block
 constexpr int BK = 64;   // contraction step
 constexpr int NT = 256;  // threads per block
 
-// Cooperative staging of a [ROWS][BK] bf16 tile into the subtiled layout.
+// Cooperative staging of a [ROWS][COLS] bf16 tile into the subtiled layout.
 // Global source is row-major with row stride `ld` elements. Rows are
 // clamped to [0, row_hi) (harmless for output-masked rows) unless
 // ZEROPAD, which writes zeros instead (for ragged contraction tails).
-template <int ROWS, bool ZEROPAD>
+template <int ROWS, int COLS, bool ZEROPAD>
 struct GGStage {
-  static constexpr int CPR = BK / 8;           // 16 B chunks per row
+  static constexpr int CPR = COLS / 8;         // 16 B chunks per row
   static constexpr int NCH = ROWS * CPR / NT;  // chunks per thread
   bf16x8_v r[NCH];
 
@@ -126,8 +126,8 @@ __global__ __launch_bounds__(gg::NT) void grouped_gemm_kernel(
 
   const int ksteps = Kin / BK;  // Kin = contraction extent
 
-  GGStage<TM, false> astage;            // token rows (output-masked)
-  GGStage<TRANS_W ? BK : TN, false> bstage;
+  GGStage<TM, BK, false> astage;        // token rows (output-masked)
+  GGStage<TRANS_W ? BK : TN, TRANS_W ? TN : BK, false> bstage;
 
   // prologue
   astage.load(x_base, Kin, grp_lo + row0, grp_hi, 0);
@@ -239,7 +239,7 @@ __global__ __launch_bounds__(gg::NT) void grouped_gemm_wgrad_kernel(
     for (int j = 0; j < 16; ++j) acc[nb][j] = 0.0f;
 
   const int nsteps = (count + BK - 1) / BK;
-  GGStage<BK, true> dystage, xstage;  // ZEROPAD the ragged tail
+  GGStage<BK, TM, true> dystage, xstage;  // ZEROPAD the ragged tail
 
   dystage.load(dY, M, grp_lo, grp_hi, m0);
   xstage.load(X, K, grp_lo, grp_hi, k0);

@@ -69,20 +69,17 @@ class MixtralSparseMoeBlock(nn.Module):
 
     def _experts_forward(self, x: torch.Tensor, expert_ids: torch.Tensor) -> torch.Tensor:
         """x [N, H] grouped so rows of the same LOCAL expert are contiguous;
-        expert_ids [N] gives each row's local expert."""
-        out = torch.zeros_like(x)
+        expert_ids [N] gives each row's local expert. One grouped-GEMM launch
+        per projection (csrc/grouped_gemm.hip) instead of a per-expert loop."""
+        from ..ops.grouped_gemm import grouped_gemm
+
         counts = torch.bincount(expert_ids, minlength=self.num_local_experts)
-        offset = 0
+        offs = [0]
         for e in range(self.num_local_experts):
-            n = int(counts[e])
-            if n == 0:
-                continue
-            seg = x[offset : offset + n]
-            gu = seg @ self.w_gate_up[e].t()
-            act = swiglu(gu)
-            out[offset : offset + n] = act @ self.w_down[e].t()
-            offset += n
-        return out
+            offs.append(offs[-1] + int(counts[e]))
+        gu = grouped_gemm(x.contiguous(), self.w_gate_up, offs)
+        act = swiglu(gu)
+        return grouped_gemm(act.contiguous(), self.w_down, offs)
 
     def forward(self, hidden: torch.Tensor) -> torch.Tensor:
         B, S, H = hidden.shape
@@ -148,10 +145,10 @@ class MixtralDecoderLayer(nn.Module):
         self.post_attention_layernorm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
         self.eps = cfg.rms_norm_eps
 
-    def forward(self, hidden, residual, rope_table):
+    def forward(self, hidden, residual, rope_table, seqlens=None):
         from ..ops import fused_add_rms_norm
 
-        attn_out = self.self_attn(hidden, rope_table)
+        attn_out = self.self_attn(hidden, rope_table, seqlens)
         hidden, residual = fused_add_rms_norm(attn_out, residual, self.post_attention_layernorm_weight, self.eps)
         mlp_out = self.mlp(hidden)
         return mlp_out, residual

@@ -1,4 +1,5 @@
 from ._kernels import has_kernels, kernels, use_hip
+from .grouped_gemm import grouped_gemm
 from .attention import attention_ref, flash_attention, flash_attention_varlen, fused_rope_attention, seqlens_from_attention_mask
 from .layernorm import layer_norm
 from .norm import fused_add_rms_norm, rms_norm, rms_norm_ref
@@ -11,6 +12,7 @@ __all__ = [
     "kernels",
     "use_hip",
     "flash_attention",
+    "grouped_gemm",
     "flash_attention_varlen",
     "seqlens_from_attention_mask",
     "attention_ref",

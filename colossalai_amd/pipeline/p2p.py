@@ -1,11 +1,14 @@
 """Stage-to-stage P2P with cached tensor metadata
-(reference: colossalai/pipeline/p2p.py:539).
+(reference: colossalai/pipeline/p2p.py:364,539).
 
-Protocol: the first exchange in each (peer, tag-direction) sends a small
-int64 header [ndim | dtype-code | dims...]; after both sides have seen one
-message the header is cached and only payloads travel (metadata is constant
-across microbatches). Payloads use blocking dist.send/recv on the default
-group — on ROCm this is RCCL point-to-point over xGMI.
+Protocol: the first exchange in each (peer, direction) sends a fixed-size
+int64 header describing the payload STRUCTURE — n_tensors and per-tensor
+(ndim, dtype, dims) — after which the header is cached and only payloads
+travel (metadata is constant across microbatches). Stage IO may be a single
+tensor or a list/tuple of tensors (multi-tensor models: encoder state +
+mask, MoE aux losses, ...); multi-tensor payloads go out as one batched
+isend group. Payload recv is blocking dist.recv on the default group — on
+ROCm this is RCCL point-to-point over xGMI.
 """
 
 from typing import Dict, Optional, Tuple
@@ -17,7 +20,10 @@ from .stage_manager import PipelineStageManager
 
 __all__ = ["PipelineP2PCommunication"]
 
-_HEADER_LEN = 10
+_MAX_TENSORS = 8
+_MAX_DIMS = 6
+_SLOT = 2 + _MAX_DIMS
+_HEADER_LEN = 1 + _MAX_TENSORS * _SLOT
 _DTYPE_CODES = {
     torch.float32: 0,
     torch.float16: 1,
@@ -50,34 +56,57 @@ class PipelineP2PCommunication:
             w.wait()
         self._pending.clear()
 
-    def _send_tensor(self, tensor: torch.Tensor, peer: int) -> None:
+    def _send_tensor(self, obj, peer: int) -> None:
         # sends are NON-BLOCKING: interleaved/1F1B warmups legitimately have
         # both neighbors sending before anyone receives (rendezvous sends
         # deadlock there). Tensors are kept alive until the work completes.
-        tensor = tensor.contiguous()
-        if not self._send_meta.get(peer, False):
+        tensors = [obj] if isinstance(obj, torch.Tensor) else list(obj)
+        assert 1 <= len(tensors) <= _MAX_TENSORS, "stage IO: 1..8 tensors"
+        tensors = [t.contiguous() for t in tensors]
+        struct = (-1 if isinstance(obj, torch.Tensor) else len(tensors))
+        prev = self._send_meta.get(peer)
+        assert prev is None or prev == struct, (
+            "stage IO structure changed mid-stream (metadata is cached after the "
+            "first microbatch); call clear_meta_cache() on both sides first"
+        )
+        if prev is None:
             header = torch.zeros(_HEADER_LEN, dtype=torch.int64, device=self._device)
-            header[0] = tensor.dim()
-            header[1] = _DTYPE_CODES[tensor.dtype]
-            for i, d in enumerate(tensor.shape):
-                header[2 + i] = d
+            header[0] = len(tensors) if not isinstance(obj, torch.Tensor) else -1
+            for k, t in enumerate(tensors):
+                assert t.dim() <= _MAX_DIMS
+                base = 1 + k * _SLOT
+                header[base] = t.dim()
+                header[base + 1] = _DTYPE_CODES[t.dtype]
+                for i, d in enumerate(t.shape):
+                    header[base + 2 + i] = d
             self._isend(header, peer)
-            self._send_meta[peer] = True
-        self._isend(tensor, peer)
+            self._send_meta[peer] = struct
+        for t in tensors:
+            self._isend(t, peer)
 
-    def _recv_tensor(self, peer: int) -> torch.Tensor:
+    def _recv_tensor(self, peer: int):
         if peer not in self._recv_meta:
             header = torch.zeros(_HEADER_LEN, dtype=torch.int64, device=self._device)
             dist.recv(header, peer)
             header = header.cpu()
-            ndim = int(header[0])
-            dtype = _CODE_DTYPES[int(header[1])]
-            shape = torch.Size(int(header[2 + i]) for i in range(ndim))
-            self._recv_meta[peer] = (shape, dtype)
-        shape, dtype = self._recv_meta[peer]
-        buf = torch.empty(shape, dtype=dtype, device=self._device)
-        dist.recv(buf, peer)
-        return buf
+            n = int(header[0])
+            single = n == -1
+            n = 1 if single else n
+            metas = []
+            for k in range(n):
+                base = 1 + k * _SLOT
+                ndim = int(header[base])
+                dtype = _CODE_DTYPES[int(header[base + 1])]
+                shape = torch.Size(int(header[base + 2 + i]) for i in range(ndim))
+                metas.append((shape, dtype))
+            self._recv_meta[peer] = (single, metas)
+        single, metas = self._recv_meta[peer]
+        bufs = []
+        for shape, dtype in metas:
+            buf = torch.empty(shape, dtype=dtype, device=self._device)
+            dist.recv(buf, peer)
+            bufs.append(buf)
+        return bufs[0] if single else bufs
 
     def clear_meta_cache(self) -> None:
         self._send_meta.clear()

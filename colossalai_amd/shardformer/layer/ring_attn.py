@@ -64,19 +64,35 @@ class RingComm:
         self.send_rank = ranks[(self.rank + 1) % self.world]
         self.recv_rank = ranks[(self.rank - 1) % self.world]
 
-    def send_recv(self, tensors):
-        """Send `tensors` to next rank, receive same-shaped from prev."""
+    def send_recv_async(self, tensors):
+        """Issue the neighbor exchange and return (recvs, works) WITHOUT
+        waiting — callers overlap the next block's compute with the ring hop
+        (reference overlaps on a second stream, attn.py:622-699; RCCL's P2P
+        runs on its own internal stream so the default-stream flash kernels
+        proceed concurrently)."""
         recvs = [torch.empty_like(t) for t in tensors]
+        sends = [t.contiguous() for t in tensors]
         ops = []
-        for t, r in zip(tensors, recvs):
+        for t, r in zip(sends, recvs):
             if self.rank % 2 == 0:
-                ops.append(dist.P2POp(dist.isend, t.contiguous(), self.send_rank, group=self.group))
+                ops.append(dist.P2POp(dist.isend, t, self.send_rank, group=self.group))
                 ops.append(dist.P2POp(dist.irecv, r, self.recv_rank, group=self.group))
             else:
                 ops.append(dist.P2POp(dist.irecv, r, self.recv_rank, group=self.group))
-                ops.append(dist.P2POp(dist.isend, t.contiguous(), self.send_rank, group=self.group))
-        for w in dist.batch_isend_irecv(ops):
+                ops.append(dist.P2POp(dist.isend, t, self.send_rank, group=self.group))
+        works = dist.batch_isend_irecv(ops)
+        self._hold = sends  # keep send buffers alive until waited
+        return recvs, works
+
+    @staticmethod
+    def wait(works):
+        for w in works:
             w.wait()
+
+    def send_recv(self, tensors):
+        """Blocking neighbor exchange."""
+        recvs, works = self.send_recv_async(tensors)
+        self.wait(works)
         return recvs
 
 
@@ -158,6 +174,10 @@ class _RingFlashAttention(torch.autograd.Function):
         cur_k, cur_v = k, v
         for step in range(world):
             src = (rank - step) % world
+            # issue the next hop FIRST so the xGMI transfer rides under the
+            # flash kernels of this step
+            if step + 1 < world:
+                nxt, works = comm.send_recv_async([cur_k, cur_v])
             if not causal:
                 o_blk, l_blk = _block_fwd(q, cur_k, cur_v, False, scale)
                 out, lse = _merge(out, lse, o_blk, l_blk.float())
@@ -178,7 +198,8 @@ class _RingFlashAttention(torch.autograd.Function):
                     out = torch.cat([out[:, :S2], o_new], dim=1)
                     lse = torch.cat([lse[:, :, :S2], l_new], dim=2)
             if step + 1 < world:
-                cur_k, cur_v = comm.send_recv([cur_k, cur_v])
+                comm.wait(works)
+                cur_k, cur_v = nxt
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.group, ctx.causal, ctx.scale, ctx.zigzag = group, causal, scale, zigzag
         return out
@@ -196,9 +217,14 @@ class _RingFlashAttention(torch.autograd.Function):
         cur_k, cur_v = k, v
         cur_dk = torch.zeros_like(k, dtype=torch.float32)
         cur_dv = torch.zeros_like(v, dtype=torch.float32)
-        # after `world` ring steps the (k, dk) pair returns to its owner
+        # after `world` ring steps the (k, dk) pair returns to its owner.
+        # k/v prefetch overlaps the block backward; dk/dv (which depend on
+        # the block compute) are exchanged in bf16 afterwards — wire format
+        # matches the grad dtype, halving ring bytes vs fp32
         for step in range(world):
             src = (rank - step) % world
+            if step + 1 < world:
+                nxt_kv, works_kv = comm.send_recv_async([cur_k, cur_v])
             if not causal or (not zigzag and src <= rank):
                 blk_causal = causal and (src == rank)
                 dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k, cur_v, out, lse, blk_causal, scale)
@@ -224,7 +250,14 @@ class _RingFlashAttention(torch.autograd.Function):
                     dq[:, S2:] += dq_b.float()
                     cur_dk += dk_b.float()
                     cur_dv += dv_b.float()
-            cur_k, cur_v, cur_dk, cur_dv = comm.send_recv([cur_k, cur_v, cur_dk, cur_dv])
+            if step + 1 < world:
+                comm.wait(works_kv)
+                dkv = comm.send_recv([cur_dk.to(k.dtype), cur_dv.to(v.dtype)])
+                cur_k, cur_v = nxt_kv
+                cur_dk, cur_dv = dkv[0].float(), dkv[1].float()
+            else:
+                dkv = comm.send_recv([cur_dk.to(k.dtype), cur_dv.to(v.dtype)])
+                cur_dk, cur_dv = dkv[0].float(), dkv[1].float()
         # one full cycle: cur_dk/cur_dv now hold this rank's own grads
         return dq.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype), None, None, None, None
 

@@ -491,3 +491,52 @@ def test_fp8_linear_gpu():
     torch.nn.functional.linear(xr, wr).backward(d.float())
     _bf16_close(x.grad, xr.grad, rtol=3e-2, atol=3e-2, frac=1e-4)
     _bf16_close(w.grad, wr.grad, rtol=3e-2, atol=3e-2, frac=1e-4)
+
+
+# ---------------------------------------------------------- grouped GEMM
+def test_grouped_gemm_fwd_bwd():
+    """MoE grouped GEMM (fwd/dgrad/wgrad) vs the per-expert fp32 loop,
+    including empty and ragged groups."""
+    torch.manual_seed(13)
+    E, M, K = 4, 256, 128
+    counts = [300, 0, 127, 37]
+    N = sum(counts)
+    offs = [0]
+    for c in counts:
+        offs.append(offs[-1] + c)
+    x = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(E, M, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    dy = torch.randn(N, M, device="cuda", dtype=torch.bfloat16)
+
+    y = _C.grouped_gemm_fwd(x, w, offs)
+    dx = _C.grouped_gemm_dgrad(dy, w, offs)
+    dw = _C.grouped_gemm_wgrad(dy, x, offs)
+
+    xf, wf, dyf = x.float(), w.float(), dy.float()
+    for g in range(E):
+        lo, hi = offs[g], offs[g + 1]
+        if hi > lo:
+            _bf16_close(y[lo:hi], xf[lo:hi] @ wf[g].t(), rtol=3e-2, atol=3e-2, frac=1e-4)
+            _bf16_close(dx[lo:hi], dyf[lo:hi] @ wf[g], rtol=3e-2, atol=3e-2, frac=1e-4)
+            torch.testing.assert_close(dw[g].cpu(), (dyf[lo:hi].t() @ xf[lo:hi]).cpu(),
+                                       rtol=3e-2, atol=3e-1)
+        else:
+            assert dw[g].abs().sum() == 0
+
+
+def test_grouped_gemm_autograd():
+    from colossalai_amd.ops import grouped_gemm
+
+    torch.manual_seed(14)
+    E, M, K = 2, 128, 64
+    offs = [0, 150, 288]
+    x = torch.randn(288, K, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = (torch.randn(E, M, K, device="cuda", dtype=torch.bfloat16) * 0.1).requires_grad_(True)
+    y = grouped_gemm(x, w, offs)
+    y.sum().backward()
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    for g in range(E):
+        (xf[offs[g]:offs[g+1]] @ wf[g].t()).sum().backward(retain_graph=True)
+    _bf16_close(x.grad, xf.grad, rtol=3e-2, atol=3e-2, frac=1e-4)
+    _bf16_close(w.grad, wf.grad, rtol=3e-2, atol=5e-1, frac=1e-3)
