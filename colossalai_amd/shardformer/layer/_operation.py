@@ -250,3 +250,187 @@ class _AllToAll(torch.autograd.Function):
 
 def all_to_all_comm(x, group, scatter_dim=2, gather_dim=1):
     return _AllToAll.apply(x, group, scatter_dim, gather_dim)
+
+
+# ------------------------------------------------- SP "ring" mode linears
+# Ring-pipelined equivalents of the split_gather collectives: instead of a
+# monolithic all-gather / reduce-scatter around the matmul, sequence shards
+# travel the sp ring one hop per step and each hop's xGMI transfer rides
+# under the partial GEMM (reference: _operation.py:418 _ring_as_gather,
+# :646 _ring_as_reducescatter). Layout-identical to split_gather; only the
+# comm schedule differs.
+
+
+class _RingGatherLinearCol(torch.autograd.Function):
+    """Column-parallel linear with ring-gathered sequence input.
+
+    forward:  y[:, full seq] = allgather_seq(x) @ W^T (+ b), computed one
+              shard per ring step.
+    backward: dx = ring-reduce-scatter of dY @ W; dW accumulated as x shards
+              travel a second ring.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, group, dim):
+        import torch.nn.functional as F
+
+        world = dist.get_world_size(group) if group is not None else 1
+        if world == 1:
+            ctx.save_for_backward(x, weight)
+            ctx.meta = (bias is not None, group, dim, 1)
+            return F.linear(x, weight, bias)
+        from .ring_attn import RingComm
+
+        comm = RingComm(group)
+        rank = comm.rank
+        shard = x.shape[dim]
+        out_shape = list(x.shape)
+        out_shape[dim] = shard * world
+        out_shape[-1] = weight.shape[0]
+        y = x.new_empty(out_shape)
+        cur = x.contiguous()
+        for step in range(world):
+            src = (rank - step) % world
+            if step + 1 < world:
+                nxt, works = comm.send_recv_async([cur])
+            y.narrow(dim, src * shard, shard).copy_(F.linear(cur, weight, bias))
+            if step + 1 < world:
+                comm.wait(works)
+                cur = nxt[0]
+        ctx.save_for_backward(x, weight)
+        ctx.meta = (bias is not None, group, dim, world)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        use_bias, group, dim, world = ctx.meta
+        dy = dy.contiguous()
+        if world == 1:
+            dy2 = dy.reshape(-1, dy.shape[-1])
+            x2 = x.reshape(-1, x.shape[-1])
+            db = dy2.sum(0) if use_bias else None
+            return dy @ weight, dy2.t() @ x2, db, None, None
+        from .ring_attn import RingComm
+
+        comm = RingComm(group)
+        rank = comm.rank
+        shard = x.shape[dim]
+
+        def contrib(j):
+            return (dy.narrow(dim, j * shard, shard) @ weight).contiguous()
+
+        # dx: ring reduce-scatter — the partial that ends on rank r is r's
+        # own seq shard; each hop overlaps the next partial GEMM
+        b = contrib((rank - 1) % world)
+        for t in range(1, world):
+            recvs, works = comm.send_recv_async([b])
+            c = contrib((rank - t - 1) % world)
+            comm.wait(works)
+            b = recvs[0] + c
+        dx = b
+
+        # dW (+ db): x shards ring-gathered, one partial accumulate per hop
+        cur = x.contiguous()
+        dw = None
+        for step in range(world):
+            src = (rank - step) % world
+            if step + 1 < world:
+                nxt, works = comm.send_recv_async([cur])
+            dys = dy.narrow(dim, src * shard, shard).reshape(-1, dy.shape[-1])
+            piece = dys.t() @ cur.reshape(-1, x.shape[-1])
+            dw = piece if dw is None else dw + piece
+            if step + 1 < world:
+                comm.wait(works)
+                cur = nxt[0]
+        db = dy.reshape(-1, dy.shape[-1]).sum(0) if use_bias else None
+        return dx, dw, db, None, None
+
+
+def ring_gather_linear_col(x, weight, bias, group, dim=1):
+    return _RingGatherLinearCol.apply(x, weight, bias, group, dim)
+
+
+class _RingReduceScatterLinearRow(torch.autograd.Function):
+    """Row-parallel linear with ring-reduce-scattered sequence output.
+
+    forward:  y_shard = reducescatter_seq(x @ W^T); one partial GEMM per
+              ring step, hop under the next GEMM. bias added once at the end.
+    backward: dX[:, full seq] = ring-gather(dy) @ W per shard; dW accumulated
+              on the same ring pass.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, group, dim):
+        import torch.nn.functional as F
+
+        world = dist.get_world_size(group) if group is not None else 1
+        ctx.save_for_backward(x, weight)
+        if world == 1:
+            ctx.meta = (bias is not None, group, dim, 1)
+            y = F.linear(x, weight)
+            return y + bias if bias is not None else y
+        from .ring_attn import RingComm
+
+        comm = RingComm(group)
+        rank = comm.rank
+        assert x.shape[dim] % world == 0
+        shard = x.shape[dim] // world
+        ctx.meta = (bias is not None, group, dim, world)
+
+        def contrib(j):
+            return F.linear(x.narrow(dim, j * shard, shard), weight).contiguous()
+
+        b = contrib((rank - 1) % world)
+        for t in range(1, world):
+            recvs, works = comm.send_recv_async([b])
+            c = contrib((rank - t - 1) % world)
+            comm.wait(works)
+            b = recvs[0] + c
+        if bias is not None:
+            b = b + bias
+        return b
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        use_bias, group, dim, world = ctx.meta
+        dy = dy.contiguous()
+        if world == 1:
+            dy2 = dy.reshape(-1, dy.shape[-1])
+            x2 = x.reshape(-1, x.shape[-1])
+            db = dy2.sum(0) if use_bias else None
+            return dy @ weight, dy2.t() @ x2, db, None, None
+        from .ring_attn import RingComm
+
+        comm = RingComm(group)
+        rank = comm.rank
+        shard = dy.shape[dim]
+        dx_shape = list(x.shape)
+        dx = x.new_empty(dx_shape)
+        dw = None
+        cur = dy
+        # ring-gather dy; per hop: dx slice GEMM + dW partial accumulate
+        for step in range(world):
+            src = (rank - step) % world
+            if step + 1 < world:
+                nxt, works = comm.send_recv_async([cur])
+            dx.narrow(dim, src * shard, shard).copy_(cur @ weight)
+            xs = x.narrow(dim, src * shard, shard).reshape(-1, x.shape[-1])
+            piece = cur.reshape(-1, dy.shape[-1]).t() @ xs
+            dw = piece if dw is None else dw + piece
+            if step + 1 < world:
+                comm.wait(works)
+                cur = nxt[0]
+        # bias is applied once per FULL output: grad sums over every shard,
+        # which equals the sum over this rank's shard all-reduced by the
+        # ring-gather above (each rank saw all dy shards)
+        db = None
+        if use_bias:
+            db = dy.reshape(-1, dy.shape[-1]).sum(0)
+            dist.all_reduce(db, group=group)
+        return dx, dw, db, None, None
+
+
+def ring_reducescatter_linear_row(x, weight, bias, group, dim=1):
+    return _RingReduceScatterLinearRow.apply(x, weight, bias, group, dim)

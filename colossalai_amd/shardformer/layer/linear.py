@@ -118,7 +118,12 @@ class Linear1D_Col(ParallelModule):
         return _gather_rows(self.weight.data, self.process_group, self.split_sizes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if self.seq_parallel_mode == "split_gather":
+        if self.seq_parallel_mode == "ring":
+            from ._operation import ring_gather_linear_col
+
+            out = ring_gather_linear_col(x, self.weight, self.bias, self.process_group,
+                                         self.seq_parallel_dim)
+        elif self.seq_parallel_mode == "split_gather":
             x = gather_forward_reduce_scatter_backward(x, self.seq_parallel_dim, self.process_group)
             out = torch.nn.functional.linear(x, self.weight, self.bias)
         else:
@@ -190,6 +195,11 @@ class Linear1D_Row(ParallelModule):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if not self.parallel_input:
             x = split_forward_gather_backward(x, -1, self.process_group)
+        if self.seq_parallel_mode == "ring":
+            from ._operation import ring_reducescatter_linear_row
+
+            return ring_reducescatter_linear_row(x, self.weight, self.bias, self.process_group,
+                                                 self.seq_parallel_dim)
         out = torch.nn.functional.linear(x, self.weight)
         if self.seq_parallel_mode == "split_gather":
             out = reduce_scatter_forward_gather_backward(out, self.seq_parallel_dim, self.process_group)

@@ -51,10 +51,10 @@ class LlamaPolicy(Policy):
             )
         sp_sg = (
             self.shard_config.enable_sequence_parallelism
-            and self.shard_config.sequence_parallelism_mode == "split_gather"
+            and self.shard_config.sequence_parallelism_mode in ("split_gather", "ring")
         )
         if sp_sg:
-            assert self.shard_config.tensor_parallel_size > 1 or True, "split_gather uses the tp/sp group"
+            assert self.shard_config.tensor_parallel_size > 1 or True, "split_gather/ring use the tp/sp group"
         if sp_a2a:
             policy[LlamaAttention] = ModulePolicyDescription(
                 attribute_replacement={
@@ -155,8 +155,10 @@ class LlamaForCausalLMPolicy(LlamaPolicy):
                 self.model.sp_group = self.shard_config.sequence_parallel_process_group
                 self.model.sp_mode = mode
                 self.model.sp_zigzag = self.shard_config.sp_zigzag and mode == "ring_attn"
-            elif mode == "split_gather":
-                # split_gather reuses the tp group (reference semantics)
+            elif mode in ("split_gather", "ring"):
+                # split_gather/ring reuse the tp group (reference semantics);
+                # ring differs only in the collective schedule inside the
+                # linears, so the model-side layout handling is identical
                 group = self.shard_config.tensor_parallel_process_group
                 self.model.sp_group = group
                 self.model.sp_mode = "split_gather"

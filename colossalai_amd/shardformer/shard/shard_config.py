@@ -8,7 +8,7 @@ from torch.distributed import ProcessGroup
 
 __all__ = ["ShardConfig"]
 
-SUPPORT_SP_MODE = ["split_gather", "all_to_all", "ring_attn"]
+SUPPORT_SP_MODE = ["split_gather", "ring", "all_to_all", "ring_attn"]
 
 
 @dataclass
@@ -38,7 +38,7 @@ class ShardConfig:
     def sequence_parallel_size(self) -> int:
         if not self.enable_sequence_parallelism:
             return 1
-        if self.sequence_parallelism_mode in ("split_gather", "ring_attn") and self.sequence_parallel_process_group is None:
+        if self.sequence_parallelism_mode in ("split_gather", "ring", "ring_attn") and self.sequence_parallel_process_group is None:
             return self.tensor_parallel_size
         if self.sequence_parallel_process_group is None:
             return 1
