@@ -24,6 +24,10 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "ViTForImageClassification": ("colossalai_amd.shardformer.policies.vit", "ViTForImageClassificationPolicy"),
     "FalconForCausalLM": ("colossalai_amd.shardformer.policies.falcon", "FalconForCausalLMPolicy"),
     "DeepseekForCausalLM": ("colossalai_amd.shardformer.policies.deepseek", "DeepseekForCausalLMPolicy"),
+    "BloomForCausalLM": ("colossalai_amd.shardformer.policies.bloom", "BloomForCausalLMPolicy"),
+    "GPTJForCausalLM": ("colossalai_amd.shardformer.policies.gptj", "GPTJForCausalLMPolicy"),
+    "WhisperForConditionalGeneration": ("colossalai_amd.shardformer.policies.whisper", "WhisperForConditionalGenerationPolicy"),
+    "CohereForCausalLM": ("colossalai_amd.shardformer.policies.cohere", "CohereForCausalLMPolicy"),
 }
 
 

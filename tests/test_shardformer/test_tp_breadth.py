@@ -1,0 +1,128 @@
+"""TP policies for the remaining native families (BLOOM, GPT-J, Whisper,
+Cohere) vs their unsharded oracles (CPU/gloo, tp=2)
+(reference registry: colossalai/shardformer/policies/auto_policy.py:30-291)."""
+
+import copy
+
+import torch
+import torch.distributed as dist
+
+import colossalai_amd
+from colossalai_amd.shardformer import ShardConfig, ShardFormer
+from colossalai_amd.testing import assert_close_loose, rerun_if_address_is_in_use, spawn
+
+
+def _shard(model):
+    sharded, _ = ShardFormer(ShardConfig(tensor_parallel_process_group=dist.group.WORLD)).optimize(model)
+    return sharded
+
+
+def run_tp_bloom(rank, world_size, port):
+    from colossalai_amd.models.bloom import BloomConfig, BloomForCausalLM
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = BloomConfig(vocab_size=256, hidden_size=64, num_hidden_layers=2, num_attention_heads=4)
+    ref = BloomForCausalLM(cfg)
+    model = _shard(copy.deepcopy(ref))
+    attn = model.transformer.h[0].self_attention
+    assert attn.num_heads == 2 and attn.slopes.numel() == 2
+
+    x = torch.randint(0, 256, (2, 16))
+    out = model(x, labels=x)
+    out_ref = ref(x, labels=x)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.transformer.h[0].ln1_w.grad, ref.transformer.h[0].ln1_w.grad,
+                       rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+def run_tp_gptj(rank, world_size, port):
+    from colossalai_amd.models.gptj import GPTJConfig, GPTJForCausalLM
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = GPTJConfig(vocab_size=256, n_embd=64, n_layer=2, n_head=4, rotary_dim=8, n_positions=64)
+    ref = GPTJForCausalLM(cfg)
+    model = _shard(copy.deepcopy(ref))
+    assert model.transformer.h[0].attn.num_heads == 2
+
+    x = torch.randint(0, 256, (2, 16))
+    out = model(x, labels=x)
+    out_ref = ref(x, labels=x)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.transformer.h[0].ln_1_weight.grad,
+                       ref.transformer.h[0].ln_1_weight.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+def run_tp_whisper(rank, world_size, port):
+    from colossalai_amd.models.whisper import WhisperConfig, WhisperForConditionalGeneration
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = WhisperConfig(vocab_size=256, num_mel_bins=16, d_model=64, encoder_layers=2,
+                        decoder_layers=2, num_heads=4, d_ff=128, max_source_positions=32,
+                        max_target_positions=32, decoder_start_token_id=1, pad_token_id=0)
+    ref = WhisperForConditionalGeneration(cfg)
+    model = _shard(copy.deepcopy(ref))
+    assert model.encoder.layers[0].self_attn.num_heads == 2
+    assert model.decoder.layers[0].cross_attn.num_heads == 2
+
+    mel = torch.randn(2, 16, 64)  # [B, mel, frames]
+    labels = torch.randint(2, 256, (2, 8))
+    out = model(mel, labels=labels)
+    out_ref = ref(mel, labels=labels)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.encoder.layers[0].self_ln_w.grad,
+                       ref.encoder.layers[0].self_ln_w.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+def run_tp_cohere(rank, world_size, port):
+    from colossalai_amd.models.cohere import CohereConfig, CohereForCausalLM
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = CohereConfig(vocab_size=256, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                       num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    ref = CohereForCausalLM(cfg)
+    model = _shard(copy.deepcopy(ref))
+    attn = model.model.layers[0].self_attn
+    assert attn.num_heads == 2 and attn.num_kv_heads == 1
+
+    x = torch.randint(0, 256, (2, 16))
+    out = model(x, labels=x)
+    out_ref = ref(x, labels=x)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.model.layers[0].ln_weight.grad,
+                       ref.model.layers[0].ln_weight.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_tp_bloom():
+    spawn(run_tp_bloom, 2)
+
+
+@rerun_if_address_is_in_use()
+def test_tp_gptj():
+    spawn(run_tp_gptj, 2)
+
+
+@rerun_if_address_is_in_use()
+def test_tp_whisper():
+    spawn(run_tp_whisper, 2)
+
+
+@rerun_if_address_is_in_use()
+def test_tp_cohere():
+    spawn(run_tp_cohere, 2)
