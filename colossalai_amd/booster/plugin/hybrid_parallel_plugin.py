@@ -432,13 +432,16 @@ class HybridParallelPlugin(Plugin):
         Convention: the model exposes ``.model.layers`` (decoder stack) — the
         native model family contract; HF models get stage-aware policies.
         """
-        inner = model.model if hasattr(model, "model") else model
-        assert hasattr(inner, "layers"), "pipeline parallelism needs a .layers decoder stack"
-        n_layers = len(inner.layers)
+        inner, layers = self._find_layer_stack(model)
+        n_layers = len(layers)
         # tied embed/lm_head detection BEFORE stubbing: first stage keeps the
-        # weight via embed_tokens, last via lm_head; grads sync over embed_group
+        # weight via the embedding, last via lm_head; grads sync over embed_group
         head = getattr(model, "lm_head", None)
-        embed = getattr(inner, "embed_tokens", None)
+        embed = None
+        for attr in ("embed_tokens", "wte", "word_embeddings"):
+            if hasattr(inner, attr):
+                embed = getattr(inner, attr)
+                break
         is_tied = (
             head is not None and embed is not None
             and getattr(head, "weight", None) is not None and head.weight is embed.weight
@@ -471,9 +474,9 @@ class HybridParallelPlugin(Plugin):
             model.stage_range = model.chunk_ranges[0]
             for i in range(n_layers):
                 if i not in held:
-                    inner.layers[i] = _StageStub()
-            if not self.stage_manager.is_first_stage() and hasattr(inner, "embed_tokens"):
-                inner.embed_tokens = _StageStub()
+                    layers[i] = _StageStub()
+            if not self.stage_manager.is_first_stage():
+                self._stub_embeddings(inner)
             if not self.stage_manager.is_last_stage(self.num_model_chunks - 1) and hasattr(model, "lm_head"):
                 model.lm_head = _StageStub()
             return
@@ -481,11 +484,29 @@ class HybridParallelPlugin(Plugin):
         model.stage_range = (start, end)
         for i in range(n_layers):
             if not (start <= i < end):
-                inner.layers[i] = _StageStub()
-        if not self.stage_manager.is_first_stage() and hasattr(inner, "embed_tokens"):
-            inner.embed_tokens = _StageStub()
+                layers[i] = _StageStub()
+        if not self.stage_manager.is_first_stage():
+            self._stub_embeddings(inner)
         if not self.stage_manager.is_last_stage() and hasattr(model, "lm_head"):
             model.lm_head = _StageStub()
+
+    @staticmethod
+    def _find_layer_stack(model):
+        """Locate the decoder stack: .model.layers (llama-family),
+        .transformer.layers (gpt2) or .transformer.h (gptj/bloom)."""
+        for path, lattr in (("model", "layers"), ("transformer", "layers"),
+                            ("transformer", "h"), (None, "layers"), (None, "h")):
+            base = getattr(model, path, None) if path else model
+            if base is not None and hasattr(base, lattr):
+                return base, getattr(base, lattr)
+        raise AssertionError("pipeline parallelism needs a decoder stack "
+                             "(.model.layers / .transformer.layers / .transformer.h)")
+
+    @staticmethod
+    def _stub_embeddings(inner) -> None:
+        for attr in ("embed_tokens", "wte", "wpe", "word_embeddings"):
+            if hasattr(inner, attr) and isinstance(getattr(inner, attr), nn.Module)                     and not isinstance(getattr(inner, attr), _StageStub):
+                setattr(inner, attr, _StageStub())
 
     @staticmethod
     def _rebuild_param_groups(optimizer: Optimizer, model: nn.Module,

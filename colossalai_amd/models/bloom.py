@@ -117,14 +117,20 @@ class BloomModel(nn.Module):
         self.ln_f_b = nn.Parameter(torch.zeros(d))
         self.gradient_checkpointing = False
 
-    def forward(self, input_ids):
-        hidden = layer_norm(self.word_embeddings(input_ids), self.emb_ln_w, self.emb_ln_b,
-                            self.cfg.layer_norm_epsilon)
-        for blk in self.h:
+    def forward(self, input_ids=None, hidden_states=None, stage_range=None):
+        start, end = stage_range if stage_range is not None else (0, len(self.h))
+        if start == 0:
+            hidden = layer_norm(self.word_embeddings(input_ids), self.emb_ln_w, self.emb_ln_b,
+                                self.cfg.layer_norm_epsilon)
+        else:
+            hidden = hidden_states
+        for blk in self.h[start:end]:
             if self.gradient_checkpointing and self.training:
                 hidden = torch.utils.checkpoint.checkpoint(blk, hidden, use_reentrant=False)
             else:
                 hidden = blk(hidden)
+        if end < len(self.h):
+            return hidden  # stage boundary
         return layer_norm(hidden, self.ln_f_w, self.ln_f_b, self.cfg.layer_norm_epsilon)
 
 
@@ -146,8 +152,12 @@ class BloomForCausalLM(nn.Module):
     def gradient_checkpointing_enable(self, ratio: float = 1.0):
         self.transformer.gradient_checkpointing = True
 
-    def forward(self, input_ids, labels: Optional[torch.Tensor] = None):
-        hidden = self.transformer(input_ids)
+    def forward(self, input_ids=None, labels: Optional[torch.Tensor] = None, hidden_states=None):
+        stage_range = getattr(self, "stage_range", None)
+        out = self.transformer(input_ids, hidden_states=hidden_states, stage_range=stage_range)
+        if stage_range is not None and stage_range[1] < len(self.transformer.h):
+            return {"hidden_states": out}
+        hidden = out
         if labels is not None:
             from ..ops.fused_ce import fused_linear_cross_entropy
 

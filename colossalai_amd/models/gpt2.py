@@ -109,15 +109,21 @@ class GPT2Model(nn.Module):
         self.ln_f_bias = nn.Parameter(torch.zeros(cfg.n_embd))
         self.gradient_checkpointing = False
 
-    def forward(self, input_ids):
-        B, S = input_ids.shape
-        pos = torch.arange(S, device=input_ids.device)
-        hidden = self.wte(input_ids) + self.wpe(pos)[None]
-        for layer in self.layers:
+    def forward(self, input_ids=None, hidden_states=None, stage_range=None):
+        start, end = stage_range if stage_range is not None else (0, len(self.layers))
+        if start == 0:
+            B, S = input_ids.shape
+            pos = torch.arange(S, device=input_ids.device)
+            hidden = self.wte(input_ids) + self.wpe(pos)[None]
+        else:
+            hidden = hidden_states
+        for layer in self.layers[start:end]:
             if self.gradient_checkpointing and self.training:
                 hidden = torch.utils.checkpoint.checkpoint(layer, hidden, use_reentrant=False)
             else:
                 hidden = layer(hidden)
+        if end < len(self.layers):
+            return hidden  # stage boundary
         return layer_norm(hidden, self.ln_f_weight, self.ln_f_bias, self.cfg.layer_norm_epsilon)
 
 
@@ -139,8 +145,12 @@ class GPT2LMHeadModel(nn.Module):
     def gradient_checkpointing_enable(self, ratio: float = 1.0):
         self.transformer.gradient_checkpointing = True
 
-    def forward(self, input_ids, labels: Optional[torch.Tensor] = None):
-        hidden = self.transformer(input_ids)
+    def forward(self, input_ids=None, labels: Optional[torch.Tensor] = None, hidden_states=None):
+        stage_range = getattr(self, "stage_range", None)
+        out = self.transformer(input_ids, hidden_states=hidden_states, stage_range=stage_range)
+        if stage_range is not None and stage_range[1] < len(self.transformer.layers):
+            return {"hidden_states": out}
+        hidden = out
         loss = None
         if labels is not None:
             from ..ops.fused_ce import fused_linear_cross_entropy

@@ -116,13 +116,16 @@ class GPTJModel(nn.Module):
         self.ln_f_bias = nn.Parameter(torch.zeros(cfg.n_embd))
         self.gradient_checkpointing = False
 
-    def forward(self, input_ids):
-        hidden = self.wte(input_ids)
-        for blk in self.h:
+    def forward(self, input_ids=None, hidden_states=None, stage_range=None):
+        start, end = stage_range if stage_range is not None else (0, len(self.h))
+        hidden = self.wte(input_ids) if start == 0 else hidden_states
+        for blk in self.h[start:end]:
             if self.gradient_checkpointing and self.training:
                 hidden = torch.utils.checkpoint.checkpoint(blk, hidden, use_reentrant=False)
             else:
                 hidden = blk(hidden)
+        if end < len(self.h):
+            return hidden  # stage boundary
         return layer_norm(hidden, self.ln_f_weight, self.ln_f_bias, self.cfg.layer_norm_epsilon)
 
 
@@ -143,9 +146,12 @@ class GPTJForCausalLM(nn.Module):
     def gradient_checkpointing_enable(self, ratio: float = 1.0):
         self.transformer.gradient_checkpointing = True
 
-    def forward(self, input_ids, labels: Optional[torch.Tensor] = None):
-        hidden = self.transformer(input_ids)
-        logits = self.lm_head(hidden)
+    def forward(self, input_ids=None, labels: Optional[torch.Tensor] = None, hidden_states=None):
+        stage_range = getattr(self, "stage_range", None)
+        out = self.transformer(input_ids, hidden_states=hidden_states, stage_range=stage_range)
+        if stage_range is not None and stage_range[1] < len(self.transformer.h):
+            return {"hidden_states": out}
+        logits = self.lm_head(out)
         loss = None
         if labels is not None:
             loss = F.cross_entropy(logits[:, :-1].float().reshape(-1, logits.shape[-1]),
