@@ -1,6 +1,7 @@
 from .adafactor import Adafactor
 from .came import CAME
 from .cpu_adam import CPUAdam
+from .distributed_factored import DistributedAdafactor, DistributedCAME, DistributedGaLoreAdamW
 from .distributed_lamb import DistributedLamb
 from .fused_adam import FusedAdam
 from .fused_sgd import FusedSGD
@@ -10,7 +11,10 @@ from .lamb import Lamb
 from .lars import Lars
 
 __all__ = ["FusedAdam",
-    "FusedSGD", "DistributedLamb", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
+    "FusedSGD", "DistributedLamb",
+    "DistributedAdafactor",
+    "DistributedCAME",
+    "DistributedGaLoreAdamW", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
 
 
 def cast_to_distributed(optimizer):
@@ -18,12 +22,14 @@ def cast_to_distributed(optimizer):
     (reference: colossalai/nn/optimizer/__init__.py cast_to_distributed).
     Returns the original optimizer unchanged otherwise; call
     ``setup_distributed(...)`` on the result before stepping."""
-    mapping = {Lamb: DistributedLamb}
+    mapping = {Lamb: DistributedLamb, Adafactor: DistributedAdafactor, CAME: DistributedCAME,
+               GaLoreAdamW: DistributedGaLoreAdamW}
     cls = mapping.get(type(optimizer))
     if cls is None:
         return optimizer
     # rebuild from the same param groups + defaults
     groups = [dict(g) for g in optimizer.param_groups]
-    kwargs = {k: v for k, v in optimizer.defaults.items() if k in ("lr", "betas", "eps", "weight_decay")}
-    return cls([{"params": g["params"], **{k: g.get(k, kwargs.get(k)) for k in kwargs}} for g in groups],
+    keys = set(optimizer.defaults)
+    kwargs = dict(optimizer.defaults)
+    return cls([{"params": g["params"], **{k: g.get(k, kwargs.get(k)) for k in keys}} for g in groups],
                **kwargs)

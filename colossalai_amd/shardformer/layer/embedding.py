@@ -41,6 +41,8 @@ class VocabParallelEmbedding1D(ParallelModule):
         layer.vocab_start = layer.rank * layer.part
         layer.vocab_end = layer.vocab_start + layer.part
         layer.weight = Parameter(module.weight.data[layer.vocab_start : layer.vocab_end].contiguous())
+        layer.weight.tp_sharded = True
+        layer.weight.tp_dim = 0
         return layer
 
     def gather_weight(self) -> torch.Tensor:

@@ -105,6 +105,7 @@ class Linear1D_Col(ParallelModule):
         layer.split_sizes = kwargs.get("split_sizes", None)
         layer.weight = Parameter(_shard_rows(module.weight.data, process_group, layer.split_sizes))
         layer.weight.tp_sharded = True  # distributed optimizers: norms need the tp all-reduce
+        layer.weight.tp_dim = 0
         layer.bias = (
             Parameter(_shard_rows(module.bias.data.unsqueeze(-1), process_group, layer.split_sizes).squeeze(-1))
             if module.bias is not None
@@ -112,6 +113,7 @@ class Linear1D_Col(ParallelModule):
         )
         if layer.bias is not None:
             layer.bias.tp_sharded = True
+            layer.bias.tp_dim = 0
         return layer
 
     def gather_weight(self) -> torch.Tensor:
@@ -180,6 +182,7 @@ class Linear1D_Row(ParallelModule):
             w = module.weight.data.chunk(layer.world, dim=1)[rank].contiguous()
         layer.weight = Parameter(w)
         layer.weight.tp_sharded = True
+        layer.weight.tp_dim = 1
         # bias applied once (after reduce), kept replicated
         layer.bias = Parameter(module.bias.data.clone()) if module.bias is not None else None
         return layer
