@@ -133,6 +133,7 @@ class GeminiPlugin(Plugin):
         chunk_config_dict: Optional[dict] = None,
         chunk_init_device: Optional[torch.device] = None,
         placement_policy: str = "static",
+        chunk_size_search: bool = False,
         shard_param_frac: float = 0.0,
         offload_optim_frac: float = 0.0,
         offload_param_frac: float = 0.0,
@@ -170,6 +171,7 @@ class GeminiPlugin(Plugin):
         self.max_norm = max_norm
         self.precision = precision
         self.placement_policy = placement_policy
+        self.chunk_size_search = chunk_size_search
         self.offload_optim_frac = offload_optim_frac
         self._memory_ratio = kwargs.get("memory_ratio", 0.9)  # HBM budget for auto placement
         # chunk size: reuse the ZeRO bucket machinery; chunks sized in MiB-elements
@@ -249,12 +251,18 @@ class GeminiPlugin(Plugin):
                 self.zero_kwargs["cpu_offload_frac"] = self.offload_optim_frac
         if self.shard_param_frac == 1.0:
             from ...zero.gemini import GeminiDDP, GeminiOptimizer
+            from ...zero.gemini.gemini_ddp import search_chunk_size
 
             if not isinstance(model, ModelWrapper):
-                model = GeminiDDP(model, chunk_size_m=self.chunk_size_m, precision=self.precision)
+                chunk_m = self.chunk_size_m
+                if self.chunk_size_search:
+                    chunk_m = max(search_chunk_size(model), self.chunk_size_m)
+                model = GeminiDDP(model, chunk_size_m=chunk_m, precision=self.precision)
             if optimizer is not None and not isinstance(optimizer, OptimizerWrapper):
                 kw = self._scaler_kwargs if self.precision == "fp16" else {}
-                optimizer = GeminiOptimizer(optimizer, model, max_norm=self.max_norm, **kw)
+                optimizer = GeminiOptimizer(optimizer, model, max_norm=self.max_norm,
+                                            auto_residency=(self.placement_policy == "auto"),
+                                            memory_ratio=self._memory_ratio, **kw)
             return model, optimizer, criterion, dataloader, lr_scheduler
         if not isinstance(model, ModelWrapper):
             model = LowLevelZeroModel(model, self.precision)

@@ -41,11 +41,46 @@ import torch.nn as nn
 
 from ...interface import ModelWrapper
 
-__all__ = ["GeminiDDP", "Chunk"]
+__all__ = ["GeminiDDP", "Chunk", "search_chunk_size"]
 
 
 def _pad(n: int, align: int) -> int:
     return (n + align - 1) // align * align
+
+
+def search_chunk_size(module: nn.Module, candidates_m=(16, 32, 48, 64, 96, 128),
+                      align: int = 64, units=None) -> int:
+    """Chunk-size search (reference: zero/gemini/chunk/search_utils.py —
+    re-derived for this design). The reference's fixed-extent chunks waste
+    tail bytes, so it searches for minimal memory waste; our chunks have
+    variable extents and waste nothing, so the objective becomes the xGMI
+    cost model instead: pick the size with the fewest UNDERSIZED collectives
+    (tail chunks below half the target pay the per-link latency term without
+    amortizing it), breaking ties toward fewer chunks."""
+    units = _auto_units(module) if units is None else units
+    best_m, best_score = candidates_m[0], None
+    for m in candidates_m:
+        target = m * 1024 * 1024
+        small = n_chunks = 0
+        for u in units:
+            direct = {id(p) for p in u.parameters(recurse=False)}
+            sizes = [_pad(p.numel(), align) for p in u.parameters()
+                     if p.requires_grad and id(p) not in direct]
+            acc = 0
+            ext = []
+            for sz in sizes:
+                acc += sz
+                if acc >= target:
+                    ext.append(acc)
+                    acc = 0
+            if acc:
+                ext.append(acc)
+            n_chunks += len(ext)
+            small += sum(1 for c in ext if c < target // 2)
+        score = (small, n_chunks)
+        if best_score is None or score < best_score:
+            best_score, best_m = score, m
+    return best_m
 
 
 class Chunk:
@@ -347,6 +382,36 @@ class GeminiDDP(ModelWrapper):
     def publish_persistent(self):
         for c in self.persistent_chunks:
             c.publish_shard(self.group)
+
+    def auto_adjust_residency(self, memory_ratio: float = 0.9,
+                              capacity_bytes: Optional[int] = None) -> int:
+        """Runtime auto placement (reference:
+        zero/gemini/placement_policy.py:128 AutoPlacementPolicy — inverted
+        for 288 GB HBM3E): after a warmup iteration, convert the measured
+        HBM headroom into chunk RESIDENCY. The reference evicts chunks when
+        the card is short; on MI355X the profitable direction is keeping
+        chunks gathered so the per-layer all-gather xGMI traffic disappears
+        where memory allows. Called by GeminiPlugin(placement_policy='auto')
+        after the first optimizer step; returns how many chunks were pinned.
+        """
+        if capacity_bytes is None:
+            if not torch.cuda.is_available():
+                return 0
+            capacity_bytes = torch.cuda.get_device_properties(self.device).total_memory
+        peak = torch.cuda.max_memory_allocated() if torch.cuda.is_available() else 0
+        headroom = capacity_bytes * memory_ratio - peak
+        pinned = 0
+        for c in self.chunks:
+            if c.persistent:
+                continue
+            cost = c.numel * c.flat.element_size()
+            if cost <= headroom:
+                c.gather(self.group)
+                c.persistent = True
+                self.persistent_chunks.append(c)  # post-step publish_shard path
+                headroom -= cost
+                pinned += 1
+        return pinned
 
     def zero_grad_shards(self):
         for c in self.chunks:

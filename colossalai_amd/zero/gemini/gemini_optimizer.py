@@ -26,11 +26,15 @@ class GeminiOptimizer(OptimizerWrapper):
         optim: Optimizer,
         model: GeminiDDP,
         max_norm: float = 0.0,
+        auto_residency: bool = False,
+        memory_ratio: float = 0.9,
         **scaler_kwargs,
     ):
         super().__init__(optim)
         self.model = model
         self.max_norm = max_norm
+        self.auto_residency = auto_residency
+        self.memory_ratio = memory_ratio
         self.world = model.world
         self.mixin = None
         if model.dtype == torch.float16:
@@ -136,6 +140,17 @@ class GeminiOptimizer(OptimizerWrapper):
         for i in idxs:
             self.model.chunks[i].grad_shard.zero_()
         self.model.publish_persistent()
+        if self.auto_residency and group["step"] == 1:
+            # after one full iteration the activation high-water is known:
+            # convert the remaining HBM headroom into pinned chunk residency
+            pinned = self.model.auto_adjust_residency(self.memory_ratio)
+            if pinned:
+                from ...logging import get_dist_logger
+
+                get_dist_logger().info(
+                    f"Gemini auto placement: pinned {pinned} chunks resident from runtime memory stats",
+                    ranks=[0],
+                )
 
     def zero_grad(self, *args, **kwargs):
         self.model.zero_grad_shards()
