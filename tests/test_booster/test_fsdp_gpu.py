@@ -25,7 +25,7 @@ def test_fsdp_boost_step(tmp_path):
         colossalai_amd.launch(0, 1, "127.0.0.1", 29511, backend="nccl", verbose=False)
 
     torch.manual_seed(0)
-    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+    cfg = LlamaConfig(vocab_size=128, hidden_size=256, intermediate_size=512, num_hidden_layers=2,
                       num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
     model = LlamaForCausalLM(cfg)
     plugin = TorchFSDPPlugin(precision="bf16")
