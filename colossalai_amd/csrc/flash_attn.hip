@@ -79,43 +79,14 @@ DEV_INLINE __bf16 f2b(float f) {
 // C-layout row for mfma_f32_32x32x16 register j (guide §3, HW-verified).
 DEV_INLINE constexpr int crow(int j, int half) { return (j & 3) + 8 * (j >> 2) + 4 * half; }
 
-// ---- subtiled row-major LDS tiles --------------------------------------
-// [COLS/16 groups][ROWS][16 bf16], 32 B row pitch, 16 B pad between groups
-// (group base bank shift of 4 keeps multi-group accesses staggered).
-template <int ROWS>
-__host__ __device__ constexpr int sub_pitch() { return ROWS * 32 + 16; }
-
-template <int ROWS>
-DEV_INLINE int sub_off(int row, int col) {
-  return (col >> 4) * sub_pitch<ROWS>() + row * 32 + (col & 15) * 2;
-}
-
-template <int ROWS, int COLS>
-__host__ __device__ constexpr int sub_bytes() { return (COLS / 16) * sub_pitch<ROWS>(); }
-
-// ds_read_b64_tr_b16: per-lane 64-bit read + 16-lane weave (see header).
-DEV_INLINE bf16x4v_t ld_tr16(const char* lds, int byte) {
-  typedef __attribute__((address_space(3))) bf16x4v_t* lds_v4p;
-  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_v4p)(lds + byte));
-}
-
-// Full 8-element column fragment from a subtiled tile: lane l receives
-// tile[k_base + (l>>5)*8 + e][cg_base*16 + (l&31)] for e = 0..7 — exactly
-// the A/B fragment of mfma_32x32x16 with the tile's ROW axis as the MFMA
-// k-dim and its COLUMN axis as the lane-owned m/n dim.
-template <int ROWS>
-DEV_INLINE bf16x8_t ld_frag_tr(const char* tile, int lane, int k_base, int cg_base) {
-  const int m = lane & 15;
-  const int cg = cg_base + ((lane >> 4) & 1);
-  const int row = k_base + ((lane >> 5) * 8) + (m >> 2);
-  const int base = cg * sub_pitch<ROWS>() + (m & 3) * 8;
-  bf16x4v_t lo = ld_tr16(tile, base + row * 32);
-  bf16x4v_t hi = ld_tr16(tile, base + (row + 4) * 32);
-  bf16x8_t r;
-  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
-  r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
-  return r;
-}
+// Subtiled row-major LDS tiles + ds_read_b64_tr_b16 column fragments live
+// in common.h (shared with grouped_gemm.hip); re-exported into this
+// namespace for the kernel bodies below.
+using cai::sub_pitch;
+using cai::sub_off;
+using cai::sub_bytes;
+using cai::ld_tr16;
+using cai::ld_frag_tr;
 
 // Synchronous staging: load + store back-to-back (transient registers).
 template <int D, int ROWS, int NT_>
@@ -450,6 +421,7 @@ __global__ __launch_bounds__(256) void fa_delta_kernel(
 //  1: same body, forced 2 waves/SIMD (256-reg cap -> ~75 spills)
 //  2: register diet for a TRUE 2 waves/SIMD: synchronous staging (no
 //     prefetch registers) + V fragments re-loaded from L1 each tile
+//  3: VAR 2 + K fragments also re-loaded (zero spill target)
 template <int D, int VAR, bool VARLEN = false>
 __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
     const unsigned short* __restrict__ Q,
@@ -504,13 +476,13 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
   // ---- K, V fragments (A-operands): lane row = kv (ln), d = sl*16+half*8+[0..7]
   const int kv_my = min(kvw + ln, Seff - 1);
   const unsigned short* vp = v_base + (long)kv_my * kts + half * 8;
-  bf16x8_t kf[DSL], vf[VAR == 2 ? 1 : DSL];
+  const unsigned short* kp = k_base + (long)kv_my * kts + half * 8;
+  bf16x8_t kf[VAR == 3 ? 1 : DSL], vf[VAR >= 2 ? 1 : DSL];
   {
-    const unsigned short* kp = k_base + (long)kv_my * kts + half * 8;
 #pragma unroll
     for (int sl = 0; sl < DSL; ++sl) {
-      kf[sl] = ld_g16(kp + sl * 16);
-      if (VAR != 2) vf[sl] = ld_g16(vp + sl * 16);
+      if (VAR != 3) kf[sl] = ld_g16(kp + sl * 16);
+      if (VAR < 2) vf[sl] = ld_g16(vp + sl * 16);
     }
   }
 
@@ -576,9 +548,10 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
           // B: lane col = q (qb*32+ln), k-rows d → row-slice read
+          bf16x8_t kfr = (VAR == 3) ? ld_g16(kp + sl * 16) : kf[VAR == 3 ? 0 : sl];
           bf16x8_t qb_frag = ld_lds16(Qlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
           __builtin_amdgcn_s_setprio(1);
-          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[sl], qb_frag, acc, 0, 0, 0);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qb_frag, acc, 0, 0, 0);
           __builtin_amdgcn_s_setprio(0);
         }
 
@@ -627,7 +600,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         for (int j = 0; j < 16; ++j) acc2[j] = 0.0f;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
-          bf16x8_t vfr = (VAR == 2) ? ld_g16(vp + sl * 16) : vf[VAR == 2 ? 0 : sl];
+          bf16x8_t vfr = (VAR >= 2) ? ld_g16(vp + sl * 16) : vf[VAR >= 2 ? 0 : sl];
           bf16x8_t dob = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
           acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dob, acc2, 0, 0, 0);
         }
@@ -1084,10 +1057,12 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
                           + 2 * QB * sizeof(float);                                                 \
     const void* dkdv_fn = dkdv_var == 0 ? (const void*)fa_bwd_dkdv_kernel<DD, 0>                    \
                         : dkdv_var == 1 ? (const void*)fa_bwd_dkdv_kernel<DD, 1>                    \
+                        : dkdv_var == 3 ? (const void*)fa_bwd_dkdv_kernel<DD, 3>                    \
                                         : (const void*)fa_bwd_dkdv_kernel<DD, 2>;                   \
     set_lds_limit(dkdv_fn, lds_kv);                                                                 \
     hipLaunchKernelGGL((dkdv_var == 0 ? fa_bwd_dkdv_kernel<DD, 0>                                   \
                         : dkdv_var == 1 ? fa_bwd_dkdv_kernel<DD, 1>                                 \
+                        : dkdv_var == 3 ? fa_bwd_dkdv_kernel<DD, 3>                                 \
                                         : fa_bwd_dkdv_kernel<DD, 2>),                               \
                        grid_kv, dim3(NTB), lds_kv, stream.stream(),                                 \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),    \

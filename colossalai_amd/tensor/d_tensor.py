@@ -19,6 +19,7 @@ import torch.distributed as dist
 __all__ = [
     "DTensorSpec", "shard_rowwise", "shard_colwise", "distribute_tensor",
     "gather_distributed", "is_distributed_tensor", "get_sharding_spec",
+    "redistribute", "comm_bytes",
 ]
 
 
@@ -94,3 +95,108 @@ def gather_distributed(tensor: torch.Tensor) -> torch.Tensor:
         full = torch.cat(parts, dim=d)
     assert tuple(full.shape) == tuple(spec.global_shape), (full.shape, spec.global_shape)
     return full
+
+
+# --------------------------------------------------------- layout conversion
+# (reference: colossalai/tensor/d_tensor/layout_converter.py + comm_spec.py —
+# re-derived: instead of enumerating one-step transform chains with a cached
+# search, the two canonical lowerings are generated directly and chosen by
+# the xGMI byte-cost model. The covered conversions are the ones the modern
+# stack performs: gather, re-shard, and same-group shard-dim moves.)
+
+
+def _a2a_move(local: torch.Tensor, src_dim: int, dst_dim: int, group) -> torch.Tensor:
+    """Move a shard from src_dim to dst_dim over the SAME group with one
+    all-to-all (gather src_dim, scatter dst_dim)."""
+    world = dist.get_world_size(group)
+    if world == 1:
+        return local
+    chunks = [c.contiguous() for c in local.chunk(world, dim=dst_dim)]
+    outs = [torch.empty_like(chunks[0]) for _ in range(world)]
+    if dist.get_backend(group) == "gloo":
+        rank = dist.get_rank(group)
+        gathered = [torch.empty_like(local) for _ in range(world)]
+        dist.all_gather(gathered, local.contiguous(), group=group)
+        outs = [g.chunk(world, dim=dst_dim)[rank].contiguous() for g in gathered]
+    else:
+        dist.all_to_all(outs, chunks, group=group)
+    return torch.cat(outs, dim=src_dim)
+
+
+def comm_bytes(src: "DTensorSpec", dst: "DTensorSpec", elem_size: int = 2) -> int:
+    """xGMI bytes each rank moves for the chosen lowering (cost model for
+    layout planning; all-gather of a shard of N bytes costs (w-1)/w * N_full,
+    an all-to-all costs (w-1)/w * N_local)."""
+    numel = 1
+    for s in src.global_shape:
+        numel *= int(s)
+    full_bytes = numel * elem_size
+    src_dims, dst_dims = dict(src.dims), dict(dst.dims)
+    moved = _moved_pair(src_dims, dst_dims)
+    cost = 0
+    if moved is not None:
+        g = src_dims[moved[0]]
+        w = dist.get_world_size(g)
+        cost += (w - 1) * full_bytes // (w * w)
+        src_dims.pop(moved[0])
+        dst_dims.pop(moved[1])
+    shard_w = 1
+    for d, g in src_dims.items():
+        shard_w *= dist.get_world_size(g)
+    for d, g in src_dims.items():
+        w = dist.get_world_size(g)
+        # all-gather along g: receive (w-1) pieces of the local shard size
+        cost += (w - 1) * full_bytes // shard_w
+    return cost
+
+
+def _moved_pair(src_dims, dst_dims):
+    """A (src_dim, dst_dim) pair sharded over the SAME group on both sides
+    but on different dims — eligible for the single-a2a fast path."""
+    for sd, sg in src_dims.items():
+        if sd in dst_dims:
+            continue
+        for dd, dg in dst_dims.items():
+            if dd not in src_dims and dg is sg:
+                return sd, dd
+    return None
+
+
+def redistribute(tensor: torch.Tensor, dst_spec: "DTensorSpec") -> torch.Tensor:
+    """Convert a spec-stamped local shard to ``dst_spec``'s layout.
+
+    Lowering: (1) same-group shard-dim moves become ONE all-to-all;
+    (2) remaining source shards all-gather; (3) remaining destination dims
+    split locally. Collective choice and order follow the byte-cost model
+    (comm_bytes)."""
+    spec = get_sharding_spec(tensor)
+    assert spec is not None, "redistribute() needs a spec-stamped tensor (distribute_tensor/shard_*)"
+    assert dst_spec.global_shape is None or tuple(dst_spec.global_shape) == tuple(spec.global_shape)
+    src_dims, dst_dims = dict(spec.dims), dict(dst_spec.dims)
+    local = tensor
+
+    moved = _moved_pair(src_dims, dst_dims)
+    while moved is not None:
+        sd, dd = moved
+        local = _a2a_move(local, sd, dd, src_dims[sd])
+        src_dims.pop(sd)
+        dst_dims.pop(dd)
+        moved = _moved_pair(src_dims, dst_dims)
+
+    # gather remaining source shards that the destination doesn't keep
+    for d in sorted([d for d in src_dims if src_dims[d] is not dst_dims.get(d)], reverse=True):
+        g = src_dims[d]
+        world = dist.get_world_size(g) if dist.is_initialized() else 1
+        if world > 1:
+            parts = [torch.empty_like(local) for _ in range(world)]
+            dist.all_gather(parts, local.contiguous(), group=g)
+            local = torch.cat(parts, dim=d)
+        src_dims.pop(d)
+
+    # split for destination dims not already sharded
+    for d, g in sorted(dst_dims.items()):
+        if src_dims.get(d) is g:
+            continue
+        local = _shard_dim(local, d, g)
+
+    return _stamp(local.contiguous(), spec.global_shape, dict(dst_spec.dims))
