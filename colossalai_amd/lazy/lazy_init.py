@@ -59,3 +59,33 @@ class LazyInitContext:
                 if hasattr(m, "reset_parameters"):
                     m.reset_parameters()
         return module
+
+    @staticmethod
+    def materialize_from_state_dict(module: nn.Module, state_dict, device: str = "cuda",
+                                    dtype: Optional[torch.dtype] = None, strict: bool = False) -> nn.Module:
+        """from_pretrained integration (reference: colossalai/lazy/pretrained.py):
+        allocate real storage and fill it straight from a checkpoint state
+        dict — the model never exists twice in memory (meta skeleton ->
+        storage -> weights). Entries may be a dict or an iterable of dicts
+        (HF-style sharded files loaded one at a time)."""
+        if any(p.is_meta for p in module.parameters()):
+            module = module.to_empty(device=device)
+        shards = [state_dict] if isinstance(state_dict, dict) else list(state_dict)
+        loaded = set()
+        for shard in shards:
+            cast = {}
+            for k, v in shard.items():
+                if torch.is_tensor(v):
+                    v = v.to(device=device)
+                    if dtype is not None and v.is_floating_point():
+                        v = v.to(dtype)
+                cast[k] = v
+            module.load_state_dict(cast, strict=False)
+            loaded.update(cast.keys())
+        if strict:
+            missing = set(k for k, _ in module.named_parameters()) - loaded
+            if missing:
+                raise RuntimeError(f"materialize_from_state_dict: missing keys {sorted(missing)[:8]}...")
+        if dtype is not None:
+            module = module.to(dtype)
+        return module

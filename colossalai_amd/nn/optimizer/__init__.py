@@ -1,6 +1,7 @@
 from .adafactor import Adafactor
 from .came import CAME
 from .cpu_adam import CPUAdam
+from .disk_offload import DiskOffloadAdam
 from .distributed_factored import DistributedAdafactor, DistributedCAME, DistributedGaLoreAdamW
 from .distributed_lamb import DistributedLamb
 from .fused_adam import FusedAdam
@@ -14,7 +15,7 @@ __all__ = ["FusedAdam",
     "FusedSGD", "DistributedLamb",
     "DistributedAdafactor",
     "DistributedCAME",
-    "DistributedGaLoreAdamW", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
+    "DistributedGaLoreAdamW", "HybridAdam", "CPUAdam", "DiskOffloadAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]
 
 
 def cast_to_distributed(optimizer):
