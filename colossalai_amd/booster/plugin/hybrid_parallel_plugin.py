@@ -237,13 +237,13 @@ class HybridParallelPlugin(Plugin):
     ):
         assert dist.is_initialized(), "launch colossalai_amd before creating HybridParallelPlugin"
         assert zero_stage in (0, 1, 2)
-        assert pp_style in ("1f1b", "interleaved", "zb"), f"unsupported pp_style {pp_style}"
+        assert pp_style in ("1f1b", "interleaved", "zb", "zbv"), f"unsupported pp_style {pp_style}"
         if pp_style == "interleaved":
             assert num_model_chunks > 1, "interleaved pipeline needs num_model_chunks > 1"
-        if pp_style == "zb":
+        if pp_style in ("zb", "zbv"):
             assert zero_stage == 0, (
                 "zero-bubble defers weight grads past AccumulateGrad, which ZeRO's "
-                "bucket hooks rely on — use pp_style='zb' with zero_stage=0"
+                "bucket hooks rely on — use pp_style='zb'/'zbv' with zero_stage=0"
             )
         world = dist.get_world_size()
         if sp_size is None:
@@ -305,6 +305,14 @@ class HybridParallelPlugin(Plugin):
                 self.scheduler = ZeroBubbleSchedule(
                     self.stage_manager, num_microbatches=num_microbatches, microbatch_size=microbatch_size
                 )
+            elif pp_style == "zbv":
+                from ...pipeline.schedule.zbv_schedule import ZeroBubbleVSchedule
+
+                self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
+                self.scheduler = ZeroBubbleVSchedule(
+                    self.stage_manager, num_microbatches=num_microbatches, microbatch_size=microbatch_size
+                )
+                self.num_model_chunks = 2
             else:
                 self.stage_manager = PipelineStageManager(self.pg_mesh, PP_AXIS)
                 self.scheduler = OneForwardOneBackwardSchedule(
@@ -376,7 +384,7 @@ class HybridParallelPlugin(Plugin):
             model = model.to(dtype)
             if torch.cuda.is_available():
                 model = model.to("cuda")
-            if self.pp_style == "zb":
+            if self.pp_style in ("zb", "zbv"):
                 from ...ops.zb_linear import convert_to_zb_linears
 
                 convert_to_zb_linears(model)
@@ -447,6 +455,30 @@ class HybridParallelPlugin(Plugin):
             and getattr(head, "weight", None) is not None and head.weight is embed.weight
         )
         model._tied_embed_param = None
+        if self.pp_style == "zbv":
+            # V placement: rank r holds vstages r (descending arm) and
+            # 2*pp-1-r (ascending arm); embeddings, LM head and the loss all
+            # live on rank 0 — tied weights need no cross-stage sync.
+            pp = self.pp_size
+            V = 2 * pp
+            per = [n_layers // V] * V
+            for i in range(n_layers % V):
+                per[i] += 1
+            starts = [sum(per[:i]) for i in range(V)]
+            vs = [self.stage_manager.stage, V - 1 - self.stage_manager.stage]
+            model.chunk_ranges = [(starts[v], starts[v] + per[v]) for v in vs]
+            model.stage_range = model.chunk_ranges[0]
+            held = set()
+            for a, b in model.chunk_ranges:
+                held.update(range(a, b))
+            for i in range(n_layers):
+                if i not in held:
+                    layers[i] = _StageStub()
+            if self.stage_manager.stage != 0:
+                self._stub_embeddings(inner)
+                if head is not None:
+                    model.lm_head = _StageStub()
+            return
         if is_tied:
             last_stage = (
                 self.stage_manager.is_last_stage(self.num_model_chunks - 1)

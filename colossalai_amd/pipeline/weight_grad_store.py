@@ -10,7 +10,7 @@ stages are waiting on — has been sent upstream.
 """
 
 from collections import deque
-from typing import Callable, List
+from typing import Callable, Dict, List, Tuple
 
 import torch
 
@@ -21,6 +21,7 @@ class WeightGradStore:
     enabled: bool = False
     _current: List[Callable] = []
     _batches: "deque[List[Callable]]" = deque()
+    _keyed: Dict[Tuple, List[Callable]] = {}
 
     @classmethod
     def put(cls, fn: Callable) -> None:
@@ -31,6 +32,23 @@ class WeightGradStore:
         """Close the current microbatch's W-batch."""
         cls._batches.append(cls._current)
         cls._current = []
+
+    @classmethod
+    def commit_key(cls, key) -> None:
+        """Close the current microbatch's W-batch under an explicit key
+        (ZB-V schedules W slots per (vstage, micro))."""
+        cls._keyed[key] = cls._current
+        cls._current = []
+
+    @classmethod
+    def pop_key(cls, key) -> None:
+        batch = cls._keyed.pop(key, None)
+        if batch:
+            import torch
+
+            with torch.no_grad():
+                for fn in batch:
+                    fn()
 
     @classmethod
     def pop(cls) -> None:
@@ -46,8 +64,11 @@ class WeightGradStore:
             cls.commit()
         while cls._batches:
             cls.pop()
+        for key in list(cls._keyed):
+            cls.pop_key(key)
 
     @classmethod
     def clear(cls) -> None:
         cls._current = []
         cls._batches.clear()
+        cls._keyed.clear()
