@@ -32,6 +32,9 @@ class SFTTrainer:
         kwargs = {}
         if batch.get("attention_mask") is not None:
             kwargs["attention_mask"] = batch["attention_mask"]
+        if batch.get("cu_seqlens") is not None:
+            # packed ragged batch from applications/chat/packing.py
+            kwargs["cu_seqlens"] = batch["cu_seqlens"]
         out = self.model(input_ids=batch["input_ids"], labels=batch["labels"], **kwargs)
         loss = self.criterion(out, batch)
         self.booster.backward(loss, self.optimizer)

@@ -59,6 +59,9 @@ std::vector<at::Tensor> moe_combine_bwd(at::Tensor dout, at::Tensor y, at::Tenso
 at::Tensor grouped_gemm_fwd(at::Tensor x, at::Tensor w, std::vector<long> offsets);
 at::Tensor grouped_gemm_dgrad(at::Tensor dy, at::Tensor w, std::vector<long> offsets);
 at::Tensor grouped_gemm_wgrad(at::Tensor dy, at::Tensor x, std::vector<long> offsets);
+std::vector<at::Tensor> moe_cumsum(at::Tensor experts, long n_experts);
+at::Tensor moe_dispatch_fwd(at::Tensor x, at::Tensor src);
+at::Tensor moe_dispatch_bwd(at::Tensor grad, at::Tensor src);
 
 // flash_attn.hip
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale,
@@ -100,6 +103,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grouped_gemm_fwd", &cai::grouped_gemm_fwd, "grouped per-expert GEMM: y = x @ w[g]^T");
   m.def("grouped_gemm_dgrad", &cai::grouped_gemm_dgrad, "grouped GEMM data grad: dx = dy @ w[g]");
   m.def("grouped_gemm_wgrad", &cai::grouped_gemm_wgrad, "grouped GEMM weight grad: dw[g] = dy_g^T @ x_g");
+  m.def("moe_cumsum", &cai::moe_cumsum, "deterministic per-expert rank/count of flat routing ids");
+  m.def("moe_dispatch_fwd", &cai::moe_dispatch_fwd, "MoE dispatch: row gather by permutation");
+  m.def("moe_dispatch_bwd", &cai::moe_dispatch_bwd, "MoE dispatch backward: row scatter (bijection)");
   m.def("flash_attn_fwd", &cai::flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA; optional right-padding seqlens)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),

@@ -122,3 +122,111 @@ std::vector<at::Tensor> moe_combine_bwd(at::Tensor dout, at::Tensor y, at::Tenso
 }
 
 }  // namespace cai
+
+// ====================================================== dispatch + cumsum
+// (reference: extensions/csrc/kernel/cuda/moe_kernel.cu:276-367
+// dispatch_forward/backward + cumsum_kernel — re-derived: instead of a
+// [T, E] mask-matrix prefix, each expert's block scans the flat routing
+// array directly, producing DETERMINISTIC within-expert positions — the
+// counting-sort ranks a stable argsort would give, without the sort.)
+
+// positions[s] = #(s' < s : expert[s'] == expert[s]); counts[e] = total.
+// One block per expert; 256 threads Hillis-Steele block scan per tile.
+__global__ __launch_bounds__(256) void moe_rank_kernel(
+    const int* __restrict__ experts,  // [N] flat expert ids
+    int* __restrict__ positions,      // [N]
+    int* __restrict__ counts,         // [E]
+    long N) {
+  __shared__ int scan[256];
+  const int e = blockIdx.x;
+  const int tid = threadIdx.x;
+  int base = 0;
+  for (long t0 = 0; t0 < N; t0 += 256) {
+    const long i = t0 + tid;
+    int m = (i < N && experts[i] == e) ? 1 : 0;
+    scan[tid] = m;
+    __syncthreads();
+    // inclusive block scan
+#pragma unroll
+    for (int off = 1; off < 256; off <<= 1) {
+      int v = (tid >= off) ? scan[tid - off] : 0;
+      __syncthreads();
+      scan[tid] += v;
+      __syncthreads();
+    }
+    if (i < N && m) positions[i] = base + scan[tid] - 1;
+    base += scan[255];
+    __syncthreads();
+  }
+  if (tid == 0) counts[e] = base;
+}
+
+// Row gather (dispatch fwd): out[d] = x[src[d]] — 16 B vector copies.
+__global__ __launch_bounds__(256) void moe_gather_rows_kernel(
+    const unsigned short* __restrict__ X, unsigned short* __restrict__ OUT,
+    const int* __restrict__ src, long n_rows, int Hc /* 16B chunks per row */) {
+  const long total = n_rows * Hc;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / Hc, ch = i % Hc;
+    const short8* s = reinterpret_cast<const short8*>(X + (long)src[row] * (Hc * 8) + ch * 8);
+    *reinterpret_cast<short8*>(OUT + row * (Hc * 8) + ch * 8) = *s;
+  }
+}
+
+// Row scatter (dispatch bwd): out[src[d]] = g[d] — src is a bijection, no atomics.
+__global__ __launch_bounds__(256) void moe_scatter_rows_kernel(
+    const unsigned short* __restrict__ G, unsigned short* __restrict__ OUT,
+    const int* __restrict__ src, long n_rows, int Hc) {
+  const long total = n_rows * Hc;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / Hc, ch = i % Hc;
+    const short8* s = reinterpret_cast<const short8*>(G + row * (Hc * 8) + ch * 8);
+    *reinterpret_cast<short8*>(OUT + (long)src[row] * (Hc * 8) + ch * 8) = *s;
+  }
+}
+
+// experts [N] int32 -> (positions [N], counts [E]); deterministic.
+std::vector<at::Tensor> moe_cumsum(at::Tensor experts, long n_experts) {
+  TORCH_CHECK(experts.scalar_type() == at::kInt && experts.is_contiguous());
+  const long N = experts.numel();
+  auto positions = at::empty({N}, experts.options());
+  auto counts = at::zeros({n_experts}, experts.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(moe_rank_kernel, dim3((int)n_experts), dim3(256), 0, stream.stream(),
+                     experts.data_ptr<int>(), positions.data_ptr<int>(), counts.data_ptr<int>(), N);
+  HIP_CHECK_LAST();
+  return {positions, counts};
+}
+
+// x [T, H] bf16, src [N] int32 -> out [N, H] with out[d] = x[src[d]].
+at::Tensor moe_dispatch_fwd(at::Tensor x, at::Tensor src) {
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x.size(1) % 8 == 0, "hidden size must be a multiple of 8");
+  const long N = src.numel();
+  const int Hc = (int)x.size(1) / 8;
+  auto out = at::empty({N, x.size(1)}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = capped_grid((N * Hc + 255) / 256, 4096);
+  hipLaunchKernelGGL(moe_gather_rows_kernel, dim3(grid), dim3(256), 0, stream.stream(),
+                     (const unsigned short*)x.data_ptr(), (unsigned short*)out.data_ptr(),
+                     src.data_ptr<int>(), N, Hc);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// grad [N, H], src [N] (bijection into N rows) -> out [N, H] with out[src[d]] = grad[d].
+at::Tensor moe_dispatch_bwd(at::Tensor grad, at::Tensor src) {
+  TORCH_CHECK(grad.dim() == 2 && grad.is_contiguous() && grad.scalar_type() == at::kBFloat16);
+  const long N = src.numel();
+  const int Hc = (int)grad.size(1) / 8;
+  auto out = at::empty_like(grad);
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = capped_grid((N * Hc + 255) / 256, 4096);
+  hipLaunchKernelGGL(moe_scatter_rows_kernel, dim3(grid), dim3(256), 0, stream.stream(),
+                     (const unsigned short*)grad.data_ptr(), (unsigned short*)out.data_ptr(),
+                     src.data_ptr<int>(), N, Hc);
+  HIP_CHECK_LAST();
+  return out;
+}

@@ -137,10 +137,30 @@ class LlamaAttention(nn.Module):
             self.norm_eps = cfg.rms_norm_eps
 
     def forward(self, hidden: torch.Tensor, rope_table: torch.Tensor,
-                seqlens: Optional[torch.Tensor] = None) -> torch.Tensor:
+                seqlens: Optional[torch.Tensor] = None,
+                cu_seqlens: Optional[torch.Tensor] = None) -> torch.Tensor:
         B, S, _ = hidden.shape
         qkv = self.qkv_proj(hidden)
         sp_mode = getattr(self, "sp_mode", None)
+        if cu_seqlens is not None:
+            # packed ragged batch ([1, total, ...]): per-sequence RoPE
+            # positions + the varlen flash kernels (ops/attention.py)
+            from ..ops import apply_rope
+            from ..ops.attention import flash_attention_varlen
+
+            assert sp_mode is None and B == 1, "varlen packing: B=1, no SP"
+            assert not self.qk_norm and self.pir_factor == 0
+            Hq, Hkv, D = self.num_heads, self.num_kv_heads, self.head_dim
+            q = qkv[:, :, : Hq * D].reshape(B, S, Hq, D).contiguous()
+            k = qkv[:, :, Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D).contiguous()
+            v = qkv[:, :, (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
+            lens = cu_seqlens[1:] - cu_seqlens[:-1]
+            positions = torch.cat(
+                [torch.arange(int(n), device=hidden.device) for n in lens]).int()
+            q, k = apply_rope(q, k, rope_table, positions)
+            attn = flash_attention_varlen(q[0], k[0], v[0].contiguous(), cu_seqlens,
+                                          causal=True, scale=self.scale)
+            return self.o_proj(attn.reshape(B, S, Hq * D))
         if sp_mode is not None:
             assert not self.qk_norm and self.pir_factor == 0, (
                 "qk-norm / partial-rotary variants are not wired into the SP "
@@ -371,14 +391,15 @@ class LlamaDecoderLayer(nn.Module):
         self.eps = cfg.rms_norm_eps
 
     def forward(self, hidden: torch.Tensor, residual: Optional[torch.Tensor], rope_table: torch.Tensor,
-                seqlens: Optional[torch.Tensor] = None):
+                seqlens: Optional[torch.Tensor] = None,
+                cu_seqlens: Optional[torch.Tensor] = None):
         """hidden = normed input to attention; residual = running stream.
 
         Returns (mlp_out, residual') where residual' = residual + attn_out:
         the NEXT junction (residual'' = residual' + mlp_out, then norm) is
         fused by the caller so norm weights stay owned by their layer.
         """
-        attn_out = self.self_attn(hidden, rope_table, seqlens)
+        attn_out = self.self_attn(hidden, rope_table, seqlens, cu_seqlens)
         hidden, residual = fused_add_rms_norm(attn_out, residual, self.post_attention_layernorm_weight, self.eps)
         mlp_out = self.mlp(hidden)
         return mlp_out, residual
@@ -409,6 +430,7 @@ class LlamaModel(nn.Module):
         hidden_states: Optional[torch.Tensor] = None,
         stage_range: Optional[tuple] = None,
         seqlens: Optional[torch.Tensor] = None,
+        cu_seqlens: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         """Full forward, or a pipeline-stage slice when ``stage_range`` is set.
 
@@ -442,10 +464,10 @@ class LlamaModel(nn.Module):
             layer = self.layers[i]
             if self.gradient_checkpointing and self.training and i < n_ckpt:
                 out, residual = torch.utils.checkpoint.checkpoint(
-                    layer, hidden, residual, table, seqlens, use_reentrant=False
+                    layer, hidden, residual, table, seqlens, cu_seqlens, use_reentrant=False
                 )
             else:
-                out, residual = layer(hidden, residual, table, seqlens)
+                out, residual = layer(hidden, residual, table, seqlens, cu_seqlens)
             if i + 1 < end:
                 hidden, residual = fused_add_rms_norm(out, residual, self.layers[i + 1].input_layernorm_weight, self.eps)
             elif is_last:
@@ -490,6 +512,7 @@ class LlamaForCausalLM(nn.Module):
         hidden_states: Optional[torch.Tensor] = None,
         pp_chunk: Optional[int] = None,
         attention_mask: Optional[torch.Tensor] = None,
+        cu_seqlens: Optional[torch.Tensor] = None,
     ):
         seqlens = None
         if attention_mask is not None:
@@ -536,8 +559,12 @@ class LlamaForCausalLM(nn.Module):
                     self._sp_labels_shifted = True
                 input_ids = (zigzag_split(input_ids, sp, rank) if zigzag
                              else input_ids[:, rank * shard : (rank + 1) * shard])
+        if cu_seqlens is not None:
+            cu_seqlens = cu_seqlens.to(torch.int32)
+            if input_ids is not None:
+                cu_seqlens = cu_seqlens.to(input_ids.device)
         out = self.model(input_ids, hidden_states=hidden_states, stage_range=stage_range,
-                         seqlens=seqlens)
+                         seqlens=seqlens, cu_seqlens=cu_seqlens)
         if stage_range is not None and stage_range[1] < len(self.model.layers):
             return {"hidden_states": out}
         hidden = out

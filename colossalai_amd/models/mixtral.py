@@ -89,12 +89,15 @@ class MixtralSparseMoeBlock(nn.Module):
         probs = F.softmax(logits, dim=-1)
         topw, topi = self._route(probs)  # fp32 routing weights
 
+        from ..ops import moe_dispatch, moe_route
+
         flat_expert = topi.reshape(-1)  # [T*k]
-        order = torch.argsort(flat_expert, stable=True)
+        # deterministic counting-sort routing + fused row gather
+        # (csrc/moe.hip moe_rank/moe_gather kernels)
+        order, counts = moe_route(flat_expert, self.num_experts)
         token_of_slot = order // self.top_k
         sorted_expert = flat_expert[order]
-        counts = torch.bincount(flat_expert, minlength=self.num_experts)
-        x_disp = x[token_of_slot]  # [T*k, H] sorted by destination expert
+        x_disp = moe_dispatch(x, token_of_slot)  # [T*k, H] sorted by destination expert
 
         if self.ep_size > 1:
             import torch.distributed as dist

@@ -4,7 +4,7 @@ from .attention import attention_ref, flash_attention, flash_attention_varlen, f
 from .layernorm import layer_norm
 from .norm import fused_add_rms_norm, rms_norm, rms_norm_ref
 from .rope import apply_rope, apply_rope_ref, build_rope_table
-from .moe import moe_combine
+from .moe import moe_combine, moe_dispatch, moe_route
 from .swiglu import swiglu, swiglu_ref
 
 __all__ = [
@@ -24,6 +24,8 @@ __all__ = [
     "apply_rope_ref",
     "build_rope_table",
     "moe_combine",
+    "moe_dispatch",
+    "moe_route",
     "swiglu",
     "swiglu_ref",
 ]

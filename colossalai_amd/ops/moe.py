@@ -44,3 +44,46 @@ def moe_combine(y: torch.Tensor, inv: torch.Tensor, topw: torch.Tensor) -> torch
     """y [T*k, H] expert outputs in dispatch order; inv [T*k]; topw [T, k]
     -> combined [T, H] in token order."""
     return _MoeCombine.apply(y, inv, topw)
+
+
+def moe_route(flat_experts: torch.Tensor, num_experts: int):
+    """Deterministic counting-sort routing (replaces stable argsort):
+    -> (order [N] long: slot index feeding sorted position d,
+        counts [E] long) — backed by csrc/moe.hip moe_rank_kernel
+    (reference: moe_kernel.cu:367 cumsum_kernel)."""
+    N = flat_experts.numel()
+    if flat_experts.is_cuda and has_kernels():
+        e_i = flat_experts.int().contiguous()
+        positions, counts = kernels().moe_cumsum(e_i, num_experts)
+        offsets = counts.cumsum(0, dtype=torch.int64) - counts.long()
+        dest = offsets[flat_experts.long()] + positions.long()
+        order = torch.empty(N, dtype=torch.int64, device=flat_experts.device)
+        order.scatter_(0, dest, torch.arange(N, device=flat_experts.device))
+        return order, counts.long()
+    order = torch.argsort(flat_experts, stable=True)
+    counts = torch.bincount(flat_experts, minlength=num_experts)
+    return order, counts
+
+
+class _MoeDispatch(torch.autograd.Function):
+    """out[d] = x[src[d]] — HIP row-gather forward; backward index-adds the
+    k routed copies back onto each token row."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, src: torch.Tensor):
+        ctx.save_for_backward(src)
+        ctx.n_tokens = x.shape[0]
+        if x.is_cuda and x.dtype == torch.bfloat16 and has_kernels() and x.shape[1] % 8 == 0:
+            return kernels().moe_dispatch_fwd(x.contiguous(), src.int().contiguous())
+        return x[src.long()]
+
+    @staticmethod
+    def backward(ctx, grad):
+        (src,) = ctx.saved_tensors
+        dx = torch.zeros(ctx.n_tokens, grad.shape[1], dtype=torch.float32, device=grad.device)
+        dx.index_add_(0, src.long(), grad.float())
+        return dx.to(grad.dtype), None
+
+
+def moe_dispatch(x: torch.Tensor, src: torch.Tensor) -> torch.Tensor:
+    return _MoeDispatch.apply(x, src)

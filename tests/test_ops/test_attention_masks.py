@@ -105,3 +105,33 @@ def test_model_padded_batch_loss():
         tot_loss += float(ref["loss"]) * n_tok
         tot_tok += n_tok
     torch.testing.assert_close(out["loss"], torch.tensor(tot_loss / tot_tok), rtol=1e-4, atol=1e-5)
+
+
+def test_model_packed_varlen_loss():
+    """Packed cu_seqlens batch == per-sequence losses (token-weighted):
+    the varlen path through LlamaForCausalLM + the SFT packer."""
+    import sys
+
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
+    sys.path.insert(0, "applications")
+    from chat.packing import pack_sft_samples
+
+    torch.manual_seed(4)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    model = LlamaForCausalLM(cfg).float()
+    lens = [13, 24, 7]
+    samples = [{"input_ids": torch.randint(0, 128, (L,)), "labels": None} for L in lens]
+    for s in samples:
+        s["labels"] = s["input_ids"].clone()
+    batches = pack_sft_samples(samples, max_tokens=64)
+    assert len(batches) == 1 and batches[0]["cu_seqlens"].tolist() == [0, 13, 37, 44]
+
+    out = model(**batches[0])
+    tot, n = 0.0, 0
+    for s, L in zip(samples, lens):
+        ref = model(input_ids=s["input_ids"].unsqueeze(0), labels=s["labels"].unsqueeze(0))
+        tot += float(ref["loss"]) * (L - 1)
+        n += L - 1
+    torch.testing.assert_close(out["loss"], torch.tensor(tot / n), rtol=1e-4, atol=1e-5)

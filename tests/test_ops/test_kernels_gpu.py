@@ -540,3 +540,35 @@ def test_grouped_gemm_autograd():
         (xf[offs[g]:offs[g+1]] @ wf[g].t()).sum().backward(retain_graph=True)
     _bf16_close(x.grad, xf.grad, rtol=3e-2, atol=3e-2, frac=1e-4)
     _bf16_close(w.grad, wf.grad, rtol=3e-2, atol=5e-1, frac=1e-3)
+
+
+# -------------------------------------------------- MoE dispatch / cumsum
+def test_moe_cumsum_dispatch():
+    """Deterministic counting-sort routing + row gather/scatter kernels vs
+    torch stable-argsort reference."""
+    from colossalai_amd.ops import moe_dispatch, moe_route
+
+    torch.manual_seed(21)
+    T, k, E, H = 997, 2, 8, 256
+    topi = torch.randint(0, E, (T, k), device="cuda")
+    flat = topi.reshape(-1)
+    order, counts = moe_route(flat, E)
+    ref_order = torch.argsort(flat, stable=True)
+    assert torch.equal(order.cpu(), ref_order.cpu()), "counting sort must equal stable argsort"
+    assert torch.equal(counts.cpu(), torch.bincount(flat, minlength=E).cpu())
+
+    x = torch.randn(T, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    src = (order // k)
+    out = moe_dispatch(x, src)
+    torch.testing.assert_close(out, x.detach()[src])
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    ref = torch.zeros(T, H, device="cuda", dtype=torch.float32)
+    ref.index_add_(0, src, dout.float())
+    torch.testing.assert_close(x.grad.float(), ref.to(x.grad.dtype).float(), rtol=2e-2, atol=2e-2)
+
+    # bijection scatter kernel
+    perm = torch.randperm(T, device="cuda").int()
+    y = torch.randn(T, H, device="cuda", dtype=torch.bfloat16)
+    scat = kernels().moe_dispatch_bwd(y, perm)
+    torch.testing.assert_close(scat[perm.long()], y)
