@@ -121,7 +121,7 @@ std::vector<at::Tensor> moe_combine_bwd(at::Tensor dout, at::Tensor y, at::Tenso
   return {dy, dw};
 }
 
-}  // namespace cai
+
 
 // ====================================================== dispatch + cumsum
 // (reference: extensions/csrc/kernel/cuda/moe_kernel.cu:276-367
@@ -230,3 +230,5 @@ at::Tensor moe_dispatch_bwd(at::Tensor grad, at::Tensor src) {
   HIP_CHECK_LAST();
   return out;
 }
+
+}  // namespace cai
