@@ -167,6 +167,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   constexpr int KB_BYTES = sub_bytes<KVB, D>();
   char* Klds = smem;
   char* Vlds = smem + KB_BYTES;
+  char* Pw = smem + 2 * KB_BYTES;  // + wave*QW*KVB*2
 
   const int lane = threadIdx.x & 63;
   const int w = threadIdx.x >> 6;
@@ -195,6 +196,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   const long o_stride = (long)Hq * D;  // O/LSE are always packed
 
   constexpr int DSL = D / 16;  // MFMA K-slices over the head dim
+  char* P = Pw + w * (QW * KVB * 2);
 
   // ---- Q fragments (B-operand of S^T = K·Q^T): lane holds q-col (ln),
   //      d-rows half*8+[0..7] per 16-d slice — one 16 B load per slice.
@@ -288,24 +290,17 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     m_run = m_new;
     l_run = l_run * corr + rowsum;
 
-    // ---- T12 in-register softmax fragments (guide §5.5): pack P into
-    // bf16 pairs and permlane32_swap the lane halves — each lane then owns
-    // its q-row's FULL 64 kv values, so the P·V A-fragments assemble from
-    // registers and the per-wave P LDS round-trip disappears.
-    // u32_own[kb][i]: C-regs j = 4*(i>>1)+2*(i&1)+{0,1} = kv run
-    // 8*(i>>1)+4*half+2*(i&1)+{0,1}; after swap h0/h1 hold the half-0/1
-    // lane's packs for q-row ln.
-    unsigned int h0[KB][8], h1[KB][8];
+    // ---- stage P (bf16) into swizzled per-wave LDS: P[q=ln][k]
+    // C-regs j=0..3 within a group are 4 consecutive k values → 8 B packed.
 #pragma unroll
     for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int j = 4 * (i >> 1) + 2 * (i & 1);
-        const unsigned int own =
-            (unsigned int)f2bf(p[kb][j]) | ((unsigned int)f2bf(p[kb][j + 1]) << 16);
-        auto pr = __builtin_amdgcn_permlane32_swap((int)own, (int)own, false, false);
-        h0[kb][i] = (unsigned int)pr[0];
-        h1[kb][i] = (unsigned int)pr[1];
+      for (int jj = 0; jj < 4; ++jj) {
+        const int kcol = kb * 32 + 8 * jj + 4 * half;
+        short4v pk;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) pk[e] = (short)f2bf(p[kb][jj * 4 + e]);
+        *reinterpret_cast<short4v*>(P + swz<KVB * 2>(ln, kcol * 2)) = pk;
       }
 
     // ---- rescale O accumulator by corr (per q-row, via shfl broadcast);
@@ -320,20 +315,11 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
       }
     }
 
-    // ---- O += P · V  (A = P assembled in registers, B = V tr-reads)
+    // ---- O += P · V  (A = P from LDS, B = V column-fragments via tr-read)
 #pragma unroll
     for (int ks = 0; ks < KVB / 16; ++ks) {
-      // A: lane row = q (ln), k = ks*16 + half*8 + [0..7]. kv_local =
-      // 8*(2s+half) + e8 with s = ks&1: u32 pair index (2s)*2 + {0,1} from
-      // the e8<4 source (h0) and the e8>=4 source (h1), selected by half.
-      const int kb2 = ks >> 1;
-      const int s2 = ks & 1;
-      union { unsigned int u[4]; bf16x8_t v; } fragcvt;
-      fragcvt.u[0] = half ? h0[kb2][4 * s2 + 2] : h0[kb2][4 * s2 + 0];
-      fragcvt.u[1] = half ? h0[kb2][4 * s2 + 3] : h0[kb2][4 * s2 + 1];
-      fragcvt.u[2] = half ? h1[kb2][4 * s2 + 2] : h1[kb2][4 * s2 + 0];
-      fragcvt.u[3] = half ? h1[kb2][4 * s2 + 3] : h1[kb2][4 * s2 + 1];
-      bf16x8_t pa = fragcvt.v;
+      // A: lane row = q (ln), k-cols = ks*16 + half*8 + [0..7]
+      bf16x8_t pa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
 #pragma unroll
       for (int nb = 0; nb < D / 32; ++nb) {
         // B: lane col = d (nb*32+ln), k-rows kv = ks*16 + half*8 + [0..7]
@@ -932,7 +918,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 
 #define LAUNCH_FWD(DD)                                                                            \
   do {                                                                                            \
-    const size_t lds = 2 * sub_bytes<KVB, DD>();                                                  \
+    const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
     set_lds_limit((const void*)fa_fwd_kernel<DD, false>, lds);                                    \
     hipLaunchKernelGGL((fa_fwd_kernel<DD, false>), grid, dim3(NT), lds, stream.stream(),          \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
@@ -971,7 +957,7 @@ std::vector<at::Tensor> flash_attn_varlen_fwd(at::Tensor q, at::Tensor k, at::Te
 
 #define LAUNCH_VFWD(DD)                                                                           \
   do {                                                                                            \
-    const size_t lds = 2 * sub_bytes<KVB, DD>();                                                  \
+    const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
     set_lds_limit((const void*)fa_fwd_kernel<DD, true>, lds);                                     \
     hipLaunchKernelGGL((fa_fwd_kernel<DD, true>), grid, dim3(NT), lds, stream.stream(),           \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
