@@ -148,10 +148,10 @@ class MixtralDecoderLayer(nn.Module):
         self.post_attention_layernorm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
         self.eps = cfg.rms_norm_eps
 
-    def forward(self, hidden, residual, rope_table, seqlens=None):
+    def forward(self, hidden, residual, rope_table, seqlens=None, cu_seqlens=None):
         from ..ops import fused_add_rms_norm
 
-        attn_out = self.self_attn(hidden, rope_table, seqlens)
+        attn_out = self.self_attn(hidden, rope_table, seqlens, cu_seqlens)
         hidden, residual = fused_add_rms_norm(attn_out, residual, self.post_attention_layernorm_weight, self.eps)
         mlp_out = self.mlp(hidden)
         return mlp_out, residual

@@ -11,9 +11,13 @@ from typing import List
 
 import torch
 
+import os
+
 from ._kernels import kernels, use_hip
 
 __all__ = ["grouped_gemm"]
+
+_DISABLED = os.environ.get("CAI_MOE_GG", "1") == "0"  # A/B: per-expert loop
 
 
 def _loop_fwd(x, w, offs):
@@ -31,7 +35,7 @@ class _GroupedGemm(torch.autograd.Function):
         offs = list(offs_tuple)
         ctx.offs = offs
         ctx.save_for_backward(x, w)
-        if use_hip(x, w) and x.shape[1] % 64 == 0 and w.shape[1] % 128 == 0:
+        if not _DISABLED and use_hip(x, w) and x.shape[1] % 64 == 0 and w.shape[1] % 128 == 0:
             ctx.hip = True
             return kernels().grouped_gemm_fwd(x.contiguous(), w.contiguous(), offs)
         ctx.hip = False
