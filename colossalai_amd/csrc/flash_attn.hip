@@ -238,23 +238,31 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     const bool active = !(causal && k0 > qw + QW - 1) && (qw < Seff);
     if (active) {
     // ---- S^T = K · Q^T : KVB/32 32-k blocks. C: col = q (ln), row = k (crow).
+    // The kb accumulator chains are independent: interleaving them keeps
+    // two MFMAs in flight instead of one serial 8-deep chain per block.
     constexpr int KB = KVB / 32;
     float p[KB][16];
+    {
+      f32x16 acc[KB];
 #pragma unroll
-    for (int kb = 0; kb < KB; ++kb) {
-      f32x16 acc;
+      for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
-      for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+        for (int j = 0; j < 16; ++j) acc[kb][j] = 0.0f;
 #pragma unroll
       for (int sl = 0; sl < DSL; ++sl) {
-        // A-operand: K rows. lane: row = kb*32+ln, cols d = sl*16 + half*8 + [0..7]
-        bf16x8_t kf = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
         __builtin_amdgcn_s_setprio(1);
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[sl], acc, 0, 0, 0);
+#pragma unroll
+        for (int kb = 0; kb < KB; ++kb) {
+          // A-operand: K rows. lane: row = kb*32+ln, cols d = sl*16 + half*8 + [0..7]
+          bf16x8_t kf = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
+          acc[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[sl], acc[kb], 0, 0, 0);
+        }
         __builtin_amdgcn_s_setprio(0);
       }
 #pragma unroll
-      for (int j = 0; j < 16; ++j) p[kb][j] = acc[j];
+      for (int kb = 0; kb < KB; ++kb)
+#pragma unroll
+        for (int j = 0; j < 16; ++j) p[kb][j] = acc[kb][j];
     }
 
     // ---- online softmax over this tile's 64 scores of q-row (qw+ln)

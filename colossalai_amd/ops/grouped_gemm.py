@@ -17,7 +17,22 @@ from ._kernels import kernels, use_hip
 
 __all__ = ["grouped_gemm"]
 
-_DISABLED = os.environ.get("CAI_MOE_GG", "1") == "0"  # A/B: per-expert loop
+# Size-aware dispatch, measured on MI355X (mixtral-small bs8 seq4096,
+# ~8k rows/expert): hipBLASLt per-expert GEMMs reach ~1.1 PF and beat the
+# hand-written 128-tile kernel end-to-end (11.9 vs 9.7 samples/s), so the
+# grouped kernel serves the SMALL-rows regime where per-expert launch
+# overhead and tail tiles dominate the library (decode / small prefill).
+# CAI_MOE_GG: "0" forces the loop, "1" forces the kernel, unset = by size.
+_FORCE = os.environ.get("CAI_MOE_GG")
+_SMALL_ROWS = int(os.environ.get("CAI_MOE_GG_ROWS", "1024"))
+
+
+def _use_kernel(n_rows: int, n_groups: int) -> bool:
+    if _FORCE == "0":
+        return False
+    if _FORCE == "1":
+        return True
+    return n_rows <= _SMALL_ROWS * n_groups
 
 
 def _loop_fwd(x, w, offs):
@@ -35,7 +50,8 @@ class _GroupedGemm(torch.autograd.Function):
         offs = list(offs_tuple)
         ctx.offs = offs
         ctx.save_for_backward(x, w)
-        if not _DISABLED and use_hip(x, w) and x.shape[1] % 64 == 0 and w.shape[1] % 128 == 0:
+        if _use_kernel(x.shape[0], w.shape[0]) and use_hip(x, w) \
+                and x.shape[1] % 64 == 0 and w.shape[1] % 128 == 0:
             ctx.hip = True
             return kernels().grouped_gemm_fwd(x.contiguous(), w.contiguous(), offs)
         ctx.hip = False
