@@ -115,10 +115,12 @@ class OPTModel(nn.Module):
         self.final_layer_norm_bias = nn.Parameter(torch.zeros(cfg.hidden_size))
         self.gradient_checkpointing = False
 
-    def forward(self, input_ids):
-        B, S = input_ids.shape
-        pos = torch.arange(2, S + 2, device=input_ids.device)
-        hidden = self.embed_tokens(input_ids) + self.embed_positions(pos)[None]
+    def forward(self, input_ids=None, inputs_embeds=None):
+        if inputs_embeds is None:
+            inputs_embeds = self.embed_tokens(input_ids)
+        B, S = inputs_embeds.shape[:2]
+        pos = torch.arange(2, S + 2, device=inputs_embeds.device)
+        hidden = inputs_embeds + self.embed_positions(pos)[None]
         for layer in self.layers:
             if self.gradient_checkpointing and self.training:
                 hidden = torch.utils.checkpoint.checkpoint(layer, hidden, use_reentrant=False)
@@ -146,8 +148,8 @@ class OPTForCausalLM(nn.Module):
     def gradient_checkpointing_enable(self, ratio: float = 1.0):
         self.model.gradient_checkpointing = True
 
-    def forward(self, input_ids, labels: Optional[torch.Tensor] = None):
-        hidden = self.model(input_ids)
+    def forward(self, input_ids=None, labels: Optional[torch.Tensor] = None, inputs_embeds=None):
+        hidden = self.model(input_ids, inputs_embeds=inputs_embeds)
         if labels is not None:
             from ..ops.fused_ce import fused_linear_cross_entropy
 

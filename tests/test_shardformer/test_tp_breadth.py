@@ -126,3 +126,40 @@ def test_tp_whisper():
 @rerun_if_address_is_in_use()
 def test_tp_cohere():
     spawn(run_tp_cohere, 2)
+
+
+def run_tp_blip2(rank, world_size, port):
+    from colossalai_amd.models.blip2 import Blip2Config, Blip2ForConditionalGeneration
+    from colossalai_amd.models.opt import OPTConfig
+    from colossalai_amd.models.vit import ViTConfig
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = Blip2Config(
+        vision=ViTConfig(image_size=16, patch_size=8, num_channels=3, hidden_size=64,
+                         num_hidden_layers=2, num_attention_heads=4, intermediate_size=128),
+        text=OPTConfig(vocab_size=256, hidden_size=64, ffn_dim=128, num_hidden_layers=2,
+                       num_attention_heads=4, max_position_embeddings=64),
+        qformer_hidden=64, qformer_layers=2, qformer_heads=4, num_query_tokens=4,
+    )
+    ref = Blip2ForConditionalGeneration(cfg)
+    model = _shard(copy.deepcopy(ref))
+    assert model.vision_model.layers[0].attention.num_heads == 2
+    assert model.qformer_layers[0].self_attn.num_heads == 2
+    assert model.language_model.model.layers[0].self_attn.num_heads == 2
+
+    px = torch.randn(2, 3, 16, 16)
+    ids = torch.randint(0, 256, (2, 12))
+    out = model(px, ids, labels=ids)
+    out_ref = ref(px, ids, labels=ids)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.qformer_ln_w.grad, ref.qformer_ln_w.grad, rtol=1e-3, atol=1e-5)
+    assert_close_loose(model.query_tokens.grad, ref.query_tokens.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_tp_blip2():
+    spawn(run_tp_blip2, 2)
