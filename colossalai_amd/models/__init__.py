@@ -1,3 +1,4 @@
+from .deepseek_v3 import DEEPSEEK_V3_CONFIGS, DeepseekV3Config, DeepseekV3ForCausalLM
 from .chatglm2 import CHATGLM_CONFIGS, ChatGLMConfig, ChatGLMForConditionalGeneration
 from .blip2 import Blip2Config, Blip2ForConditionalGeneration
 from .llama import LLAMA_CONFIGS, LlamaConfig, LlamaForCausalLM, llama_flops_per_token

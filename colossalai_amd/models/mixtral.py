@@ -67,6 +67,12 @@ class MixtralSparseMoeBlock(nn.Module):
         topw, topi = torch.topk(probs, self.top_k, dim=-1)
         return topw / topw.sum(-1, keepdim=True), topi
 
+    def _gate_and_route(self, x: torch.Tensor):
+        """x [T, H] -> (topw [T, k] fp32, topi [T, k]). Subclasses override
+        for non-softmax routers (DeepSeek-V3 sigmoid/noaux-tc)."""
+        probs = F.softmax(self.gate(x).float(), dim=-1)
+        return self._route(probs)
+
     def _experts_forward(self, x: torch.Tensor, expert_ids: torch.Tensor) -> torch.Tensor:
         """x [N, H] grouped so rows of the same LOCAL expert are contiguous;
         expert_ids [N] gives each row's local expert. One grouped-GEMM launch
@@ -85,9 +91,7 @@ class MixtralSparseMoeBlock(nn.Module):
         B, S, H = hidden.shape
         x = hidden.reshape(-1, H)
         T = x.shape[0]
-        logits = self.gate(x).float()
-        probs = F.softmax(logits, dim=-1)
-        topw, topi = self._route(probs)  # fp32 routing weights
+        topw, topi = self._gate_and_route(x)  # fp32 routing weights
 
         from ..ops import moe_dispatch, moe_route
 

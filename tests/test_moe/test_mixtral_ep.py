@@ -112,3 +112,45 @@ def run_deepseek_ep(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_deepseek_ep2():
     spawn(run_deepseek_ep, 2)
+
+
+def run_deepseek_v3_ep(rank, world_size, port):
+    """DeepSeek-V3: MLA + noaux-tc routing, EP-sharded vs dense oracle."""
+    import copy
+
+    from colossalai_amd import Booster
+    from colossalai_amd.booster.plugin import MoeHybridParallelPlugin
+    from colossalai_amd.models.deepseek_v3 import DEEPSEEK_V3_CONFIGS, DeepseekV3ForCausalLM
+    from colossalai_amd.nn import FusedAdam
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    model = DeepseekV3ForCausalLM(DEEPSEEK_V3_CONFIGS["deepseek-v3-tiny"])
+    ref = copy.deepcopy(model).float()
+
+    plugin = MoeHybridParallelPlugin(ep_size=2, zero_stage=1, precision="fp32",
+                                     overlap_communication=False)
+    booster = Booster(plugin=plugin)
+    opt = FusedAdam(model.parameters(), lr=1e-3)
+    model_b, opt_b, *_ = booster.boost(model, opt)
+
+    moe = model_b.module.model.layers[-1].mlp
+    assert moe.num_local_experts == 4 and moe.ep_size == 2
+
+    torch.manual_seed(9)
+    xs = [torch.randint(0, 128, (2, 16)) for _ in range(world_size)]
+    out = model_b(input_ids=xs[rank], labels=xs[rank])
+    losses_ref = [ref(input_ids=x, labels=x)["loss"] for x in xs]
+    assert_close_loose(out["loss"], losses_ref[rank], rtol=1e-4, atol=1e-5)
+
+    opt_b.backward(out["loss"])
+    opt_b.step()
+    opt_b.zero_grad()
+    out2 = model_b(input_ids=xs[rank], labels=xs[rank])
+    assert torch.isfinite(out2["loss"])
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_deepseek_v3_ep2():
+    spawn(run_deepseek_v3_ep, 2)

@@ -191,3 +191,30 @@ def run_tp_chatglm(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_tp_chatglm():
     spawn(run_tp_chatglm, 2)
+
+
+def run_tp_deepseek_v3(rank, world_size, port):
+    """MLA attention under TP: per-head shards of q_b/kv_b, replicated latents."""
+    from colossalai_amd.models.deepseek_v3 import DEEPSEEK_V3_CONFIGS, DeepseekV3ForCausalLM
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    ref = DeepseekV3ForCausalLM(DEEPSEEK_V3_CONFIGS["deepseek-v3-tiny"])
+    model = _shard(copy.deepcopy(ref))
+    attn = model.model.layers[0].self_attn
+    assert attn.num_heads == 2 and attn.q_b_proj.weight.shape[0] == 2 * (16 + 8)
+
+    x = torch.randint(0, 128, (2, 16))
+    out = model(x, labels=x)
+    out_ref = ref(x, labels=x)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.model.layers[0].self_attn.kv_a_ln_w.grad,
+                       ref.model.layers[0].self_attn.kv_a_ln_w.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_tp_deepseek_v3():
+    spawn(run_tp_deepseek_v3, 2)
