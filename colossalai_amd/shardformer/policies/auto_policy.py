@@ -28,6 +28,7 @@ POLICY_REGISTRY: Dict[str, tuple] = {
     "GPTJForCausalLM": ("colossalai_amd.shardformer.policies.gptj", "GPTJForCausalLMPolicy"),
     "WhisperForConditionalGeneration": ("colossalai_amd.shardformer.policies.whisper", "WhisperForConditionalGenerationPolicy"),
     "CohereForCausalLM": ("colossalai_amd.shardformer.policies.cohere", "CohereForCausalLMPolicy"),
+    "SamModel": ("colossalai_amd.shardformer.policies.sam", "SamModelPolicy"),
     "DeepseekV3ForCausalLM": ("colossalai_amd.shardformer.policies.deepseek_v3", "DeepseekV3ForCausalLMPolicy"),
     "ChatGLMForConditionalGeneration": ("colossalai_amd.shardformer.policies.chatglm2", "ChatGLMForConditionalGenerationPolicy"),
     "Blip2ForConditionalGeneration": ("colossalai_amd.shardformer.policies.blip2", "Blip2ForConditionalGenerationPolicy"),

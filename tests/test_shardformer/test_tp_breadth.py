@@ -218,3 +218,46 @@ def run_tp_deepseek_v3(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_tp_deepseek_v3():
     spawn(run_tp_deepseek_v3, 2)
+
+
+def run_tp_sam(rank, world_size, port):
+    from colossalai_amd.models.sam import SamConfig, SamModel, SamVisionConfig
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = SamConfig(
+        vision=SamVisionConfig(image_size=64, patch_size=8, hidden_size=64, num_hidden_layers=3,
+                               num_attention_heads=4, window_size=4, global_attn_indexes=(1,),
+                               output_channels=32),
+        prompt_embed_dim=32, decoder_hidden=32, decoder_heads=4, decoder_layers=2,
+        decoder_mlp_dim=64)
+    ref = SamModel(cfg)
+    model = _shard(copy.deepcopy(ref))
+    assert model.vision_encoder.layers[0].attn.num_heads == 2
+    assert model.mask_decoder.layers[0].self_attn.num_heads == 2
+
+    px = torch.randn(2, 3, 64, 64)
+    pts = torch.rand(2, 3, 2)
+    lbl = torch.randint(0, 2, (2, 3))
+    tgt = torch.rand(2, 32, 32) > 0.5
+    out = model(px, pts, lbl, mask_labels=tgt)
+    out_ref = ref(px, pts, lbl, mask_labels=tgt)
+    assert_close_loose(out["loss"], out_ref["loss"], rtol=1e-4, atol=1e-5)
+    out["loss"].backward()
+    out_ref["loss"].backward()
+    assert_close_loose(model.vision_encoder.layers[0].ln1_w.grad,
+                       ref.vision_encoder.layers[0].ln1_w.grad, rtol=1e-3, atol=1e-5)
+    # rel-pos grads are partial per rank (local heads only) and marked
+    # _sp_partial_grad; the plugin all-reduces them — emulate that here
+    rel = model.vision_encoder.layers[0].attn.rel_pos_h
+    assert getattr(rel, "_sp_partial_grad", False)
+    g = rel.grad.clone()
+    dist.all_reduce(g)
+    assert_close_loose(g, ref.vision_encoder.layers[0].attn.rel_pos_h.grad,
+                       rtol=1e-3, atol=1e-8)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_tp_sam():
+    spawn(run_tp_sam, 2)
