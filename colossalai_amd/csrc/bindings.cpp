@@ -47,9 +47,10 @@ std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor
 
 // decode_attn.hip
 at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, at::Tensor seq_lens,
-                            double scale);
+                            double scale, int64_t n_splits);
 at::Tensor decode_attention_paged(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
-                                  at::Tensor block_tables, at::Tensor seq_lens, double scale);
+                                  at::Tensor block_tables, at::Tensor seq_lens, double scale,
+                                  int64_t n_splits);
 
 // moe.hip
 at::Tensor moe_combine_fwd(at::Tensor y, at::Tensor inv, at::Tensor topw);
@@ -95,9 +96,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scaled_masked_softmax_fwd", &cai::scaled_masked_softmax_fwd, "fused scale+mask+softmax fwd");
   m.def("scaled_masked_softmax_bwd", &cai::scaled_masked_softmax_bwd, "fused scale+mask+softmax bwd");
   m.def("mfma_selftest", &cai::mfma_selftest, "MFMA layout self-test probes");
-  m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache");
+  m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache",
+        py::arg("q"), py::arg("kcache"), py::arg("vcache"), py::arg("seq_lens"), py::arg("scale"),
+        py::arg("n_splits") = 0);
   m.def("decode_attention_paged", &cai::decode_attention_paged,
-        "single-token attention over a paged (block-table) KV pool");
+        "single-token attention over a paged (block-table) KV pool",
+        py::arg("q"), py::arg("kpool"), py::arg("vpool"), py::arg("block_tables"),
+        py::arg("seq_lens"), py::arg("scale"), py::arg("n_splits") = 0);
   m.def("moe_combine_fwd", &cai::moe_combine_fwd, "fused MoE un-permute + weighted top-k sum");
   m.def("moe_combine_bwd", &cai::moe_combine_bwd, "MoE combine backward (dy + routing-weight grads)");
   m.def("grouped_gemm_fwd", &cai::grouped_gemm_fwd, "grouped per-expert GEMM: y = x @ w[g]^T");

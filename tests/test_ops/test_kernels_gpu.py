@@ -430,6 +430,23 @@ def test_decode_attention_paged_matches_contiguous(D):
     torch.testing.assert_close(out, ref, rtol=0.0, atol=0.0)
 
 
+@pytest.mark.parametrize("D", [64, 128])
+def test_decode_attention_splitkv(D):
+    """Flash-decoding v2: split-KV partials + LSE reduce must match the
+    single-pass kernel at long context / small batch."""
+    torch.manual_seed(14)
+    B, Smax, Hq, Hkv = 2, 4096, 8, 2
+    lens = torch.tensor([4096, 1531], dtype=torch.int32, device="cuda")
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.randn(B, Smax, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(B, Smax, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    ref = _C.decode_attention(q.contiguous(), kc, vc, lens, scale, 1)
+    for ns in (0, 2, 8, 64):  # 0 = auto (B*Hq=16 blocks -> deep split)
+        out = _C.decode_attention(q.contiguous(), kc, vc, lens, scale, ns)
+        torch.testing.assert_close(out, ref, rtol=1e-2, atol=1e-2)
+
+
 def test_multi_tensor_sgd():
     torch.manual_seed(13)
     shapes = [(1000,), (333,), (512, 64)]
