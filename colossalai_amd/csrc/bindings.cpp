@@ -13,6 +13,11 @@ void multi_tensor_adam(std::vector<at::Tensor> grads, std::vector<at::Tensor> pa
 void multi_tensor_scale(std::vector<at::Tensor> inputs, std::vector<at::Tensor> outputs, double scale,
                         long chunk_size);
 at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> inputs, long chunk_size);
+void multi_tensor_lamb(std::vector<at::Tensor> grads, std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+                       std::vector<at::Tensor> param_outs, double lr, double beta1, double beta2,
+                       double eps, long step, bool bias_correction, double weight_decay,
+                       double div_scale, long chunk_size);
 void multi_tensor_sgd(std::vector<at::Tensor> grads, std::vector<at::Tensor> params,
                       std::vector<at::Tensor> momentum_bufs, std::vector<at::Tensor> param_outs,
                       double lr, double momentum, double dampening, double weight_decay,
@@ -85,6 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_scale", &cai::multi_tensor_scale, "fused multi-tensor scale");
   m.def("multi_tensor_l2norm", &cai::multi_tensor_l2norm, "fused multi-tensor L2 norm");
   m.def("multi_tensor_sgd", &cai::multi_tensor_sgd, "fused multi-tensor SGD (momentum/nesterov)");
+  m.def("multi_tensor_lamb", &cai::multi_tensor_lamb,
+        "fused multi-tensor LAMB (two-stage: Adam update + trust-ratio scale)");
   m.def("rmsnorm_fwd", &cai::rmsnorm_fwd, "RMSNorm forward");
   m.def("rmsnorm_fused_add_fwd", &cai::rmsnorm_fused_add_fwd, "fused residual-add RMSNorm forward");
   m.def("rmsnorm_bwd", &cai::rmsnorm_bwd, "RMSNorm backward");

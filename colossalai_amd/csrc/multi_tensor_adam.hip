@@ -35,6 +35,7 @@ struct TensorListMeta {
   long sizes[MTA_TENSORS];
   short block_to_tensor[MTA_BLOCKS];
   int block_to_chunk[MTA_BLOCKS];
+  int global_idx[MTA_TENSORS];  // position in the FULL tensor list (LAMB norms)
 };
 
 template <typename T>
@@ -213,6 +214,7 @@ void multi_tensor_apply(const std::vector<std::vector<at::Tensor>>& lists, long 
       meta.addresses[d][t_in_meta] = lists[d][t].data_ptr();
     }
     meta.sizes[t_in_meta] = numel;
+    meta.global_idx[t_in_meta] = (int)t;
     const long nchunks = (numel + chunk_size - 1) / chunk_size;
     for (long c = 0; c < nchunks; ++c) {
       meta.block_to_tensor[b_in_meta] = (short)t_in_meta;
@@ -228,6 +230,7 @@ void multi_tensor_apply(const std::vector<std::vector<at::Tensor>>& lists, long 
           // continue the same tensor in a fresh batch
           for (int d = 0; d < depth; ++d) meta.addresses[d][0] = lists[d][t].data_ptr();
           meta.sizes[0] = numel;
+          meta.global_idx[0] = (int)t;
           t_in_meta = 0;
         } else if (!last) {
           t_in_meta = -1;  // next tensor becomes index 0
@@ -390,6 +393,146 @@ void multi_tensor_sgd(
   else if (p_bf16) DISPATCH_SGD(float, US);
   else DISPATCH_SGD(float, float);
 #undef DISPATCH_SGD
+}
+
+// --------------------------------------------------------------------- lamb
+// Two MTA passes (reference: multi_tensor_lamb_kernel.cu two-stage scheme):
+// stage 1 runs the Adam-style moment update, writes the raw update u into a
+// scratch list and block-atomically accumulates per-tensor ||p||^2 / ||u||^2;
+// stage 2 applies p -= lr * trust * u with trust = ||p|| / ||u|| read from
+// the accumulators (computed per block, no host sync).
+template <typename GT, typename PT>
+__global__ __launch_bounds__(ADAM_BLOCK) void multi_tensor_lamb_stage1_kernel(
+    TensorListMeta meta, long chunk_size, float beta1, float beta2, float bc1, float bc2,
+    float eps, float weight_decay, float inv_div_scale, float* __restrict__ norms) {
+  const int tensor_id = meta.block_to_tensor[blockIdx.x];
+  const long chunk_id = meta.block_to_chunk[blockIdx.x];
+  const long offset = chunk_id * chunk_size;
+  const long n = min(chunk_size, meta.sizes[tensor_id] - offset);
+
+  const GT* g = reinterpret_cast<const GT*>(meta.addresses[0][tensor_id]) + offset;
+  const PT* p = reinterpret_cast<const PT*>(meta.addresses[1][tensor_id]) + offset;
+  float* m = reinterpret_cast<float*>(meta.addresses[2][tensor_id]) + offset;
+  float* v = reinterpret_cast<float*>(meta.addresses[3][tensor_id]) + offset;
+  float* u = reinterpret_cast<float*>(meta.addresses[4][tensor_id]) + offset;
+
+  float pn = 0.0f, un = 0.0f;
+  for (long i = threadIdx.x; i < n; i += blockDim.x) {
+    const float gk = VecIO<GT>::load1(g + i) * inv_div_scale;
+    const float pk = VecIO<PT>::load1(p + i);
+    float mk = beta1 * m[i] + (1.0f - beta1) * gk;
+    float vk = beta2 * v[i] + (1.0f - beta2) * gk * gk;
+    m[i] = mk;
+    v[i] = vk;
+    float uk = (mk * bc1) / (sqrtf(vk * bc2) + eps) + weight_decay * pk;
+    u[i] = uk;
+    pn += pk * pk;
+    un += uk * uk;
+  }
+  // block reduce (wave shuffle + LDS across the 4 waves)
+  __shared__ float sm[2][4];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    pn += __shfl_xor(pn, off);
+    un += __shfl_xor(un, off);
+  }
+  const int wave = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) { sm[0][wave] = pn; sm[1][wave] = un; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    pn = sm[0][0] + sm[0][1] + sm[0][2] + sm[0][3];
+    un = sm[1][0] + sm[1][1] + sm[1][2] + sm[1][3];
+    const int gi = meta.global_idx[tensor_id];
+    atomicAdd(norms + 2 * gi, pn);
+    atomicAdd(norms + 2 * gi + 1, un);
+  }
+}
+
+template <typename PT, bool HAS_OUT>
+__global__ __launch_bounds__(ADAM_BLOCK) void multi_tensor_lamb_stage2_kernel(
+    TensorListMeta meta, long chunk_size, float lr, const float* __restrict__ norms) {
+  const int tensor_id = meta.block_to_tensor[blockIdx.x];
+  const long chunk_id = meta.block_to_chunk[blockIdx.x];
+  const long offset = chunk_id * chunk_size;
+  const long n = min(chunk_size, meta.sizes[tensor_id] - offset);
+
+  const float* u = reinterpret_cast<const float*>(meta.addresses[0][tensor_id]) + offset;
+  PT* p = reinterpret_cast<PT*>(meta.addresses[1][tensor_id]) + offset;
+  unsigned short* out =
+      HAS_OUT ? reinterpret_cast<unsigned short*>(meta.addresses[2][tensor_id]) + offset : nullptr;
+
+  const int gi = meta.global_idx[tensor_id];
+  const float pn = sqrtf(norms[2 * gi]);
+  const float un = sqrtf(norms[2 * gi + 1]);
+  const float trust = (pn > 0.0f && un > 0.0f) ? pn / un : 1.0f;
+  const float step_sz = lr * trust;
+  for (long i = threadIdx.x; i < n; i += blockDim.x) {
+    const float pk = VecIO<PT>::load1(p + i) - step_sz * u[i];
+    VecIO<PT>::store1(p + i, pk);
+    if (HAS_OUT) VecIO<unsigned short>::store1(out + i, pk);
+  }
+}
+
+void multi_tensor_lamb(
+    std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> params,
+    std::vector<at::Tensor> exp_avgs,
+    std::vector<at::Tensor> exp_avg_sqs,
+    std::vector<at::Tensor> param_outs,  // empty, or bf16 working copies
+    double lr, double beta1, double beta2, double eps, long step,
+    bool bias_correction, double weight_decay, double div_scale, long chunk_size) {
+  TORCH_CHECK(!grads.empty(), "multi_tensor_lamb: empty tensor list");
+  const bool has_out = !param_outs.empty();
+  float bc1 = 1.0f, bc2 = 1.0f;
+  if (bias_correction) {
+    bc1 = 1.0f / (1.0f - powf((float)beta1, (float)step));
+    bc2 = 1.0f / (1.0f - powf((float)beta2, (float)step));
+  }
+  const float inv_div_scale = (float)(1.0 / div_scale);
+  auto stream = at::hip::getCurrentHIPStream();
+
+  const long nt = (long)grads.size();
+  auto norms = at::zeros({2 * nt}, params[0].options().dtype(at::kFloat));
+  float* nptr = norms.data_ptr<float>();
+  std::vector<at::Tensor> scratch;
+  scratch.reserve(nt);
+  for (auto& gt : grads) scratch.push_back(at::empty_like(gt, gt.options().dtype(at::kFloat)));
+
+  const bool g_bf16 = is_bf16(grads[0]);
+  const bool p_bf16 = is_bf16(params[0]);
+  std::vector<std::vector<at::Tensor>> l1 = {grads, params, exp_avgs, exp_avg_sqs, scratch};
+  auto run1 = [&](auto gt, auto pt) {
+    using GT = decltype(gt);
+    using PT = decltype(pt);
+    multi_tensor_apply(l1, chunk_size, [&](const TensorListMeta& meta, int nblocks) {
+      hipLaunchKernelGGL((multi_tensor_lamb_stage1_kernel<GT, PT>), dim3(nblocks),
+                         dim3(ADAM_BLOCK), 0, stream.stream(), meta, chunk_size, (float)beta1,
+                         (float)beta2, bc1, bc2, (float)eps, (float)weight_decay, inv_div_scale,
+                         nptr);
+      HIP_CHECK_LAST();
+    });
+  };
+  using US = unsigned short;
+  if (g_bf16 && p_bf16) run1(US{}, US{});
+  else if (g_bf16) run1(US{}, float{});
+  else if (p_bf16) run1(float{}, US{});
+  else run1(float{}, float{});
+
+  std::vector<std::vector<at::Tensor>> l2 = {scratch, params};
+  if (has_out) l2.push_back(param_outs);
+  auto run2 = [&](auto pt, auto out_c) {
+    using PT = decltype(pt);
+    multi_tensor_apply(l2, chunk_size, [&](const TensorListMeta& meta, int nblocks) {
+      hipLaunchKernelGGL((multi_tensor_lamb_stage2_kernel<PT, decltype(out_c)::value>),
+                         dim3(nblocks), dim3(ADAM_BLOCK), 0, stream.stream(), meta, chunk_size,
+                         (float)lr, nptr);
+      HIP_CHECK_LAST();
+    });
+  };
+  using TrueT = std::integral_constant<bool, true>;
+  using FalseT = std::integral_constant<bool, false>;
+  if (p_bf16) { if (has_out) run2(US{}, TrueT{}); else run2(US{}, FalseT{}); }
+  else { if (has_out) run2(float{}, TrueT{}); else run2(float{}, FalseT{}); }
 }
 
 void multi_tensor_scale(std::vector<at::Tensor> inputs, std::vector<at::Tensor> outputs, double scale,

@@ -1,3 +1,3 @@
-from .optimizer import CAME, Adafactor, CPUAdam, DistributedLamb, FusedAdam, FusedSGD, GaLoreAdamW, HybridAdam, Lamb, Lars
+from .optimizer import CAME, Adafactor, CPUAdam, DistributedLamb, FusedAdam, FusedLAMB, FusedSGD, GaLoreAdamW, HybridAdam, Lamb, Lars
 
-__all__ = ["FusedAdam", "FusedSGD", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW", "DistributedLamb"]
+__all__ = ["FusedAdam", "FusedSGD", "FusedLAMB", "HybridAdam", "CPUAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW", "DistributedLamb"]

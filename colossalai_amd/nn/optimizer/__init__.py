@@ -5,6 +5,7 @@ from .disk_offload import DiskOffloadAdam
 from .distributed_factored import DistributedAdafactor, DistributedCAME, DistributedGaLoreAdamW
 from .distributed_lamb import DistributedLamb
 from .fused_adam import FusedAdam
+from .fused_lamb import FusedLAMB
 from .fused_sgd import FusedSGD
 from .galore import GaLoreAdamW
 from .hybrid_adam import HybridAdam
@@ -12,7 +13,7 @@ from .lamb import Lamb
 from .lars import Lars
 
 __all__ = ["FusedAdam",
-    "FusedSGD", "DistributedLamb",
+    "FusedSGD", "FusedLAMB", "DistributedLamb",
     "DistributedAdafactor",
     "DistributedCAME",
     "DistributedGaLoreAdamW", "HybridAdam", "CPUAdam", "DiskOffloadAdam", "Lamb", "Lars", "Adafactor", "CAME", "GaLoreAdamW"]

@@ -467,6 +467,32 @@ def test_multi_tensor_sgd():
         torch.testing.assert_close(p.cpu(), pr.cpu(), rtol=1e-5, atol=1e-6)
 
 
+def test_multi_tensor_lamb():
+    """Two-stage fused LAMB vs the CPU Lamb math (incl. trust ratio) —
+    many tensors to exercise the per-tensor norm accumulators across
+    MTA batches."""
+    from colossalai_amd.nn.optimizer.fused_lamb import _lamb_step_cpu
+
+    torch.manual_seed(15)
+    shapes = [(1000,), (333,), (512, 64), (7,)] * 8  # 32 tensors > MTA_TENSORS
+    lr, wd, eps = 1e-2, 0.01, 1e-6
+    gs, ps, ms, vs, refs = [], [], [], [], []
+    for sh in shapes:
+        g = torch.randn(sh, device="cuda", dtype=torch.bfloat16)
+        p = torch.randn(sh, device="cuda", dtype=torch.float32)
+        m = torch.randn(sh, device="cuda").abs() * 0.1
+        v = torch.randn(sh, device="cuda").abs() * 0.01
+        refs.append((p.clone(), g.clone(), m.clone(), v.clone()))
+        gs.append(g); ps.append(p); ms.append(m); vs.append(v)
+    _C.multi_tensor_lamb(gs, ps, ms, vs, [], lr, 0.9, 0.999, eps, 1, False, wd, 1.0, 65536)
+    for (p0, g0, m0, v0), p, m, v in zip(refs, ps, ms, vs):
+        pr = p0.clone()
+        _lamb_step_cpu(pr, g0, m0, v0, lr, 0.9, 0.999, eps, wd, 1, False)
+        torch.testing.assert_close(m.cpu(), m0.cpu(), rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(v.cpu(), v0.cpu(), rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(p.cpu(), pr.cpu(), rtol=1e-4, atol=1e-5)
+
+
 def test_moe_combine():
     from colossalai_amd.ops.moe import _MoeCombine
 
