@@ -160,6 +160,22 @@ inline int capped_grid(long total_blocks, int cap = 2048) {
   return (int)(total_blocks < cap ? total_blocks : cap);
 }
 
+
+// XCD-aware blockIdx remap (guide T1, bijective m204 form): dispatch-order
+// block f lands on XCD f%8; remapping logical ids so each XCD owns a
+// CONTIGUOUS chunk of the (y-major) grid makes neighbor blocks — which
+// share K/V (attention) or operand panels (GEMM) — hit that XCD's
+// private 4 MiB L2 instead of re-streaming HBM.
+DEV_INLINE void xcd_swizzle_xy(int& x, int& y) {
+  const int gx = (int)gridDim.x;
+  const int f = (int)blockIdx.y * gx + (int)blockIdx.x;
+  const int nwg = gx * (int)gridDim.y;
+  const int q = nwg >> 3, r = nwg & 7, xcd = f & 7, i = f >> 3;
+  const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  x = swz % gx;
+  y = swz / gx;
+}
+
 }  // namespace cai
 
 #define HIP_CHECK_LAST()                                                        \

@@ -174,10 +174,12 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   const int half = lane >> 5;
   const int ln = lane & 31;
 
-  const int bh = blockIdx.y;
+  int swz_x, swz_y;
+  xcd_swizzle_xy(swz_x, swz_y);  // same-bh tiles share one XCD's L2 (K/V reuse)
+  const int bh = swz_y;
   const int b = bh / Hq, h = bh % Hq;
   const int hk = h / (Hq / Hkv);
-  const int q0 = blockIdx.x * (NW * QW);
+  const int q0 = swz_x * (NW * QW);
   const int qw = q0 + w * QW;  // this wave's first q row
 
   long seq0 = 0;
@@ -459,10 +461,12 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
   const int half = lane >> 5;
   const int ln = lane & 31;
 
-  const int bh = blockIdx.y;
+  int swz_x, swz_y;
+  xcd_swizzle_xy(swz_x, swz_y);  // same-bh kv tiles share an XCD's L2 (Q/dO stream reuse)
+  const int bh = swz_y;
   const int b = bh / Hkv, hk = bh % Hkv;
   const int G = Hq / Hkv;
-  const int kv0 = blockIdx.x * (NWB * QW);
+  const int kv0 = swz_x * (NWB * QW);
   const int kvw = kv0 + w * QW;  // this wave's first kv row
 
   long seq0 = 0;
@@ -717,10 +721,12 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
   const int half = lane >> 5;
   const int ln = lane & 31;
 
-  const int bh = blockIdx.y;
+  int swz_x, swz_y;
+  xcd_swizzle_xy(swz_x, swz_y);  // same-bh q tiles share an XCD's L2 (K/V stream reuse)
+  const int bh = swz_y;
   const int b = bh / Hq, h = bh % Hq;
   const int hk = h / (Hq / Hkv);
-  const int q0 = blockIdx.x * (NWB * QW);
+  const int q0 = swz_x * (NWB * QW);
   const int qw = q0 + w * QW;
 
   long seq0 = 0;

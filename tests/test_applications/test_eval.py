@@ -51,3 +51,58 @@ def test_multiple_choice_prefers_trained_continuation():
     assert sequence_loglikelihood(model, prompt, good) > sequence_loglikelihood(model, prompt, bad)
     res = evaluate_multiple_choice(model, [{"prompt": prompt, "choices": [bad, good], "answer": 1}])
     assert res["accuracy"] == 1.0
+
+
+def test_metrics():
+    from applications.eval import exact_match, extract_numeric_answer, f1_score, first_choice, numeric_match
+
+    assert exact_match("The Answer!", "answer") == 1.0
+    assert 0.0 < f1_score("paris is the capital", "the capital is paris city") < 1.0
+    assert first_choice("I think (B) is right") == "B"
+    assert extract_numeric_answer("so 3+4 = 7 dollars #### 7") == "7"
+    assert numeric_match("the total is 1,234.0", "#### 1234") == 1.0
+    assert numeric_match("no idea", "#### 5") == 0.0
+
+
+def test_dataset_loaders_and_generation(tmp_path):
+    import json
+
+    import torch
+
+    from applications.eval import (
+        build_choice_examples,
+        build_generation_examples,
+        evaluate_generation,
+        evaluate_multiple_choice,
+        greedy_generate,
+        load_choice_csv,
+        load_qa_jsonl,
+        numeric_match,
+    )
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
+    csv_p = tmp_path / "mmlu.csv"
+    csv_p.write_text('what is 2+2?,three,four,five,six,B\npick A,yes,no,maybe,never,A\n')
+    rows = load_choice_csv(csv_p)
+    assert len(rows) == 2 and rows[0]["answer"] == 1 and len(rows[0]["choices"]) == 4
+
+    jl = tmp_path / "gsm.jsonl"
+    jl.write_text(json.dumps({"question": "2+2?", "answer": "#### 4"}) + "\n")
+    qa = load_qa_jsonl(jl)
+    assert qa[0]["answer"] == "#### 4"
+
+    tokenize = lambda s: [min(ord(c), 127) for c in s][:32]
+    detok = lambda ids: "".join(chr(i % 128) for i in ids)
+    mc = build_choice_examples(rows, tokenize)
+    gen = build_generation_examples(qa, tokenize)
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=128))
+    res = evaluate_multiple_choice(model, mc)
+    assert 0.0 <= res["accuracy"] <= 1.0 and res["n"] == 2
+    out = greedy_generate(model, gen[0]["prompt"], max_new_tokens=4)
+    assert len(out) == 4
+    score = evaluate_generation(model, gen, detok, numeric_match, max_new_tokens=4)
+    assert score["n"] == 1
