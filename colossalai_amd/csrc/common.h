@@ -166,6 +166,19 @@ inline int capped_grid(long total_blocks, int cap = 2048) {
 // CONTIGUOUS chunk of the (y-major) grid makes neighbor blocks — which
 // share K/V (attention) or operand panels (GEMM) — hit that XCD's
 // private 4 MiB L2 instead of re-streaming HBM.
+// 3-D variant: z (e.g. expert) is the slowest logical axis, so contiguous
+// per-XCD chunks stay within one z and share its operand panels.
+DEV_INLINE void xcd_swizzle_xyz(int& x, int& y, int& z) {
+  const int gx = (int)gridDim.x, gy = (int)gridDim.y;
+  const int f = ((int)blockIdx.z * gy + (int)blockIdx.y) * gx + (int)blockIdx.x;
+  const int nwg = gx * gy * (int)gridDim.z;
+  const int q = nwg >> 3, r = nwg & 7, xcd = f & 7, i = f >> 3;
+  const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
+  x = swz % gx;
+  y = (swz / gx) % gy;
+  z = swz / (gx * gy);
+}
+
 DEV_INLINE void xcd_swizzle_xy(int& x, int& y) {
   const int gx = (int)gridDim.x;
   const int f = (int)blockIdx.y * gx + (int)blockIdx.x;

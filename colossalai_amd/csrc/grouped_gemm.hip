@@ -110,10 +110,12 @@ __global__ __launch_bounds__(gg::NT) void grouped_gemm_kernel(
   const int half = lane >> 5;
   const int ln = lane & 31;
 
-  const int g = tile_g[blockIdx.x];
-  const int row0 = tile_row0[blockIdx.x];
+  int swz_x, swz_y;
+  xcd_swizzle_xy(swz_x, swz_y);  // neighbor tiles share w[g] panels in the XCD L2
+  const int g = tile_g[swz_x];
+  const int row0 = tile_row0[swz_x];
   const int grp_lo = offs[g], grp_hi = offs[g + 1];
-  const int n0 = blockIdx.y * TN;  // first output col
+  const int n0 = swz_y * TN;  // first output col
 
   const unsigned short* x_base = X;
   const unsigned short* w_base = W + (long)g * w_group_stride;
@@ -216,9 +218,11 @@ __global__ __launch_bounds__(gg::NT) void grouped_gemm_wgrad_kernel(
   const int half = lane >> 5;
   const int ln = lane & 31;
 
-  const int g = blockIdx.z;
-  const int m0 = blockIdx.x * TM;
-  const int k0 = blockIdx.y * TN;
+  int swz_x, swz_y, swz_z;
+  xcd_swizzle_xyz(swz_x, swz_y, swz_z);  // same-expert tiles share dy/x panels
+  const int g = swz_z;
+  const int m0 = swz_x * TM;
+  const int k0 = swz_y * TN;
   const int grp_lo = offs[g], grp_hi = offs[g + 1];
   const int count = grp_hi - grp_lo;
   if (count == 0) {
