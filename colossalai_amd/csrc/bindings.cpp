@@ -50,6 +50,9 @@ at::Tensor scaled_masked_softmax_bwd(at::Tensor dy, at::Tensor y, double scale);
 // mfma_selftest.hip
 std::vector<at::Tensor> mfma_selftest(at::Tensor A16, at::Tensor B16, at::Tensor A32, at::Tensor B32);
 
+void kv_cache_append(at::Tensor k, at::Tensor v, at::Tensor kpool, at::Tensor vpool,
+                     at::Tensor slot_rows);
+
 // decode_attn.hip
 at::Tensor decode_attention(at::Tensor q, at::Tensor kcache, at::Tensor vcache, at::Tensor seq_lens,
                             double scale, int64_t n_splits);
@@ -103,6 +106,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scaled_masked_softmax_fwd", &cai::scaled_masked_softmax_fwd, "fused scale+mask+softmax fwd");
   m.def("scaled_masked_softmax_bwd", &cai::scaled_masked_softmax_bwd, "fused scale+mask+softmax bwd");
   m.def("mfma_selftest", &cai::mfma_selftest, "MFMA layout self-test probes");
+  m.def("kv_cache_append", &cai::kv_cache_append,
+        "decode-step K+V scatter into paged pools (one launch)");
   m.def("decode_attention", &cai::decode_attention, "single-token attention over KV cache",
         py::arg("q"), py::arg("kcache"), py::arg("vcache"), py::arg("seq_lens"), py::arg("scale"),
         py::arg("n_splits") = 0);

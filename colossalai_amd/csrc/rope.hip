@@ -18,6 +18,8 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <algorithm>
+
 #include "common.h"
 
 namespace cai {
@@ -80,6 +82,43 @@ __global__ __launch_bounds__(ROPE_BLOCK) void rope_kernel(
     *reinterpret_cast<short8*>(base + vec * 8) = olo;
     *reinterpret_cast<short8*>(base + half + vec * 8) = ohi;
   }
+}
+
+// Decode-step KV append: k/v [B, Hkv, D] current-token states scatter into
+// the paged pools at per-sequence physical rows — both tensors in ONE
+// launch (reference: decode_kv_cache_memcpy_kernel.cu). 16 B/lane copies.
+__global__ __launch_bounds__(ROPE_BLOCK) void kv_cache_append_kernel(
+    const unsigned short* __restrict__ k, const unsigned short* __restrict__ v,
+    unsigned short* __restrict__ kpool, unsigned short* __restrict__ vpool,
+    const int* __restrict__ slot_rows, long units, int hd8) {
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < 2 * units;
+       idx += (long)gridDim.x * blockDim.x) {
+    const bool is_k = idx < units;
+    const long u = is_k ? idx : idx - units;
+    const long b = u / hd8;
+    const int off = (int)(u % hd8) * 8;
+    const long dst = (long)slot_rows[b] * hd8 * 8 + off;
+    const short8 val = *reinterpret_cast<const short8*>((is_k ? k : v) + u * 8);
+    *reinterpret_cast<short8*>((is_k ? kpool : vpool) + dst) = val;
+  }
+}
+
+void kv_cache_append(at::Tensor k, at::Tensor v, at::Tensor kpool, at::Tensor vpool,
+                     at::Tensor slot_rows) {
+  TORCH_CHECK(k.dim() == 3 && k.is_contiguous() && v.is_contiguous(), "kv [B,Hkv,D] contiguous");
+  TORCH_CHECK(k.scalar_type() == at::kBFloat16 && kpool.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(slot_rows.scalar_type() == at::kInt && slot_rows.is_contiguous());
+  const int B = (int)k.size(0), Hkv = (int)k.size(1), D = (int)k.size(2);
+  TORCH_CHECK(D % 8 == 0);
+  const int hd8 = Hkv * D / 8;
+  const long units = (long)B * hd8;
+  auto stream = at::hip::getCurrentHIPStream();
+  const int blocks = (int)std::min((2 * units + ROPE_BLOCK - 1) / ROPE_BLOCK, (long)1024);
+  hipLaunchKernelGGL(kv_cache_append_kernel, dim3(blocks), dim3(ROPE_BLOCK), 0, stream.stream(),
+                     (const unsigned short*)k.data_ptr(), (const unsigned short*)v.data_ptr(),
+                     (unsigned short*)kpool.data_ptr(), (unsigned short*)vpool.data_ptr(),
+                     slot_rows.data_ptr<int>(), units, hd8);
+  HIP_CHECK_LAST();
 }
 
 // In-place RoPE on q (and optionally k), both [B,S,H,D] views (strided B/S ok).

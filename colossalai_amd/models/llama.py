@@ -346,10 +346,19 @@ class LlamaAttention(nn.Module):
                 out = attention_ref(q, k, v, causal=True, scale=self.scale)
         else:
             if slot_rows is not None:
-                # one scatter per layer: slot_rows [B] = pool row of each
-                # sequence's current token (precomputed once per step)
-                kv.k_pools[layer_idx].view(-1, Hkv, D)[slot_rows] = k[:, 0]
-                kv.v_pools[layer_idx].view(-1, Hkv, D)[slot_rows] = v[:, 0]
+                # one launch per layer: slot_rows [B] = pool row of each
+                # sequence's current token (precomputed once per step);
+                # csrc/rope.hip kv_cache_append scatters K and V together
+                if use_hip_path:
+                    from ..ops import kernels
+
+                    kernels().kv_cache_append(
+                        k[:, 0].contiguous(), v[:, 0].contiguous(),
+                        kv.k_pools[layer_idx], kv.v_pools[layer_idx],
+                        slot_rows.to(torch.int32))
+                else:
+                    kv.k_pools[layer_idx].view(-1, Hkv, D)[slot_rows] = k[:, 0]
+                    kv.v_pools[layer_idx].view(-1, Hkv, D)[slot_rows] = v[:, 0]
             else:
                 for i, sid in enumerate(seq_ids):
                     kv.write_token(layer_idx, sid, int(seq_lens[i]) - 1, k[i, 0], v[i, 0])

@@ -467,6 +467,23 @@ def test_multi_tensor_sgd():
         torch.testing.assert_close(p.cpu(), pr.cpu(), rtol=1e-5, atol=1e-6)
 
 
+def test_kv_cache_append():
+    torch.manual_seed(16)
+    B, Hkv, D, rows = 5, 4, 128, 64
+    k = torch.randn(B, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    kpool = torch.zeros(rows, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    vpool = torch.zeros_like(kpool)
+    slots = torch.tensor([3, 17, 0, 63, 40], dtype=torch.int32, device="cuda")
+    _C.kv_cache_append(k, v, kpool, vpool, slots)
+    for b in range(B):
+        torch.testing.assert_close(kpool[int(slots[b])], k[b], rtol=0.0, atol=0.0)
+        torch.testing.assert_close(vpool[int(slots[b])], v[b], rtol=0.0, atol=0.0)
+    untouched = torch.ones(rows, dtype=torch.bool)
+    untouched[slots.long().cpu()] = False
+    assert kpool[untouched.to("cuda")].abs().sum() == 0
+
+
 def test_multi_tensor_lamb():
     """Two-stage fused LAMB vs the CPU Lamb math (incl. trust ratio) —
     many tensors to exercise the per-tensor norm accumulators across
