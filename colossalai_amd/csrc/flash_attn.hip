@@ -45,7 +45,10 @@ namespace fa {
 
 constexpr int KVB = 64;     // kv rows per tile
 constexpr int QW = 32;      // q rows per wave (fwd / dq); kv rows per wave (dkdv)
-constexpr int NW = 8;       // waves per block (forward)
+constexpr int NW = 8;       // waves per block (forward). NW=4 (two independent
+                            // blocks/CU for natural stagger) measured 319 vs 519
+                            // TF — the doubled K/V staging work loses more than
+                            // the barrier decoupling gains; struck.
 constexpr int NT = 512;     // threads per block (forward)
 // Backward kernels carry dK+dV (or dQ) accumulators plus K/V (or Q/dO)
 // fragments in registers (~380-450 VGPR+AGPR): 1 wave/SIMD — 4-wave blocks.

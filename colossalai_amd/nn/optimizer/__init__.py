@@ -29,9 +29,14 @@ def cast_to_distributed(optimizer):
     cls = mapping.get(type(optimizer))
     if cls is None:
         return optimizer
-    # rebuild from the same param groups + defaults
+    # rebuild from the same param groups + defaults, keeping only the
+    # kwargs the distributed variant accepts (e.g. Lamb's ``adam`` flag has
+    # no DistributedLamb counterpart)
+    import inspect
+
+    accepted = set(inspect.signature(cls.__init__).parameters) - {"self", "params"}
     groups = [dict(g) for g in optimizer.param_groups]
-    keys = set(optimizer.defaults)
-    kwargs = dict(optimizer.defaults)
+    keys = set(optimizer.defaults) & accepted
+    kwargs = {k: v for k, v in optimizer.defaults.items() if k in accepted}
     return cls([{"params": g["params"], **{k: g.get(k, kwargs.get(k)) for k in keys}} for g in groups],
                **kwargs)
