@@ -28,18 +28,21 @@ class LLMEngine:
         self._graphs = {}  # batch size -> (hipGraph, input buffers, output)
 
     def _ensure_caches(self, batch_size: int):
-        cfg = self.model.config
         Smax = self.config.max_seq_len
         if self._caches is not None and self._cache_bs >= batch_size:
             return
+        # size caches from the LAYER attributes, not the config: under TP
+        # (ShardFormer-sharded model, gather_output LM head) each rank holds
+        # num_kv_heads / tp local heads and caches only those
+        attn0 = self.model.model.layers[0].self_attn
         self._caches = [
             (
-                torch.zeros(batch_size, Smax, cfg.num_key_value_heads, cfg.head_dim,
+                torch.zeros(batch_size, Smax, attn0.num_kv_heads, attn0.head_dim,
                             device=self.device, dtype=self.dtype),
-                torch.zeros(batch_size, Smax, cfg.num_key_value_heads, cfg.head_dim,
+                torch.zeros(batch_size, Smax, attn0.num_kv_heads, attn0.head_dim,
                             device=self.device, dtype=self.dtype),
             )
-            for _ in range(cfg.num_hidden_layers)
+            for _ in range(len(self.model.model.layers))
         ]
         self._cache_bs = batch_size
 

@@ -68,3 +68,35 @@ def test_engine_hip_graph_matches_eager():
                                            use_hip_graph=True)
                     ).generate(prompts, GenerationConfig(max_new_tokens=12))
     assert out == ref, f"graph {out} vs eager {ref}"
+
+
+def _run_tp_engine(rank, world_size, port):
+    """TP-sharded serving: tp2 engine decode must match the unsharded engine
+    token-for-token (reference: inference tp via its own plugin/rpc executor)."""
+    import torch.distributed as dist
+
+    import colossalai_amd
+    from colossalai_amd.shardformer import ShardConfig, ShardFormer
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(_tiny()).eval()
+    ref_engine = LLMEngine(model, InferenceConfig(max_batch_size=2, max_input_len=32, max_output_len=16))
+    prompts = [[5, 17, 42, 7], [99, 3]]
+    ref_out = ref_engine.generate(prompts, GenerationConfig(max_new_tokens=8))
+
+    import copy
+
+    sharded, _ = ShardFormer(ShardConfig(tensor_parallel_process_group=dist.group.WORLD,
+                                         parallel_output=False)).optimize(copy.deepcopy(model))
+    assert sharded.model.layers[0].self_attn.num_heads * 2 == model.model.layers[0].self_attn.num_heads
+    engine = LLMEngine(sharded.eval(), InferenceConfig(max_batch_size=2, max_input_len=32, max_output_len=16))
+    out = engine.generate(prompts, GenerationConfig(max_new_tokens=8))
+    assert out == ref_out, f"tp engine {out} vs ref {ref_out}"
+    dist.destroy_process_group()
+
+
+def test_engine_tp2_matches_unsharded():
+    from colossalai_amd.testing import rerun_if_address_is_in_use, spawn
+
+    rerun_if_address_is_in_use()(lambda: spawn(_run_tp_engine, 2))()
