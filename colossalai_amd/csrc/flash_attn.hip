@@ -844,32 +844,44 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
       // ---- dS = scale * P ⊙ (dP - delta[q]) with P = exp(scale*s - lse[q]);
       // C-regs j in a 4-group are 4 consecutive q rows → packed 8 B store
       // into the per-wave [kv 32][q 32] subtiled tile.
+      // interior kb-halves (all q/k in range, below the diagonal) take a
+      // maskless dS loop — branching at loop level keeps each path's
+      // register footprint tight
+      const bool blk_full = (k0 + kb * 32 + 32 <= Seff) && (qw + QW <= Seff) &&
+                            (!causal || k0 + kb * 32 + 31 <= qw);
+      if (blk_full) {
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+          const int q_loc0 = 8 * jj + 4 * half;
+          short4v dk4;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const int r = q_loc0 + e;
+            const float pv = __expf(acc[jj * 4 + e] * scale - lse_lds[w * QW + r]);
+            const float ds = scale * pv * (acc2[jj * 4 + e] - dta_lds[w * QW + r]);
+            dk4[e] = (short)f2bf(ds);
+          }
+          *reinterpret_cast<short4v*>(P + sub_off<32>(ln, q_loc0)) = dk4;
+        }
+      } else {
 #pragma unroll
       for (int jj = 0; jj < 4; ++jj) {
         const int q_loc0 = 8 * jj + 4 * half;
         short4v dk4;
-        // interior kb-halves (all q/k in range, below the diagonal) skip
-        // the per-element mask chain
-        const bool blk_full = (k0 + kb * 32 + 32 <= Seff) && (qw + QW <= Seff) &&
-                              (!causal || k0 + kb * 32 + 31 <= qw);
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
           const int r = q_loc0 + e;
+          const int q_abs = qw + r;
+          const int k_abs = k0 + kb * 32 + ln;
           const float lse = lse_lds[w * QW + r];
           const float dta = dta_lds[w * QW + r];
-          float pv;
-          if (blk_full) {
-            pv = __expf(acc[jj * 4 + e] * scale - lse);
-          } else {
-            const int q_abs = qw + r;
-            const int k_abs = k0 + kb * 32 + ln;
-            const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
-            pv = valid ? __expf(acc[jj * 4 + e] * scale - lse) : 0.0f;
-          }
+          const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+          const float pv = valid ? __expf(acc[jj * 4 + e] * scale - lse) : 0.0f;
           const float ds = scale * pv * (acc2[jj * 4 + e] - dta);
           dk4[e] = (short)f2bf(ds);
         }
         *reinterpret_cast<short4v*>(P + sub_off<32>(ln, q_loc0)) = dk4;
+      }
       }
 
       // ---- dQ += dS · K   (A = dS tr-read: lane row = q, k = kv-local;
