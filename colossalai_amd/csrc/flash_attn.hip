@@ -213,6 +213,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     for (int sl = 0; sl < DSL; ++sl) qf[sl] = ld_g16(qp + sl * 16);
   }
 
+  const float scale2 = scale * 1.44269504f;  // scale * log2(e)
   f32x16 oacc[D / 32];
 #pragma unroll
   for (int nb = 0; nb < D / 32; ++nb)
@@ -264,10 +265,14 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
         }
         __builtin_amdgcn_s_setprio(0);
       }
+      // exp2-domain softmax: v_exp_f32 IS exp2, so fold scale*log2e into
+      // the copy-out and every __expf's hidden *log2e mul disappears.
+      // m_run / l_run / rowmax live in the log2 domain; LSE converts back
+      // at the epilogue.
 #pragma unroll
       for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
-        for (int j = 0; j < 16; ++j) p[kb][j] = acc[kb][j];
+        for (int j = 0; j < 16; ++j) p[kb][j] = acc[kb][j] * scale2;
     }
 
     // ---- online softmax over this tile's 64 scores of q-row (qw+ln)
@@ -281,10 +286,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
 #pragma unroll
       for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          p[kb][j] *= scale;
-          rowmax = fmaxf(rowmax, p[kb][j]);
-        }
+        for (int j = 0; j < 16; ++j) rowmax = fmaxf(rowmax, p[kb][j]);
     } else {
 #pragma unroll
     for (int kb = 0; kb < KB; ++kb)
@@ -292,7 +294,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
       for (int j = 0; j < 16; ++j) {
         const int k_abs = k0 + kb * 32 + crow(j, half);
         const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs);
-        p[kb][j] = valid ? p[kb][j] * scale : -INFINITY;
+        if (!valid) p[kb][j] = -INFINITY;
         rowmax = fmaxf(rowmax, p[kb][j]);
       }
     }
@@ -300,17 +302,17 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     // defer-max (guide T13): if this tile's max is within 8 of the running
     // max, keep the old max (P bounded by e^8, fine in fp32/bf16) and skip
     // the O-rescale pass entirely when every row defers.
-    const bool defer = (m_run != -INFINITY) && (rowmax - m_run <= 8.0f);
+    const bool defer = (m_run != -INFINITY) && (rowmax - m_run <= 11.5416f);  // 8 nats in log2
     const float m_new = defer ? m_run : fmaxf(m_run, rowmax);
     const float msafe = (m_new == -INFINITY) ? 0.0f : m_new;
     const float corr =
-        defer ? 1.0f : ((m_run == -INFINITY) ? ((m_new == -INFINITY) ? 1.0f : 0.0f) : __expf(m_run - m_new));
+        defer ? 1.0f : ((m_run == -INFINITY) ? ((m_new == -INFINITY) ? 1.0f : 0.0f) : __builtin_amdgcn_exp2f(m_run - m_new));
     float rowsum = 0.0f;
 #pragma unroll
     for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
-        p[kb][j] = __expf(p[kb][j] - msafe);
+        p[kb][j] = __builtin_amdgcn_exp2f(p[kb][j] - msafe);
         rowsum += p[kb][j];
       }
     rowsum += __shfl_xor(rowsum, 32);
@@ -387,7 +389,8 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     }
   }
   if (lane < QW && qw + lane < row_lim) {
-    const float lse = (l_run > 0.0f) ? m_run + __logf(l_run) : -INFINITY;
+    // back to nats: m_run is log2-domain
+    const float lse = (l_run > 0.0f) ? m_run * 0.6931472f + __logf(l_run) : -INFINITY;
     lse_base[qw + lane] = lse;
   }
 }
@@ -590,17 +593,21 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         {
           const int q_abs = qt0 + qb * 32 + ln;
           const float lse = lse_lds[qb * 32 + ln];
+          // exp2 domain: exp(a*s - l) = exp2(fma(a, s*log2e, -l*log2e))
+          const float scale2 = scale * 1.44269504f;
+          const float lse2 = lse * 1.44269504f;
           const bool blk_full = (qt0 + qb * 32 + 32 <= Seff) && (kvw + QW <= Seff) &&
                                 (!causal || kvw + QW - 1 <= qt0 + qb * 32);
           if (blk_full) {
 #pragma unroll
-            for (int j = 0; j < 16; ++j) p[j] = __expf(acc[j] * scale - lse);
+            for (int j = 0; j < 16; ++j)
+              p[j] = __builtin_amdgcn_exp2f(fmaf(acc[j], scale2, -lse2));
           } else {
 #pragma unroll
           for (int j = 0; j < 16; ++j) {
             const int k_abs = kvw + crow(j, half);
             const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
-            p[j] = valid ? __expf(acc[j] * scale - lse) : 0.0f;
+            p[j] = valid ? __builtin_amdgcn_exp2f(fmaf(acc[j], scale2, -lse2)) : 0.0f;
           }
           }
         }
@@ -857,7 +864,8 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
 #pragma unroll
           for (int e = 0; e < 4; ++e) {
             const int r = q_loc0 + e;
-            const float pv = __expf(acc[jj * 4 + e] * scale - lse_lds[w * QW + r]);
+            const float pv = __builtin_amdgcn_exp2f(
+                fmaf(acc[jj * 4 + e], scale * 1.44269504f, -lse_lds[w * QW + r] * 1.44269504f));
             const float ds = scale * pv * (acc2[jj * 4 + e] - dta_lds[w * QW + r]);
             dk4[e] = (short)f2bf(ds);
           }
@@ -876,7 +884,9 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
           const float lse = lse_lds[w * QW + r];
           const float dta = dta_lds[w * QW + r];
           const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
-          const float pv = valid ? __expf(acc[jj * 4 + e] * scale - lse) : 0.0f;
+          const float pv = valid
+              ? __builtin_amdgcn_exp2f(fmaf(acc[jj * 4 + e], scale * 1.44269504f, -lse * 1.44269504f))
+              : 0.0f;
           const float ds = scale * pv * (acc2[jj * 4 + e] - dta);
           dk4[e] = (short)f2bf(ds);
         }
