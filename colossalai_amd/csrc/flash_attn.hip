@@ -254,16 +254,27 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
       for (int kb = 0; kb < KB; ++kb)
 #pragma unroll
         for (int j = 0; j < 16; ++j) acc[kb][j] = 0.0f;
+      // software-pipelined A-operand loads: slice sl+1's LDS reads issue
+      // UNDER slice sl's MFMAs (the ld->mfma dependency chain was the
+      // limiter — each b128 LDS read is ~64 cycles the MFMA waited out)
+      bf16x8_t kf_cur[KB], kf_nxt[KB];
+#pragma unroll
+      for (int kb = 0; kb < KB; ++kb)
+        kf_cur[kb] = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, half * 8));
 #pragma unroll
       for (int sl = 0; sl < DSL; ++sl) {
+        if (sl + 1 < DSL) {
+#pragma unroll
+          for (int kb = 0; kb < KB; ++kb)
+            kf_nxt[kb] = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, (sl + 1) * 16 + half * 8));
+        }
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int kb = 0; kb < KB; ++kb) {
-          // A-operand: K rows. lane: row = kb*32+ln, cols d = sl*16 + half*8 + [0..7]
-          bf16x8_t kf = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
-          acc[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[sl], acc[kb], 0, 0, 0);
-        }
+        for (int kb = 0; kb < KB; ++kb)
+          acc[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_cur[kb], qf[sl], acc[kb], 0, 0, 0);
         __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int kb = 0; kb < KB; ++kb) kf_cur[kb] = kf_nxt[kb];
       }
       // exp2-domain softmax: v_exp_f32 IS exp2, so fold scale*log2e into
       // the copy-out and every __expf's hidden *log2e mul disappears.
@@ -345,17 +356,29 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     }
 
     // ---- O += P · V  (A = P from LDS, B = V column-fragments via tr-read)
+    // software-pipelined: the NEXT (pa, vb) pair's LDS reads issue under
+    // the current MFMA so the ld->mfma chain never serializes
+    {
+      constexpr int KS = KVB / 16, NB = D / 32;
+      bf16x8_t pa_cur = ld_lds16(P, swz<KVB * 2>(ln, (half * 8) * 2));
+      bf16x8_t vb_cur = ld_frag_tr<KVB>(Vlds, lane, 0, 0);
+      bf16x8_t pa_nxt, vb_nxt;
 #pragma unroll
-    for (int ks = 0; ks < KVB / 16; ++ks) {
-      // A: lane row = q (ln), k-cols = ks*16 + half*8 + [0..7]
-      bf16x8_t pa = ld_lds16(P, swz<KVB * 2>(ln, (ks * 16 + half * 8) * 2));
+      for (int ks = 0; ks < KS; ++ks) {
 #pragma unroll
-      for (int nb = 0; nb < D / 32; ++nb) {
-        // B: lane col = d (nb*32+ln), k-rows kv = ks*16 + half*8 + [0..7]
-        bf16x8_t vb = ld_frag_tr<KVB>(Vlds, lane, ks * 16, nb * 2);
-        __builtin_amdgcn_s_setprio(1);
-        oacc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, oacc[nb], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+        for (int nb = 0; nb < NB; ++nb) {
+          if (nb + 1 < NB) {
+            vb_nxt = ld_frag_tr<KVB>(Vlds, lane, ks * 16, (nb + 1) * 2);
+          } else if (ks + 1 < KS) {
+            pa_nxt = ld_lds16(P, swz<KVB * 2>(ln, ((ks + 1) * 16 + half * 8) * 2));
+            vb_nxt = ld_frag_tr<KVB>(Vlds, lane, (ks + 1) * 16, 0);
+          }
+          __builtin_amdgcn_s_setprio(1);
+          oacc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa_cur, vb_cur, oacc[nb], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+          vb_cur = vb_nxt;
+          if (nb + 1 == NB) pa_cur = pa_nxt;
+        }
       }
     }
     }  // active
