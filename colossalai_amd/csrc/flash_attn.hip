@@ -600,14 +600,18 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         f32x16 acc;
 #pragma unroll
         for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+        // pipelined: slice sl+1's Q row-slice read issues under slice sl's MFMA
+        bf16x8_t qb_cur = ld_lds16(Qlds, sub_off<QB>(qb * 32 + ln, half * 8));
+        bf16x8_t qb_nxt;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
-          // B: lane col = q (qb*32+ln), k-rows d → row-slice read
+          if (sl + 1 < DSL)
+            qb_nxt = ld_lds16(Qlds, sub_off<QB>(qb * 32 + ln, (sl + 1) * 16 + half * 8));
           bf16x8_t kfr = (VAR == 3) ? ld_g16(kp + sl * 16) : kf[VAR == 3 ? 0 : sl];
-          bf16x8_t qb_frag = ld_lds16(Qlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
           __builtin_amdgcn_s_setprio(1);
-          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qb_frag, acc, 0, 0, 0);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qb_cur, acc, 0, 0, 0);
           __builtin_amdgcn_s_setprio(0);
+          qb_cur = qb_nxt;
         }
 
         // ---- P' = exp(scale*s - lse[q]), masked (interior blocks skip
@@ -648,15 +652,27 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
 
         // ---- dV += P' · dO  (A = P' via tr-read: lane row = kv, k = q-local;
         //      B = dO column-fragments, k = q-global)
+        {
+          constexpr int NB = D / 32;
+          bf16x8_t pa_cur = ld_frag_tr<32>(P, lane, 0, 0);
+          bf16x8_t dob_cur = ld_frag_tr<QB>(dOlds, lane, qb * 32, 0);
+          bf16x8_t pa_nxt, dob_nxt;
 #pragma unroll
-        for (int ks2 = 0; ks2 < 2; ++ks2) {
-          bf16x8_t pa = ld_frag_tr<32>(P, lane, ks2 * 16, 0);
+          for (int ks2 = 0; ks2 < 2; ++ks2) {
 #pragma unroll
-          for (int nb = 0; nb < D / 32; ++nb) {
-            bf16x8_t dob = ld_frag_tr<QB>(dOlds, lane, qb * 32 + ks2 * 16, nb * 2);
-            __builtin_amdgcn_s_setprio(1);
-            dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[nb], 0, 0, 0);
-            __builtin_amdgcn_s_setprio(0);
+            for (int nb = 0; nb < NB; ++nb) {
+              if (nb + 1 < NB) {
+                dob_nxt = ld_frag_tr<QB>(dOlds, lane, qb * 32 + ks2 * 16, (nb + 1) * 2);
+              } else if (ks2 == 0) {
+                pa_nxt = ld_frag_tr<32>(P, lane, 16, 0);
+                dob_nxt = ld_frag_tr<QB>(dOlds, lane, qb * 32 + 16, 0);
+              }
+              __builtin_amdgcn_s_setprio(1);
+              dv_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa_cur, dob_cur, dv_acc[nb], 0, 0, 0);
+              __builtin_amdgcn_s_setprio(0);
+              dob_cur = dob_nxt;
+              if (nb + 1 == NB) pa_cur = pa_nxt;
+            }
           }
         }
 
@@ -665,11 +681,15 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         f32x16 acc2;
 #pragma unroll
         for (int j = 0; j < 16; ++j) acc2[j] = 0.0f;
+        bf16x8_t do_cur = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, half * 8));
+        bf16x8_t do_nxt;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
+          if (sl + 1 < DSL)
+            do_nxt = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, (sl + 1) * 16 + half * 8));
           bf16x8_t vfr = (VAR >= 2) ? ld_g16(vp + sl * 16) : vf[VAR >= 2 ? 0 : sl];
-          bf16x8_t dob = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, sl * 16 + half * 8));
-          acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dob, acc2, 0, 0, 0);
+          acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, do_cur, acc2, 0, 0, 0);
+          do_cur = do_nxt;
         }
 
         // ---- dS' = scale * P' ⊙ (dP' - delta[q]) → overwrite P tile
@@ -689,16 +709,28 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         }
 
         // ---- dK += dS' · Q   (A = dS' tr-read, B = Q column-fragments)
+        {
+          constexpr int NB = D / 32;
+          bf16x8_t dsa_cur = ld_frag_tr<32>(P, lane, 0, 0);
+          bf16x8_t qtb_cur = ld_frag_tr<QB>(Qlds, lane, qb * 32, 0);
+          bf16x8_t dsa_nxt, qtb_nxt;
 #pragma unroll
         for (int ks2 = 0; ks2 < 2; ++ks2) {
-          bf16x8_t dsa = ld_frag_tr<32>(P, lane, ks2 * 16, 0);
 #pragma unroll
-          for (int nb = 0; nb < D / 32; ++nb) {
-            bf16x8_t qtb = ld_frag_tr<QB>(Qlds, lane, qb * 32 + ks2 * 16, nb * 2);
+          for (int nb = 0; nb < NB; ++nb) {
+            if (nb + 1 < NB) {
+              qtb_nxt = ld_frag_tr<QB>(Qlds, lane, qb * 32 + ks2 * 16, (nb + 1) * 2);
+            } else if (ks2 == 0) {
+              dsa_nxt = ld_frag_tr<32>(P, lane, 16, 0);
+              qtb_nxt = ld_frag_tr<QB>(Qlds, lane, qb * 32 + 16, 0);
+            }
             __builtin_amdgcn_s_setprio(1);
-            dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb, dk_acc[nb], 0, 0, 0);
+            dk_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa_cur, qtb_cur, dk_acc[nb], 0, 0, 0);
             __builtin_amdgcn_s_setprio(0);
+            qtb_cur = qtb_nxt;
+            if (nb + 1 == NB) dsa_cur = dsa_nxt;
           }
+        }
         }
       }  // qb halves
       }  // active
@@ -861,14 +893,21 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
       f32x16 acc, acc2;
 #pragma unroll
       for (int j = 0; j < 16; ++j) { acc[j] = 0.0f; acc2[j] = 0.0f; }
+      // software-pipelined: slice sl+1's K/V row-slice reads issue under
+      // slice sl's MFMAs (same restructure as the forward, +10% there)
+      bf16x8_t kb_cur = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, half * 8));
+      bf16x8_t vb_cur = ld_lds16(Vlds, sub_off<KVB>(kb * 32 + ln, half * 8));
+      bf16x8_t kb_nxt, vb_nxt;
 #pragma unroll
       for (int sl = 0; sl < DSL; ++sl) {
-        // B = K^T: lane col = kv (kb*32+ln), rows d → row-slice read
-        bf16x8_t kb_frag = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa[sl], kb_frag, acc, 0, 0, 0);
-        // B = V^T: same pattern → dP = dO · V^T
-        bf16x8_t vb_frag = ld_lds16(Vlds, sub_off<KVB>(kb * 32 + ln, sl * 16 + half * 8));
-        acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa[sl], vb_frag, acc2, 0, 0, 0);
+        if (sl + 1 < DSL) {
+          kb_nxt = ld_lds16(Klds, sub_off<KVB>(kb * 32 + ln, (sl + 1) * 16 + half * 8));
+          vb_nxt = ld_lds16(Vlds, sub_off<KVB>(kb * 32 + ln, (sl + 1) * 16 + half * 8));
+        }
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa[sl], kb_cur, acc, 0, 0, 0);
+        acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(doa[sl], vb_cur, acc2, 0, 0, 0);
+        kb_cur = kb_nxt;
+        vb_cur = vb_nxt;
       }
 
       // ---- dS = scale * P ⊙ (dP - delta[q]) with P = exp(scale*s - lse[q]);
@@ -919,15 +958,27 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
 
       // ---- dQ += dS · K   (A = dS tr-read: lane row = q, k = kv-local;
       //      B = K column-fragments, k = kv-global)
+      {
+        constexpr int NB = D / 32;
+        bf16x8_t dsa_cur = ld_frag_tr<32>(P, lane, 0, 0);
+        bf16x8_t ktb_cur = ld_frag_tr<KVB>(Klds, lane, kb * 32, 0);
+        bf16x8_t dsa_nxt, ktb_nxt;
 #pragma unroll
-      for (int ks2 = 0; ks2 < 2; ++ks2) {
-        bf16x8_t dsa = ld_frag_tr<32>(P, lane, ks2 * 16, 0);
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
 #pragma unroll
-        for (int nb = 0; nb < D / 32; ++nb) {
-          bf16x8_t ktb = ld_frag_tr<KVB>(Klds, lane, kb * 32 + ks2 * 16, nb * 2);
-          __builtin_amdgcn_s_setprio(1);
-          dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, ktb, dq_acc[nb], 0, 0, 0);
-          __builtin_amdgcn_s_setprio(0);
+          for (int nb = 0; nb < NB; ++nb) {
+            if (nb + 1 < NB) {
+              ktb_nxt = ld_frag_tr<KVB>(Klds, lane, kb * 32 + ks2 * 16, (nb + 1) * 2);
+            } else if (ks2 == 0) {
+              dsa_nxt = ld_frag_tr<32>(P, lane, 16, 0);
+              ktb_nxt = ld_frag_tr<KVB>(Klds, lane, kb * 32 + 16, 0);
+            }
+            __builtin_amdgcn_s_setprio(1);
+            dq_acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa_cur, ktb_cur, dq_acc[nb], 0, 0, 0);
+            __builtin_amdgcn_s_setprio(0);
+            ktb_cur = ktb_nxt;
+            if (nb + 1 == NB) dsa_cur = dsa_nxt;
+          }
         }
       }
     }  // kb halves
