@@ -153,23 +153,33 @@ __global__ __launch_bounds__(gg::NT) void grouped_gemm_kernel(
       else bstage.load(w_base, Kin, n0, 1 << 30, k0 + BK);
     }
 
+    // software-pipelined (as in flash_attn.hip): the next (af, bf) LDS
+    // reads issue under the current MFMA
+    {
+      constexpr int KS = BK / 16, NB = TN / 32;
+      auto ldb = [&](int ks, int nb) {
+        return TRANS_W ? ld_frag_tr<BK>(Blds, lane, ks * 16, nb * 2)
+                       : ld_lds16b(Blds, sub_off<TN>(nb * 32 + ln, ks * 16 + half * 8));
+      };
+      bf16x8_v af_cur = ld_lds16b(Alds, sub_off<TM>(w * 32 + ln, half * 8));
+      bf16x8_v bf_cur = ldb(0, 0);
+      bf16x8_v af_nxt, bf_nxt;
 #pragma unroll
-    for (int ks = 0; ks < BK / 16; ++ks) {
-      // A-frag: lane row = token (w*32+ln), 8 consecutive contraction elems
-      bf16x8_v af = ld_lds16b(Alds, sub_off<TM>(w * 32 + ln, ks * 16 + half * 8));
+      for (int ks = 0; ks < KS; ++ks) {
 #pragma unroll
-      for (int nb = 0; nb < TN / 32; ++nb) {
-        bf16x8_v bf;
-        if (TRANS_W) {
-          // B [BK contraction rows][TN out cols]: column fragment
-          bf = ld_frag_tr<BK>(Blds, lane, ks * 16, nb * 2);
-        } else {
-          // B [TN out rows][BK]: row-slice at out row nb*32+ln
-          bf = ld_lds16b(Blds, sub_off<TN>(nb * 32 + ln, ks * 16 + half * 8));
+        for (int nb = 0; nb < NB; ++nb) {
+          if (nb + 1 < NB) {
+            bf_nxt = ldb(ks, nb + 1);
+          } else if (ks + 1 < KS) {
+            af_nxt = ld_lds16b(Alds, sub_off<TM>(w * 32 + ln, (ks + 1) * 16 + half * 8));
+            bf_nxt = ldb(ks + 1, 0);
+          }
+          __builtin_amdgcn_s_setprio(1);
+          acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af_cur, bf_cur, acc[nb], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+          bf_cur = bf_nxt;
+          if (nb + 1 == NB) af_cur = af_nxt;
         }
-        __builtin_amdgcn_s_setprio(1);
-        acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc[nb], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
       }
     }
 
@@ -261,15 +271,27 @@ __global__ __launch_bounds__(gg::NT) void grouped_gemm_wgrad_kernel(
     // A-frag col index = m = w*32 + (lane&31); ld_frag_tr's internal
     // cg = cg_base + ((lane>>4)&1) covers 32 consecutive cols from
     // cg_base*16 — cg_base = w*2 selects this wave's 32-m window.
+    {
+      constexpr int KS = BK / 16, NB = TN / 32;
+      bf16x8_v af_cur = ld_frag_tr<BK>(DYlds, lane, 0, w * 2);
+      bf16x8_v bf_cur = ld_frag_tr<BK>(Xlds, lane, 0, 0);
+      bf16x8_v af_nxt, bf_nxt;
 #pragma unroll
-    for (int ks = 0; ks < BK / 16; ++ks) {
-      bf16x8_v af = ld_frag_tr<BK>(DYlds, lane, ks * 16, w * 2);
+      for (int ks = 0; ks < KS; ++ks) {
 #pragma unroll
-      for (int nb = 0; nb < TN / 32; ++nb) {
-        bf16x8_v bf = ld_frag_tr<BK>(Xlds, lane, ks * 16, nb * 2);
-        __builtin_amdgcn_s_setprio(1);
-        acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc[nb], 0, 0, 0);
-        __builtin_amdgcn_s_setprio(0);
+        for (int nb = 0; nb < NB; ++nb) {
+          if (nb + 1 < NB) {
+            bf_nxt = ld_frag_tr<BK>(Xlds, lane, ks * 16, (nb + 1) * 2);
+          } else if (ks + 1 < KS) {
+            af_nxt = ld_frag_tr<BK>(DYlds, lane, (ks + 1) * 16, w * 2);
+            bf_nxt = ld_frag_tr<BK>(Xlds, lane, (ks + 1) * 16, 0);
+          }
+          __builtin_amdgcn_s_setprio(1);
+          acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af_cur, bf_cur, acc[nb], 0, 0, 0);
+          __builtin_amdgcn_s_setprio(0);
+          bf_cur = bf_nxt;
+          if (nb + 1 == NB) af_cur = af_nxt;
+        }
       }
     }
 

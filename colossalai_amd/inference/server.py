@@ -78,6 +78,41 @@ def create_app(engine: ContinuousBatchEngine, tokenizer=None):
                           "completion_tokens": sum(len(c["token_ids"]) for c in choices),
                           "total_tokens": len(ids) + sum(len(c["token_ids"]) for c in choices)}}
 
+    class ChatMessage(BaseModel):
+        role: str
+        content: str
+
+    class ChatRequest(BaseModel):
+        model: str = "colossalai_amd"
+        messages: List[ChatMessage]
+        max_tokens: int = 64
+        temperature: float = 1.0
+        top_p: float = 1.0
+
+    @app.post("/v1/chat/completions")
+    def chat_completions(req: ChatRequest):
+        """OpenAI-compatible chat endpoint. Uses the tokenizer's chat
+        template when available, else a plain role-tagged transcript."""
+        if tokenizer is None:
+            raise HTTPException(400, "chat completions need a tokenizer attached")
+        msgs = [{"role": m.role, "content": m.content} for m in req.messages]
+        if hasattr(tokenizer, "apply_chat_template"):
+            ids = tokenizer.apply_chat_template(msgs, add_generation_prompt=True)
+        else:
+            text = "".join(f"<|{m['role']}|>\n{m['content']}\n" for m in msgs) + "<|assistant|>\n"
+            ids = tokenizer(text)["input_ids"]
+        gen = GenerationConfig(max_new_tokens=req.max_tokens, do_sample=req.temperature > 0,
+                               temperature=max(req.temperature, 1e-5), top_p=req.top_p)
+        out = engine.generate([list(ids)], gen)[0]
+        new = out[len(ids):]
+        return {"id": f"chatcmpl-{id(out) & 0xFFFFFF:x}", "object": "chat.completion",
+                "model": req.model,
+                "choices": [{"index": 0,
+                             "message": {"role": "assistant", "content": tokenizer.decode(new)},
+                             "finish_reason": "length" if len(new) >= req.max_tokens else "stop"}],
+                "usage": {"prompt_tokens": len(ids), "completion_tokens": len(new),
+                          "total_tokens": len(ids) + len(new)}}
+
     @app.post("/generate", response_model=GenerateResponse)
     def generate(req: GenerateRequest):
         if req.prompt_ids is None:

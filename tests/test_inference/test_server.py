@@ -51,3 +51,38 @@ def test_openai_completions_endpoint():
     assert len(body["choices"]) == 1
     assert len(body["choices"][0]["token_ids"]) == 6
     assert body["usage"]["total_tokens"] == 9
+
+
+def test_openai_chat_endpoint():
+    from starlette.testclient import TestClient
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=128)
+    model = LlamaForCausalLM(cfg).eval()
+    engine = ContinuousBatchEngine(model, InferenceConfig(max_batch_size=2, max_input_len=64,
+                                                          max_output_len=16), block_size=4)
+
+    class ToyTok:
+        """char-level stand-in with no chat template"""
+
+        def __call__(self, text):
+            return {"input_ids": [min(ord(c), 127) for c in text][:40]}
+
+        def decode(self, ids):
+            return "".join(chr(i % 128) for i in ids)
+
+    client = TestClient(create_app(engine, tokenizer=ToyTok()))
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "hi"}], "max_tokens": 5, "temperature": 0})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    assert body["usage"]["completion_tokens"] == 5
+
+    # no tokenizer -> 400
+    client2 = TestClient(create_app(engine))
+    r = client2.post("/v1/chat/completions", json={
+        "messages": [{"role": "user", "content": "hi"}]})
+    assert r.status_code == 400
