@@ -74,11 +74,13 @@ at::Tensor moe_dispatch_bwd(at::Tensor grad, at::Tensor src);
 
 // flash_attn.hip
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale,
-                                       c10::optional<at::Tensor> seqlens);
+                                       c10::optional<at::Tensor> seqlens,
+                                       c10::optional<at::Tensor> seqlens_k);
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
                                        at::Tensor out, at::Tensor lse, bool causal, double scale,
                                        at::Tensor dq, at::Tensor dk, at::Tensor dv,
-                                       c10::optional<at::Tensor> seqlens);
+                                       c10::optional<at::Tensor> seqlens,
+                                       c10::optional<at::Tensor> seqlens_k);
 std::vector<at::Tensor> flash_attn_varlen_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                               at::Tensor cu_seqlens, long max_seqlen,
                                               bool causal, double scale);
@@ -126,11 +128,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &cai::flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA; optional right-padding seqlens)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),
-        py::arg("seqlens") = py::none());
+        py::arg("seqlens") = py::none(), py::arg("seqlens_k") = py::none());
   m.def("flash_attn_bwd", &cai::flash_attn_bwd, "flash attention backward",
         py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("out"), py::arg("lse"),
         py::arg("causal"), py::arg("scale"), py::arg("dq"), py::arg("dk"), py::arg("dv"),
-        py::arg("seqlens") = py::none());
+        py::arg("seqlens") = py::none(), py::arg("seqlens_k") = py::none());
   m.def("flash_attn_varlen_fwd", &cai::flash_attn_varlen_fwd,
         "flash attention forward over a packed ragged batch (cu_seqlens)");
   m.def("flash_attn_varlen_bwd", &cai::flash_attn_varlen_bwd, "varlen flash attention backward");

@@ -111,7 +111,7 @@ struct RegStage {
     for (int i = 0; i < NCH; ++i) {
       const int c = (int)threadIdx.x + i * NT_;
       const int row = c / CPR, ch = c % CPR;
-      const int grow = min(row0 + row, S - 1);
+      const int grow = max(min(row0 + row, S - 1), 0);
       r[i] = ld_g16(base + (long)grow * tok_stride + ch * 8);
     }
   }
@@ -155,6 +155,9 @@ DEV_INLINE int swz(int row, int byte_in_row) {
 //  - VARLEN=true, CU=cu_seqlens[n+1]: packed ragged batch [total,H,D]
 //    passed with S=total and grid.y = n_seq*Hq; sequence b spans tokens
 //    [CU[b], CU[b+1]); LSE/delta are [Hq, total].
+//  - CUK (padded path only): SEPARATE kv-side valid counts — ring-attention
+//    pieces where the q chunk and the arriving kv chunk cover different
+//    parts of each padded sequence. null -> kv counts = CU.
 template <int D, bool VARLEN>
 __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     const unsigned short* __restrict__ Q,
@@ -163,8 +166,9 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     unsigned short* __restrict__ O,
     float* __restrict__ LSE,
     const int* __restrict__ CU,
+    const int* __restrict__ CUK,
     long qbs, long qts, long kbs, long kts,
-    int B, int S, int Hq, int Hkv, float scale, int causal) {
+    int B, int S, int Sk, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
   extern __shared__ char smem[];
   constexpr int KB_BYTES = sub_bytes<KVB, D>();
@@ -194,6 +198,9 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   } else if (CU != nullptr) {
     Seff = CU[b];
   }
+  // kv extent: its own tensor length Sk (ring blocks where kv is a piece of
+  // the sequence), further clipped by per-batch valid counts
+  int SeffK = VARLEN ? Seff : (CUK != nullptr ? CUK[b] : min(Seff, Sk));
 
   const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
   const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
@@ -221,24 +228,26 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     for (int j = 0; j < 16; ++j) oacc[nb][j] = 0.0f;
   float m_run = -INFINITY, l_run = 0.0f;
 
-  const int kv_tiles_all = (Seff + KVB - 1) / KVB;
+  const int kv_tiles_all = (SeffK + KVB - 1) / KVB;
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NW * QW + KVB - 1) / KVB) : kv_tiles_all;
 
   // async-split staging (T14): tile 0 staged synchronously, tile t+1's global
   // loads issue before tile t's compute and land in LDS after the barrier.
   RegStage<D, KVB, NT> kstage, vstage;
-  kstage.load(k_base, kts, 0, Seff);
-  vstage.load(v_base, kts, 0, Seff);
-  kstage.store_subtiled(Klds);
-  vstage.store_subtiled(Vlds);
+  if (kv_tiles > 0) {
+    kstage.load(k_base, kts, 0, SeffK);
+    vstage.load(v_base, kts, 0, SeffK);
+    kstage.store_subtiled(Klds);
+    vstage.store_subtiled(Vlds);
+  }
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
     const int k0 = t * KVB;
     const bool has_next = (t + 1 < kv_tiles);
     if (has_next) {
-      kstage.load(k_base, kts, k0 + KVB, Seff);
-      vstage.load(v_base, kts, k0 + KVB, Seff);
+      kstage.load(k_base, kts, k0 + KVB, SeffK);
+      vstage.load(v_base, kts, k0 + KVB, SeffK);
     }
     // waves entirely above the diagonal produce nothing (barriers stay uniform)
     const bool active = !(causal && k0 > qw + QW - 1) && (qw < Seff);
@@ -292,7 +301,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     // interior tiles (every row/col in range, strictly below the causal
     // diagonal) skip the per-element mask math — at S=4096 that is ~94%
     // of tiles, and the mask chain was ~4 VALU ops per score
-    const bool full = (k0 + KVB <= Seff) && (qw + QW <= Seff) && (!causal || k0 + KVB - 1 <= qw);
+    const bool full = (k0 + KVB <= SeffK) && (qw + QW <= Seff) && (!causal || k0 + KVB - 1 <= qw);
     if (full) {
 #pragma unroll
       for (int kb = 0; kb < KB; ++kb)
@@ -304,7 +313,7 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const int k_abs = k0 + kb * 32 + crow(j, half);
-        const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs);
+        const bool valid = (q_abs < Seff) && (k_abs < SeffK) && (!causal || k_abs <= q_abs);
         if (!valid) p[kb][j] = -INFINITY;
         rowmax = fmaxf(rowmax, p[kb][j]);
       }
@@ -486,8 +495,9 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
     unsigned short* __restrict__ dK,
     unsigned short* __restrict__ dV,
     const int* __restrict__ CU,
+    const int* __restrict__ CUK,  // padded ring pieces: kv-side valid counts
     long qbs, long qts, long kbs, long kts, long dkbs, long dkts,
-    int B, int S, int Hq, int Hkv, float scale, int causal) {
+    int B, int S, int Sk, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
   constexpr int QB = 64;  // q rows per staged tile
   extern __shared__ char smem[];
@@ -521,6 +531,8 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
   } else if (CU != nullptr) {
     Seff = CU[b];
   }
+  // this kernel's OWN rows are kv (extent Sk); q streams over Seff
+  int SeffK = VARLEN ? Seff : (CUK != nullptr ? CUK[b] : min(Seff, Sk));
 
   const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
   const unsigned short* v_base = VARLEN ? V + seq0 * kts + (long)hk * D : V + (long)b * kbs + (long)hk * D;
@@ -529,7 +541,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
   char* P = Pw + w * PB;
 
   // ---- K, V fragments (A-operands): lane row = kv (ln), d = sl*16+half*8+[0..7]
-  const int kv_my = min(kvw + ln, Seff - 1);
+  const int kv_my = max(min(kvw + ln, SeffK - 1), 0);
   const unsigned short* vp = v_base + (long)kv_my * kts + half * 8;
   const unsigned short* kp = k_base + (long)kv_my * kts + half * 8;
   bf16x8_t kf[VAR == 3 ? 1 : DSL], vf[VAR >= 2 ? 1 : DSL];
@@ -577,7 +589,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
       stage_direct<D, QB, NTB>(dOlds, do_base, do_stride, qt_first * QB, Seff);
     }
     if (threadIdx.x < QB) {
-      const int qr = min(qt_first * QB + (int)threadIdx.x, Seff - 1);
+      const int qr = max(min(qt_first * QB + (int)threadIdx.x, Seff - 1), 0);
       lse_lds[threadIdx.x] = lse_base[qr];
       dta_lds[threadIdx.x] = dta_base[qr];
     }
@@ -590,7 +602,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         qstage.load(q_base, qts, qt0 + QB, Seff);
         dostage.load(do_base, do_stride, qt0 + QB, Seff);
       }
-      const bool active = !(causal && qt0 + QB - 1 < kvw) && (kvw < Seff);
+      const bool active = !(causal && qt0 + QB - 1 < kvw) && (kvw < SeffK);
       if (active) {
 
       // 32-q halves: live score state is p[16]+dp[16] instead of [2][16].
@@ -623,7 +635,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
           // exp2 domain: exp(a*s - l) = exp2(fma(a, s*log2e, -l*log2e))
           const float scale2 = scale * 1.44269504f;
           const float lse2 = lse * 1.44269504f;
-          const bool blk_full = (qt0 + qb * 32 + 32 <= Seff) && (kvw + QW <= Seff) &&
+          const bool blk_full = (qt0 + qb * 32 + 32 <= Seff) && (kvw + QW <= SeffK) &&
                                 (!causal || kvw + QW - 1 <= qt0 + qb * 32);
           if (blk_full) {
 #pragma unroll
@@ -633,7 +645,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
 #pragma unroll
           for (int j = 0; j < 16; ++j) {
             const int k_abs = kvw + crow(j, half);
-            const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+            const bool valid = (q_abs < Seff) && (k_abs < SeffK) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
             p[j] = valid ? __builtin_amdgcn_exp2f(fmaf(acc[j], scale2, -lse2)) : 0.0f;
           }
           }
@@ -745,7 +757,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
           stage_direct<D, QB, NTB>(dOlds, do_base, do_stride, qt0 + QB, Seff);
         }
         if (threadIdx.x < QB) {
-          const int qr = min(qt0 + QB + (int)threadIdx.x, Seff - 1);
+          const int qr = max(min(qt0 + QB + (int)threadIdx.x, Seff - 1), 0);
           lse_lds[threadIdx.x] = lse_base[qr];
           dta_lds[threadIdx.x] = dta_base[qr];
         }
@@ -756,7 +768,7 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
 
   // ---- epilogue: write dK, dV (each kv row owned by exactly one block).
   // Padded rows accumulate nothing and are written as zeros.
-  const int kv_lim = VARLEN ? Seff : S;
+  const int kv_lim = VARLEN ? Seff : Sk;
   if (kvw >= kv_lim) return;
   unsigned short* dk_base = VARLEN ? dK + seq0 * dkts + (long)hk * D : dK + (long)b * dkbs + (long)hk * D;
   unsigned short* dv_base = VARLEN ? dV + seq0 * dkts + (long)hk * D : dV + (long)b * dkbs + (long)hk * D;
@@ -791,8 +803,9 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     const float* __restrict__ DELTA,
     unsigned short* __restrict__ dQ,
     const int* __restrict__ CU,
+    const int* __restrict__ CUK,  // padded ring pieces: kv-side valid counts
     long qbs, long qts, long kbs, long kts, long dqbs, long dqts,
-    int B, int S, int Hq, int Hkv, float scale, int causal) {
+    int B, int S, int Sk, int Hq, int Hkv, float scale, int causal) {
   using namespace fa;
   extern __shared__ char smem[];
   constexpr int KB_BYTES = sub_bytes<KVB, D>();
@@ -825,6 +838,7 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
   } else if (CU != nullptr) {
     Seff = CU[b];
   }
+  const int SeffK = VARLEN ? Seff : (CUK != nullptr ? CUK[b] : min(Seff, Sk));
 
   const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
   const unsigned short* do_base =
@@ -841,14 +855,14 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
     const float* lse_base = VARLEN ? LSE + (long)h * S + seq0 : LSE + ((long)b * Hq + h) * S;
     const float* dta_base = VARLEN ? DELTA + (long)h * S + seq0 : DELTA + ((long)b * Hq + h) * S;
     for (int i = threadIdx.x; i < NWB * QW; i += NTB) {
-      const int qr = min(q0 + i, Seff - 1);
+      const int qr = max(min(q0 + i, Seff - 1), 0);
       lse_lds[i] = lse_base[qr];
       dta_lds[i] = dta_base[qr];
     }
   }
 
   // ---- Q, dO fragments (A-operands): lane row = q (ln), d cols
-  const int q_my = min(qw + ln, Seff - 1);
+  const int q_my = max(min(qw + ln, Seff - 1), 0);
   bf16x8_t qa[DSL], doa[DSL];
   {
     const unsigned short* qp = q_base + (long)q_my * qts + half * 8;
@@ -866,22 +880,24 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
 #pragma unroll
     for (int j = 0; j < 16; ++j) dq_acc[nb][j] = 0.0f;
 
-  const int kv_tiles_all = (Seff + KVB - 1) / KVB;
+  const int kv_tiles_all = (SeffK + KVB - 1) / KVB;
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NWB * QW + KVB - 1) / KVB) : kv_tiles_all;
 
   RegStage<D, KVB, NTB> kstage, vstage;
-  kstage.load(k_base, kts, 0, Seff);
-  vstage.load(v_base, kts, 0, Seff);
-  kstage.store_subtiled(Klds);
-  vstage.store_subtiled(Vlds);
+  if (kv_tiles > 0) {
+    kstage.load(k_base, kts, 0, SeffK);
+    vstage.load(v_base, kts, 0, SeffK);
+    kstage.store_subtiled(Klds);
+    vstage.store_subtiled(Vlds);
+  }
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
     const int k0 = t * KVB;
     const bool has_next = (t + 1 < kv_tiles);
     if (has_next) {
-      kstage.load(k_base, kts, k0 + KVB, Seff);
-      vstage.load(v_base, kts, k0 + KVB, Seff);
+      kstage.load(k_base, kts, k0 + KVB, SeffK);
+      vstage.load(v_base, kts, k0 + KVB, SeffK);
     }
     const bool active = !(causal && k0 > qw + QW - 1) && (qw < Seff);
     if (active) {
@@ -916,7 +932,7 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
       // interior kb-halves (all q/k in range, below the diagonal) take a
       // maskless dS loop — branching at loop level keeps each path's
       // register footprint tight
-      const bool blk_full = (k0 + kb * 32 + 32 <= Seff) && (qw + QW <= Seff) &&
+      const bool blk_full = (k0 + kb * 32 + 32 <= SeffK) && (qw + QW <= Seff) &&
                             (!causal || k0 + kb * 32 + 31 <= qw);
       if (blk_full) {
 #pragma unroll
@@ -945,7 +961,7 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
           const int k_abs = k0 + kb * 32 + ln;
           const float lse = lse_lds[w * QW + r];
           const float dta = dta_lds[w * QW + r];
-          const bool valid = (q_abs < Seff) && (k_abs < Seff) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
+          const bool valid = (q_abs < Seff) && (k_abs < SeffK) && (!causal || k_abs <= q_abs) && (lse != -INFINITY);
           const float pv = valid
               ? __builtin_amdgcn_exp2f(fmaf(acc[jj * 4 + e], scale * 1.44269504f, -lse * 1.44269504f))
               : 0.0f;
@@ -1042,17 +1058,26 @@ void check_fa_inputs(const at::Tensor& q, const at::Tensor& k, const at::Tensor&
 }  // namespace
 
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal, double scale,
-                                       c10::optional<at::Tensor> seqlens_opt) {
+                                       c10::optional<at::Tensor> seqlens_opt,
+                                       c10::optional<at::Tensor> seqlens_k_opt) {
   at::Tensor seqlens = seqlens_opt.has_value() ? *seqlens_opt : at::Tensor();
+  at::Tensor seqlens_k = seqlens_k_opt.has_value() ? *seqlens_k_opt : at::Tensor();
   using namespace fa;
   check_fa_inputs(q, k, v);
   const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2), D = (int)q.size(3);
   const int Hkv = (int)k.size(2);
+  const int Sk = (int)k.size(1);  // kv may be a shorter ring piece
   const int* cu = nullptr;
+  const int* cuk = nullptr;
   if (seqlens.defined() && seqlens.numel() > 0) {
     TORCH_CHECK(seqlens.scalar_type() == at::kInt && seqlens.numel() == B && seqlens.is_cuda(),
                 "flash_attn: seqlens must be int32 [B] on device");
     cu = seqlens.data_ptr<int>();
+  }
+  if (seqlens_k.defined() && seqlens_k.numel() > 0) {
+    TORCH_CHECK(cu != nullptr, "flash_attn: seqlens_k requires seqlens");
+    TORCH_CHECK(seqlens_k.scalar_type() == at::kInt && seqlens_k.numel() == B && seqlens_k.is_cuda());
+    cuk = seqlens_k.data_ptr<int>();
   }
   auto o = at::empty({B, S, Hq, D}, q.options());
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
@@ -1067,8 +1092,8 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
     hipLaunchKernelGGL((fa_fwd_kernel<DD, false>), grid, dim3(NT), lds, stream.stream(),          \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
                        (const unsigned short*)v.data_ptr(), (unsigned short*)o.data_ptr(),        \
-                       lse.data_ptr<float>(), cu, qbs, qts, kbs, kts,                             \
-                       B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                              \
+                       lse.data_ptr<float>(), cu, cuk, qbs, qts, kbs, kts,                        \
+                       B, S, Sk, Hq, Hkv, (float)scale, causal ? 1 : 0);                              \
   } while (0)
 
   if (D == 128) LAUNCH_FWD(128);
@@ -1106,8 +1131,9 @@ std::vector<at::Tensor> flash_attn_varlen_fwd(at::Tensor q, at::Tensor k, at::Te
     hipLaunchKernelGGL((fa_fwd_kernel<DD, true>), grid, dim3(NT), lds, stream.stream(),           \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
                        (const unsigned short*)v.data_ptr(), (unsigned short*)o.data_ptr(),        \
-                       lse.data_ptr<float>(), cu_seqlens.data_ptr<int>(), 0L, qts, 0L, kts,       \
-                       n_seq, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                          \
+                       lse.data_ptr<float>(), cu_seqlens.data_ptr<int>(), nullptr,                \
+                       0L, qts, 0L, kts,                                                          \
+                       n_seq, T, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                          \
   } while (0)
 
   if (D == 128) LAUNCH_VFWD(128);
@@ -1149,8 +1175,10 @@ int dkdv_variant() {
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
                                        at::Tensor out, at::Tensor lse, bool causal, double scale,
                                        at::Tensor dq, at::Tensor dk, at::Tensor dv,
-                                       c10::optional<at::Tensor> seqlens_opt) {
+                                       c10::optional<at::Tensor> seqlens_opt,
+                                       c10::optional<at::Tensor> seqlens_k_opt) {
   at::Tensor seqlens = seqlens_opt.has_value() ? *seqlens_opt : at::Tensor();
+  at::Tensor seqlens_k = seqlens_k_opt.has_value() ? *seqlens_k_opt : at::Tensor();
   using namespace fa;
   check_fa_inputs(q, k, v);
   TORCH_CHECK(dout.is_contiguous() && out.is_contiguous() && lse.is_contiguous(),
@@ -1158,14 +1186,21 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
   const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2), D = (int)q.size(3);
   const int Hkv = (int)k.size(2);
   const int* cu = nullptr;
+  const int* cuk = nullptr;
   if (seqlens.defined() && seqlens.numel() > 0) {
     TORCH_CHECK(seqlens.scalar_type() == at::kInt && seqlens.numel() == B && seqlens.is_cuda(),
                 "flash_attn: seqlens must be int32 [B] on device");
     cu = seqlens.data_ptr<int>();
   }
+  if (seqlens_k.defined() && seqlens_k.numel() > 0) {
+    TORCH_CHECK(cu != nullptr, "flash_attn: seqlens_k requires seqlens");
+    TORCH_CHECK(seqlens_k.scalar_type() == at::kInt && seqlens_k.numel() == B && seqlens_k.is_cuda());
+    cuk = seqlens_k.data_ptr<int>();
+  }
+  const int Sk = (int)k.size(1);  // kv may be a shorter ring piece
   if (dq.numel() == 0) dq = at::empty({B, S, Hq, D}, q.options());
-  if (dk.numel() == 0) dk = at::empty({B, S, Hkv, D}, k.options());
-  if (dv.numel() == 0) dv = at::empty({B, S, Hkv, D}, v.options());
+  if (dk.numel() == 0) dk = at::empty({B, Sk, Hkv, D}, k.options());
+  if (dv.numel() == 0) dv = at::empty({B, Sk, Hkv, D}, v.options());
   check_fa_view(dq, "dq");
   check_fa_view(dk, "dk");
   TORCH_CHECK(dk.strides() == dv.strides(), "flash_attn_bwd: dk/dv strides must match");
@@ -1181,7 +1216,7 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
     HIP_CHECK_LAST();
   }
 
-  const dim3 grid_kv((S + NWB * QW - 1) / (NWB * QW), B * Hkv);
+  const dim3 grid_kv((Sk + NWB * QW - 1) / (NWB * QW), B * Hkv);
   const dim3 grid_q((S + NWB * QW - 1) / (NWB * QW), B * Hq);
   const long qbs = q.stride(0), qts = q.stride(1), kbs = k.stride(0), kts = k.stride(1);
   const long dqbs = dq.stride(0), dqts = dq.stride(1), dkbs = dk.stride(0), dkts = dk.stride(1);
@@ -1213,8 +1248,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
                        (unsigned short*)dk.data_ptr(), (unsigned short*)dv.data_ptr(),              \
-                       cu, qbs, qts, kbs, kts, dkbs, dkts,                                          \
-                       B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
+                       cu, cuk, qbs, qts, kbs, kts, dkbs, dkts,                                    \
+                       B, S, Sk, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
     HIP_CHECK_LAST();                                                                               \
     const size_t lds_q = 2 * sub_bytes<KVB, DD>() + NWB * sub_bytes<32, 32>()                       \
                          + 2 * NWB * QW * sizeof(float);                                            \
@@ -1224,8 +1259,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
                        (unsigned short*)dq.data_ptr(),                                              \
-                       cu, qbs, qts, kbs, kts, dqbs, dqts,                                          \
-                       B, S, Hq, Hkv, (float)scale, causal ? 1 : 0);                                \
+                       cu, cuk, qbs, qts, kbs, kts, dqbs, dqts,                                    \
+                       B, S, Sk, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
     HIP_CHECK_LAST();                                                                               \
   } while (0)
 
@@ -1284,8 +1319,8 @@ std::vector<at::Tensor> flash_attn_varlen_bwd(at::Tensor dout, at::Tensor q, at:
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
                        (unsigned short*)dk.data_ptr(), (unsigned short*)dv.data_ptr(),              \
-                       cu, 0L, qts, 0L, kts, 0L, dkts,                                              \
-                       n_seq, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
+                       cu, nullptr, 0L, qts, 0L, kts, 0L, dkts,                                              \
+                       n_seq, T, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
     HIP_CHECK_LAST();                                                                               \
     const size_t lds_q = 2 * sub_bytes<KVB, DD>() + NWB * sub_bytes<32, 32>()                       \
                          + 2 * NWB * QW * sizeof(float);                                            \
@@ -1295,8 +1330,8 @@ std::vector<at::Tensor> flash_attn_varlen_bwd(at::Tensor dout, at::Tensor q, at:
                        (const unsigned short*)v.data_ptr(), (const unsigned short*)dout.data_ptr(), \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),                              \
                        (unsigned short*)dq.data_ptr(),                                              \
-                       cu, 0L, qts, 0L, kts, 0L, dqts,                                              \
-                       n_seq, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
+                       cu, nullptr, 0L, qts, 0L, kts, 0L, dqts,                                              \
+                       n_seq, T, T, Hq, Hkv, (float)scale, causal ? 1 : 0);                            \
     HIP_CHECK_LAST();                                                                               \
   } while (0)
 

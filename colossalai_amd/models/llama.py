@@ -166,9 +166,9 @@ class LlamaAttention(nn.Module):
                 "qk-norm / partial-rotary variants are not wired into the SP "
                 "attention branches yet — run these models without SP"
             )
-            assert seqlens is None, (
-                "padded attention_mask under sequence parallelism is not wired "
-                "yet — pack the batch (varlen) or drop the mask"
+            assert seqlens is None or (sp_mode == "ring_attn" and getattr(self, "sp_zigzag", False)), (
+                "padded attention_mask under SP is wired only for zigzag "
+                "ring_attn — pack the batch (varlen) or drop the mask"
             )
         if sp_mode == "ring_attn":
             # context parallelism: Q stays, K/V blocks travel the xGMI ring
@@ -193,7 +193,8 @@ class LlamaAttention(nn.Module):
             else:
                 positions = (torch.arange(S, device=hidden.device) + rank * S).repeat(B).int()
             q, k = apply_rope(q, k, rope_table, positions)
-            attn = ring_flash_attention(q, k, v, sp_group, causal=True, scale=self.scale, zigzag=zigzag)
+            attn = ring_flash_attention(q, k, v, sp_group, causal=True, scale=self.scale,
+                                        zigzag=zigzag, seqlens=seqlens)
             return self.o_proj(attn.reshape(B, S, -1))
         if sp_mode == "all_to_all":
             # Ulysses: scatter heads / gather sequence around attention

@@ -25,11 +25,14 @@ __all__ = ["flash_attention", "flash_attention_varlen", "fused_rope_attention", 
 def attention_ref(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, causal: bool = True, scale: Optional[float] = None,
     upcast: bool = True, bias: Optional[torch.Tensor] = None, seqlens: Optional[torch.Tensor] = None,
+    seqlens_k: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """fp32 reference attention on [B,S,H,D] bshd tensors (GQA-aware).
 
     ``seqlens`` [B] masks right-padded rows/cols: pad queries produce ZERO
     output rows (matching the HIP kernel, which writes O=0 / LSE=-inf there).
+    ``seqlens_k`` optionally gives the kv side its OWN valid counts (ring
+    pieces where q and kv cover different parts of each sequence).
     """
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
@@ -50,7 +53,9 @@ def attention_ref(
         mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device=q.device), diagonal=1)
         scores = scores.masked_fill(mask, float("-inf"))
     if seqlens is not None:
-        kv_valid = torch.arange(S, device=q.device).view(1, 1, 1, S) < seqlens.view(B, 1, 1, 1)
+        sk = seqlens_k if seqlens_k is not None else seqlens
+        Sk = k.shape[1]
+        kv_valid = torch.arange(Sk, device=q.device).view(1, 1, 1, Sk) < sk.view(B, 1, 1, 1)
         scores = scores.masked_fill(~kv_valid, float("-inf"))
     p = torch.softmax(scores.float(), dim=-1).to(dt)
     if seqlens is not None:
@@ -77,14 +82,16 @@ def seqlens_from_attention_mask(attention_mask: torch.Tensor) -> torch.Tensor:
 
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale, seqlens):
+    def forward(ctx, q, k, v, causal, scale, seqlens, seqlens_k=None):
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
         ctx.causal, ctx.scale = causal, scale
         if use_hip(q, k, v):
-            sl = seqlens if seqlens is not None else torch.empty(0, dtype=torch.int32, device=q.device)
-            out, lse = kernels().flash_attn_fwd(q, k, v, causal, scale, sl)
-            ctx.save_for_backward(q, k, v, out, lse, sl)
+            empty = torch.empty(0, dtype=torch.int32, device=q.device)
+            sl = seqlens if seqlens is not None else empty
+            slk = seqlens_k if seqlens_k is not None else empty
+            out, lse = kernels().flash_attn_fwd(q, k, v, causal, scale, sl, slk)
+            ctx.save_for_backward(q, k, v, out, lse, sl, slk)
             ctx.hip = True
             return out
         ctx.hip = False
@@ -92,30 +99,32 @@ class _FlashAttention(torch.autograd.Function):
             qd = q.detach().requires_grad_(True)
             kd = k.detach().requires_grad_(True)
             vd = v.detach().requires_grad_(True)
-            out = attention_ref(qd, kd, vd, causal, scale, upcast=False, seqlens=seqlens)
+            out = attention_ref(qd, kd, vd, causal, scale, upcast=False, seqlens=seqlens,
+                                seqlens_k=seqlens_k)
         ctx.ref = (qd, kd, vd, out)
         return out.detach()
 
     @staticmethod
     def backward(ctx, dout):
         if ctx.hip:
-            q, k, v, out, lse, sl = ctx.saved_tensors
+            q, k, v, out, lse, sl, slk = ctx.saved_tensors
             dq, dk, dv = kernels().flash_attn_bwd(
                 dout.contiguous(), q, k, v, out, lse, ctx.causal, ctx.scale,
                 torch.empty(0, dtype=q.dtype, device=q.device),
                 torch.empty(0, dtype=q.dtype, device=q.device),
                 torch.empty(0, dtype=q.dtype, device=q.device),
-                sl,
+                sl, slk,
             )
-            return dq, dk, dv, None, None, None
+            return dq, dk, dv, None, None, None, None
         qd, kd, vd, out = ctx.ref
         torch.autograd.backward(out, dout)
-        return qd.grad, kd.grad, vd.grad, None, None, None
+        return qd.grad, kd.grad, vd.grad, None, None, None, None
 
 
 def flash_attention(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, causal: bool = True, scale: Optional[float] = None,
     attention_mask: Optional[torch.Tensor] = None, seqlens: Optional[torch.Tensor] = None,
+    seqlens_k: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """q [B,S,Hq,D], k/v [B,S,Hkv,D] bf16 (bshd) -> out [B,S,Hq,D].
 
@@ -126,7 +135,7 @@ def flash_attention(
     """
     if seqlens is None and attention_mask is not None:
         seqlens = seqlens_from_attention_mask(attention_mask).to(q.device)
-    return _FlashAttention.apply(q, k, v, causal, scale, seqlens)
+    return _FlashAttention.apply(q, k, v, causal, scale, seqlens, seqlens_k)
 
 
 class _FlashAttentionVarlen(torch.autograd.Function):

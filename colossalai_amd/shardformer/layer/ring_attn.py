@@ -96,37 +96,51 @@ class RingComm:
         return recvs
 
 
-def _block_fwd(q, k, v, causal, scale):
-    """-> (out [B,S,H,D], lse [B,H,S] fp32)."""
+def _block_fwd(q, k, v, causal, scale, sq=None, sk=None):
+    """-> (out [B,S,H,D], lse [B,H,S] fp32). ``sq``/``sk`` int32 [B]: valid
+    (right-padded) row counts of the q / kv slices (padded ring pieces)."""
     if q.is_cuda and has_kernels():
         from ...ops import kernels
 
-        return kernels().flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale)
+        e = torch.empty(0, dtype=torch.int32, device=q.device)
+        return kernels().flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale,
+                                        sq if sq is not None else e, sk if sk is not None else e)
     # fp32 reference with lse
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
+    Sk = k.shape[1]
     rep = Hq // Hkv
     qt = q.float().permute(0, 2, 1, 3)
     kt = k.float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
     vt = v.float().permute(0, 2, 1, 3).repeat_interleave(rep, 1)
     scores = qt @ kt.transpose(-1, -2) * scale
     if causal:
-        mask = torch.triu(torch.ones(S, k.shape[1], dtype=torch.bool, device=q.device), 1)
+        mask = torch.triu(torch.ones(S, Sk, dtype=torch.bool, device=q.device), 1)
         scores = scores.masked_fill(mask, float("-inf"))
+    if sq is not None:
+        kv_ok = torch.arange(Sk, device=q.device).view(1, 1, 1, Sk) < (sk if sk is not None else sq).view(B, 1, 1, 1)
+        scores = scores.masked_fill(~kv_ok, float("-inf"))
     lse = torch.logsumexp(scores, dim=-1)  # [B,H,S]
     out = torch.exp(scores - lse.unsqueeze(-1)) @ vt
+    out = torch.nan_to_num(out)
+    if sq is not None:
+        q_ok = torch.arange(S, device=q.device).view(1, 1, S, 1) < sq.view(B, 1, 1, 1)
+        out = out * q_ok
+        lse = torch.where(q_ok.view(B, 1, S), lse, torch.full_like(lse, float("-inf")))
     return out.permute(0, 2, 1, 3).to(q.dtype), lse
 
 
-def _block_bwd(dout, q, k, v, out, lse, causal, scale):
+def _block_bwd(dout, q, k, v, out, lse, causal, scale, sq=None, sk=None):
     """Per-block FA2 backward with the GLOBAL out/lse -> (dq, dk, dv)."""
     if q.is_cuda and has_kernels():
         from ...ops import kernels
 
         e = torch.empty(0, dtype=q.dtype, device=q.device)
+        ei = torch.empty(0, dtype=torch.int32, device=q.device)
         return kernels().flash_attn_bwd(dout.contiguous(), q.contiguous(), k.contiguous(), v.contiguous(),
                                         out.contiguous(), lse.contiguous(), causal, scale,
-                                        e.clone(), e.clone(), e.clone())
+                                        e.clone(), e.clone(), e.clone(),
+                                        sq if sq is not None else ei, sk if sk is not None else ei)
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
     rep = Hq // Hkv
@@ -139,7 +153,17 @@ def _block_bwd(dout, q, k, v, out, lse, causal, scale):
     if causal:
         mask = torch.triu(torch.ones(S, k.shape[1], dtype=torch.bool, device=q.device), 1)
         scores = scores.masked_fill(mask, float("-inf"))
+    if sq is not None:
+        Sk = k.shape[1]
+        skv = sk if sk is not None else sq
+        kv_ok = torch.arange(Sk, device=q.device).view(1, 1, 1, Sk) < skv.view(B, 1, 1, 1)
+        scores = scores.masked_fill(~kv_ok, float("-inf"))
     p = torch.exp(scores - lse.unsqueeze(-1))
+    if sq is not None:
+        # pad q rows: scores finite but lse may be -inf -> inf/nan; hard-zero
+        q_ok = torch.arange(S, device=q.device).view(1, 1, S, 1) < sq.view(B, 1, 1, 1)
+        p = p.masked_fill(~q_ok.expand_as(p), 0.0)
+    p = torch.nan_to_num(p)
     delta = (dot * ot).sum(-1, keepdim=True)
     dv = p.transpose(-1, -2) @ dot
     dp = dot @ vt.transpose(-1, -2)
@@ -154,20 +178,36 @@ def _block_bwd(dout, q, k, v, out, lse, causal, scale):
 
 
 def _merge(out_a, lse_a, out_b, lse_b):
-    """Merge two partial attention results (LSE rescale)."""
+    """Merge two partial attention results (LSE rescale). Rows masked on
+    BOTH sides (-inf lse, padded ring pieces) stay zero."""
     lse = torch.logaddexp(lse_a, lse_b)  # [B,H,S]
     wa = torch.exp(lse_a - lse).permute(0, 2, 1).unsqueeze(-1)  # [B,S,H,1]
     wb = torch.exp(lse_b - lse).permute(0, 2, 1).unsqueeze(-1)
-    return (out_a.float() * wa + out_b.float() * wb).to(out_a.dtype), lse
+    merged = torch.nan_to_num(out_a.float() * wa + out_b.float() * wb)
+    return merged.to(out_a.dtype), lse
+
+
+def _pc(seqlens, P, idx):
+    """valid rows of zigzag piece ``idx`` (absolute start idx*P) per batch."""
+    return (seqlens - idx * P).clamp(0, P).to(torch.int32)
 
 
 class _RingFlashAttention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, group, causal, scale, zigzag):
+    def forward(ctx, q, k, v, group, causal, scale, zigzag, seqlens):
         comm = RingComm(group)
         rank, world = comm.rank, comm.world
         B, S, Hq, D = q.shape
         S2 = S // 2
+        if seqlens is not None:
+            assert zigzag and causal, "padded ring attention requires zigzag causal"
+            seqlens = seqlens.to(q.device)
+            # zigzag validity is prefix-shaped per CHUNK (monotone padding):
+            # rank r holds pieces (r, 2*world-1-r) of size S2 each
+            sq_self = _pc(seqlens, S2, rank) + _pc(seqlens, S2, 2 * world - 1 - rank)
+            sq_tail = _pc(seqlens, S2, 2 * world - 1 - rank)
+        else:
+            sq_self = sq_tail = None
         # -inf lse + zero out make partial-row merges uniform
         out = torch.zeros(B, S, Hq, D, dtype=q.dtype, device=q.device)
         lse = torch.full((B, Hq, S), float("-inf"), dtype=torch.float32, device=q.device)
@@ -186,32 +226,45 @@ class _RingFlashAttention(torch.autograd.Function):
                     o_blk, l_blk = _block_fwd(q, cur_k, cur_v, src == rank, scale)
                     out, lse = _merge(out, lse, o_blk, l_blk.float())
             else:
+                sk_src = None if seqlens is None else \
+                    _pc(seqlens, S2, src) + _pc(seqlens, S2, 2 * world - 1 - src)
                 if src == rank:  # diagonal: plain causal over the concat shard
-                    o_blk, l_blk = _block_fwd(q, cur_k, cur_v, True, scale)
+                    o_blk, l_blk = _block_fwd(q, cur_k, cur_v, True, scale, sq_self, sq_self)
                     out, lse = _merge(out, lse, o_blk, l_blk.float())
                 elif src < rank:  # earlier src: only its first chunk is visible
-                    o_blk, l_blk = _block_fwd(q, cur_k[:, :S2], cur_v[:, :S2], False, scale)
+                    sk1 = None if seqlens is None else _pc(seqlens, S2, src)
+                    o_blk, l_blk = _block_fwd(q, cur_k[:, :S2], cur_v[:, :S2], False, scale,
+                                              sq_self, sk1)
                     out, lse = _merge(out, lse, o_blk, l_blk.float())
                 else:  # later src: visible only to the shard's second half of Q
-                    o_blk, l_blk = _block_fwd(q[:, S2:].contiguous(), cur_k, cur_v, False, scale)
+                    o_blk, l_blk = _block_fwd(q[:, S2:].contiguous(), cur_k, cur_v, False, scale,
+                                              sq_tail, sk_src)
                     o_new, l_new = _merge(out[:, S2:], lse[:, :, S2:], o_blk, l_blk.float())
                     out = torch.cat([out[:, :S2], o_new], dim=1)
                     lse = torch.cat([lse[:, :, :S2], l_new], dim=2)
             if step + 1 < world:
                 comm.wait(works)
                 cur_k, cur_v = nxt
-        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.save_for_backward(q, k, v, out, lse,
+                              seqlens if seqlens is not None else torch.empty(0, dtype=torch.int32))
         ctx.group, ctx.causal, ctx.scale, ctx.zigzag = group, causal, scale, zigzag
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, out, lse = ctx.saved_tensors
+        q, k, v, out, lse, seqlens = ctx.saved_tensors
         group, causal, scale = ctx.group, ctx.causal, ctx.scale
         zigzag = ctx.zigzag
         comm = RingComm(group)
         rank, world = comm.rank, comm.world
         S2 = q.shape[1] // 2
+        if seqlens.numel():
+            seqlens = seqlens.to(q.device)
+            sq_self = _pc(seqlens, S2, rank) + _pc(seqlens, S2, 2 * world - 1 - rank)
+            sq_tail = _pc(seqlens, S2, 2 * world - 1 - rank)
+        else:
+            seqlens = None
+            sq_self = sq_tail = None
         dout = dout.contiguous()
         dq = torch.zeros_like(q, dtype=torch.float32)
         cur_k, cur_v = k, v
@@ -232,21 +285,27 @@ class _RingFlashAttention(torch.autograd.Function):
                 cur_dk += dk_b.float()
                 cur_dv += dv_b.float()
             elif zigzag:
+                sk_src = None if seqlens is None else \
+                    _pc(seqlens, S2, src) + _pc(seqlens, S2, 2 * world - 1 - src)
                 if src == rank:
-                    dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k, cur_v, out, lse, True, scale)
+                    dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k, cur_v, out, lse, True, scale,
+                                                  sq_self, sq_self)
                     dq += dq_b.float()
                     cur_dk += dk_b.float()
                     cur_dv += dv_b.float()
                 elif src < rank:
+                    sk1 = None if seqlens is None else _pc(seqlens, S2, src)
                     dq_b, dk_b, dv_b = _block_bwd(dout, q, cur_k[:, :S2].contiguous(),
-                                                  cur_v[:, :S2].contiguous(), out, lse, False, scale)
+                                                  cur_v[:, :S2].contiguous(), out, lse, False, scale,
+                                                  sq_self, sk1)
                     dq += dq_b.float()
                     cur_dk[:, :S2] += dk_b.float()
                     cur_dv[:, :S2] += dv_b.float()
                 else:
                     dq_b, dk_b, dv_b = _block_bwd(
                         dout[:, S2:].contiguous(), q[:, S2:].contiguous(), cur_k, cur_v,
-                        out[:, S2:].contiguous(), lse[:, :, S2:].contiguous(), False, scale)
+                        out[:, S2:].contiguous(), lse[:, :, S2:].contiguous(), False, scale,
+                        sq_tail, sk_src)
                     dq[:, S2:] += dq_b.float()
                     cur_dk += dk_b.float()
                     cur_dv += dv_b.float()
@@ -259,16 +318,20 @@ class _RingFlashAttention(torch.autograd.Function):
                 dkv = comm.send_recv([cur_dk.to(k.dtype), cur_dv.to(v.dtype)])
                 cur_dk, cur_dv = dkv[0].float(), dkv[1].float()
         # one full cycle: cur_dk/cur_dv now hold this rank's own grads
-        return dq.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype), None, None, None, None
+        return dq.to(q.dtype), cur_dk.to(k.dtype), cur_dv.to(v.dtype), None, None, None, None, None
 
 
 def ring_flash_attention(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, group, causal: bool = True,
     scale: Optional[float] = None, zigzag: bool = False,
+    seqlens: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """q/k/v [B, S/sp, H, D] sequence shards -> local out shard.
     ``zigzag=True`` expects shards laid out by ``zigzag_split`` and
-    balances causal work exactly across the ring."""
+    balances causal work exactly across the ring. ``seqlens`` int32 [B]
+    gives GLOBAL right-padded valid lengths (ragged batches under context
+    parallelism — the reference's prepare_varlen_batch role); requires
+    zigzag causal."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    return _RingFlashAttention.apply(q, k, v, group, causal, scale, zigzag)
+    return _RingFlashAttention.apply(q, k, v, group, causal, scale, zigzag, seqlens)

@@ -467,6 +467,39 @@ def test_multi_tensor_sgd():
         torch.testing.assert_close(p.cpu(), pr.cpu(), rtol=1e-5, atol=1e-6)
 
 
+def test_flash_attn_asymmetric_kv():
+    """Ring-piece shapes: kv shorter than q, separate q/k valid counts
+    (seqlens_k) — fwd + bwd vs the masked fp32 reference."""
+    from colossalai_amd.ops.attention import attention_ref, flash_attention
+
+    torch.manual_seed(17)
+    B, S, Sk, Hq, Hkv, D = 2, 256, 128, 4, 2, 128
+    q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Sk, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Sk, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    sq = torch.tensor([256, 180], dtype=torch.int32, device="cuda")
+    sk = torch.tensor([128, 55], dtype=torch.int32, device="cuda")
+
+    for (qq, kk, vv) in [(q, k, v)]:
+        qd = qq.clone().requires_grad_(True)
+        kd = kk.clone().requires_grad_(True)
+        vd = vv.clone().requires_grad_(True)
+        out = flash_attention(qd, kd, vd, causal=False, seqlens=sq, seqlens_k=sk)
+        qr = qq.float().clone().requires_grad_(True)
+        kr = kk.float().clone().requires_grad_(True)
+        vr = vv.float().clone().requires_grad_(True)
+        ref = attention_ref(qr, kr, vr, causal=False, seqlens=sq, seqlens_k=sk)
+        _bf16_close(out, ref, rtol=3e-2, atol=3e-2, frac=1e-3)
+        dout = torch.randn_like(out)
+        for b in range(B):
+            dout[b, int(sq[b]):] = 0
+        out.backward(dout)
+        ref.backward(dout.float())
+        _bf16_close(qd.grad, qr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+        _bf16_close(kd.grad, kr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+        _bf16_close(vd.grad, vr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+
+
 def test_kv_cache_append():
     torch.manual_seed(16)
     B, Hkv, D, rows = 5, 4, 128, 64
