@@ -234,12 +234,11 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   // async-split staging (T14): tile 0 staged synchronously, tile t+1's global
   // loads issue before tile t's compute and land in LDS after the barrier.
   RegStage<D, KVB, NT> kstage, vstage;
-  if (kv_tiles > 0) {
-    kstage.load(k_base, kts, 0, SeffK);
-    vstage.load(v_base, kts, 0, SeffK);
-    kstage.store_subtiled(Klds);
-    vstage.store_subtiled(Vlds);
-  }
+  // SeffK == 0 blocks skip the tile loop; the clamped loads stay in-bounds
+  kstage.load(k_base, kts, 0, SeffK);
+  vstage.load(v_base, kts, 0, SeffK);
+  kstage.store_subtiled(Klds);
+  vstage.store_subtiled(Vlds);
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
@@ -884,12 +883,11 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
   const int kv_tiles = causal ? min(kv_tiles_all, (q0 + NWB * QW + KVB - 1) / KVB) : kv_tiles_all;
 
   RegStage<D, KVB, NTB> kstage, vstage;
-  if (kv_tiles > 0) {
-    kstage.load(k_base, kts, 0, SeffK);
-    vstage.load(v_base, kts, 0, SeffK);
-    kstage.store_subtiled(Klds);
-    vstage.store_subtiled(Vlds);
-  }
+  // SeffK == 0 blocks skip the tile loop; the clamped loads stay in-bounds
+  kstage.load(k_base, kts, 0, SeffK);
+  vstage.load(v_base, kts, 0, SeffK);
+  kstage.store_subtiled(Klds);
+  vstage.store_subtiled(Vlds);
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
