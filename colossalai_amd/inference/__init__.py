@@ -1,9 +1,10 @@
 from .config import GenerationConfig, InferenceConfig
+from .diffusion import DiffusionEngine, PatchParallelDiT, ddim_sample
 from .engine import LLMEngine
 from .kv_cache import KVCacheManager
 from .paged_engine import ContinuousBatchEngine
 from .request_manager import Request, RequestManager, RequestStatus
 from .spec_decode import BatchedSpeculativeEngine, SpeculativeEngine
 
-__all__ = ["InferenceConfig", "GenerationConfig", "LLMEngine", "ContinuousBatchEngine",
+__all__ = ["InferenceConfig", "GenerationConfig", "LLMEngine", "ContinuousBatchEngine", "DiffusionEngine", "PatchParallelDiT", "ddim_sample",
            "KVCacheManager", "SpeculativeEngine", "BatchedSpeculativeEngine", "RequestManager", "Request", "RequestStatus"]

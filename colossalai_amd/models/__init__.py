@@ -1,4 +1,5 @@
 from .sam import SAM_CONFIGS, SamConfig, SamModel
+from .dit import DIT_CONFIGS, DiT, DiTConfig
 from .deepseek_v3 import DEEPSEEK_V3_CONFIGS, DeepseekV3Config, DeepseekV3ForCausalLM
 from .chatglm2 import CHATGLM_CONFIGS, ChatGLMConfig, ChatGLMForConditionalGeneration
 from .blip2 import Blip2Config, Blip2ForConditionalGeneration
