@@ -126,8 +126,13 @@ class DeepseekV3Attention(nn.Module):
         kv = kv.view(B, S, H, self.nope_dim + self.v_dim)
         k_nope, v = kv.split([self.nope_dim, self.v_dim], dim=-1)
 
-        q_pe, k_pe = apply_rope(q_pe.contiguous(), k_pe.view(B, S, 1, self.rope_dim).contiguous(),
-                                self._table(hidden.device))
+        k_pe = k_pe.view(B, S, 1, self.rope_dim)
+        if self.rope_dim % 16 == 0:
+            q_pe, k_pe = apply_rope(q_pe.contiguous(), k_pe.contiguous(), self._table(hidden.device))
+        else:  # tiny configs: the HIP rope kernel needs rope_dim % 16 == 0
+            from ..ops.rope import apply_rope_ref
+
+            q_pe, k_pe = apply_rope_ref(q_pe, k_pe, self._table(hidden.device), None, S)
         q = torch.cat([q_nope, q_pe], dim=-1)
         k = torch.cat([k_nope, k_pe.expand(B, S, H, self.rope_dim)], dim=-1)
         out = attention_ref(q, k, v, causal=True, scale=self.scale, upcast=False)

@@ -133,7 +133,8 @@ class DiT(nn.Module):
                 m.weight.data.normal_(0.0, self.config.initializer_range)
 
     def cond(self, t, labels, B, device):
-        c = self.t_mlp(timestep_embedding(t, self.config.hidden_size))
+        dtype = self.label_embed.weight.dtype
+        c = self.t_mlp(timestep_embedding(t, self.config.hidden_size).to(dtype))
         if labels is None:
             labels = torch.full((B,), self.config.num_classes, dtype=torch.long, device=device)
         return c + self.label_embed(labels)

@@ -217,8 +217,9 @@ class SamPromptEncoder(nn.Module):
         self.no_mask_embed = nn.Embedding(1, d)
 
     def _pe(self, coords):  # coords in [0,1], [..., 2]
-        proj = (2 * coords - 1) @ self.pe_gaussian * 2 * math.pi
-        return torch.cat([proj.sin(), proj.cos()], dim=-1)
+        g = self.pe_gaussian.float()
+        proj = (2 * coords.float() - 1) @ g * 2 * math.pi
+        return torch.cat([proj.sin(), proj.cos()], dim=-1).to(self.pe_gaussian.dtype)
 
     def forward(self, points: Optional[torch.Tensor], labels: Optional[torch.Tensor], B: int,
                 boxes: Optional[torch.Tensor] = None):
