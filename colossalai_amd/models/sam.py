@@ -238,8 +238,8 @@ class SamPromptEncoder(nn.Module):
                 [self.point_embeds[2].weight[0], self.point_embeds[3].weight[0]])
             parts.append(corners.view(boxes.shape[0], -1, self.embed_dim))
         if not parts:
-            dev = self.no_mask_embed.weight.device
-            return torch.zeros(B, 0, self.embed_dim, device=dev)
+            w = self.no_mask_embed.weight
+            return torch.zeros(B, 0, self.embed_dim, device=w.device, dtype=w.dtype)
         return torch.cat(parts, dim=1)
 
     def dense_pe(self, h, w, device):

@@ -27,7 +27,8 @@ class _RMSNorm(torch.autograd.Function):
     def forward(ctx, x, weight, eps):
         if use_hip(x, weight):
             # save the normalization input + inv_rms for backward
-            out, inv_rms = kernels().rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps, True)
+            x = x.contiguous()  # split/slice views (e.g. MLA latents) arrive here
+            out, inv_rms = kernels().rmsnorm_fwd(x, weight.contiguous(), eps, True)
             ctx.save_for_backward(x, weight, inv_rms)
             ctx.eps = eps
             return out
