@@ -94,3 +94,34 @@ def test_paged_engine_gpu_matches_contiguous_engine():
     out = ContinuousBatchEngine(model, icfg, block_size=16).generate(
         prompts, GenerationConfig(max_new_tokens=12))
     assert out == ref, f"paged {out} vs contiguous {ref}"
+
+
+def test_async_engine_concurrent():
+    """Concurrent awaits share the continuous batch; results match the
+    synchronous engine token-for-token."""
+    import asyncio
+
+    from colossalai_amd.inference import AsyncInferenceEngine, ContinuousBatchEngine, InferenceConfig
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=128)
+    model = LlamaForCausalLM(cfg).eval()
+    prompts = [[5, 17, 42], [99, 3, 4, 7], [1, 2]]
+
+    from colossalai_amd.inference import GenerationConfig
+
+    ref_engine = ContinuousBatchEngine(model, InferenceConfig(max_batch_size=4, max_input_len=16,
+                                                              max_output_len=8), block_size=4)
+    ref = ref_engine.generate(prompts, GenerationConfig(max_new_tokens=8))
+
+    engine = ContinuousBatchEngine(model, InferenceConfig(max_batch_size=4, max_input_len=16,
+                                                          max_output_len=8), block_size=4)
+    aeng = AsyncInferenceEngine(engine)
+
+    async def main():
+        return await asyncio.gather(*(aeng.submit(p, 8) for p in prompts))
+
+    outs = asyncio.run(main())
+    assert outs == ref
