@@ -374,3 +374,90 @@ class BatchedSpeculativeEngine:
     @property
     def acceptance_rate(self) -> float:
         return self.accepted / max(self.proposed, 1)
+
+
+class GlideCrossAttention(torch.nn.Module):
+    """GLIDE's glance: the drafter cross-attends to the TARGET model's KV
+    cache so a tiny draft stays anchored to the big model's context
+    (reference: colossalai/inference/modeling/models/glide_llama.py).
+    Output projection is zero-initialized — an untrained glance is a
+    no-op, training moves it off the identity."""
+
+    def __init__(self, d_draft: int, t_kv_heads: int, t_head_dim: int):
+        super().__init__()
+        import torch.nn as nn
+
+        self.t_kv_heads = t_kv_heads
+        self.t_head_dim = t_head_dim
+        inner = t_kv_heads * t_head_dim
+        self.q_proj = nn.Linear(d_draft, inner, bias=False)
+        self.o_proj = nn.Linear(inner, d_draft, bias=False)
+        nn.init.zeros_(self.o_proj.weight)
+        self.scale = t_head_dim ** -0.5
+
+    def forward(self, hidden, kcache, vcache, prefix_len: int):
+        """hidden [1, S, d]; kcache/vcache [1, max, Hkv, D] target cache.
+        The cache holds only verified prefix rows, so attending to ALL of
+        it is causally safe for every drafted position."""
+        if prefix_len == 0:
+            return hidden
+        B, S, _ = hidden.shape
+        H, D = self.t_kv_heads, self.t_head_dim
+        q = self.q_proj(hidden).view(B, S, H, D).permute(0, 2, 1, 3)
+        k = kcache[:, :prefix_len].permute(0, 2, 1, 3).to(q.dtype)
+        v = vcache[:, :prefix_len].permute(0, 2, 1, 3).to(q.dtype)
+        attn = torch.softmax((q @ k.transpose(-1, -2)) * self.scale, dim=-1) @ v
+        out = self.o_proj(attn.permute(0, 2, 1, 3).reshape(B, S, H * D))
+        return hidden + out
+
+
+class _GlideCachedModel(_CachedModel):
+    """Draft wrapper that glances at the target's last-layer KV before the
+    LM head."""
+
+    def __init__(self, model: LlamaForCausalLM, cross: GlideCrossAttention,
+                 target: _CachedModel, max_seq_len: int):
+        super().__init__(model, max_seq_len)
+        self.cross = cross.to(self.device, self.dtype)
+        self.target = target
+
+    @torch.inference_mode()
+    def forward_chunk(self, token_ids: List[int]) -> torch.Tensor:
+        from ..ops import fused_add_rms_norm, rms_norm
+
+        m = self.model.model
+        ids = torch.tensor([token_ids], device=self.device)
+        S = ids.shape[1]
+        positions = (torch.arange(S, device=self.device) + self.len).int()
+        table = m.rope_table(self.device)
+        residual = m.embed_tokens(ids)
+        hidden = rms_norm(residual, m.layers[0].input_layernorm_weight, m.eps)
+        n = len(m.layers)
+        for i, layer in enumerate(m.layers):
+            kc, vc = self.caches[i]
+            attn_out = _chunk_attend(layer.self_attn, hidden, table, kc, vc, self.len, positions)
+            hidden, residual = fused_add_rms_norm(attn_out, residual,
+                                                  layer.post_attention_layernorm_weight, m.eps)
+            mlp_out = layer.mlp(hidden)
+            next_w = m.layers[i + 1].input_layernorm_weight if i + 1 < n else m.norm_weight
+            hidden, residual = fused_add_rms_norm(mlp_out, residual, next_w, m.eps)
+        # GLIDE glance: target's LAST layer cache over the verified prefix
+        t_kc, t_vc = self.target.caches[-1]
+        hidden = self.cross(hidden, t_kc, t_vc, self.target.len)
+        self.len += S
+        return self.model.lm_head(hidden[0]).float()
+
+
+class GlideSpeculativeEngine(SpeculativeEngine):
+    """Speculative decoding with a GLIDE drafter: identical accept/verify
+    loop (still lossless under greedy), but the draft model reuses the
+    target's KV cache through a cross-attention glance."""
+
+    def __init__(self, target: LlamaForCausalLM, draft: LlamaForCausalLM,
+                 cross: Optional[GlideCrossAttention] = None,
+                 config: Optional[InferenceConfig] = None, gamma: int = 4):
+        super().__init__(target, draft, config, gamma)
+        if cross is None:
+            tcfg, dcfg = target.config, draft.config
+            cross = GlideCrossAttention(dcfg.hidden_size, tcfg.num_key_value_heads, tcfg.head_dim)
+        self.draft = _GlideCachedModel(draft, cross, self.target, self.config.max_seq_len)

@@ -138,3 +138,27 @@ def test_batched_spec_decode_gpu():
     assert eng.proposed > 0 and 0.0 <= eng.acceptance_rate <= 1.0
     # (self-draft acceptance is asserted on CPU fp32 only: a random-init
     # bf16 model's near-flat logits make cross-path argmax ties common)
+
+
+def test_glide_spec_decode_lossless():
+    """GLIDE drafter (cross-attention glance at the target KV): greedy
+    speculative output must equal plain greedy from the target, and the
+    glance must actually see the target cache."""
+    from colossalai_amd.inference import GlideSpeculativeEngine, InferenceConfig, GenerationConfig
+
+    torch.manual_seed(0)
+    tcfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                       num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=256)
+    dcfg = LlamaConfig(vocab_size=128, hidden_size=32, intermediate_size=64, num_hidden_layers=1,
+                       num_attention_heads=2, num_key_value_heads=2, max_position_embeddings=256)
+    target = LlamaForCausalLM(tcfg).eval()
+    draft = LlamaForCausalLM(dcfg).eval()
+    eng = GlideSpeculativeEngine(target, draft, config=InferenceConfig(max_input_len=64, max_output_len=64), gamma=3)
+    # make the glance non-trivial (zero-init o_proj would be a no-op)
+    torch.nn.init.normal_(eng.draft.cross.o_proj.weight, 0.0, 0.02)
+
+    prompt = [5, 17, 42, 7]
+    out = eng.generate(prompt, GenerationConfig(max_new_tokens=12))
+    ref = _oracle(target, prompt, 12)
+    assert out == ref, f"glide spec {out} vs greedy {ref}"
+    assert eng.proposed > 0
