@@ -198,9 +198,11 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   } else if (CU != nullptr) {
     Seff = CU[b];
   }
-  // kv extent: its own tensor length Sk (ring blocks where kv is a piece of
-  // the sequence), further clipped by per-batch valid counts
-  int SeffK = VARLEN ? Seff : (CUK != nullptr ? CUK[b] : min(Seff, Sk));
+  // kv extent: its own tensor length Sk (ring blocks where kv covers a
+  // DIFFERENT — possibly longer — span than q), clipped by per-batch valid
+  // counts: CUK when given, else CU only in the symmetric padded case
+  int SeffK = VARLEN ? Seff
+            : (CUK != nullptr ? CUK[b] : (CU != nullptr ? min(CU[b], Sk) : Sk));
 
   const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
   const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
@@ -531,7 +533,8 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
     Seff = CU[b];
   }
   // this kernel's OWN rows are kv (extent Sk); q streams over Seff
-  int SeffK = VARLEN ? Seff : (CUK != nullptr ? CUK[b] : min(Seff, Sk));
+  int SeffK = VARLEN ? Seff
+            : (CUK != nullptr ? CUK[b] : (CU != nullptr ? min(CU[b], Sk) : Sk));
 
   const unsigned short* k_base = VARLEN ? K + seq0 * kts + (long)hk * D : K + (long)b * kbs + (long)hk * D;
   const unsigned short* v_base = VARLEN ? V + seq0 * kts + (long)hk * D : V + (long)b * kbs + (long)hk * D;
@@ -837,7 +840,8 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
   } else if (CU != nullptr) {
     Seff = CU[b];
   }
-  const int SeffK = VARLEN ? Seff : (CUK != nullptr ? CUK[b] : min(Seff, Sk));
+  const int SeffK = VARLEN ? Seff
+            : (CUK != nullptr ? CUK[b] : (CU != nullptr ? min(CU[b], Sk) : Sk));
 
   const unsigned short* q_base = VARLEN ? Q + seq0 * qts + (long)h * D : Q + (long)b * qbs + (long)h * D;
   const unsigned short* do_base =

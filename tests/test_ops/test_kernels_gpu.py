@@ -500,6 +500,33 @@ def test_flash_attn_asymmetric_kv():
         _bf16_close(vd.grad, vr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
 
 
+def test_flash_attn_kv_longer_than_q():
+    """Ring "later src" blocks: kv tensor LONGER than q (dense, no seqlens)
+    must attend the FULL kv extent."""
+    from colossalai_amd.ops.attention import attention_ref, flash_attention
+
+    torch.manual_seed(18)
+    B, S, Sk, Hq, Hkv, D = 2, 128, 256, 4, 2, 128
+    q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Sk, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Sk, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    qd = q.clone().requires_grad_(True)
+    kd = k.clone().requires_grad_(True)
+    vd = v.clone().requires_grad_(True)
+    out = flash_attention(qd, kd, vd, causal=False)
+    qr = q.float().clone().requires_grad_(True)
+    kr = k.float().clone().requires_grad_(True)
+    vr = v.float().clone().requires_grad_(True)
+    ref = attention_ref(qr, kr, vr, causal=False)
+    _bf16_close(out, ref, rtol=3e-2, atol=3e-2, frac=1e-3)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    ref.backward(dout.float())
+    _bf16_close(qd.grad, qr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+    _bf16_close(kd.grad, kr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+    _bf16_close(vd.grad, vr.grad, rtol=5e-2, atol=5e-2, frac=2e-3)
+
+
 def test_kv_cache_append():
     torch.manual_seed(16)
     B, Hkv, D, rows = 5, 4, 128, 64
