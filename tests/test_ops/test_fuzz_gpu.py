@@ -106,3 +106,44 @@ def test_flash_attn_varlen_fuzz():
             _close(k.grad[sl], kr.grad[0], what + " dk")
             _close(v.grad[sl], vr.grad[0], what + " dv")
             lo += L
+
+
+def test_grouped_gemm_fuzz():
+    if not torch.cuda.is_available():
+        pytest.skip("GPU only")
+    import os
+
+    os.environ["CAI_MOE_GG"] = "1"
+    from colossalai_amd.ops.grouped_gemm import grouped_gemm
+
+    rng = random.Random(99)
+    for trial in range(8):
+        E = rng.choice([2, 4, 8])
+        K = rng.choice([64, 128, 256])
+        M = rng.choice([128, 256])
+        counts = [rng.choice([0, 1, 7, 33, 130]) for _ in range(E)]
+        offs = [0]
+        for c in counts:
+            offs.append(offs[-1] + c)
+        N = offs[-1]
+        if N == 0:
+            continue
+        torch.manual_seed(2000 + trial)
+        x = torch.randn(N, K, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+        w = (torch.randn(E, M, K, device="cuda", dtype=torch.bfloat16) * 0.05).requires_grad_(True)
+        y = grouped_gemm(x, w, offs)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        what = f"gg trial{trial} E{E} K{K} M{M} counts={counts}"
+        # fp32 loop reference
+        xr = x.detach().float().requires_grad_(True)
+        wr = w.detach().float().requires_grad_(True)
+        yr = torch.zeros(N, M, device="cuda")
+        for g in range(E):
+            lo, hi = offs[g], offs[g + 1]
+            if hi > lo:
+                yr[lo:hi] = xr[lo:hi] @ wr[g].t()
+        yr.backward(dy.float())
+        _close(y, yr, what)
+        _close(x.grad, xr.grad, what + " dx")
+        _close(w.grad, wr.grad, what + " dw")
