@@ -24,13 +24,16 @@ class LoraConfig:
 
 
 class LoraLinear(nn.Module):
-    def __init__(self, base: nn.Linear, r: int, alpha: int, dropout: float):
+    def __init__(self, base: nn.Module, r: int, alpha: int, dropout: float):
         super().__init__()
         self.base = base
         for p in self.base.parameters():
             p.requires_grad_(False)
-        dtype = base.weight.dtype
-        device = base.weight.device
+        # base may be a quantized linear (W8Linear / NF4Linear — QLoRA):
+        # those expose in/out_features but keep weights as int buffers
+        w = getattr(base, "weight", None)
+        dtype = w.dtype if isinstance(w, torch.Tensor) and w.is_floating_point() else torch.float32
+        device = next(iter(base.state_dict().values())).device
         self.lora_A = nn.Parameter(torch.zeros(r, base.in_features, dtype=dtype, device=device))
         self.lora_B = nn.Parameter(torch.zeros(base.out_features, r, dtype=dtype, device=device))
         nn.init.kaiming_uniform_(self.lora_A, a=math.sqrt(5))
@@ -44,7 +47,8 @@ class LoraLinear(nn.Module):
 
     @torch.no_grad()
     def merge(self) -> nn.Linear:
-        self.base.weight.data += (self.lora_B @ self.lora_A) * self.scaling
+        assert isinstance(self.base, nn.Linear), "merge() needs an unquantized base"
+        self.base.weight.data += ((self.lora_B @ self.lora_A) * self.scaling).to(self.base.weight.dtype)
         return self.base
 
 
@@ -53,10 +57,12 @@ def apply_lora(model: nn.Module, config: Optional[LoraConfig] = None) -> nn.Modu
     config = config or LoraConfig()
     for p in model.parameters():
         p.requires_grad_(False)
+    from ..quantization.weight_quant import NF4Linear, W8Linear
+
     pattern = re.compile("|".join(re.escape(t) for t in config.target_modules))
     for name, module in list(model.named_modules()):
         for child_name, child in list(module.named_children()):
-            if isinstance(child, nn.Linear) and pattern.search(child_name):
+            if isinstance(child, (nn.Linear, W8Linear, NF4Linear)) and pattern.search(child_name):
                 setattr(module, child_name, LoraLinear(child, config.r, config.lora_alpha, config.lora_dropout))
     return model
 

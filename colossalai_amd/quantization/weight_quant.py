@@ -32,6 +32,7 @@ class W8Linear(nn.Module):
 
     def __init__(self, weight: torch.Tensor, bias: Optional[torch.Tensor]):
         super().__init__()
+        weight = weight.detach()  # buffers must not carry the source graph
         scale = weight.abs().amax(dim=1, keepdim=True).clamp_min(1e-8) / 127.0
         q = torch.round(weight.float() / scale).clamp(-127, 127).to(torch.int8)
         self.register_buffer("qweight", q)

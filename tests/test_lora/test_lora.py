@@ -46,3 +46,33 @@ def test_booster_enable_lora():
     booster = Booster(plugin=TorchDDPPlugin())
     model = booster.enable_lora(model)
     assert any(p.requires_grad and "lora" in n for n, p in model.named_parameters())
+
+
+def test_qlora_int8_base():
+    """QLoRA flow (reference: bnb-quantized base + LoRA adapters): int8
+    frozen base, trainable adapters, loss decreases."""
+    from colossalai_amd.lora import LoraConfig, apply_lora
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+    from colossalai_amd.quantization import W8Linear, quantize_model
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    model = quantize_model(LlamaForCausalLM(cfg), bits=8)
+    model = apply_lora(model, LoraConfig(r=4, lora_alpha=8))
+    # adapters wrap the quantized projections
+    attn = model.model.layers[0].self_attn
+    assert isinstance(attn.qkv_proj.base, W8Linear)
+    trainable = [p for p in model.parameters() if p.requires_grad]
+    assert all(p.shape[0] == 4 or p.shape[1] == 4 for p in trainable)
+
+    opt = torch.optim.AdamW(trainable, lr=1e-2)
+    x = torch.randint(0, 128, (2, 16))
+    losses = []
+    for _ in range(6):
+        out = model(x, labels=x)
+        out["loss"].backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(out["loss"]))
+    assert losses[-1] < losses[0]
