@@ -697,13 +697,20 @@ __global__ __launch_bounds__(fa::NTB, VAR >= 1 ? 2 : 1) void fa_bwd_dkdv_kernel(
         for (int j = 0; j < 16; ++j) acc2[j] = 0.0f;
         bf16x8_t do_cur = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, half * 8));
         bf16x8_t do_nxt;
+        // VAR>=2 reloads V fragments from L1 — pipeline those too, the
+        // global-load latency was the serial chain here (PMC: both pipes idle)
+        bf16x8_t vf_cur = (VAR >= 2) ? ld_g16(vp) : vf[0];
+        bf16x8_t vf_nxt;
 #pragma unroll
         for (int sl = 0; sl < DSL; ++sl) {
-          if (sl + 1 < DSL)
+          if (sl + 1 < DSL) {
             do_nxt = ld_lds16(dOlds, sub_off<QB>(qb * 32 + ln, (sl + 1) * 16 + half * 8));
-          bf16x8_t vfr = (VAR >= 2) ? ld_g16(vp + sl * 16) : vf[VAR >= 2 ? 0 : sl];
+            if (VAR >= 2) vf_nxt = ld_g16(vp + (sl + 1) * 16);
+          }
+          bf16x8_t vfr = (VAR >= 2) ? vf_cur : vf[VAR >= 2 ? 0 : sl];
           acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, do_cur, acc2, 0, 0, 0);
           do_cur = do_nxt;
+          vf_cur = vf_nxt;
         }
 
         // ---- dS' = scale * P' ⊙ (dP' - delta[q]) → overwrite P tile
