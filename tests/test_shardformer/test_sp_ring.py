@@ -196,3 +196,61 @@ def run_ring_zigzag_padded_op(rank, world_size, port):
 @rerun_if_address_is_in_use()
 def test_sp_ring_zigzag_padded():
     spawn(run_ring_zigzag_padded_op, 4)
+
+
+def run_ring_zigzag_padded_model(rank, world_size, port):
+    """Full llama under zigzag ring SP with a right-padded attention_mask
+    vs the unsharded padded oracle (loss + grads)."""
+    from colossalai_amd.shardformer.layer.ring_attn import zigzag_split
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    ref = LlamaForCausalLM(cfg)
+    model = copy.deepcopy(ref)
+
+    plugin = HybridParallelPlugin(
+        tp_size=1, pp_size=1, sp_size=2, precision="fp32", zero_stage=0,
+        enable_sequence_parallelism=True, sequence_parallelism_mode="ring_attn", sp_zigzag=True,
+    )
+    booster = Booster(plugin=plugin)
+    optimizer = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    model_b, optimizer_b, *_ = booster.boost(model, optimizer)
+
+    torch.manual_seed(7)
+    S = 16
+    seqlens = [16, 9]
+    x = torch.randint(0, 128, (2, S))
+    mask = torch.zeros(2, S, dtype=torch.long)
+    labels = torch.full_like(x, -100)
+    for b, L in enumerate(seqlens):
+        mask[b, :L] = 1
+        labels[b, :L] = x[b, :L]
+
+    out = model_b(input_ids=x, attention_mask=mask, labels=labels)
+
+    logits_ref = ref(x, attention_mask=mask)["logits"]
+    shifted = torch.full_like(x, -100)
+    shifted[:, :-1] = labels[:, 1:]
+    local_losses = []
+    for r in range(world_size):
+        lg = zigzag_split(logits_ref, world_size, r)
+        lb = zigzag_split(shifted, world_size, r)
+        local_losses.append(F.cross_entropy(lg.reshape(-1, 128).float(), lb.reshape(-1), ignore_index=-100))
+    ref_loss = sum(local_losses) / world_size
+
+    assert_close_loose(out["loss"], local_losses[rank], rtol=1e-4, atol=1e-5)
+
+    booster.backward(out["loss"], optimizer_b)
+    ref_loss.backward()
+    assert_close_loose(model_b.module.model.layers[0].input_layernorm_weight.grad,
+                       ref.model.layers[0].input_layernorm_weight.grad, rtol=1e-3, atol=1e-5)
+    assert_close_loose(model_b.module.model.layers[1].self_attn.qkv_proj.weight.grad,
+                       ref.model.layers[1].self_attn.qkv_proj.weight.grad, rtol=1e-3, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@rerun_if_address_is_in_use()
+def test_sp_ring_zigzag_padded_model():
+    spawn(run_ring_zigzag_padded_model, 2)
