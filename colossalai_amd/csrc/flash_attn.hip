@@ -172,9 +172,10 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   using namespace fa;
   extern __shared__ char smem[];
   constexpr int KB_BYTES = sub_bytes<KVB, D>();
-  char* Klds = smem;
-  char* Vlds = smem + KB_BYTES;
-  char* Pw = smem + 2 * KB_BYTES;  // + wave*QW*KVB*2
+  // double-buffered K/V: tile t+1 stages into the OTHER buffer while tile
+  // t is being consumed — one barrier per tile instead of two
+  // (no pointer array: LDS addrspace casts cannot form a static initializer)
+  char* Pw = smem + 4 * KB_BYTES;  // + wave*QW*KVB*2
 
   const int lane = threadIdx.x & 63;
   const int w = threadIdx.x >> 6;
@@ -239,13 +240,15 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
   // SeffK == 0 blocks skip the tile loop; the clamped loads stay in-bounds
   kstage.load(k_base, kts, 0, SeffK);
   vstage.load(v_base, kts, 0, SeffK);
-  kstage.store_subtiled(Klds);
-  vstage.store_subtiled(Vlds);
+  kstage.store_subtiled(smem);
+  vstage.store_subtiled(smem + KB_BYTES);
   __syncthreads();
 
   for (int t = 0; t < kv_tiles; ++t) {
     const int k0 = t * KVB;
     const bool has_next = (t + 1 < kv_tiles);
+    char* Klds = smem + (t & 1) * 2 * KB_BYTES;
+    char* Vlds = Klds + KB_BYTES;
     if (has_next) {
       kstage.load(k_base, kts, k0 + KVB, SeffK);
       vstage.load(v_base, kts, k0 + KVB, SeffK);
@@ -393,10 +396,13 @@ __global__ __launch_bounds__(fa::NT) void fa_fwd_kernel(
     }
     }  // active
 
-    __syncthreads();  // all waves done reading Klds/Vlds for tile t
+    // stores target the OTHER buffer: no barrier needed before them, and
+    // the single barrier below covers both "stores visible" and "everyone
+    // done with this tile's buffer before it is reused two tiles on"
     if (has_next) {
-      kstage.store_subtiled(Klds);
-      vstage.store_subtiled(Vlds);
+      char* Kn = smem + ((t + 1) & 1) * 2 * KB_BYTES;
+      kstage.store_subtiled(Kn);
+      vstage.store_subtiled(Kn + KB_BYTES);
     }
     __syncthreads();
   }
@@ -1096,7 +1102,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 
 #define LAUNCH_FWD(DD)                                                                            \
   do {                                                                                            \
-    const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
+    const size_t lds = 4 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
     set_lds_limit((const void*)fa_fwd_kernel<DD, false>, lds);                                    \
     hipLaunchKernelGGL((fa_fwd_kernel<DD, false>), grid, dim3(NT), lds, stream.stream(),          \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
@@ -1135,7 +1141,7 @@ std::vector<at::Tensor> flash_attn_varlen_fwd(at::Tensor q, at::Tensor k, at::Te
 
 #define LAUNCH_VFWD(DD)                                                                           \
   do {                                                                                            \
-    const size_t lds = 2 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
+    const size_t lds = 4 * sub_bytes<KVB, DD>() + NW * (QW * KVB * 2);                            \
     set_lds_limit((const void*)fa_fwd_kernel<DD, true>, lds);                                     \
     hipLaunchKernelGGL((fa_fwd_kernel<DD, true>), grid, dim3(NT), lds, stream.stream(),           \
                        (const unsigned short*)q.data_ptr(), (const unsigned short*)k.data_ptr(),  \
