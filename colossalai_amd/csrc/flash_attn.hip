@@ -825,6 +825,8 @@ __global__ __launch_bounds__(fa::NTB, 2) void fa_bwd_dq_kernel(
   extern __shared__ char smem[];
   constexpr int KB_BYTES = sub_bytes<KVB, D>();
   constexpr int PB = sub_bytes<32, 32>();  // per-wave dS tile: [kv 32][q 32]
+  // single-buffered here: double-buffering (as in the forward) pushes the
+  // block past the 80 KB needed for 2 blocks/CU and measured -13%
   char* Klds = smem;
   char* Vlds = smem + KB_BYTES;
   char* Pw = smem + 2 * KB_BYTES;  // + wave*PB
