@@ -59,3 +59,21 @@ def test_quantize_model_nf4_runs():
     with torch.no_grad():
         out = q(torch.randint(0, 128, (2, 16)))["logits"]
     assert torch.isfinite(out).all()
+
+
+def test_quantized_state_dict_roundtrip():
+    """Quantized model state dicts restore exactly (buffers + bias)."""
+    import copy
+
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    qa = quantize_model(LlamaForCausalLM(cfg), bits=8).eval()
+    torch.manual_seed(9)
+    qb = quantize_model(LlamaForCausalLM(cfg), bits=8).eval()
+    qb.load_state_dict(qa.state_dict())
+    x = torch.randint(0, 128, (2, 12))
+    with torch.no_grad():
+        torch.testing.assert_close(qa(x)["logits"], qb(x)["logits"], rtol=0.0, atol=0.0)
