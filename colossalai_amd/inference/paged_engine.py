@@ -41,7 +41,10 @@ class ContinuousBatchEngine:
             num_blocks = self.config.max_batch_size * (
                 (self.config.max_seq_len + block_size - 1) // block_size
             )
-        self.kv = KVCacheManager(cfg.num_hidden_layers, cfg.num_key_value_heads, cfg.head_dim,
+        # per-rank layer attributes, not the config: under TP each rank
+        # caches only its local KV heads (same rule as LLMEngine)
+        attn0 = model.model.layers[0].self_attn
+        self.kv = KVCacheManager(cfg.num_hidden_layers, attn0.num_kv_heads, attn0.head_dim,
                                  num_blocks=num_blocks, block_size=block_size,
                                  device=self.device, dtype=self.dtype)
         self.rm = RequestManager(self.kv, max_batch_size=self.config.max_batch_size)

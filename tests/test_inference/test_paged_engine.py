@@ -125,3 +125,38 @@ def test_async_engine_concurrent():
 
     outs = asyncio.run(main())
     assert outs == ref
+
+
+def _run_tp_paged(rank, world_size, port):
+    """TP-sharded continuous batching matches the unsharded engine."""
+    import copy
+
+    import torch.distributed as dist
+
+    import colossalai_amd
+    from colossalai_amd.inference import ContinuousBatchEngine, GenerationConfig, InferenceConfig
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+    from colossalai_amd.shardformer import ShardConfig, ShardFormer
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=128)
+    model = LlamaForCausalLM(cfg).eval()
+    prompts = [[5, 17, 42], [99, 3, 4, 7]]
+    icfg = InferenceConfig(max_batch_size=2, max_input_len=16, max_output_len=8)
+    ref = ContinuousBatchEngine(model, icfg, block_size=4).generate(
+        prompts, GenerationConfig(max_new_tokens=6))
+
+    sharded, _ = ShardFormer(ShardConfig(tensor_parallel_process_group=dist.group.WORLD,
+                                         parallel_output=False)).optimize(copy.deepcopy(model))
+    eng = ContinuousBatchEngine(sharded.eval(), icfg, block_size=4)
+    out = eng.generate(prompts, GenerationConfig(max_new_tokens=6))
+    assert out == ref, f"tp paged {out} vs ref {ref}"
+    dist.destroy_process_group()
+
+
+def test_paged_engine_tp2():
+    from colossalai_amd.testing import rerun_if_address_is_in_use, spawn
+
+    rerun_if_address_is_in_use()(lambda: spawn(_run_tp_paged, 2))()
