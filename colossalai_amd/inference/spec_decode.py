@@ -61,10 +61,11 @@ class _CachedModel:
         self.device = next(model.parameters()).device
         self.dtype = next(model.parameters()).dtype
         cfg = model.config
+        attn0 = model.model.layers[0].self_attn  # per-rank heads under TP
         self.caches = [
-            (torch.zeros(1, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+            (torch.zeros(1, max_seq_len, attn0.num_kv_heads, attn0.head_dim,
                          device=self.device, dtype=self.dtype),
-             torch.zeros(1, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+             torch.zeros(1, max_seq_len, attn0.num_kv_heads, attn0.head_dim,
                          device=self.device, dtype=self.dtype))
             for _ in range(cfg.num_hidden_layers)
         ]
@@ -202,10 +203,11 @@ class _BatchedCachedModel:
         self.device = next(model.parameters()).device
         self.dtype = next(model.parameters()).dtype
         cfg = model.config
+        attn0 = model.model.layers[0].self_attn  # per-rank heads under TP
         self.caches = [
-            (torch.zeros(batch, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+            (torch.zeros(batch, max_seq_len, attn0.num_kv_heads, attn0.head_dim,
                          device=self.device, dtype=self.dtype),
-             torch.zeros(batch, max_seq_len, cfg.num_key_value_heads, cfg.head_dim,
+             torch.zeros(batch, max_seq_len, attn0.num_kv_heads, attn0.head_dim,
                          device=self.device, dtype=self.dtype))
             for _ in range(cfg.num_hidden_layers)
         ]
