@@ -106,3 +106,38 @@ def test_dataset_loaders_and_generation(tmp_path):
     assert len(out) == 4
     score = evaluate_generation(model, gen, detok, numeric_match, max_new_tokens=4)
     assert score["n"] == 1
+
+
+def _run_eval_tp(rank, world_size, port):
+    """Loglikelihood eval on a TP-sharded model matches the unsharded one."""
+    import copy
+
+    import torch
+    import torch.distributed as dist
+
+    import colossalai_amd
+    from applications.eval import evaluate_multiple_choice
+    from colossalai_amd.models import LlamaConfig, LlamaForCausalLM
+    from colossalai_amd.shardformer import ShardConfig, ShardFormer
+
+    colossalai_amd.launch(rank, world_size, "127.0.0.1", port, backend="gloo", verbose=False)
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+                      num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64)
+    ref = LlamaForCausalLM(cfg).eval()
+    sharded, _ = ShardFormer(ShardConfig(tensor_parallel_process_group=dist.group.WORLD,
+                                         parallel_output=False)).optimize(copy.deepcopy(ref))
+    examples = [
+        {"prompt": [5, 9, 11], "choices": [[3, 4], [7], [2, 2, 2]], "answer": 1},
+        {"prompt": [1, 2], "choices": [[8, 8], [9]], "answer": 0},
+    ]
+    a = evaluate_multiple_choice(ref, examples)
+    b = evaluate_multiple_choice(sharded.eval(), examples)
+    assert a == b
+    dist.destroy_process_group()
+
+
+def test_eval_tp2_matches():
+    from colossalai_amd.testing import rerun_if_address_is_in_use, spawn
+
+    rerun_if_address_is_in_use()(lambda: spawn(_run_eval_tp, 2))()
