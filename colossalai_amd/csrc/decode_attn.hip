@@ -32,7 +32,10 @@ constexpr int GROUPS = DEC_BLOCK / 16;  // 16-lane groups
 // kernel LSE-merges them. B*Hq blocks alone can't fill 256 CUs at decode
 // batch sizes; splitting the sequence restores the parallelism
 // (reference: flash_decoding_attention_kernel.cu v2 + reduce at :558).
-template <int D, bool PAGED, bool SPLIT>
+// GQ = Hq/Hkv: one block serves ALL GQ query heads of its kv head, so the
+// K/V stream is read ONCE instead of GQ times (decode is pure KV
+// bandwidth; GQA models were paying a GQ-fold read tax).
+template <int D, bool PAGED, bool SPLIT, int GQ>
 __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
     const unsigned short* __restrict__ Q,   // [B, Hq, D]
     const unsigned short* __restrict__ KC,
@@ -42,6 +45,7 @@ __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
     const int* __restrict__ block_tables,   // [B, max_blocks] (PAGED only)
     int max_blocks, int lbs,                // log2(block_size)
     int B, int Smax, int Hq, int Hkv, float scale,
+    int Hg, int gpk,  // Hg = Hq/GQ head-groups per batch; gpk = Hg/Hkv
     int n_splits, float* __restrict__ part_m,  // [B, Hq, NS]
     float* __restrict__ part_l,                // [B, Hq, NS]
     float* __restrict__ part_o) {              // [B, Hq, NS, D]
@@ -50,23 +54,27 @@ __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
   __shared__ float sm_l[GROUPS];
   __shared__ float sm_o[GROUPS][D];
 
-  const int bh = blockIdx.x;
-  const int b = bh / Hq, h = bh % Hq;
-  const int hk = h / (Hq / Hkv);
+  const int bh = blockIdx.x;                // over B * Hg
+  const int b = bh / Hg;
+  const int gidx = bh % Hg;                 // q-head group
+  const int hk = gidx / gpk;                // its kv head
   const int S = seq_lens[b];
   const int g = threadIdx.x / 16;   // group id
   const int e = threadIdx.x % 16;   // lane-in-group: owns d = e*EPL..+EPL
 
-  const unsigned short* q = Q + ((long)b * Hq + h) * D + e * EPL;
   const long kv_tok = (long)Hkv * D;
   const long seq_off = PAGED ? 0 : (long)b * Smax * kv_tok;
   const unsigned short* kbase = KC + seq_off + (long)hk * D + e * EPL;
   const unsigned short* vbase = VC + seq_off + (long)hk * D + e * EPL;
   const int* bt = PAGED ? block_tables + (long)b * max_blocks : nullptr;
 
-  float qf[EPL];
+  float qf[GQ][EPL];
 #pragma unroll
-  for (int j = 0; j < EPL; ++j) qf[j] = bf2f(q[j]);
+  for (int qh = 0; qh < GQ; ++qh) {
+    const unsigned short* q = Q + ((long)b * Hq + gidx * GQ + qh) * D + e * EPL;
+#pragma unroll
+    for (int j = 0; j < EPL; ++j) qf[qh][j] = bf2f(q[j]);
+  }
 
   int lo = 0, hi = S;
   if constexpr (SPLIT) {
@@ -75,82 +83,100 @@ __global__ __launch_bounds__(DEC_BLOCK) void decode_attn_kernel(
     hi = min(S, lo + chunk);
   }
 
-  float m = -INFINITY, l = 0.0f, o[EPL];
+  float m[GQ], l[GQ], o[GQ][EPL];
 #pragma unroll
-  for (int j = 0; j < EPL; ++j) o[j] = 0.0f;
+  for (int qh = 0; qh < GQ; ++qh) {
+    m[qh] = -INFINITY;
+    l[qh] = 0.0f;
+#pragma unroll
+    for (int j = 0; j < EPL; ++j) o[qh][j] = 0.0f;
+  }
 
   for (int s = lo + g; s < hi; s += GROUPS) {
     const long row = PAGED ? (((long)bt[s >> lbs] << lbs) | (s & ((1 << lbs) - 1))) : (long)s;
     const unsigned short* kp = kbase + row * kv_tok;
-    float dot = 0.0f;
+    float kfv[EPL];
     if constexpr (EPL == 8) {
       short8 kv = *reinterpret_cast<const short8*>(kp);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) dot += qf[j] * bf2f((unsigned short)kv[j]);
+      for (int j = 0; j < 8; ++j) kfv[j] = bf2f((unsigned short)kv[j]);
     } else {
       short4v kv = *reinterpret_cast<const short4v*>(kp);
 #pragma unroll
-      for (int j = 0; j < EPL; ++j) dot += qf[j] * bf2f((unsigned short)kv[j]);
+      for (int j = 0; j < EPL; ++j) kfv[j] = bf2f((unsigned short)kv[j]);
     }
-    // sum over the 16 lanes of the group
-#pragma unroll
-    for (int off = 8; off > 0; off >>= 1) dot += __shfl_xor(dot, off);
-    const float sc = dot * scale;
-    const float m_new = fmaxf(m, sc);
-    const float corr = (m == -INFINITY) ? 0.0f : __expf(m - m_new);
-    const float p = __expf(sc - m_new);
-    l = l * corr + p;
+    float vfv[EPL];
     const unsigned short* vp = vbase + row * kv_tok;
     if constexpr (EPL == 8) {
       short8 vv = *reinterpret_cast<const short8*>(vp);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) o[j] = o[j] * corr + p * bf2f((unsigned short)vv[j]);
+      for (int j = 0; j < 8; ++j) vfv[j] = bf2f((unsigned short)vv[j]);
     } else {
       short4v vv = *reinterpret_cast<const short4v*>(vp);
 #pragma unroll
-      for (int j = 0; j < EPL; ++j) o[j] = o[j] * corr + p * bf2f((unsigned short)vv[j]);
+      for (int j = 0; j < EPL; ++j) vfv[j] = bf2f((unsigned short)vv[j]);
     }
-    m = m_new;
+#pragma unroll
+    for (int qh = 0; qh < GQ; ++qh) {
+      float dot = 0.0f;
+#pragma unroll
+      for (int j = 0; j < EPL; ++j) dot += qf[qh][j] * kfv[j];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) dot += __shfl_xor(dot, off);
+      const float sc = dot * scale;
+      const float m_new = fmaxf(m[qh], sc);
+      const float corr = (m[qh] == -INFINITY) ? 0.0f : __expf(m[qh] - m_new);
+      const float pp = __expf(sc - m_new);
+      l[qh] = l[qh] * corr + pp;
+#pragma unroll
+      for (int j = 0; j < EPL; ++j) o[qh][j] = o[qh][j] * corr + pp * vfv[j];
+      m[qh] = m_new;
+    }
   }
 
-  // merge the GROUPS partials
-  if (e == 0) {
-    sm_m[g] = m;
-    sm_l[g] = l;
-  }
+  // merge the GROUPS partials, one q-head at a time (shared LDS scratch)
 #pragma unroll
-  for (int j = 0; j < EPL; ++j) sm_o[g][e * EPL + j] = o[j];
-  __syncthreads();
-
-  if (g == 0) {
-    float m_all = -INFINITY;
-#pragma unroll
-    for (int i = 0; i < GROUPS; ++i) m_all = fmaxf(m_all, sm_m[i]);
-    float l_all = 0.0f;
-    float acc[EPL];
-#pragma unroll
-    for (int j = 0; j < EPL; ++j) acc[j] = 0.0f;
-#pragma unroll
-    for (int i = 0; i < GROUPS; ++i) {
-      const float c = (sm_m[i] == -INFINITY) ? 0.0f : __expf(sm_m[i] - m_all);
-      l_all += sm_l[i] * c;
-#pragma unroll
-      for (int j = 0; j < EPL; ++j) acc[j] += sm_o[i][e * EPL + j] * c;
+  for (int qh = 0; qh < GQ; ++qh) {
+    if (qh > 0) __syncthreads();  // previous head's reduction done
+    if (e == 0) {
+      sm_m[g] = m[qh];
+      sm_l[g] = l[qh];
     }
-    if constexpr (SPLIT) {
-      const long pslot = ((long)b * Hq + h) * n_splits + blockIdx.y;
-      if (e == 0) {
-        part_m[pslot] = m_all;
-        part_l[pslot] = l_all;
+#pragma unroll
+    for (int j = 0; j < EPL; ++j) sm_o[g][e * EPL + j] = o[qh][j];
+    __syncthreads();
+
+    if (g == 0) {
+      float m_all = -INFINITY;
+#pragma unroll
+      for (int i = 0; i < GROUPS; ++i) m_all = fmaxf(m_all, sm_m[i]);
+      float l_all = 0.0f;
+      float acc[EPL];
+#pragma unroll
+      for (int j = 0; j < EPL; ++j) acc[j] = 0.0f;
+#pragma unroll
+      for (int i = 0; i < GROUPS; ++i) {
+        const float c = (sm_m[i] == -INFINITY) ? 0.0f : __expf(sm_m[i] - m_all);
+        l_all += sm_l[i] * c;
+#pragma unroll
+        for (int j = 0; j < EPL; ++j) acc[j] += sm_o[i][e * EPL + j] * c;
       }
-      float* po = part_o + pslot * D + e * EPL;
+      const int h = gidx * GQ + qh;
+      if constexpr (SPLIT) {
+        const long pslot = ((long)b * Hq + h) * n_splits + blockIdx.y;
+        if (e == 0) {
+          part_m[pslot] = m_all;
+          part_l[pslot] = l_all;
+        }
+        float* po = part_o + pslot * D + e * EPL;
 #pragma unroll
-      for (int j = 0; j < EPL; ++j) po[j] = acc[j];
-    } else {
-      const float inv_l = l_all > 0.0f ? 1.0f / l_all : 0.0f;
-      unsigned short* op = O + ((long)b * Hq + h) * D + e * EPL;
+        for (int j = 0; j < EPL; ++j) po[j] = acc[j];
+      } else {
+        const float inv_l = l_all > 0.0f ? 1.0f / l_all : 0.0f;
+        unsigned short* op = O + ((long)b * Hq + h) * D + e * EPL;
 #pragma unroll
-      for (int j = 0; j < EPL; ++j) op[j] = f2bf(acc[j] * inv_l);
+        for (int j = 0; j < EPL; ++j) op[j] = f2bf(acc[j] * inv_l);
+      }
     }
   }
 }
@@ -191,9 +217,16 @@ static at::Tensor decode_launch(at::Tensor q, at::Tensor kc, at::Tensor vc,
                                 at::Tensor seq_lens, double scale, int n_splits) {
   const int B = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
   const int Hkv = (int)kc.size(2);
+  const int G = Hq / Hkv;
+  // largest supported group factor: those query heads share one block's
+  // K/V stream (decode is KV bandwidth; GQA was paying a G-fold read tax)
+  const int GQ = (G % 8 == 0) ? 8 : (G % 4 == 0) ? 4 : (G % 2 == 0) ? 2 : 1;
+  const int Hg = Hq / GQ;         // head-groups per batch
+  const int gpk = Hg / Hkv;       // groups per kv head
+  const int n_bh = B * Hg;
   auto out = at::empty_like(q);
   auto stream = at::hip::getCurrentHIPStream();
-  if (n_splits <= 0) n_splits = pick_splits(B * Hq, Smax > 0 ? Smax : 4096);
+  if (n_splits <= 0) n_splits = pick_splits(n_bh, Smax > 0 ? Smax : 4096);
   at::Tensor pm, pl, po;
   float *pm_p = nullptr, *pl_p = nullptr, *po_p = nullptr;
   if (n_splits > 1) {
@@ -203,15 +236,23 @@ static at::Tensor decode_launch(at::Tensor q, at::Tensor kc, at::Tensor vc,
     po = at::empty({(long)B * Hq * n_splits * D}, opts);
     pm_p = pm.data_ptr<float>(); pl_p = pl.data_ptr<float>(); po_p = po.data_ptr<float>();
   }
-#define LAUNCH_DEC(DD, SPL, GRID)                                                                  \
-  hipLaunchKernelGGL((decode_attn_kernel<DD, PAGED, SPL>), GRID, dim3(DEC_BLOCK), 0,               \
+#define LAUNCH_DEC(DD, SPL, GQC, GRID)                                                             \
+  hipLaunchKernelGGL((decode_attn_kernel<DD, PAGED, SPL, GQC>), GRID, dim3(DEC_BLOCK), 0,          \
                      stream.stream(), (const unsigned short*)q.data_ptr(),                         \
                      (const unsigned short*)kc.data_ptr(), (const unsigned short*)vc.data_ptr(),   \
                      (unsigned short*)out.data_ptr(), seq_lens.data_ptr<int>(), bt_ptr,            \
-                     max_blocks, lbs, B, Smax, Hq, Hkv, (float)scale, n_splits, pm_p, pl_p, po_p)
+                     max_blocks, lbs, B, Smax, Hq, Hkv, (float)scale, Hg, gpk, n_splits,           \
+                     pm_p, pl_p, po_p)
+#define DISPATCH_GQ(DD, SPL, GRID)                                                                 \
+  do {                                                                                             \
+    if (GQ == 8) LAUNCH_DEC(DD, SPL, 8, GRID);                                                     \
+    else if (GQ == 4) LAUNCH_DEC(DD, SPL, 4, GRID);                                                \
+    else if (GQ == 2) LAUNCH_DEC(DD, SPL, 2, GRID);                                                \
+    else LAUNCH_DEC(DD, SPL, 1, GRID);                                                             \
+  } while (0)
   if (n_splits > 1) {
-    const dim3 grid(B * Hq, n_splits);
-    if (D == 128) LAUNCH_DEC(128, true, grid); else LAUNCH_DEC(64, true, grid);
+    const dim3 grid(n_bh, n_splits);
+    if (D == 128) DISPATCH_GQ(128, true, grid); else DISPATCH_GQ(64, true, grid);
     HIP_CHECK_LAST();
     const dim3 rgrid(B * Hq);
     if (D == 128)
@@ -221,9 +262,10 @@ static at::Tensor decode_launch(at::Tensor q, at::Tensor kc, at::Tensor vc,
       hipLaunchKernelGGL((decode_reduce_kernel<64>), rgrid, dim3(64), 0, stream.stream(),
                          pm_p, pl_p, po_p, (unsigned short*)out.data_ptr(), Hq, n_splits);
   } else {
-    const dim3 grid(B * Hq);
-    if (D == 128) LAUNCH_DEC(128, false, grid); else LAUNCH_DEC(64, false, grid);
+    const dim3 grid(n_bh);
+    if (D == 128) DISPATCH_GQ(128, false, grid); else DISPATCH_GQ(64, false, grid);
   }
+#undef DISPATCH_GQ
 #undef LAUNCH_DEC
   HIP_CHECK_LAST();
   return out;
