@@ -72,3 +72,21 @@ torch.cuda.synchronize()
 dt = time.perf_counter() - t0
 total_new = sum(len(o) - 128 for o in out)
 print(f"llama-7b bf16 bs8 hipGraph: {total_new} tokens in {dt:.2f}s = {total_new/dt:.1f} tok/s")
+
+
+# ----- GQA model (llama3-8b: Hq32/Hkv8) — exercises the grouped decode kernel
+cfg3 = LLAMA_CONFIGS["llama3-8b"]
+with torch.device("meta"):
+    model3 = LlamaForCausalLM(cfg3)
+model3 = model3.to_empty(device="cuda").to(torch.bfloat16)
+with torch.no_grad():
+    for p_ in model3.parameters():
+        p_.normal_(0, 0.02)
+eng3 = LLMEngine(model3, InferenceConfig(max_batch_size=8, max_input_len=256, max_output_len=160))
+out = eng3.generate(prompts, GenerationConfig(max_new_tokens=8))  # warm
+t0 = time.perf_counter()
+out = eng3.generate(prompts, GenerationConfig(max_new_tokens=128))
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+total_new = sum(len(o) - len(p) for o, p in zip(out, prompts))
+print(f"llama3-8b (GQA 32/8) bf16 bs8: {total_new} tokens in {dt:.2f}s = {total_new/dt:.1f} tok/s")
